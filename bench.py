@@ -1,0 +1,170 @@
+#!/usr/bin/env python3
+"""Flagship benchmark for the driver contract.
+
+Default: llama2-7b LoRA fine-tune step (DP over RCCL/xGMI), bf16,
+synthetic token data, random-init weights — the BASELINE.json metric
+"finetune samples/sec, llama2-7b at 1/2/4/8 MI355X" (the reference
+publishes no numbers; this bench ESTABLISHES the baseline,
+BASELINE.md).  --mode serve measures the companion metric
+"Server tokens/sec" (paged-KV decode throughput on the same model).
+
+Launch (driver):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+
+Weak scaling: per-GPU work is fixed (micro batch x seq per rank for
+train; the serve bench is single-GPU TP=1 per rank... serve mode is
+rank-0 only). Rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from runbooks_amd.parallel import comm
+
+
+def _sync():
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+
+
+def _max_over_ranks(x: float) -> float:
+    if comm.is_dist():
+        t = torch.tensor([x], dtype=torch.float64)
+        if torch.cuda.is_available():
+            t = t.cuda()
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        return float(t.item())
+    return x
+
+
+def bench_train(args) -> dict:
+    from runbooks_amd.train import TrainConfig, Trainer
+    from runbooks_amd.train.data import SyntheticTokens
+
+    cfg = TrainConfig(model=args.model, seq_len=args.seq_len,
+                      micro_batch=args.micro_batch,
+                      num_train_steps=args.warmup + args.steps,
+                      dtype="bfloat16" if torch.cuda.is_available() else "float32",
+                      save_steps=0, seed=17)
+    trainer = Trainer(cfg)
+    vocab = trainer.model.cfg.vocab_size
+    ds = SyntheticTokens(vocab, args.seq_len + 1, n=4096, seed=comm.rank())
+    batches = [torch.stack([ds[i * args.micro_batch + j]
+                            for j in range(args.micro_batch)])
+               for i in range(min(8, 4096 // args.micro_batch))]
+
+    for i in range(args.warmup):
+        trainer.train_step(batches[i % len(batches)])
+    comm.barrier()
+    _sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        trainer.train_step(batches[i % len(batches)])
+    comm.barrier()
+    _sync()
+    dt = _max_over_ranks(time.perf_counter() - t0)
+
+    n = comm.world_size()
+    samples = args.steps * args.micro_batch * n
+    return {
+        "metric": "finetune_samples_per_sec",
+        "value": samples / dt,
+        "unit": "samples/sec",
+        "n_gpus": n,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1e3,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if torch.cuda.is_available() else "float32",
+        "data": "synthetic",
+        "config": {"model": args.model, "global_batch": args.micro_batch * n,
+                   "seq_len": args.seq_len,
+                   "parallelism": f"dp{n}", "lora_r": cfg.lora_r},
+    }
+
+
+def bench_serve(args) -> dict:
+    from runbooks_amd.serve import Engine
+
+    dtype = torch.bfloat16 if torch.cuda.is_available() else torch.float32
+    eng = Engine(args.model, dtype=dtype,
+                 kv_blocks=None if torch.cuda.is_available() else 4096,
+                 max_batch=args.serve_batch, seed=17)
+    vocab = eng.cfg.vocab_size
+    g = torch.Generator().manual_seed(17)
+
+    def new_req():
+        prompt = torch.randint(0, vocab, (args.prompt_len,), generator=g).tolist()
+        eng.submit(prompt, max_new_tokens=1 << 30)  # run until bench ends
+
+    for _ in range(args.serve_batch):
+        new_req()
+    # warmup: prefill everyone + W decode steps
+    for _ in range(args.serve_batch + args.warmup):
+        eng.step()
+    assert len(eng.running) == args.serve_batch
+    _sync()
+    t0 = time.perf_counter()
+    tokens = 0
+    for _ in range(args.steps):
+        before = sum(len(r.output_ids) for r in eng.running)
+        eng.step()
+        after = sum(len(r.output_ids) for r in eng.running)
+        tokens += after - before
+    _sync()
+    dt = time.perf_counter() - t0
+    return {
+        "metric": "serve_tokens_per_sec",
+        "value": tokens / dt,
+        "unit": "tokens/sec",
+        "n_gpus": comm.world_size(),
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1e3,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if torch.cuda.is_available() else "float32",
+        "data": "synthetic",
+        "config": {"model": args.model, "global_batch": args.serve_batch,
+                   "seq_len": args.prompt_len,
+                   "parallelism": f"tp{comm.world_size()}"},
+    }
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--mode", choices=["train", "serve"], default="train")
+    p.add_argument("--model", default=None)
+    p.add_argument("--seq-len", type=int, default=512)
+    p.add_argument("--micro-batch", type=int, default=4)
+    p.add_argument("--serve-batch", type=int, default=32)
+    p.add_argument("--prompt-len", type=int, default=128)
+    args = p.parse_args()
+    if args.model is None:
+        args.model = "llama2-7b" if torch.cuda.is_available() else "tiny-llama"
+
+    comm.init_from_env()
+    torch.manual_seed(17)
+
+    result = bench_train(args) if args.mode == "train" else bench_serve(args)
+    if comm.rank() == 0:
+        print(json.dumps(result), flush=True)
+    if comm.is_dist():
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

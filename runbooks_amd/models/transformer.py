@@ -1,0 +1,250 @@
+"""TP-aware decoder-only transformer covering the llama / falcon / OPT
+families the reference platform serves.
+
+One implementation, three family knobs (ModelConfig):
+  * norm: RMSNorm (fused gfx950 kernel) or LayerNorm
+  * act: SwiGLU (llama), GELU (falcon), ReLU (OPT)
+  * pos: RoPE (fused kernel) or learned embeddings (OPT)
+  * parallel_residual: falcon's attn+mlp-in-parallel block
+  * GQA/MQA via num_kv_heads (llama2-70b 8, falcon-40b 8, falcon-7b 1)
+
+Three forward modes:
+  * ``forward(tokens)``             — training, full causal attention
+  * ``prefill(...)``                — fills the paged KV cache, returns
+                                      last-position logits
+  * ``decode(...)``                 — one token per sequence against the
+                                      paged cache (gfx950 decode kernel)
+
+TP: attention QKV is column-parallel (head-sharded), the output
+projection row-parallel (one xGMI all-reduce); MLP up column / down row
+(second all-reduce). KV heads shard when divisible by tp, otherwise the
+KV projection is replicated (MQA at tp>1) and each rank keeps a full KV
+cache — 288 GB HBM per GPU makes that the right trade on MI355X.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+from torch import nn
+
+from .. import ops
+from ..parallel import ColumnParallelLinear, RowParallelLinear, comm
+from .config import ModelConfig
+
+
+def _norm_module(cfg: ModelConfig, dtype):
+    if cfg.norm == "rmsnorm":
+        return ops.RMSNorm(cfg.hidden_size, eps=cfg.norm_eps, dtype=dtype)
+    return nn.LayerNorm(cfg.hidden_size, eps=cfg.norm_eps, dtype=dtype)
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype, tp: int):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp
+        self.dh = cfg.head_dim
+        assert cfg.num_heads % tp == 0, "num_heads must divide tp"
+        self.hq = cfg.num_heads // tp
+        self.kv_sharded = cfg.num_kv_heads % tp == 0
+        self.hkv = cfg.num_kv_heads // tp if self.kv_sharded else cfg.num_kv_heads
+        self.scale = 1.0 / math.sqrt(self.dh)
+
+        q_out = cfg.num_heads * self.dh
+        kv_out = cfg.num_kv_heads * self.dh
+        self.q_proj = ColumnParallelLinear(cfg.hidden_size, q_out,
+                                           bias=cfg.attn_bias, tp_size=tp, dtype=dtype)
+        if self.kv_sharded:
+            self.k_proj = ColumnParallelLinear(cfg.hidden_size, kv_out,
+                                               bias=cfg.attn_bias, tp_size=tp,
+                                               dtype=dtype, gather_input=False)
+            self.v_proj = ColumnParallelLinear(cfg.hidden_size, kv_out,
+                                               bias=cfg.attn_bias, tp_size=tp,
+                                               dtype=dtype, gather_input=False)
+        else:  # replicate KV (MQA with tp > num_kv_heads)
+            self.k_proj = nn.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
+                                    dtype=dtype)
+            self.v_proj = nn.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
+                                    dtype=dtype)
+        self.o_proj = RowParallelLinear(cfg.num_heads * self.dh, cfg.hidden_size,
+                                        bias=cfg.attn_bias, tp_size=tp, dtype=dtype)
+
+    def _qkv(self, x, cos, sin, positions):
+        # x: [T, hidden] token-major
+        T = x.shape[0]
+        q = self.q_proj(x).view(T, self.hq, self.dh)
+        k = self.k_proj(x).view(T, self.hkv, self.dh)
+        v = self.v_proj(x).view(T, self.hkv, self.dh)
+        if self.cfg.pos == "rope":
+            q = ops.rope(q, cos, sin, positions)
+            k = ops.rope(k, cos, sin, positions)
+        return q, k, v
+
+    def forward_train(self, x, cos, sin, positions, B, S):
+        q, k, v = self._qkv(x, cos, sin, positions)
+        q = q.view(B, S, self.hq, self.dh)
+        k = k.view(B, S, self.hkv, self.dh)
+        v = v.view(B, S, self.hkv, self.dh)
+        o = ops.causal_attention(q, k, v, scale=self.scale)
+        return self.o_proj(o.reshape(B * S, self.hq * self.dh))
+
+    def forward_prefill(self, x, cos, sin, positions, kc, vc, slot_mapping, B, S):
+        q, k, v = self._qkv(x, cos, sin, positions)
+        ops.kv_append(k, v, kc, vc, slot_mapping)
+        q = q.view(B, S, self.hq, self.dh)
+        o = ops.causal_attention(q, k.view(B, S, self.hkv, self.dh),
+                                 v.view(B, S, self.hkv, self.dh), scale=self.scale)
+        return self.o_proj(o.reshape(B * S, self.hq * self.dh))
+
+    def forward_decode(self, x, cos, sin, positions, kc, vc, slot_mapping,
+                       block_tables, seq_lens):
+        q, k, v = self._qkv(x, cos, sin, positions)
+        ops.kv_append(k, v, kc, vc, slot_mapping)
+        o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale)
+        B = x.shape[0]
+        return self.o_proj(o.reshape(B, self.hq * self.dh))
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype, tp: int):
+        super().__init__()
+        self.act = cfg.act
+        if cfg.act == "silu_glu":
+            self.gate_proj = ColumnParallelLinear(cfg.hidden_size, cfg.intermediate_size,
+                                                  bias=cfg.mlp_bias, tp_size=tp, dtype=dtype)
+            self.up_proj = ColumnParallelLinear(cfg.hidden_size, cfg.intermediate_size,
+                                                bias=cfg.mlp_bias, tp_size=tp, dtype=dtype,
+                                                gather_input=False)
+        else:
+            self.fc1 = ColumnParallelLinear(cfg.hidden_size, cfg.intermediate_size,
+                                            bias=cfg.mlp_bias, tp_size=tp, dtype=dtype)
+        self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size,
+                                           bias=cfg.mlp_bias, tp_size=tp, dtype=dtype)
+
+    def forward(self, x):
+        if self.act == "silu_glu":
+            return self.down_proj(torch.nn.functional.silu(self.gate_proj(x))
+                                  * self.up_proj(x))
+        h = self.fc1(x)
+        h = torch.nn.functional.gelu(h) if self.act == "gelu" else torch.relu(h)
+        return self.down_proj(h)
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype, tp: int):
+        super().__init__()
+        self.cfg = cfg
+        self.attn = Attention(cfg, dtype, tp)
+        self.mlp = MLP(cfg, dtype, tp)
+        self.norm1 = _norm_module(cfg, dtype)
+        self.norm2 = _norm_module(cfg, dtype)
+
+    def forward(self, x, attn_fn):
+        """attn_fn(h) runs the right attention mode on normed input."""
+        if self.cfg.parallel_residual:
+            # falcon: x + attn(ln1(x)) + mlp(ln2(x))
+            return x + attn_fn(self.norm1(x)) + self.mlp(self.norm2(x))
+        x = x + attn_fn(self.norm1(x))
+        return x + self.mlp(self.norm2(x))
+
+
+class Transformer(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16,
+                 tp: int | None = None, device=None):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp if tp is not None else comm.world_size()
+        self.dtype = dtype
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size, dtype=dtype)
+        if cfg.pos == "learned":
+            self.embed_pos = nn.Embedding(cfg.max_seq_len, cfg.hidden_size, dtype=dtype)
+        self.blocks = nn.ModuleList(Block(cfg, dtype, self.tp)
+                                    for _ in range(cfg.num_layers))
+        self.norm_f = _norm_module(cfg, dtype)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False, dtype=dtype)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.embed.weight
+        if cfg.pos == "rope":
+            cos, sin = ops.rope_tables(cfg.head_dim, cfg.max_seq_len, cfg.rope_theta)
+            self.register_buffer("rope_cos", cos, persistent=False)
+            self.register_buffer("rope_sin", sin, persistent=False)
+        else:
+            self.rope_cos = self.rope_sin = None
+        self.reset_parameters()
+        if device is not None:
+            self.to(device)
+
+    def reset_parameters(self):
+        for m in self.modules():
+            if isinstance(m, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):
+                nn.init.normal_(m.weight, std=0.02)
+                if getattr(m, "bias", None) is not None:
+                    nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, std=0.02)
+
+    # -- shared ----------------------------------------------------------------
+    def _embed(self, tokens, positions):
+        x = self.embed(tokens.reshape(-1))
+        if self.cfg.pos == "learned":
+            x = x + self.embed_pos(positions.long())
+        return x
+
+    # -- training --------------------------------------------------------------
+    def forward(self, tokens: torch.Tensor) -> torch.Tensor:
+        """tokens [B, S] -> logits [B, S, V] (training / eval, no cache)."""
+        B, S = tokens.shape
+        positions = torch.arange(S, device=tokens.device, dtype=torch.int32)
+        positions = positions.repeat(B)
+        x = self._embed(tokens, positions)
+        for blk in self.blocks:
+            fn = lambda h, b=blk: b.attn.forward_train(  # noqa: E731
+                h, self.rope_cos, self.rope_sin, positions, B, S)
+            x = blk(x, fn)
+        x = self.norm_f(x)
+        return self.lm_head(x).view(B, S, -1)
+
+    # -- serving ---------------------------------------------------------------
+    @torch.no_grad()
+    def prefill(self, tokens, positions, caches, slot_mapping):
+        """tokens [B, S]; caches: list of (k_cache, v_cache) per layer.
+        Returns last-position logits [B, V]."""
+        B, S = tokens.shape
+        x = self._embed(tokens, positions)
+        for blk, (kc, vc) in zip(self.blocks, caches):
+            fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_prefill(  # noqa: E731
+                h, self.rope_cos, self.rope_sin, positions, kc, vc, slot_mapping, B, S)
+            x = blk(x, fn)
+        x = self.norm_f(x.view(B, S, -1)[:, -1])
+        return self.lm_head(x)
+
+    @torch.no_grad()
+    def decode(self, tokens, positions, caches, slot_mapping, block_tables, seq_lens):
+        """tokens [B] (one new token per sequence) -> logits [B, V]."""
+        B = tokens.shape[0]
+        x = self._embed(tokens.view(B, 1), positions)
+        for blk, (kc, vc) in zip(self.blocks, caches):
+            fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
+                h, self.rope_cos, self.rope_sin, positions, kc, vc, slot_mapping,
+                block_tables, seq_lens)
+            x = blk(x, fn)
+        x = self.norm_f(x)
+        return self.lm_head(x)
+
+    # -- cache helpers ----------------------------------------------------------
+    def local_kv_heads(self) -> int:
+        return self.blocks[0].attn.hkv
+
+    def alloc_caches(self, num_blocks: int, device):
+        return [ops.alloc_kv_cache(num_blocks, self.local_kv_heads(),
+                                   self.cfg.head_dim, device, dtype=self.dtype)
+                for _ in range(self.cfg.num_layers)]
+
+
+def build_model(name_or_cfg, dtype=torch.bfloat16, tp: int | None = None,
+                device=None, seed: int = 0) -> Transformer:
+    from .config import get_config
+    cfg = name_or_cfg if isinstance(name_or_cfg, ModelConfig) else get_config(name_or_cfg)
+    torch.manual_seed(seed)
+    return Transformer(cfg, dtype=dtype, tp=tp, device=device)

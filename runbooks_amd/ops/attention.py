@@ -1,0 +1,93 @@
+"""Attention ops.
+
+* ``causal_attention``: batch-uniform causal GQA attention used by the
+  trainer forward/backward. Built from explicit GEMMs (hipBLASLt/rocBLAS
+  via torch.matmul — plain library GEMMs) + fused softmax, chunked over
+  query blocks to bound the S^2 score memory. No Triton, no SDPA dispatch.
+  A hand-written MFMA flash-attention prefill kernel supersedes this on
+  the serving path (see csrc/attention_prefill.hip).
+
+* ``paged_decode``: single-token decode against the paged KV cache —
+  gfx950 HIP kernel on GPU (csrc/attention_decode.hip), fp32 reference on
+  CPU (also the numerics oracle for the GPU test).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+from . import _backend
+
+
+def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     scale: float | None = None, q_block: int = 256) -> torch.Tensor:
+    """q [B, S, Hq, Dh]; k, v [B, S, Hkv, Dh] -> [B, S, Hq, Dh].
+
+    Differentiable; softmax in fp32.
+    """
+    B, S, Hq, Dh = q.shape
+    Hkv = k.shape[2]
+    if scale is None:
+        scale = 1.0 / math.sqrt(Dh)
+    g = Hq // Hkv
+    if g > 1:
+        k = k.repeat_interleave(g, dim=2)
+        v = v.repeat_interleave(g, dim=2)
+    qt = q.transpose(1, 2)  # [B, H, S, D]
+    kt = k.transpose(1, 2)
+    vt = v.transpose(1, 2)
+
+    outs = []
+    for s0 in range(0, S, q_block):
+        s1 = min(S, s0 + q_block)
+        qb = qt[:, :, s0:s1]                              # [B,H,bs,D]
+        scores = torch.matmul(qb, kt[:, :, :s1].transpose(-1, -2)).float() * scale
+        mask = torch.ones(s1 - s0, s1, dtype=torch.bool, device=q.device)
+        mask = torch.triu(mask, diagonal=s0 + 1)
+        scores = scores.masked_fill(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1).to(vt.dtype)
+        outs.append(torch.matmul(p, vt[:, :, :s1]))
+    return torch.cat(outs, dim=2).transpose(1, 2).contiguous()
+
+
+def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):
+    """fp32 reference decode: q [B, Hq, Dh]."""
+    B, Hq, Dh = q.shape
+    _, Hkv, BS, _ = k_cache.shape
+    g = Hq // Hkv
+    out = torch.empty_like(q)
+    for b in range(B):
+        n = int(seq_lens[b])
+        blocks = block_tables[b, : (n + BS - 1) // BS].long()
+        k = k_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[:n].float()
+        v = v_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[:n].float()
+        for h in range(Hq):
+            hk = h // g
+            s = (k[:, hk] @ q[b, h].float()) * scale
+            p = torch.softmax(s, dim=0)
+            out[b, h] = (p @ v[:, hk]).to(q.dtype)
+    return out
+
+
+def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
+                 scale: float | None = None, nsplit: int | None = None):
+    """Decode attention over the paged cache. q [B, Hq, Dh] bf16."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _backend.use_hip(q):
+        if nsplit is None:
+            # Fill 256 CUs: B * Hkv workgroups per split.
+            B, hkv = q.shape[0], k_cache.shape[1]
+            base = B * hkv
+            max_len = int(seq_lens.max())
+            nsplit = 1
+            if base < 256 and max_len > 256:
+                nsplit = min(16, max(1, (2 * 256) // max(base, 1)))
+                nsplit = min(nsplit, (max_len + 255) // 256)
+                nsplit = max(nsplit, 1)
+        return _backend.ext().paged_decode(
+            q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
+            int(nsplit), float(scale),
+        )
+    return paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale)

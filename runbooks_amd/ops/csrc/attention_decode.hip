@@ -1,0 +1,293 @@
+// Paged decode attention (one query token per sequence) for CDNA4 (gfx950).
+//
+// The serving hot loop: memory-bound on the KV read (288 GB HBM3E at
+// ~6.3 TB/s achievable). Design, MI355X-first:
+//  * wave64 split into four 16-lane groups; each group owns ONE kv token
+//    per step at 8 (Dh=128) or 4 (Dh=64) elems/lane, so every wave streams
+//    4 contiguous KV rows per iteration at 16/8 B per lane.
+//  * GQA/MQA-native: a workgroup handles one (seq, kv_head); its G query
+//    heads share each K/V read (falcon-40b MQA G=16, llama2-70b G=8,
+//    llama2-7b MHA G=1 — SURVEY.md §2b "server image").
+//  * flash-style online softmax entirely in registers; group partials
+//    merged via shfl_xor(16/32), wave partials via LDS.
+//  * kv-split ("flash-decoding") for long sequences / small batch: grid z
+//    = nsplit partitions, fp32 partials (m, l, acc) merged by a tiny
+//    second kernel — sized so batch*Hkv*nsplit >> 256 CUs.
+//
+// q:       [B, Hq, Dh] bf16     out: [B, Hq, Dh] bf16
+// caches:  [num_blocks, Hkv, BS, Dh] bf16 (see kvcache.hip)
+// block_tables: [B, max_blocks] int32;  seq_lens: [B] int32
+// partial: [B, Hq, nsplit, Dh + 2] fp32 workspace when nsplit > 1
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;    // 4 waves
+constexpr int NW = 4;
+
+template <int DH, int G, bool SPLIT>
+__global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
+    const uint16_t *__restrict__ q, const uint16_t *__restrict__ k_cache,
+    const uint16_t *__restrict__ v_cache, const int32_t *__restrict__ block_tables,
+    const int32_t *__restrict__ seq_lens, uint16_t *__restrict__ out,
+    float *__restrict__ partial, int hkv, int bs, int max_blocks, int nsplit,
+    float scale) {
+  constexpr int VE = DH / 16;              // elems per lane (16 lanes cover Dh)
+  const int b = blockIdx.x;
+  const int h_kv = blockIdx.y;
+  const int split = SPLIT ? blockIdx.z : 0;
+  const int hq0 = h_kv * G;
+  const int Hq = hkv * G;
+
+  const int seq_len = seq_lens[b];
+  int t_begin = 0, t_end = seq_len;
+  if (SPLIT) {
+    const int chunk = (seq_len + nsplit - 1) / nsplit;
+    t_begin = split * chunk;
+    t_end = min(seq_len, t_begin + chunk);
+  }
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int grp = lane >> 4;               // 16-lane group: token sub-index
+  const int gl = lane & 15;                // lane within group
+  const int d0 = gl * VE;                  // this lane's Dh slice
+
+  // Q for all G heads, pre-scaled (softmax scale folded into q).
+  float qreg[G][VE];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    const uint16_t *qp = q + ((int64_t)b * Hq + hq0 + g) * DH + d0;
+#pragma unroll
+    for (int e = 0; e < VE; ++e) qreg[g][e] = 0.f;
+    {
+      float tmp[VE];
+      if (VE == 8) rb::VIO<uint16_t>::load(qp, tmp);
+      else {
+#pragma unroll
+        for (int e = 0; e < VE; ++e) tmp[e] = rb::bf16_to_f32(qp[e]);
+      }
+#pragma unroll
+      for (int e = 0; e < VE; ++e) qreg[g][e] = tmp[e] * scale;
+    }
+  }
+
+  float m[G], l[G], acc[G][VE];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    m[g] = -INFINITY; l[g] = 0.f;
+#pragma unroll
+    for (int e = 0; e < VE; ++e) acc[g][e] = 0.f;
+  }
+
+  const int32_t *bt = block_tables + (int64_t)b * max_blocks;
+
+  // Each wave covers tokens t_begin + i*16 + wid*4 + grp for i = 0,1,...
+  for (int t = t_begin + wid * 4 + grp; t < t_end; t += NW * 4) {
+    const bool valid = true;
+    const int tt = t;
+    const int blk = bt[tt / bs];
+    const int64_t base = (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * DH + d0;
+
+    float kf[VE], vf[VE];
+    if (VE == 8) {
+      rb::VIO<uint16_t>::load(k_cache + base, kf);
+      rb::VIO<uint16_t>::load(v_cache + base, vf);
+    } else {
+#pragma unroll
+      for (int e = 0; e < VE; ++e) {
+        kf[e] = rb::bf16_to_f32(k_cache[base + e]);
+        vf[e] = rb::bf16_to_f32(v_cache[base + e]);
+      }
+    }
+
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      float s = 0.f;
+#pragma unroll
+      for (int e = 0; e < VE; ++e) s += qreg[g][e] * kf[e];
+      // reduce across the 16-lane group
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+      if (!valid) s = -INFINITY;
+      const float mn = fmaxf(m[g], s);
+      if (mn != -INFINITY) {
+        const float alpha = __expf(m[g] - mn);
+        const float p = __expf(s - mn);
+        l[g] = l[g] * alpha + p;
+#pragma unroll
+        for (int e = 0; e < VE; ++e) acc[g][e] = acc[g][e] * alpha + p * vf[e];
+        m[g] = mn;
+      }
+    }
+  }
+
+  // Merge the 4 groups inside each wave: lanes l and l^16 / l^32 hold the
+  // same d0 slice, so shfl_xor merges matching elements.
+#pragma unroll
+  for (int off = 16; off <= 32; off <<= 1) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const float mo = __shfl_xor(m[g], off, 64);
+      const float lo = __shfl_xor(l[g], off, 64);
+      const float mn = fmaxf(m[g], mo);
+      if (mn == -INFINITY) continue;
+      const float a1 = __expf(m[g] - mn);
+      const float a2 = __expf(mo - mn);
+#pragma unroll
+      for (int e = 0; e < VE; ++e) {
+        const float ao = __shfl_xor(acc[g][e], off, 64);
+        acc[g][e] = acc[g][e] * a1 + ao * a2;
+      }
+      l[g] = l[g] * a1 + lo * a2;
+      m[g] = mn;
+    }
+  }
+
+  // Merge across the 4 waves via LDS. Layout per (wave, head):
+  // [DH acc][m][l] floats.
+  __shared__ float lds[NW][G][DH + 2];
+  if (grp == 0) {   // lanes 0..15 of each wave hold the wave's merged state
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+#pragma unroll
+      for (int e = 0; e < VE; ++e) lds[wid][g][d0 + e] = acc[g][e];
+      if (gl == 0) { lds[wid][g][DH] = m[g]; lds[wid][g][DH + 1] = l[g]; }
+    }
+  }
+  __syncthreads();
+
+  // Wave `wid` finalizes heads g = wid, wid+NW, ... (lanes 0..15 active).
+  for (int g = wid; g < G; g += NW) {
+    if (grp != 0) continue;
+    float mm = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) mm = fmaxf(mm, lds[w][g][DH]);
+    float ll = 0.f, av[VE];
+#pragma unroll
+    for (int e = 0; e < VE; ++e) av[e] = 0.f;
+    if (mm != -INFINITY) {
+#pragma unroll
+      for (int w = 0; w < NW; ++w) {
+        const float a = __expf(lds[w][g][DH] - mm);
+        ll += lds[w][g][DH + 1] * a;
+#pragma unroll
+        for (int e = 0; e < VE; ++e) av[e] += lds[w][g][d0 + e] * a;
+      }
+    }
+    if (SPLIT) {
+      float *pp = partial + (((int64_t)b * Hq + hq0 + g) * nsplit + split) * (DH + 2);
+#pragma unroll
+      for (int e = 0; e < VE; ++e) pp[d0 + e] = av[e];
+      if (gl == 0) { pp[DH] = mm; pp[DH + 1] = ll; }
+    } else {
+      const float inv_l = (ll > 0.f) ? 1.0f / ll : 0.f;
+      uint16_t *op = out + ((int64_t)b * Hq + hq0 + g) * DH + d0;
+#pragma unroll
+      for (int e = 0; e < VE; ++e) op[e] = rb::f32_to_bf16(av[e] * inv_l);
+    }
+  }
+}
+
+// Second pass for SPLIT mode: merge nsplit partials per (b, hq) row.
+__global__ void decode_combine_kernel(const float *__restrict__ partial,
+                                      uint16_t *__restrict__ out, int nsplit,
+                                      int dh) {
+  const int64_t row = blockIdx.x;          // b * Hq + hq
+  const float *p = partial + row * (int64_t)nsplit * (dh + 2);
+  float mm = -INFINITY;
+  for (int s = 0; s < nsplit; ++s) mm = fmaxf(mm, p[s * (dh + 2) + dh]);
+  float ll = 0.f;
+  for (int s = 0; s < nsplit; ++s)
+    if (p[s * (dh + 2) + dh] != -INFINITY)
+      ll += p[s * (dh + 2) + dh + 1] * __expf(p[s * (dh + 2) + dh] - mm);
+  const float inv_l = (ll > 0.f) ? 1.0f / ll : 0.f;
+  for (int d = threadIdx.x; d < dh; d += blockDim.x) {
+    float o = 0.f;
+    for (int s = 0; s < nsplit; ++s) {
+      const float ms = p[s * (dh + 2) + dh];
+      if (ms != -INFINITY) o += p[s * (dh + 2) + d] * __expf(ms - mm);
+    }
+    out[row * dh + d] = rb::f32_to_bf16(o * inv_l);
+  }
+}
+
+template <int DH, int G>
+void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
+                   const at::Tensor &v_cache, const at::Tensor &block_tables,
+                   const at::Tensor &seq_lens, at::Tensor &out, int nsplit,
+                   float scale, hipStream_t stream) {
+  const int B = (int)q.size(0);
+  const int hkv = (int)k_cache.size(1);
+  const int bs = (int)k_cache.size(2);
+  const int max_blocks = (int)block_tables.size(1);
+  if (nsplit <= 1) {
+    hipLaunchKernelGGL((paged_decode_kernel<DH, G, false>), dim3(B, hkv, 1),
+                       dim3(BLOCK), 0, stream, (const uint16_t *)q.data_ptr(),
+                       (const uint16_t *)k_cache.data_ptr(),
+                       (const uint16_t *)v_cache.data_ptr(),
+                       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+                       (uint16_t *)out.data_ptr(), nullptr, hkv, bs, max_blocks, 1, scale);
+  } else {
+    const int Hq = hkv * G;
+    auto partial = at::empty({B, Hq, nsplit, DH + 2},
+                             q.options().dtype(at::kFloat));
+    hipLaunchKernelGGL((paged_decode_kernel<DH, G, true>), dim3(B, hkv, nsplit),
+                       dim3(BLOCK), 0, stream, (const uint16_t *)q.data_ptr(),
+                       (const uint16_t *)k_cache.data_ptr(),
+                       (const uint16_t *)v_cache.data_ptr(),
+                       block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+                       nullptr, partial.data_ptr<float>(), hkv, bs, max_blocks, nsplit,
+                       scale);
+    hipLaunchKernelGGL(decode_combine_kernel, dim3(B * Hq), dim3(256), 0, stream,
+                       partial.data_ptr<float>(), (uint16_t *)out.data_ptr(), nsplit, DH);
+  }
+}
+
+}  // namespace
+
+at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
+                        at::Tensor block_tables, at::Tensor seq_lens,
+                        int64_t nsplit, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous(), "paged_decode: q");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "paged_decode: bf16 only");
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt && seq_lens.scalar_type() == at::kInt,
+              "paged_decode: int32 metadata");
+  const int Hq = (int)q.size(1);
+  const int dh = (int)q.size(2);
+  const int hkv = (int)k_cache.size(1);
+  TORCH_CHECK(Hq % hkv == 0, "paged_decode: Hq % Hkv");
+  const int G = Hq / hkv;
+  auto out = at::empty_like(q);
+  auto stream = at::hip::getCurrentHIPStream();
+
+#define RB_DEC(DHV, GV)                                                        \
+  launch_decode<DHV, GV>(q, k_cache, v_cache, block_tables, seq_lens, out,     \
+                         (int)nsplit, (float)scale, stream)
+  if (dh == 128) {
+    switch (G) {
+      case 1: RB_DEC(128, 1); break;
+      case 2: RB_DEC(128, 2); break;
+      case 4: RB_DEC(128, 4); break;
+      case 8: RB_DEC(128, 8); break;
+      default: TORCH_CHECK(false, "paged_decode: unsupported G=", G);
+    }
+  } else if (dh == 64) {
+    switch (G) {
+      case 1: RB_DEC(64, 1); break;
+      case 2: RB_DEC(64, 2); break;
+      case 4: RB_DEC(64, 4); break;
+      case 8: RB_DEC(64, 8); break;
+      case 16: RB_DEC(64, 16); break;
+      default: TORCH_CHECK(false, "paged_decode: unsupported G=", G);
+    }
+  } else {
+    TORCH_CHECK(false, "paged_decode: Dh must be 64 or 128, got ", dh);
+  }
+#undef RB_DEC
+  return out;
+}

@@ -1,0 +1,32 @@
+// Python bindings for the runbooks_amd CDNA4 (gfx950) kernel library.
+
+#include <torch/extension.h>
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
+                                    at::Tensor inv_rms);
+at::Tensor rope_fwd(at::Tensor x, at::Tensor cos, at::Tensor sin,
+                    at::Tensor positions, int64_t heads);
+at::Tensor rope_bwd(at::Tensor dy, at::Tensor cos, at::Tensor sin,
+                    at::Tensor positions, int64_t heads);
+void adamw_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                double lr, double beta1, double beta2, double eps, double wd,
+                int64_t step);
+at::Tensor sample_tokens(at::Tensor logits, double temperature, int64_t seed);
+void kv_append(at::Tensor k, at::Tensor v, at::Tensor k_cache, at::Tensor v_cache,
+               at::Tensor slot_mapping);
+at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
+                        at::Tensor block_tables, at::Tensor seq_lens,
+                        int64_t nsplit, double scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "runbooks_amd gfx950 HIP kernels";
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm forward (bf16/f32)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm backward");
+  m.def("rope_fwd", &rope_fwd, "RoPE rotate-half forward");
+  m.def("rope_bwd", &rope_bwd, "RoPE rotate-half backward (inverse rotation)");
+  m.def("adamw_step", &adamw_step, "fused AdamW single-tensor step");
+  m.def("sample_tokens", &sample_tokens, "greedy / Gumbel-max sampling");
+  m.def("kv_append", &kv_append, "paged KV-cache append");
+  m.def("paged_decode", &paged_decode, "paged GQA/MQA decode attention");
+}

@@ -1,0 +1,147 @@
+// Common device helpers for runbooks_amd CDNA4 (gfx950) kernels.
+//
+// Written MI355X-first: wave64 everywhere, vectorized bf16 access in
+// 8/16-byte units (hipcc does not auto-vectorize bf16 loads), grid-stride
+// loops capped so the scheduler has room (256 CUs x 8 XCDs).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define RB_WAVE 64
+#define RB_DEV __device__ __forceinline__
+
+namespace rb {
+
+// ---------------------------------------------------------------------------
+// bf16 <-> f32 (explicit bit ops; avoids header/format portability issues)
+// ---------------------------------------------------------------------------
+RB_DEV float bf16_to_f32(uint16_t u) {
+  union { uint32_t u32; float f; } c;
+  c.u32 = (uint32_t)u << 16;
+  return c.f;
+}
+
+RB_DEV uint16_t f32_to_bf16(float f) {
+  union { float f; uint32_t u32; } c;
+  c.f = f;
+  uint32_t u = c.u32;
+  // round-to-nearest-even
+  uint32_t rounding_bias = 0x7FFF + ((u >> 16) & 1);
+  u += rounding_bias;
+  return (uint16_t)(u >> 16);
+}
+
+// 8 x bf16 = 16 bytes: the coalescing sweet spot for wave64 streams.
+struct bf16x8 { uint16_t v[8]; };
+struct f32x8 { float v[8]; };
+
+RB_DEV f32x8 to_f32(const bf16x8 &a) {
+  f32x8 r;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) r.v[i] = bf16_to_f32(a.v[i]);
+  return r;
+}
+
+RB_DEV bf16x8 to_bf16(const f32x8 &a) {
+  bf16x8 r;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) r.v[i] = f32_to_bf16(a.v[i]);
+  return r;
+}
+
+// Scalar dtype-generic helpers (bf16-as-u16 or f32).
+RB_DEV float bf16_to_f32_or_id(uint16_t u) { return bf16_to_f32(u); }
+RB_DEV float bf16_to_f32_or_id(float f) { return f; }
+RB_DEV void store_scalar(uint16_t *p, float f) { *p = f32_to_bf16(f); }
+RB_DEV void store_scalar(float *p, float f) { *p = f; }
+
+// ---------------------------------------------------------------------------
+// Wave-level reductions (64 lanes).
+// ---------------------------------------------------------------------------
+RB_DEV float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, 64);
+  return x;
+}
+
+RB_DEV float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, 64));
+  return x;
+}
+
+// Block reduction via LDS; `tmp` must hold >= blockDim.x/64 floats.
+// Result is valid on every thread.
+RB_DEV float block_reduce_sum(float x, float *tmp) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = (blockDim.x + 63) >> 6;
+  x = wave_reduce_sum(x);
+  if (lane == 0) tmp[wid] = x;
+  __syncthreads();
+  float r = (lane < nwaves) ? tmp[lane] : 0.0f;
+  r = wave_reduce_sum(r);   // nwaves <= 16, fits one wave
+  __syncthreads();
+  return r;
+}
+
+RB_DEV float block_reduce_max(float x, float *tmp) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = (blockDim.x + 63) >> 6;
+  x = wave_reduce_max(x);
+  if (lane == 0) tmp[wid] = x;
+  __syncthreads();
+  float r = (lane < nwaves) ? tmp[lane] : -INFINITY;
+  r = wave_reduce_max(r);
+  __syncthreads();
+  return r;
+}
+
+// Grid sizing for memory-bound grid-stride kernels: enough blocks to fill
+// 256 CUs x several blocks each, capped so tail effects stay small.
+static inline int rb_grid_1d(int64_t total, int block) {
+  int64_t blocks = (total + block - 1) / block;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+}  // namespace rb
+
+// ---------------------------------------------------------------------------
+// Vectorized IO traits: bf16 in 16B (8 elems), f32 in 16B (4 elems).
+// Both move 16 bytes per lane per instruction (global_load_dwordx4).
+// ---------------------------------------------------------------------------
+namespace rb {
+
+template <typename T> struct VIO;
+
+template <> struct VIO<uint16_t> {  // bf16 carried as raw u16
+  static constexpr int W = 8;
+  RB_DEV static void load(const uint16_t *p, float *f) {
+    bf16x8 v = *reinterpret_cast<const bf16x8 *>(p);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) f[i] = bf16_to_f32(v.v[i]);
+  }
+  RB_DEV static void store(uint16_t *p, const float *f) {
+    bf16x8 v;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v.v[i] = f32_to_bf16(f[i]);
+    *reinterpret_cast<bf16x8 *>(p) = v;
+  }
+};
+
+template <> struct VIO<float> {
+  static constexpr int W = 4;
+  RB_DEV static void load(const float *p, float *f) {
+    float4 v = *reinterpret_cast<const float4 *>(p);
+    f[0] = v.x; f[1] = v.y; f[2] = v.z; f[3] = v.w;
+  }
+  RB_DEV static void store(float *p, const float *f) {
+    *reinterpret_cast<float4 *>(p) = make_float4(f[0], f[1], f[2], f[3]);
+  }
+};
+
+}  // namespace rb

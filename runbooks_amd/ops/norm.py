@@ -1,0 +1,47 @@
+"""RMSNorm with a fused gfx950 HIP kernel (GPU) / fp32 reference (CPU)."""
+from __future__ import annotations
+
+import torch
+
+from . import _backend
+
+
+def rmsnorm_ref(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    """Plain PyTorch fp32 reference (llama-family RMSNorm semantics)."""
+    xf = x.float()
+    var = xf.pow(2).mean(-1, keepdim=True)
+    y = xf * torch.rsqrt(var + eps)
+    return (y * weight.float()).to(x.dtype)
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        y, inv_rms = _backend.ext().rmsnorm_fwd(x, weight, eps)
+        ctx.save_for_backward(x, weight, inv_rms)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, inv_rms = ctx.saved_tensors
+        dx, dw = _backend.ext().rmsnorm_bwd(x, weight, dy.contiguous(), inv_rms)
+        return dx, dw.to(weight.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if _backend.use_hip(x):
+        return _RMSNormFn.apply(x.contiguous(), weight.contiguous(), eps)
+    return rmsnorm_ref(x, weight, eps)
+
+
+class RMSNorm(torch.nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-5, dtype=None):
+        super().__init__()
+        self.weight = torch.nn.Parameter(torch.ones(dim, dtype=dtype))
+        self.eps = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return rmsnorm(x, self.weight, self.eps)
+
+    def extra_repr(self) -> str:
+        return f"dim={self.weight.numel()}, eps={self.eps}"

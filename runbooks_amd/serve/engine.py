@@ -1,0 +1,195 @@
+"""Serving engine: paged KV cache + continuous batching.
+
+Native replacement for the reference's external basaran server image
+(SURVEY.md §2b "server image"): greedy/temperature decode over a paged
+KV cache, sized for the MI355X's 288 GB HBM3E (the cache pool is
+allocated from measured free memory, not a guess), TP across up to 8
+GPUs over xGMI.
+
+Scheduling: one prefill (whole prompt in one pass) is admitted per step
+when capacity allows, then all running sequences decode as one batch —
+decode batches hit the gfx950 paged-decode kernel
+(ops/csrc/attention_decode.hip).
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+
+import torch
+
+from .. import ops
+from ..models import Transformer, build_model
+from ..parallel import comm
+
+
+class BlockAllocator:
+    def __init__(self, num_blocks: int):
+        self.free = list(range(num_blocks - 1, -1, -1))
+        self.num_blocks = num_blocks
+
+    def alloc(self, n: int) -> list[int]:
+        if n > len(self.free):
+            raise RuntimeError("KV cache exhausted")
+        return [self.free.pop() for _ in range(n)]
+
+    def release(self, blocks: list[int]):
+        self.free.extend(blocks)
+
+
+@dataclass
+class Request:
+    request_id: int
+    prompt_ids: list[int]
+    max_new_tokens: int = 64
+    temperature: float = 0.0
+    stop_ids: tuple[int, ...] = ()
+    # state
+    output_ids: list[int] = field(default_factory=list)
+    blocks: list[int] = field(default_factory=list)
+    finished: bool = False
+    created: float = field(default_factory=time.time)
+
+    @property
+    def seq_len(self) -> int:
+        return len(self.prompt_ids) + len(self.output_ids)
+
+
+class Engine:
+    def __init__(self, model: Transformer | str, device=None,
+                 dtype=torch.bfloat16, kv_blocks: int | None = None,
+                 max_batch: int = 64, mem_fraction: float = 0.85, seed: int = 0):
+        self.device = device if device is not None else (
+            f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
+        if isinstance(model, str):
+            model = build_model(model, dtype=dtype, seed=seed)
+        self.model = model.to(self.device).eval()
+        self.cfg = model.cfg
+        self.bs = ops.BLOCK_SIZE
+        self.max_batch = max_batch
+        self.seed = seed
+        if kv_blocks is None:
+            kv_blocks = self._auto_kv_blocks(mem_fraction)
+        self.allocator = BlockAllocator(kv_blocks)
+        self.caches = self.model.alloc_caches(kv_blocks, self.device)
+        self.waiting: list[Request] = []
+        self.running: list[Request] = []
+        self._next_id = 0
+        self.max_blocks_per_seq = min(
+            kv_blocks, (self.cfg.max_seq_len + self.bs - 1) // self.bs)
+
+    def _auto_kv_blocks(self, mem_fraction: float) -> int:
+        bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
+                           * self.bs * self.cfg.head_dim * 2)
+        if torch.cuda.is_available():
+            free, _total = torch.cuda.mem_get_info(self.device)
+            budget = int(free * mem_fraction) - (2 << 30)  # activations headroom
+        else:
+            budget = 64 << 20
+        return max(16, budget // bytes_per_block)
+
+    # -- request API ------------------------------------------------------------
+    def submit(self, prompt_ids: list[int], max_new_tokens: int = 64,
+               temperature: float = 0.0) -> Request:
+        req = Request(self._next_id, list(prompt_ids), max_new_tokens, temperature)
+        self._next_id += 1
+        self.waiting.append(req)
+        return req
+
+    def _admit(self) -> Request | None:
+        if not self.waiting or len(self.running) >= self.max_batch:
+            return None
+        req = self.waiting[0]
+        need = (len(req.prompt_ids) + self.bs - 1) // self.bs + 1
+        if need > len(self.allocator.free):
+            return None
+        self.waiting.pop(0)
+        req.blocks = self.allocator.alloc(need)
+        return req
+
+    # -- model invocations ------------------------------------------------------
+    def _prefill(self, req: Request) -> int:
+        S = len(req.prompt_ids)
+        tokens = torch.tensor([req.prompt_ids], dtype=torch.long, device=self.device)
+        positions = torch.arange(S, dtype=torch.int32, device=self.device)
+        slots = torch.tensor(
+            [req.blocks[i // self.bs] * self.bs + i % self.bs for i in range(S)],
+            dtype=torch.int32, device=self.device)
+        logits = self.model.prefill(tokens, positions, self.caches, slots)
+        tok = ops.sample_tokens(logits, req.temperature,
+                                seed=self.seed + req.request_id * 65537 + S)
+        return int(tok[0])
+
+    def _decode_batch(self, reqs: list[Request]) -> list[int]:
+        B = len(reqs)
+        dev = self.device
+        last = [r.prompt_ids[-1] if not r.output_ids else r.output_ids[-1]
+                for r in reqs]
+        # position of the token being generated = current seq_len - 1 for the
+        # query; its kv slot appends at position seq_len - 1.
+        pos = [r.seq_len - 1 for r in reqs]
+        slots = []
+        for r, p in zip(reqs, pos):
+            blk = r.blocks[p // self.bs]
+            slots.append(blk * self.bs + p % self.bs)
+        maxb = max(len(r.blocks) for r in reqs)
+        bt = torch.zeros(B, maxb, dtype=torch.int32)
+        for i, r in enumerate(reqs):
+            bt[i, :len(r.blocks)] = torch.tensor(r.blocks, dtype=torch.int32)
+        tokens = torch.tensor(last, dtype=torch.long, device=dev)
+        positions = torch.tensor(pos, dtype=torch.int32, device=dev)
+        slot_t = torch.tensor(slots, dtype=torch.int32, device=dev)
+        seq_lens = torch.tensor([p + 1 for p in pos], dtype=torch.int32, device=dev)
+        logits = self.model.decode(tokens, positions, self.caches, slot_t,
+                                   bt.to(dev), seq_lens)
+        toks = ops.sample_tokens(
+            logits, reqs[0].temperature,
+            seed=self.seed + 1_000_003 * reqs[0].seq_len)
+        return [int(t) for t in toks]
+
+    # -- scheduler step ----------------------------------------------------------
+    def step(self) -> list[Request]:
+        """One engine iteration. Returns requests finished this step."""
+        finished = []
+        req = self._admit()
+        if req is None and not self.running and self.waiting:
+            need = (len(self.waiting[0].prompt_ids) + self.bs - 1) // self.bs + 1
+            if need > self.allocator.num_blocks:
+                r = self.waiting.pop(0)
+                r.finished = True
+                raise RuntimeError(
+                    f"prompt of {len(r.prompt_ids)} tokens cannot fit the KV "
+                    f"cache ({self.allocator.num_blocks} blocks)")
+        if req is not None:
+            first = self._prefill(req)
+            req.output_ids.append(first)
+            self.running.append(req)
+        elif self.running:
+            # ensure every running seq has a block for the next position
+            for r in self.running:
+                if r.seq_len >= len(r.blocks) * self.bs:
+                    r.blocks.extend(self.allocator.alloc(1))
+            toks = self._decode_batch(self.running)
+            for r, t in zip(self.running, toks):
+                r.output_ids.append(t)
+        for r in list(self.running):
+            if (len(r.output_ids) >= r.max_new_tokens
+                    or (r.stop_ids and r.output_ids[-1] in r.stop_ids)
+                    or r.seq_len >= self.cfg.max_seq_len):
+                r.finished = True
+                self.allocator.release(r.blocks)
+                r.blocks = []
+                self.running.remove(r)
+                finished.append(r)
+        return finished
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    # -- convenience --------------------------------------------------------------
+    def generate(self, prompt_ids: list[int], max_new_tokens: int = 64,
+                 temperature: float = 0.0) -> list[int]:
+        req = self.submit(prompt_ids, max_new_tokens, temperature)
+        while not req.finished:
+            self.step()
+        return req.output_ids

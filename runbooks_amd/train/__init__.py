@@ -1,0 +1,5 @@
+"""Training: LoRA fine-tune loop, checkpointing, data."""
+from .checkpoint import latest_checkpoint, load_checkpoint, save_checkpoint  # noqa: F401
+from .data import JsonlTextDataset, SyntheticTokens, data_loader  # noqa: F401
+from .lora import LoRALinear, apply_lora, lora_state_dict, merge_lora  # noqa: F401
+from .trainer import TrainConfig, Trainer  # noqa: F401

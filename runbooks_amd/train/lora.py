@@ -1,0 +1,88 @@
+"""LoRA adapters (Hu et al. 2021) over the transformer's linear layers.
+
+The reference platform's fine-tune path is LoRA via the external HF
+trainer image (reference examples/llama2-7b/finetuned-model.yaml,
+SURVEY.md §2b "trainer image"); here it is native: frozen base weights,
+trainable low-rank A/B pairs on the attention + MLP projections, with
+the skinny adapter GEMMs running through hipBLASLt-backed torch.matmul
+(a fused gfx950 LoRA kernel is the planned upgrade on this path).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+from torch import nn
+
+from ..parallel.tp import ColumnParallelLinear, RowParallelLinear
+
+DEFAULT_TARGETS = ("q_proj", "k_proj", "v_proj", "o_proj",
+                   "gate_proj", "up_proj", "down_proj", "fc1", "fc2")
+
+
+class LoRALinear(nn.Module):
+    """Wraps a (possibly TP-sharded) linear: y = base(x) + scale * B(A(x)).
+
+    A: [r, in_shard], B: [out_shard, r] — sharded the same way as the base
+    weight so TP stays consistent; A/B are the only trainable params.
+    """
+
+    def __init__(self, base: nn.Module, r: int = 16, alpha: int = 32,
+                 dtype=None):
+        super().__init__()
+        self.base = base
+        w = base.weight
+        out_f, in_f = w.shape
+        dtype = dtype or w.dtype
+        self.r = r
+        self.scale = alpha / r
+        self.lora_a = nn.Parameter(torch.empty(r, in_f, dtype=dtype, device=w.device))
+        self.lora_b = nn.Parameter(torch.zeros(out_f, r, dtype=dtype, device=w.device))
+        nn.init.kaiming_uniform_(self.lora_a, a=math.sqrt(5))
+        base.weight.requires_grad_(False)
+        if getattr(base, "bias", None) is not None and isinstance(base.bias, nn.Parameter):
+            base.bias.requires_grad_(False)
+
+    def forward(self, x):
+        y = self.base(x)
+        adapter = torch.nn.functional.linear(
+            torch.nn.functional.linear(x, self.lora_a), self.lora_b)
+        return y + self.scale * adapter
+
+    @property
+    def weight(self):  # so init / inspection code keeps working
+        return self.base.weight
+
+
+def apply_lora(model: nn.Module, r: int = 16, alpha: int = 32,
+               targets=DEFAULT_TARGETS) -> list[str]:
+    """Freeze the model and wrap target linears with LoRA. Returns wrapped
+    module names."""
+    for p in model.parameters():
+        p.requires_grad_(False)
+    wrapped = []
+    for name, module in model.named_modules():
+        for child_name, child in list(module.named_children()):
+            if child_name in targets and isinstance(
+                    child, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):
+                setattr(module, child_name, LoRALinear(child, r=r, alpha=alpha))
+                wrapped.append(f"{name}.{child_name}" if name else child_name)
+    if not wrapped:
+        raise ValueError("apply_lora: no target modules found")
+    return wrapped
+
+
+def lora_state_dict(model: nn.Module) -> dict[str, torch.Tensor]:
+    return {k: v for k, v in model.state_dict().items()
+            if "lora_a" in k or "lora_b" in k}
+
+
+def merge_lora(model: nn.Module) -> None:
+    """Fold adapters into base weights (for serving a fine-tuned model)."""
+    for module in model.modules():
+        if isinstance(module, LoRALinear):
+            with torch.no_grad():
+                delta = (module.lora_b.float() @ module.lora_a.float())
+                module.base.weight.add_(delta.to(module.base.weight.dtype),
+                                        alpha=module.scale)
+                module.lora_b.zero_()

@@ -1,0 +1,117 @@
+"""The fine-tune loop: LoRA (or full) training with DP over RCCL.
+
+Native replacement for the reference's external HF trainer image
+(SURVEY.md §2b "trainer image"): params arrive via PARAM_* env /
+params.json (utils.params), data from /content/data, checkpoints to
+/content/artifacts.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass
+
+import torch
+
+from ..models import build_model
+from ..ops import FusedAdamW
+from ..parallel import DataParallel, comm
+from . import checkpoint as ckpt_mod
+from .data import SyntheticTokens, data_loader
+from .lora import apply_lora, lora_state_dict
+
+
+@dataclass
+class TrainConfig:
+    model: str = "llama2-7b"
+    seq_len: int = 512
+    micro_batch: int = 4
+    lr: float = 2e-4
+    weight_decay: float = 0.0
+    num_train_steps: int = 100
+    save_steps: int = 0                 # 0 = no checkpoints
+    lora_r: int = 16
+    lora_alpha: int = 32
+    full_finetune: bool = False
+    grad_clip: float = 1.0
+    dtype: str = "bfloat16"
+    seed: int = 0
+    output_dir: str = "/content/artifacts"
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig, device=None):
+        self.cfg = cfg
+        self.device = device if device is not None else (
+            f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
+        dtype = getattr(torch, cfg.dtype)
+        model = build_model(cfg.model, dtype=dtype, tp=1, seed=cfg.seed)
+        if not cfg.full_finetune:
+            apply_lora(model, r=cfg.lora_r, alpha=cfg.lora_alpha)
+        model = model.to(self.device)
+        self.ddp = DataParallel(model)
+        self.model = model
+        trainable = [p for p in model.parameters() if p.requires_grad]
+        self.optimizer = FusedAdamW(trainable, lr=cfg.lr,
+                                    weight_decay=cfg.weight_decay)
+        self.step_num = 0
+
+    # -- core step -------------------------------------------------------------
+    def train_step(self, tokens: torch.Tensor) -> float:
+        """tokens [B, S+1] (inputs + shifted labels). Returns loss."""
+        inputs = tokens[:, :-1].to(self.device)
+        labels = tokens[:, 1:].to(self.device)
+        logits = self.ddp(inputs)
+        loss = torch.nn.functional.cross_entropy(
+            logits.float().view(-1, logits.shape[-1]), labels.reshape(-1))
+        loss.backward()
+        self.ddp.finish_backward()
+        if self.cfg.grad_clip > 0:
+            torch.nn.utils.clip_grad_norm_(
+                [p for p in self.model.parameters() if p.requires_grad],
+                self.cfg.grad_clip)
+        self.optimizer.step()
+        self.ddp.zero_grad()
+        self.step_num += 1
+        return float(loss.detach())
+
+    # -- full loop -------------------------------------------------------------
+    def fit(self, dataset=None, log_every: int = 10):
+        cfg = self.cfg
+        if dataset is None:
+            dataset = SyntheticTokens(self.model.cfg.vocab_size, cfg.seq_len + 1)
+        loader = data_loader(dataset, cfg.micro_batch, rank=comm.rank(),
+                             world=comm.world_size(), seed=cfg.seed)
+        it = iter(loader)
+        t0 = time.time()
+        while self.step_num < cfg.num_train_steps:
+            try:
+                batch = next(it)
+            except StopIteration:
+                it = iter(loader)
+                batch = next(it)
+            loss = self.train_step(batch)
+            if cfg.save_steps and self.step_num % cfg.save_steps == 0:
+                self.save()
+            if comm.rank() == 0 and self.step_num % log_every == 0:
+                dt = time.time() - t0
+                sps = self.step_num * cfg.micro_batch * comm.world_size() / max(dt, 1e-9)
+                print(f"step {self.step_num} loss {loss:.4f} "
+                      f"samples/sec {sps:.2f}", flush=True)
+        if cfg.save_steps:
+            self.save()
+
+    def save(self):
+        if comm.rank() != 0:
+            return
+        state = (self.model.state_dict() if self.cfg.full_finetune
+                 else lora_state_dict(self.model))
+        ckpt_mod.save_checkpoint(self.cfg.output_dir, self.step_num, state)
+
+    def resume(self) -> bool:
+        latest = ckpt_mod.latest_checkpoint(self.cfg.output_dir)
+        if latest is None:
+            return False
+        state, _, step = ckpt_mod.load_checkpoint(latest)
+        self.model.load_state_dict(state, strict=False)
+        self.step_num = step
+        return True

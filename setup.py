@@ -1,0 +1,35 @@
+"""In-tree build of the runbooks_amd gfx950 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built `.so` lands next to the Python package (runbooks_amd/ops/) so it
+travels with repo snapshots; there is no JIT cache dependency.
+"""
+import os
+from pathlib import Path
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils import cpp_extension  # noqa: E402
+
+ROOT = Path(__file__).parent
+CSRC = ROOT / "runbooks_amd" / "ops" / "csrc"
+
+sources = sorted(str(p) for p in CSRC.glob("*.hip")) + [str(CSRC / "bindings.cpp")]
+
+setup(
+    name="runbooks_amd_kernels",
+    ext_modules=[
+        cpp_extension.CUDAExtension(
+            name="runbooks_amd.ops._hip",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": cpp_extension.BuildExtension.with_options(use_ninja=True)},
+)

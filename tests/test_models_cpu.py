@@ -1,0 +1,92 @@
+"""Model-family + engine + LoRA behavior on CPU (fp32 references)."""
+import pytest
+import torch
+
+from runbooks_amd.models import build_model, get_config, list_configs
+from runbooks_amd.serve import Engine
+from runbooks_amd.train import apply_lora, lora_state_dict, merge_lora
+
+
+@pytest.mark.parametrize("name", ["tiny-llama", "tiny-falcon", "tiny-opt"])
+def test_forward_shapes(name):
+    m = build_model(name, dtype=torch.float32)
+    out = m(torch.randint(0, 256, (2, 16)))
+    assert out.shape == (2, 16, m.cfg.vocab_size)
+    assert torch.isfinite(out).all()
+
+
+def test_registry_has_flagship_configs():
+    for name in ("llama2-7b", "llama2-70b", "falcon-40b", "falcon-7b", "opt-125m"):
+        cfg = get_config(name)
+        assert cfg.num_heads % cfg.num_kv_heads == 0
+    # llama2-7b parameter count sanity (±10%)
+    assert abs(get_config("llama2-7b").params_b - 6.7) < 0.7
+    assert abs(get_config("llama2-70b").params_b - 69) < 7
+    assert abs(get_config("opt-125m").params_b - 0.125) < 0.05
+
+
+@pytest.mark.parametrize("name", ["tiny-llama", "tiny-falcon", "tiny-opt"])
+def test_engine_matches_full_forward(name):
+    """Paged prefill+decode must equal the no-cache forward (greedy)."""
+    m = build_model(name, dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    prompt = [5, 9, 2, 7, 1]
+    out = eng.generate(list(prompt), max_new_tokens=6)
+    seq = list(prompt)
+    for _ in range(6):
+        logits = m(torch.tensor([seq]))
+        seq.append(int(logits[0, -1].argmax()))
+    assert out == seq[len(prompt):], f"{name}: {out} != {seq[len(prompt):]}"
+
+
+def test_engine_batched_decode_matches_serial():
+    m = build_model("tiny-llama", dtype=torch.float32)
+    prompts = [[1, 2, 3], [9, 8, 7, 6], [4, 4]]
+    serial = []
+    for p in prompts:
+        eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+        serial.append(eng.generate(list(p), max_new_tokens=4))
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, max_batch=4)
+    reqs = [eng.submit(list(p), max_new_tokens=4) for p in prompts]
+    while eng.has_work():
+        eng.step()
+    for r, s in zip(reqs, serial):
+        assert r.output_ids == s
+
+
+def test_lora_freezes_base_and_merges():
+    torch.manual_seed(0)
+    m = build_model("tiny-llama", dtype=torch.float32)
+    wrapped = apply_lora(m, r=4, alpha=8)
+    assert len(wrapped) > 0
+    trainable = [n for n, p in m.named_parameters() if p.requires_grad]
+    assert all("lora_" in n for n in trainable)
+    assert len(lora_state_dict(m)) == 2 * len(wrapped)
+
+    x = torch.randint(0, 256, (1, 8))
+    before = m(x)
+    # train the adapters a little so B != 0
+    opt = torch.optim.SGD([p for p in m.parameters() if p.requires_grad], lr=0.1)
+    loss = m(x).square().mean()
+    loss.backward()
+    opt.step()
+    after_train = m(x)
+    assert not torch.allclose(before, after_train)
+    merge_lora(m)
+    merged = m(x)
+    assert torch.allclose(after_train, merged, atol=1e-4)
+
+
+def test_kv_cache_exhaustion_raises():
+    m = build_model("tiny-llama", dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=2)
+    with pytest.raises(RuntimeError):
+        eng.generate(list(range(40)), max_new_tokens=4)
+
+
+def test_block_allocator_release():
+    m = build_model("tiny-llama", dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    free0 = len(eng.allocator.free)
+    eng.generate([1, 2, 3], max_new_tokens=3)
+    assert len(eng.allocator.free) == free0, "blocks leaked"

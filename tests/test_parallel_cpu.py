@@ -1,0 +1,154 @@
+"""Multi-process CPU tests (gloo, world_size 2) for DP and TP — the
+distributed paths that run over RCCL/xGMI on the GPU node."""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+PORT = 29781
+
+
+def _run_dist(fn, world=2, port=PORT):
+    ctx = mp.get_context("spawn")
+    procs = []
+    for r in range(world):
+        p = ctx.Process(target=_entry, args=(fn.__name__, r, world, port))
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(180)
+    for p in procs:
+        assert p.exitcode == 0, f"worker failed with {p.exitcode}"
+
+
+def _entry(fn_name, rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        globals()[fn_name](rank, world)
+    finally:
+        dist.destroy_process_group()
+
+
+# --- worker bodies ----------------------------------------------------------
+
+def _dp_grads_average(rank, world):
+    from runbooks_amd.models import build_model
+    from runbooks_amd.parallel import DataParallel
+
+    torch.manual_seed(0)
+    model = build_model("tiny-llama", dtype=torch.float32, tp=1)
+    ddp = DataParallel(model, bucket_mb=1)
+    torch.manual_seed(100 + rank)
+    x = torch.randint(0, 256, (2, 9))
+    logits = ddp(x[:, :-1])
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, logits.shape[-1]), x[:, 1:].reshape(-1))
+    loss.backward()
+    ddp.finish_backward()
+
+    # compute expected: average of per-rank grads on a fresh replica
+    ref = build_model("tiny-llama", dtype=torch.float32, tp=1)
+    ref.load_state_dict(model.state_dict())
+    grads = {}
+    for r in range(world):
+        ref.zero_grad()
+        torch.manual_seed(100 + r)
+        xr = torch.randint(0, 256, (2, 9))
+        lg = ref(xr[:, :-1])
+        ls = torch.nn.functional.cross_entropy(
+            lg.reshape(-1, lg.shape[-1]), xr[:, 1:].reshape(-1))
+        ls.backward()
+        for n, p in ref.named_parameters():
+            grads[n] = grads.get(n, 0) + p.grad / world
+    for n, p in model.named_parameters():
+        assert torch.allclose(p.grad, grads[n], atol=1e-5), n
+
+
+def _tp_linear_matches_dense(rank, world):
+    from runbooks_amd.parallel.tp import ColumnParallelLinear, RowParallelLinear
+
+    torch.manual_seed(0)
+    IN, HID, OUT, B = 16, 32, 12, 5
+    w1 = torch.randn(HID, IN)
+    w2 = torch.randn(OUT, HID)
+    x = torch.randn(B, IN)
+
+    col = ColumnParallelLinear(IN, HID, tp_size=world)
+    row = RowParallelLinear(HID, OUT, tp_size=world)
+    shard = HID // world
+    with torch.no_grad():
+        col.weight.copy_(w1[rank * shard:(rank + 1) * shard])
+        row.weight.copy_(w2[:, rank * shard:(rank + 1) * shard])
+    y = row(torch.relu(col(x)))
+    expected = torch.relu(x @ w1.T) @ w2.T
+    assert torch.allclose(y, expected, atol=1e-4), (y - expected).abs().max()
+
+
+def _tp_model_matches_single(rank, world):
+    from runbooks_amd.models import build_model
+
+    torch.manual_seed(0)
+    tp_model = build_model("tiny-llama", dtype=torch.float32, tp=world, seed=3)
+    single = build_model("tiny-llama", dtype=torch.float32, tp=1, seed=3)
+    # shard the single model's weights into this rank's tp model
+    sd = single.state_dict()
+    tsd = tp_model.state_dict()
+    for name, t in tsd.items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:      # column shard
+            s = t.shape[0]
+            t.copy_(full[rank * s:(rank + 1) * s])
+        elif t.shape[1] * world == full.shape[1]:      # row shard
+            s = t.shape[1]
+            t.copy_(full[:, rank * s:(rank + 1) * s])
+        else:
+            raise AssertionError(f"unexpected shape {name}: {t.shape} vs {full.shape}")
+    x = torch.randint(0, 256, (2, 10))
+    y_tp = tp_model(x)
+    y_single = single(x)
+    assert torch.allclose(y_tp, y_single, atol=1e-4), (y_tp - y_single).abs().max()
+
+
+def _bucket_overlap_many_buckets(rank, world):
+    from runbooks_amd.parallel.ddp import DataParallel
+
+    torch.manual_seed(1)
+    m = torch.nn.Sequential(*[torch.nn.Linear(64, 64) for _ in range(8)])
+    ddp = DataParallel(m, bucket_mb=0)  # ~1 bucket per param
+    assert len(ddp._buckets) >= 8
+    x = torch.randn(4, 64)
+    ddp(x).sum().backward()
+    ddp.finish_backward()
+    g = m[0].weight.grad.clone()
+    # all ranks saw the same input (same seed) -> grad equals local grad
+    m2 = torch.nn.Sequential(*[torch.nn.Linear(64, 64) for _ in range(8)])
+    torch.manual_seed(1)
+    m3 = torch.nn.Sequential(*[torch.nn.Linear(64, 64) for _ in range(8)])
+    m3(x).sum().backward()
+    assert torch.allclose(g, m3[0].weight.grad, atol=1e-6)
+
+
+# --- test entries -----------------------------------------------------------
+
+def test_dp_gradient_allreduce():
+    _run_dist(_dp_grads_average, port=PORT)
+
+
+def test_tp_linear():
+    _run_dist(_tp_linear_matches_dense, port=PORT + 1)
+
+
+def test_tp_model_full():
+    _run_dist(_tp_model_matches_single, port=PORT + 2)
+
+
+def test_ddp_many_buckets():
+    _run_dist(_bucket_overlap_many_buckets, port=PORT + 3)
