@@ -45,9 +45,11 @@ def _max_over_ranks(x: float) -> float:
 
 
 def bench_train(args) -> dict:
+    from runbooks_amd.models import get_config
     from runbooks_amd.train import TrainConfig, Trainer
     from runbooks_amd.train.data import SyntheticTokens
 
+    args.seq_len = min(args.seq_len, get_config(args.model).max_seq_len - 1)
     cfg = TrainConfig(model=args.model, seq_len=args.seq_len,
                       micro_batch=args.micro_batch,
                       num_train_steps=args.warmup + args.steps,
