@@ -66,12 +66,8 @@ def test_rope_fwd_bwd(dtype):
     assert torch.allclose(x.grad.cpu().float(), ref_grad, atol=tol, rtol=tol)
 
 
-@pytest.mark.parametrize("pdtype,gdtype", [
-    (torch.float32, torch.float32),
-    (torch.float32, torch.bfloat16),
-    (torch.bfloat16, torch.bfloat16),
-])
-def test_fused_adamw(pdtype, gdtype):
+@pytest.mark.parametrize("pdtype", [torch.float32, torch.bfloat16])
+def test_fused_adamw_optimizer(pdtype):
     _assert_hip()
     torch.manual_seed(0)
     n = 4097
@@ -80,7 +76,7 @@ def test_fused_adamw(pdtype, gdtype):
     o_gpu = ops.FusedAdamW([p_gpu], lr=1e-2, weight_decay=0.1)
     o_cpu = ops.FusedAdamW([p_cpu], lr=1e-2, weight_decay=0.1)
     for i in range(4):
-        g = torch.randn(n, dtype=gdtype)
+        g = torch.randn(n, dtype=pdtype)
         p_gpu.grad = g.to(DEV)
         p_cpu.grad = g.clone()
         o_gpu.step()
@@ -88,6 +84,27 @@ def test_fused_adamw(pdtype, gdtype):
     tol = 3e-2 if pdtype == torch.bfloat16 else 1e-5
     assert torch.allclose(p_gpu.detach().cpu().float(), p_cpu.detach().float(),
                           atol=tol, rtol=tol)
+
+
+def test_fused_adamw_kernel_mixed_dtypes():
+    """fp32 master params + bf16 grads straight through the kernel."""
+    _assert_hip()
+    torch.manual_seed(0)
+    n = 2049
+    p = torch.randn(n, dtype=torch.float32, device=DEV)
+    g = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+    m = torch.zeros(n, dtype=torch.float32, device=DEV)
+    v = torch.zeros(n, dtype=torch.float32, device=DEV)
+    p_ref, m_ref, v_ref = p.cpu().clone(), m.cpu().clone(), v.cpu().clone()
+    for step in (1, 2):
+        ops.ext().adamw_step(p, g, m, v, 1e-2, 0.9, 0.999, 1e-8, 0.1, step)
+        # CPU reference math
+        gf = g.cpu().float()
+        m_ref.mul_(0.9).add_(gf, alpha=0.1)
+        v_ref.mul_(0.999).addcmul_(gf, gf, value=0.001)
+        bc1, bc2 = 1 - 0.9 ** step, 1 - 0.999 ** step
+        p_ref -= 1e-2 * ((m_ref / bc1) / ((v_ref / bc2).sqrt() + 1e-8) + 0.1 * p_ref)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5, rtol=1e-5)
 
 
 def test_sample_greedy_matches_argmax():
