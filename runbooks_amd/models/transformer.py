@@ -93,8 +93,8 @@ class Attention(nn.Module):
         q, k, v = self._qkv(x, cos, sin, positions)
         ops.kv_append(k, v, kc, vc, slot_mapping)
         q = q.view(B, S, self.hq, self.dh)
-        o = ops.causal_attention(q, k.view(B, S, self.hkv, self.dh),
-                                 v.view(B, S, self.hkv, self.dh), scale=self.scale)
+        o = ops.flash_prefill(q, k.view(B, S, self.hkv, self.dh),
+                              v.view(B, S, self.hkv, self.dh), scale=self.scale)
         return self.o_proj(o.reshape(B * S, self.hq * self.dh))
 
     def forward_decode(self, x, cos, sin, positions, kc, vc, slot_mapping,

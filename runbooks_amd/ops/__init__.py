@@ -6,7 +6,7 @@ built extension raises — there is no silent eager fallback on the GPU.
 """
 from ._backend import ext, has_hip, use_hip  # noqa: F401
 from .adamw import FusedAdamW  # noqa: F401
-from .attention import causal_attention, paged_decode, paged_decode_ref  # noqa: F401
+from .attention import causal_attention, flash_prefill, paged_decode, paged_decode_ref  # noqa: F401
 from .kvcache import BLOCK_SIZE, alloc_kv_cache, kv_append, kv_append_ref  # noqa: F401
 from .norm import RMSNorm, rmsnorm, rmsnorm_ref  # noqa: F401
 from .rope import rope, rope_ref, rope_tables  # noqa: F401

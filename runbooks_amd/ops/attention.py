@@ -51,6 +51,22 @@ def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return torch.cat(outs, dim=2).transpose(1, 2).contiguous()
 
 
+def flash_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                  scale: float | None = None) -> torch.Tensor:
+    """Causal GQA attention, inference-only (no autograd).
+
+    q [B, S, Hq, Dh] bf16 on GPU -> gfx950 MFMA flash kernel
+    (csrc/attention_prefill.hip); CPU falls back to the chunked reference.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _backend.use_hip(q) and q.shape[-1] in (64, 128) \
+            and q.dtype == torch.bfloat16:
+        return _backend.ext().flash_prefill(
+            q.contiguous(), k.contiguous(), v.contiguous(), float(scale))
+    return causal_attention(q, k, v, scale=scale)
+
+
 def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):
     """fp32 reference decode: q [B, Hq, Dh]."""
     B, Hq, Dh = q.shape

@@ -18,6 +18,8 @@ void kv_append(at::Tensor k, at::Tensor v, at::Tensor k_cache, at::Tensor v_cach
 at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                         at::Tensor block_tables, at::Tensor seq_lens,
                         int64_t nsplit, double scale);
+at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v, double scale);
+at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "runbooks_amd gfx950 HIP kernels";
@@ -29,4 +31,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_tokens", &sample_tokens, "greedy / Gumbel-max sampling");
   m.def("kv_append", &kv_append, "paged KV-cache append");
   m.def("paged_decode", &paged_decode, "paged GQA/MQA decode attention");
+  m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill (causal)");
+  m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
+        "layout probe: one 32x32x16 bf16 MFMA as a plain matmul");
 }

@@ -199,3 +199,35 @@ def test_train_step_gpu():
     losses = [tr.train_step(batch) for _ in range(8)]
     assert losses[-1] < losses[0], losses
     assert all(math.isfinite(x) for x in losses)
+
+
+def test_mfma_32x32x16_layout_probe():
+    """One MFMA vs torch matmul — pinpoints a wrong fragment map."""
+    _assert_hip()
+    torch.manual_seed(0)
+    # asymmetric inputs (guide: symmetric B passes transposed layouts)
+    a = torch.randn(32, 16, dtype=torch.bfloat16, device=DEV)
+    b = torch.arange(16 * 32, dtype=torch.float32, device=DEV).reshape(16, 32)
+    b = ((b % 13) / 6.0 - 1.0).to(torch.bfloat16)
+    c = ops.ext().mfma_probe_32x32x16(a, b)
+    ref = a.float() @ b.float()
+    assert torch.allclose(c, ref, atol=2e-2, rtol=2e-2), \
+        (c - ref).abs().max()
+
+
+@pytest.mark.parametrize("B,S,Hq,Hkv,dh", [
+    (1, 64, 4, 4, 128),      # tiny, single WG tile
+    (2, 333, 8, 2, 128),     # ragged S, GQA
+    (1, 511, 16, 1, 64),     # MQA falcon-style, Dh=64
+    (2, 1024, 8, 8, 128),    # multi-tile
+])
+def test_flash_prefill(B, S, Hq, Hkv, dh):
+    _assert_hip()
+    torch.manual_seed(B * 1000 + S)
+    q = torch.randn(B, S, Hq, dh, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(B, S, Hkv, dh, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(B, S, Hkv, dh, dtype=torch.bfloat16, device=DEV)
+    out = ops.flash_prefill(q, k, v)
+    ref = ops.causal_attention(q.cpu().float(), k.cpu().float(), v.cpu().float())
+    diff = (out.cpu().float() - ref).abs().max()
+    assert diff < 3e-2, f"max diff {diff}"
