@@ -124,8 +124,7 @@ class MLP(nn.Module):
 
     def forward(self, x):
         if self.act == "silu_glu":
-            return self.down_proj(torch.nn.functional.silu(self.gate_proj(x))
-                                  * self.up_proj(x))
+            return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
         h = self.fc1(x)
         h = torch.nn.functional.gelu(h) if self.act == "gelu" else torch.relu(h)
         return self.down_proj(h)

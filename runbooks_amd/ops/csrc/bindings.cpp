@@ -19,6 +19,12 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                         at::Tensor block_tables, at::Tensor seq_lens,
                         int64_t nsplit, double scale);
 at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v, double scale);
+at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u);
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor g, at::Tensor u);
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
+                                          int64_t ignore_index);
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                             double gscale, int64_t ignore_index);
 at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -32,6 +38,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_append", &kv_append, "paged KV-cache append");
   m.def("paged_decode", &paged_decode, "paged GQA/MQA decode attention");
   m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill (causal)");
+  m.def("swiglu_fwd", &swiglu_fwd, "fused silu(g)*u");
+  m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE: per-row loss + lse");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward -> dlogits");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "layout probe: one 32x32x16 bf16 MFMA as a plain matmul");
 }

@@ -13,6 +13,7 @@ from dataclasses import dataclass
 import torch
 
 from ..models import build_model
+from .. import ops
 from ..ops import FusedAdamW
 from ..parallel import DataParallel, comm
 from . import checkpoint as ckpt_mod
@@ -61,8 +62,7 @@ class Trainer:
         inputs = tokens[:, :-1].to(self.device)
         labels = tokens[:, 1:].to(self.device)
         logits = self.ddp(inputs)
-        loss = torch.nn.functional.cross_entropy(
-            logits.float().view(-1, logits.shape[-1]), labels.reshape(-1))
+        loss = ops.cross_entropy(logits, labels)
         loss.backward()
         self.ddp.finish_backward()
         if self.cfg.grad_clip > 0:

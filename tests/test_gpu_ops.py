@@ -231,3 +231,42 @@ def test_flash_prefill(B, S, Hq, Hkv, dh):
     ref = ops.causal_attention(q.cpu().float(), k.cpu().float(), v.cpu().float())
     diff = (out.cpu().float() - ref).abs().max()
     assert diff < 3e-2, f"max diff {diff}"
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_swiglu_fwd_bwd(dtype):
+    _assert_hip()
+    torch.manual_seed(0)
+    g = torch.randn(64, 1024, dtype=dtype, device=DEV, requires_grad=True)
+    u = torch.randn(64, 1024, dtype=dtype, device=DEV, requires_grad=True)
+    y = ops.swiglu(g, u)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    gc = g.detach().cpu().float().requires_grad_(True)
+    uc = u.detach().cpu().float().requires_grad_(True)
+    yc = torch.nn.functional.silu(gc) * uc
+    yc.backward(dy.cpu().float())
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    assert torch.allclose(y.detach().cpu().float(), yc.detach(), atol=tol, rtol=tol)
+    assert torch.allclose(g.grad.cpu().float(), gc.grad, atol=tol, rtol=tol)
+    assert torch.allclose(u.grad.cpu().float(), uc.grad, atol=tol, rtol=tol)
+
+
+@pytest.mark.parametrize("V", [32000, 50272])
+def test_cross_entropy_fwd_bwd(V):
+    _assert_hip()
+    torch.manual_seed(0)
+    N = 37
+    logits = torch.randn(N, V, dtype=torch.bfloat16, device=DEV,
+                         requires_grad=True)
+    targets = torch.randint(0, V, (N,), device=DEV)
+    targets[3] = -100  # ignore_index
+    loss = ops.cross_entropy(logits, targets)
+    loss.backward()
+    lc = logits.detach().cpu().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lc, targets.cpu().long(),
+                                            ignore_index=-100)
+    ref.backward()
+    assert abs(float(loss) - float(ref)) < 2e-2, (float(loss), float(ref))
+    assert torch.allclose(logits.grad.cpu().float(), lc.grad, atol=2e-3,
+                          rtol=2e-2), (logits.grad.cpu().float() - lc.grad).abs().max()
