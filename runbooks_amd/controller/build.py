@@ -89,7 +89,8 @@ class BuildReconciler:
             return Result()  # Job watch requeues
 
         obj.set_image(image)
-        self.kube.update(obj.to_dict())
+        self.kube.patch("substratus.ai/v1", obj.kind, obj.namespace,
+                        obj.name, {"spec": {"image": image}})
         obj.set_condition(cond.CONDITION_BUILT, True,
                           cond.REASON_JOB_COMPLETE,
                           f"Builder Job completed: {job['metadata']['name']}",

@@ -58,6 +58,7 @@ def serve(servicer: ControllerServicer, address: str = "0.0.0.0:10080",
                                                      server)
     except ImportError:
         pass  # health service optional when grpcio-health-checking is absent
-    server.add_insecure_port(address)
+    port = server.add_insecure_port(address)
     server.start()
+    server.bound_port = port  # ephemeral binds (":0") report the real port
     return server
