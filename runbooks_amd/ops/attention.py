@@ -20,12 +20,56 @@ import torch
 from . import _backend
 
 
+class _FlashAttnTrain(torch.autograd.Function):
+    """Differentiable causal GQA attention on the gfx950 MFMA kernels.
+
+    Forward: flash_fwd_train (attention_prefill.hip, saves LSE).
+    Backward: fa_bwd (attention_bwd.hip, FA2-style recompute); GQA dk/dv
+    come back per-q-head and are group-summed to the kv heads here.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+        out, lse = _backend.ext().flash_fwd_train(qc, kc, vc, float(scale))
+        ctx.save_for_backward(qc, kc, vc, out, lse)
+        ctx.scale = float(scale)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        dq, dk, dv = _backend.ext().fa_bwd(dout.contiguous(), q, k, v, out,
+                                           lse, ctx.scale)
+        hq, hkv = q.shape[2], k.shape[2]
+        if hkv != hq:  # sum the q-head group (GQA/MQA)
+            g = hq // hkv
+            b, s = dk.shape[0], dk.shape[1]
+            dk = dk.view(b, s, hkv, g, -1).sum(dim=3)
+            dv = dv.view(b, s, hkv, g, -1).sum(dim=3)
+        return dq, dk, dv, None
+
+
 def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                      scale: float | None = None, q_block: int = 256) -> torch.Tensor:
     """q [B, S, Hq, Dh]; k, v [B, S, Hkv, Dh] -> [B, S, Hq, Dh].
 
-    Differentiable; softmax in fp32.
+    Differentiable. GPU bf16 (Dh 64/128) runs the MFMA flash kernels;
+    otherwise the chunked fp32 reference below (also the numerics oracle
+    for the GPU test).
     """
+    if _backend.use_hip(q) and q.shape[-1] in (64, 128) \
+            and q.dtype == torch.bfloat16:
+        if scale is None:
+            scale = 1.0 / math.sqrt(q.shape[-1])
+        return _FlashAttnTrain.apply(q, k, v, scale)
+    return causal_attention_ref(q, k, v, scale=scale, q_block=q_block)
+
+
+def causal_attention_ref(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                         scale: float | None = None,
+                         q_block: int = 256) -> torch.Tensor:
+    """Chunked eager reference (fp32 softmax), differentiable."""
     B, S, Hq, Dh = q.shape
     Hkv = k.shape[2]
     if scale is None:
@@ -64,7 +108,7 @@ def flash_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
             and q.dtype == torch.bfloat16:
         return _backend.ext().flash_prefill(
             q.contiguous(), k.contiguous(), v.contiguous(), float(scale))
-    return causal_attention(q, k, v, scale=scale)
+    return causal_attention_ref(q, k, v, scale=scale)
 
 
 def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):

@@ -49,6 +49,7 @@ template <int DH>
 __global__ __launch_bounds__(BLOCK, 2) void flash_prefill_kernel(
     const uint16_t *__restrict__ qp, const uint16_t *__restrict__ kp,
     const uint16_t *__restrict__ vp, uint16_t *__restrict__ op,
+    float *__restrict__ lsep,  // [B, Hq, S] log-sum-exp for training bwd
     int B, int S, int Hq, int Hkv, float scale) {
   constexpr int KSTEPS = DH / 16;       // QK^T contraction steps
   constexpr int DTILES = DH / 32;       // O column tiles
@@ -216,6 +217,9 @@ __global__ __launch_bounds__(BLOCK, 2) void flash_prefill_kernel(
   if (q_valid) {
     const float l_full = l_run + __shfl_xor(l_run, 32, 64);
     const float inv_l = (l_full > 0.0f) ? 1.0f / l_full : 0.0f;
+    if (lsep != nullptr && hi == 0)
+      lsep[((int64_t)b * Hq + h) * S + qrow] =
+          (l_full > 0.0f) ? m_run + __logf(l_full) : -INFINITY;
     uint16_t *orow = op + ((int64_t)(b * S + qrow) * Hq + h) * DH;
 #pragma unroll
     for (int dt = 0; dt < DTILES; ++dt) {
@@ -235,7 +239,8 @@ __global__ __launch_bounds__(BLOCK, 2) void flash_prefill_kernel(
 
 }  // namespace
 
-at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v, double scale) {
+static at::Tensor flash_fwd_impl(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 double scale, float *lsep) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
                   v.is_contiguous(), "flash_prefill: contiguous GPU tensors");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "flash_prefill: bf16 only");
@@ -254,18 +259,33 @@ at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v, double scale)
                        stream, (const uint16_t *)q.data_ptr(),
                        (const uint16_t *)k.data_ptr(),
                        (const uint16_t *)v.data_ptr(), (uint16_t *)out.data_ptr(),
-                       B, S, Hq, Hkv, (float)scale);
+                       lsep, B, S, Hq, Hkv, (float)scale);
   } else if (DH == 64) {
     constexpr size_t shmem = KVB * (64 * 2 + 16) + 64 * (KVB * 2 + 16);
     hipLaunchKernelGGL((flash_prefill_kernel<64>), grid, dim3(BLOCK), shmem,
                        stream, (const uint16_t *)q.data_ptr(),
                        (const uint16_t *)k.data_ptr(),
                        (const uint16_t *)v.data_ptr(), (uint16_t *)out.data_ptr(),
-                       B, S, Hq, Hkv, (float)scale);
+                       lsep, B, S, Hq, Hkv, (float)scale);
   } else {
     TORCH_CHECK(false, "flash_prefill: DH must be 64 or 128, got ", DH);
   }
   return out;
+}
+
+at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v,
+                         double scale) {
+  return flash_fwd_impl(q, k, v, scale, nullptr);
+}
+
+// Training forward: additionally returns the per-row log-sum-exp
+// ([B, Hq, S] f32) consumed by fa_bwd (attention_bwd.hip).
+std::vector<at::Tensor> flash_fwd_train(at::Tensor q, at::Tensor k,
+                                        at::Tensor v, double scale) {
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  auto out = flash_fwd_impl(q, k, v, scale, lse.data_ptr<float>());
+  return {out, lse};
 }
 
 // ---------------------------------------------------------------------------

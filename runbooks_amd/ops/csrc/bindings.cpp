@@ -19,6 +19,11 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                         at::Tensor block_tables, at::Tensor seq_lens,
                         int64_t nsplit, double scale);
 at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v, double scale);
+std::vector<at::Tensor> flash_fwd_train(at::Tensor q, at::Tensor k, at::Tensor v,
+                                        double scale);
+std::vector<at::Tensor> fa_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                               at::Tensor v, at::Tensor out, at::Tensor lse,
+                               double scale);
 at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u);
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor g, at::Tensor u);
 std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
@@ -38,6 +43,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_append", &kv_append, "paged KV-cache append");
   m.def("paged_decode", &paged_decode, "paged GQA/MQA decode attention");
   m.def("flash_prefill", &flash_prefill, "MFMA flash-attention prefill (causal)");
+  m.def("flash_fwd_train", &flash_fwd_train,
+        "MFMA flash-attention forward + log-sum-exp (training)");
+  m.def("fa_bwd", &fa_bwd,
+        "MFMA flash-attention backward -> (dq, dk_perq, dv_perq)");
   m.def("swiglu_fwd", &swiglu_fwd, "fused silu(g)*u");
   m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE: per-row loss + lse");

@@ -233,6 +233,42 @@ def test_flash_prefill(B, S, Hq, Hkv, dh):
     assert diff < 3e-2, f"max diff {diff}"
 
 
+@pytest.mark.parametrize("B,S,Hq,Hkv,dh", [
+    (1, 64, 4, 4, 128),      # tiny, single WG tile
+    (2, 333, 8, 2, 128),     # ragged S, GQA (group-summed dk/dv)
+    (1, 200, 8, 1, 64),      # MQA, Dh=64
+    (1, 576, 4, 4, 128),     # multi-WG, MHA (llama2-7b shape class)
+])
+def test_flash_attention_train_bwd(B, S, Hq, Hkv, dh):
+    """fa_bwd (attention_bwd.hip) vs torch autograd through the fp32
+    reference attention."""
+    _assert_hip()
+    torch.manual_seed(B * 77 + S)
+    q = torch.randn(B, S, Hq, dh, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    k = torch.randn(B, S, Hkv, dh, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    v = torch.randn(B, S, Hkv, dh, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    out = ops.causal_attention(q, k, v)   # MFMA autograd path on GPU
+    dy = torch.randn_like(out)
+    out.backward(dy)
+
+    qc = q.detach().cpu().float().requires_grad_(True)
+    kc = k.detach().cpu().float().requires_grad_(True)
+    vc = v.detach().cpu().float().requires_grad_(True)
+    ref = ops.causal_attention(qc, kc, vc)
+    ref.backward(dy.cpu().float())
+
+    assert (out.detach().cpu().float() - ref.detach()).abs().max() < 3e-2
+    for got, want, name in ((q.grad, qc.grad, "dq"), (k.grad, kc.grad, "dk"),
+                            (v.grad, vc.grad, "dv")):
+        g = got.cpu().float()
+        d = (g - want).abs().max().item()
+        rel = d / max(want.abs().max().item(), 1e-6)
+        assert rel < 4e-2, f"{name}: max abs diff {d} rel {rel}"
+
+
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
 def test_swiglu_fwd_bwd(dtype):
     _assert_hip()
