@@ -98,10 +98,11 @@ class Attention(nn.Module):
         return self.o_proj(o.reshape(B * S, self.hq * self.dh))
 
     def forward_decode(self, x, cos, sin, positions, kc, vc, slot_mapping,
-                       block_tables, seq_lens):
+                       block_tables, seq_lens, nsplit=None):
         q, k, v = self._qkv(x, cos, sin, positions)
         ops.kv_append(k, v, kc, vc, slot_mapping)
-        o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale)
+        o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale,
+                             nsplit=nsplit)
         B = x.shape[0]
         return self.o_proj(o.reshape(B, self.hq * self.dh))
 
@@ -219,14 +220,17 @@ class Transformer(nn.Module):
         return self.lm_head(x)
 
     @torch.no_grad()
-    def decode(self, tokens, positions, caches, slot_mapping, block_tables, seq_lens):
-        """tokens [B] (one new token per sequence) -> logits [B, V]."""
+    def decode(self, tokens, positions, caches, slot_mapping, block_tables,
+               seq_lens, nsplit=None):
+        """tokens [B] (one new token per sequence) -> logits [B, V].
+        ``nsplit`` fixes the paged-decode work split (required under
+        hipGraph capture, serve/graph.py)."""
         B = tokens.shape[0]
         x = self._embed(tokens.view(B, 1), positions)
         for blk, (kc, vc) in zip(self.blocks, caches):
             fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
                 h, self.rope_cos, self.rope_sin, positions, kc, vc, slot_mapping,
-                block_tables, seq_lens)
+                block_tables, seq_lens, nsplit=nsplit)
             x = blk(x, fn)
         x = self.norm_f(x)
         return self.lm_head(x)

@@ -12,6 +12,10 @@ at::Tensor rope_bwd(at::Tensor dy, at::Tensor cos, at::Tensor sin,
 void adamw_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                 double lr, double beta1, double beta2, double eps, double wd,
                 int64_t step);
+void adamw_step_multi(at::Tensor table, int64_t nchunks, bool p32, bool g32,
+                      double lr, double beta1, double beta2, double eps,
+                      double wd, int64_t step);
+int64_t adamw_mt_chunk_elems();
 at::Tensor sample_tokens(at::Tensor logits, double temperature, int64_t seed);
 void kv_append(at::Tensor k, at::Tensor v, at::Tensor k_cache, at::Tensor v_cache,
                at::Tensor slot_mapping);
@@ -39,6 +43,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd, "RoPE rotate-half forward");
   m.def("rope_bwd", &rope_bwd, "RoPE rotate-half backward (inverse rotation)");
   m.def("adamw_step", &adamw_step, "fused AdamW single-tensor step");
+  m.def("adamw_step_multi", &adamw_step_multi,
+        "fused AdamW multi-tensor step over a cached chunk table");
+  m.def("adamw_mt_chunk_elems", &adamw_mt_chunk_elems,
+        "elements per multi-tensor chunk");
   m.def("sample_tokens", &sample_tokens, "greedy / Gumbel-max sampling");
   m.def("kv_append", &kv_append, "paged KV-cache append");
   m.def("paged_decode", &paged_decode, "paged GQA/MQA decode attention");

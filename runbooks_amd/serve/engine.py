@@ -70,13 +70,18 @@ class Engine:
         self.seed = seed
         if kv_blocks is None:
             kv_blocks = self._auto_kv_blocks(mem_fraction)
-        self.allocator = BlockAllocator(kv_blocks)
+        # last block is reserved as the dummy block hipGraph-padded decode
+        # rows read/write (serve/graph.py)
+        self.allocator = BlockAllocator(kv_blocks - 1)
         self.caches = self.model.alloc_caches(kv_blocks, self.device)
+        self.dummy_block = kv_blocks - 1
         self.waiting: list[Request] = []
         self.running: list[Request] = []
         self._next_id = 0
         self.max_blocks_per_seq = min(
             kv_blocks, (self.cfg.max_seq_len + self.bs - 1) // self.bs)
+        self._graphed = None
+        self.use_graphs = torch.cuda.is_available()
 
     def _auto_kv_blocks(self, mem_fraction: float) -> int:
         bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
@@ -132,16 +137,29 @@ class Engine:
         for r, p in zip(reqs, pos):
             blk = r.blocks[p // self.bs]
             slots.append(blk * self.bs + p % self.bs)
-        maxb = max(len(r.blocks) for r in reqs)
-        bt = torch.zeros(B, maxb, dtype=torch.int32)
-        for i, r in enumerate(reqs):
-            bt[i, :len(r.blocks)] = torch.tensor(r.blocks, dtype=torch.int32)
-        tokens = torch.tensor(last, dtype=torch.long, device=dev)
-        positions = torch.tensor(pos, dtype=torch.int32, device=dev)
-        slot_t = torch.tensor(slots, dtype=torch.int32, device=dev)
-        seq_lens = torch.tensor([p + 1 for p in pos], dtype=torch.int32, device=dev)
-        logits = self.model.decode(tokens, positions, self.caches, slot_t,
-                                   bt.to(dev), seq_lens)
+
+        if self.use_graphs and B <= self.max_batch:
+            if self._graphed is None:
+                from .graph import GraphedDecoder
+                self._graphed = GraphedDecoder(
+                    self.model, self.caches, self.max_batch,
+                    self.max_blocks_per_seq, self.dummy_block, dev)
+            logits = self._graphed.decode(last, pos, slots,
+                                          [r.blocks for r in reqs],
+                                          [p + 1 for p in pos])
+        else:
+            maxb = max(len(r.blocks) for r in reqs)
+            bt = torch.zeros(B, maxb, dtype=torch.int32)
+            for i, r in enumerate(reqs):
+                bt[i, :len(r.blocks)] = torch.tensor(r.blocks,
+                                                     dtype=torch.int32)
+            tokens = torch.tensor(last, dtype=torch.long, device=dev)
+            positions = torch.tensor(pos, dtype=torch.int32, device=dev)
+            slot_t = torch.tensor(slots, dtype=torch.int32, device=dev)
+            seq_lens = torch.tensor([p + 1 for p in pos], dtype=torch.int32,
+                                    device=dev)
+            logits = self.model.decode(tokens, positions, self.caches, slot_t,
+                                       bt.to(dev), seq_lens)
         toks = ops.sample_tokens(
             logits, reqs[0].temperature,
             seed=self.seed + 1_000_003 * reqs[0].seq_len)

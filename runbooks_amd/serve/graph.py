@@ -1,0 +1,122 @@
+"""hipGraph-captured decode step.
+
+The decode inner loop is launch-bound: ~8 small kernels x num_layers per
+step plus five host->device tensor builds. This wraps the whole
+model.decode(...) call in a hipGraph (torch.cuda.CUDAGraph is hipGraph on
+ROCm) per batch-size bucket: static device input buffers are filled from
+pinned staging with async copies, then one graph replay launches the
+whole step (graph-replay floor ~10-16 us vs ~3.5 us/launch eager,
+MI355X_MICROARCH.md price list).
+
+Batch sizes are bucketed to powers of two; short batches pad with a
+dummy sequence that reads/writes a reserved KV block, so replay shapes
+stay fixed while requests come and go.
+"""
+from __future__ import annotations
+
+import torch
+
+
+def fixed_nsplit(batch: int, hkv: int) -> int:
+    """Work-split for the paged-decode kernel, chosen per bucket at capture
+    time (the eager heuristic in ops/attention.py reads seq_lens.max() —
+    a host sync, impossible inside a graph). base = batch*hkv workgroups
+    per split; split until ~2x256 CUs are covered."""
+    base = max(1, batch * hkv)
+    if base >= 256:
+        return 1
+    return min(16, max(1, (2 * 256) // base))
+
+
+class GraphedDecoder:
+    def __init__(self, model, caches, max_batch: int, max_blocks: int,
+                 dummy_block: int, device):
+        self.model = model
+        self.caches = caches
+        self.device = device
+        self.max_blocks = max_blocks
+        self.dummy_block = dummy_block
+        self.buckets = []
+        b = 1
+        while b < max_batch:
+            self.buckets.append(b)
+            b *= 2
+        self.buckets.append(max_batch)
+
+        B = max_batch
+        dev = device
+        self.tokens = torch.zeros(B, dtype=torch.long, device=dev)
+        self.positions = torch.zeros(B, dtype=torch.int32, device=dev)
+        self.slots = torch.full((B,), dummy_block * _bs(), dtype=torch.int32,
+                                device=dev)
+        self.block_tables = torch.full((B, max_blocks), dummy_block,
+                                       dtype=torch.int32, device=dev)
+        self.seq_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        pin = dict(dtype=torch.int64, pin_memory=True)
+        self.h_staging = torch.zeros(B * 4 + B * max_blocks, **pin)
+        self.graphs = {}     # bucket -> (CUDAGraph, logits_out)
+        self._pool = None
+
+    # -- capture -----------------------------------------------------------
+    def _capture(self, b: int):
+        hkv = self.model.local_kv_heads()
+        nsplit = fixed_nsplit(b, hkv)
+        args = (self.tokens[:b], self.positions[:b], self.caches,
+                self.slots[:b], self.block_tables[:b], self.seq_lens[:b])
+        # warmup outside the graph (allocator + autotune settle)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.model.decode(*args, nsplit=nsplit)
+        torch.cuda.current_stream().wait_stream(s)
+
+        g = torch.cuda.CUDAGraph()
+        if self._pool is None:
+            with torch.cuda.graph(g):
+                out = self.model.decode(*args, nsplit=nsplit)
+            self._pool = g.pool()
+        else:
+            with torch.cuda.graph(g, pool=self._pool):
+                out = self.model.decode(*args, nsplit=nsplit)
+        self.graphs[b] = (g, out)
+
+    # -- replay ------------------------------------------------------------
+    def decode(self, tokens, positions, slots, block_rows, seq_lens):
+        """All args host lists; block_rows is a list of per-seq block lists.
+        Returns logits [len(tokens), V] (a view into the static output)."""
+        n = len(tokens)
+        b = next(x for x in self.buckets if x >= n)
+        if b not in self.graphs:
+            self._capture(b)
+        g, out = self.graphs[b]
+
+        mb = self.max_blocks
+        h = self.h_staging
+        h[0:n] = torch.tensor(tokens, dtype=torch.int64)
+        h[b:b + n] = torch.tensor(positions, dtype=torch.int64)
+        h[2 * b:2 * b + n] = torch.tensor(slots, dtype=torch.int64)
+        h[3 * b:3 * b + n] = torch.tensor(seq_lens, dtype=torch.int64)
+        bt = h[4 * b:4 * b + b * mb].view(b, mb)
+        bt.fill_(self.dummy_block)
+        for i, row in enumerate(block_rows):
+            bt[i, :len(row)] = torch.tensor(row, dtype=torch.int64)
+        # pad rows beyond n: dummy sequence at position 0, length 1
+        if n < b:
+            h[n:b] = 0
+            h[b + n:2 * b] = 0
+            h[2 * b + n:3 * b] = self.dummy_block * _bs()
+            h[3 * b + n:4 * b] = 1
+
+        self.tokens[:b].copy_(h[0:b], non_blocking=True)
+        self.positions[:b].copy_(h[b:2 * b], non_blocking=True)
+        self.slots[:b].copy_(h[2 * b:3 * b], non_blocking=True)
+        self.seq_lens[:b].copy_(h[3 * b:4 * b], non_blocking=True)
+        self.block_tables[:b].copy_(bt, non_blocking=True)
+        g.replay()
+        return out[:n]
+
+
+def _bs() -> int:
+    from .. import ops
+    return ops.BLOCK_SIZE

@@ -45,9 +45,14 @@ class LoRALinear(nn.Module):
 
     def forward(self, x):
         y = self.base(x)
-        adapter = torch.nn.functional.linear(
-            torch.nn.functional.linear(x, self.lora_a), self.lora_b)
-        return y + self.scale * adapter
+        # adapter fused into the second GEMM's epilogue:
+        # y = 1*y + scale * (x A^T) B^T — addmm keeps it in hipBLASLt,
+        # no materialized adapter tensor, no separate mul/add kernels.
+        t = torch.nn.functional.linear(x, self.lora_a)
+        out_f = y.shape[-1]
+        return torch.addmm(y.reshape(-1, out_f), t.reshape(-1, self.r),
+                           self.lora_b.t(), beta=1.0,
+                           alpha=self.scale).view(y.shape)
 
     @property
     def weight(self):  # so init / inspection code keeps working

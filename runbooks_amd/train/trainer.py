@@ -66,9 +66,14 @@ class Trainer:
         loss.backward()
         self.ddp.finish_backward()
         if self.cfg.grad_clip > 0:
-            torch.nn.utils.clip_grad_norm_(
-                [p for p in self.model.parameters() if p.requires_grad],
-                self.cfg.grad_clip)
+            # grads are views into the DDP flat buckets — clip the few flat
+            # buffers instead of hundreds of per-tensor norms.
+            bufs = self.ddp.grad_buffers()
+            norms = torch._foreach_norm(bufs)
+            total = torch.linalg.vector_norm(torch.stack(norms))
+            scale = self.cfg.grad_clip / (total + 1e-6)
+            if float(scale) < 1.0:
+                torch._foreach_mul_(bufs, scale)
         self.optimizer.step()
         self.ddp.zero_grad()
         self.step_num += 1
