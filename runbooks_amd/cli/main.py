@@ -1,0 +1,201 @@
+"""`sub` command implementations.
+
+Parity with the reference command surface (internal/cli/root.go:9-24):
+  sub apply -f PATH        server-side apply manifests (+ upload builds)
+  sub run PATH [-i|-r]     upload build context, create Model, wait ready
+  sub notebook PATH        dev loop: notebook from manifest, sync, forward
+  sub serve PATH           apply a Server and wait for readiness
+  sub get KIND [NAME]      list/get substratus objects
+  sub delete KIND NAME     delete an object
+"""
+from __future__ import annotations
+
+import sys
+import threading
+import uuid
+
+import click
+
+from ..api.types import KINDS
+from ..k8s import HTTPKubeClient
+from .. import client as sclient
+from .manifests import find_manifests, next_version_name
+
+try:
+    from rich.console import Console
+    _console = Console()
+
+    def _say(msg):
+        _console.print(msg)
+except ImportError:  # rich is in the image, but degrade gracefully
+    def _say(msg):
+        print(msg)
+
+
+def _kube():
+    return HTTPKubeClient()
+
+
+def _report_conditions(raw):
+    conds = (raw.get("status") or {}).get("conditions") or []
+    for c in conds:
+        mark = "[green]✓[/green]" if c["status"] == "True" else "[yellow]…[/yellow]"
+        _say(f"  {mark} {c['type']}: {c.get('reason', '')}")
+
+
+def _apply_with_optional_upload(kube, obj, build_path=None, wait=False):
+    if build_path is not None:
+        tb = sclient.prepare_image_tarball(build_path,
+                                           progress=lambda p: None)
+        sclient.set_upload_container_spec(obj, tb, uuid.uuid4().hex)
+        sclient.clear_image(obj)
+        kube.apply(obj.to_dict())
+        _say(f"[bold]{obj.kind}/{obj.name}[/bold]: uploading build context "
+             f"(md5 {tb.md5_checksum[:12]}…)")
+        sclient.upload(kube, obj, tb)
+    else:
+        kube.apply(obj.to_dict())
+        _say(f"[bold]{obj.kind}/{obj.name}[/bold]: applied")
+    if wait:
+        _say(f"waiting for {obj.kind}/{obj.name} to become ready…")
+        raw = sclient.wait_ready(kube, obj)
+        _report_conditions(raw)
+
+
+@click.group()
+def main():
+    """Substratus-compatible CLI for the MI355X-native platform."""
+
+
+@main.command()
+@click.option("-f", "--filename", default=".", help="manifest file or dir")
+@click.option("-n", "--namespace", default="default")
+@click.option("--build", "build_path", default=None,
+              help="directory with Dockerfile to upload as build context")
+@click.option("--wait/--no-wait", default=False)
+def apply(filename, namespace, build_path, wait):
+    kube = _kube()
+    objs = find_manifests(filename)
+    if not objs:
+        _say("[red]no substratus manifests found[/red]")
+        sys.exit(1)
+    for obj in objs:
+        obj.namespace = namespace
+        _apply_with_optional_upload(kube, obj, build_path, wait)
+
+
+@main.command()
+@click.argument("path", default=".")
+@click.option("-n", "--namespace", default="default")
+@click.option("-i", "--increment", is_flag=True,
+              help="create a new auto-versioned name-N object")
+@click.option("-r", "--replace", is_flag=True, help="replace existing object")
+def run(path, namespace, increment, replace):
+    """Upload PATH (with Dockerfile) and run it as a Model build+train."""
+    kube = _kube()
+    objs = find_manifests(path, kind_filter="Model") or \
+        find_manifests(path)
+    if not objs:
+        _say("[red]no substratus manifests found[/red]")
+        sys.exit(1)
+    obj = objs[0]
+    obj.namespace = namespace
+    if increment:
+        obj.name = next_version_name(kube, obj.kind, namespace, obj.name)
+        _say(f"auto-versioned name: {obj.name}")
+    elif replace:
+        kube.delete("substratus.ai/v1", obj.kind, namespace, obj.name)
+    _apply_with_optional_upload(kube, obj, path, wait=True)
+
+
+@main.command()
+@click.argument("path", default=".")
+@click.option("-n", "--namespace", default="default")
+@click.option("--sync-dir", default=".",
+              help="local directory mirrored from the notebook")
+@click.option("--port", default=8888)
+@click.option("--no-sync", is_flag=True)
+def notebook(path, namespace, sync_dir, port, no_sync):
+    """Dev loop: build a Notebook from the first manifest in PATH, upload
+    the dir as its image context, wait, sync files, port-forward 8888."""
+    kube = _kube()
+    objs = find_manifests(path)
+    if not objs:
+        _say("[red]no substratus manifests found[/red]")
+        sys.exit(1)
+    nb = sclient.notebook_for_object(objs[0])
+    nb.namespace = namespace
+    _apply_with_optional_upload(kube, nb, path, wait=True)
+
+    ns, pod = sclient.pod_for_notebook(nb)
+    fwd = sclient.sync.port_forward(ns, pod, port, 8888)
+    _say(f"[green]notebook ready[/green] → http://localhost:{port} "
+         f"(token: default)")
+    stop = threading.Event()
+    try:
+        if no_sync:
+            fwd.wait()
+        else:
+            sclient.sync_files_from_notebook(
+                ns, pod, sync_dir, stop=stop,
+                on_event=lambda e: _say(f"  sync {e['op']} {e['path']}"))
+    except KeyboardInterrupt:
+        pass
+    finally:
+        stop.set()
+        fwd.terminate()
+
+
+@main.command()
+@click.argument("path", default=".")
+@click.option("-n", "--namespace", default="default")
+def serve(path, namespace):
+    """Apply a Server manifest and wait for it to serve."""
+    kube = _kube()
+    objs = find_manifests(path, kind_filter="Server")
+    if not objs:
+        _say("[red]no Server manifest found[/red]")
+        sys.exit(1)
+    for obj in objs:
+        obj.namespace = namespace
+        _apply_with_optional_upload(kube, obj, None, wait=True)
+        _say(f"Server {obj.name} ready on service {obj.name}-server:8080")
+
+
+@main.command()
+@click.argument("kind")
+@click.argument("name", required=False)
+@click.option("-n", "--namespace", default="default")
+def get(kind, name, namespace):
+    kube = _kube()
+    kind = {k.lower(): k for k in KINDS}.get(kind.rstrip("s").lower(), kind)
+    if name:
+        raw = kube.get("substratus.ai/v1", kind, namespace, name)
+        if raw is None:
+            _say(f"[red]{kind}/{name} not found[/red]")
+            sys.exit(1)
+        import json
+        print(json.dumps(raw, indent=2))
+        return
+    rows = kube.list("substratus.ai/v1", kind, namespace)
+    for o in rows:
+        ready = (o.get("status") or {}).get("ready", False)
+        print(f"{o['metadata']['name']}\tready={ready}")
+
+
+@main.command()
+@click.argument("kind")
+@click.argument("name")
+@click.option("-n", "--namespace", default="default")
+def delete(kind, name, namespace):
+    kube = _kube()
+    kind = {k.lower(): k for k in KINDS}.get(kind.rstrip("s").lower(), kind)
+    if kube.delete("substratus.ai/v1", kind, namespace, name):
+        _say(f"deleted {kind}/{name}")
+    else:
+        _say(f"[red]{kind}/{name} not found[/red]")
+        sys.exit(1)
+
+
+if __name__ == "__main__":
+    main()

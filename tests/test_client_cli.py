@@ -1,0 +1,156 @@
+"""Client library + CLI tests, including the full upload handshake
+"system test": CLI client <-> controllers <-> real kind SCI (gRPC + HTTP)
+against the in-memory API server — the offline analog of the reference's
+test/system.sh kind flow.
+"""
+import os
+import threading
+
+import pytest
+
+from runbooks_amd import client as sclient
+from runbooks_amd.api.types import Dataset, Model, Notebook, ObjectRef, Server
+from runbooks_amd.cloud import new_cloud
+from runbooks_amd.controller import ControllerManager
+from runbooks_amd.k8s import MemoryKubeClient
+from runbooks_amd.sci import ControllerClient
+from runbooks_amd.sci.kind_server import KindSCI, make_http_server
+from runbooks_amd.sci.server import serve as sci_serve
+
+
+def test_prepare_tarball_requires_dockerfile(tmp_path):
+    with pytest.raises(FileNotFoundError):
+        sclient.prepare_image_tarball(str(tmp_path))
+    (tmp_path / "Dockerfile").write_text("FROM scratch\n")
+    (tmp_path / "train.py").write_text("print('hi')\n")
+    tb = sclient.prepare_image_tarball(str(tmp_path))
+    assert len(tb.md5_checksum) == 32
+    assert os.path.exists(tb.path)
+
+
+def test_notebook_for_object_conversions():
+    m = Model(name="m", image="img:1", model=ObjectRef("base"),
+              dataset=ObjectRef("d"), params={"x": 1})
+    nb = sclient.notebook_for_object(m)
+    assert isinstance(nb, Notebook)
+    assert nb.name == "m-model" and nb.model.name == "base"
+    s = Server(name="s", image="img:2", model=ObjectRef("m"))
+    nb2 = sclient.notebook_for_object(s)
+    assert nb2.name == "s-server" and nb2.model.name == "m"
+    d = Dataset(name="d", image="img:3")
+    assert sclient.notebook_for_object(d).name == "d-dataset"
+
+
+def test_upload_handshake_system(tmp_path):
+    """End-to-end: tarball -> apply -> controller signed URL (real kind SCI
+    over gRPC) -> HTTP PUT -> controller verifies md5 -> Built image."""
+    kube = MemoryKubeClient()
+    cloud = new_cloud({"CLOUD": "kind", "CLUSTER_NAME": "kind",
+                       "REGISTRY_PORT_5000_TCP_ADDR": "10.0.0.9"})
+
+    sci_impl = KindSCI(root=str(tmp_path))
+    httpd = make_http_server(sci_impl, port=0)
+    sci_impl.signed_url_address = f"http://127.0.0.1:{httpd.server_address[1]}"
+    grpc_server = sci_serve(sci_impl, "127.0.0.1:0")
+    sci_client = ControllerClient(f"127.0.0.1:{grpc_server.bound_port}")
+
+    mgr = ControllerManager(kube, cloud, sci_client)
+
+    build_dir = tmp_path / "ctx"
+    build_dir.mkdir()
+    (build_dir / "Dockerfile").write_text("FROM scratch\nCOPY . /src\n")
+    (build_dir / "main.py").write_text("print('model')\n")
+
+    obj = Model(name="up1")
+    tb = sclient.prepare_image_tarball(str(build_dir))
+    sclient.set_upload_container_spec(obj, tb, request_id="req-abc")
+    kube.apply(obj.to_dict())
+
+    # background reconciler thread stands in for the running operator
+    stop = threading.Event()
+
+    def loop():
+        while not stop.is_set():
+            mgr.reconcile_all(rounds=1)
+            stop.wait(0.05)
+
+    t = threading.Thread(target=loop, daemon=True)
+    t.start()
+    try:
+        sclient.upload(kube, obj, tb, timeout=20)
+        # controller verifies the stored md5 and marks Uploaded
+        deadline = threading.Event()
+        for _ in range(100):
+            raw = kube.get("substratus.ai/v1", "Model", "default", "up1")
+            got = Model.from_dict(raw)
+            if got.is_condition_true("Uploaded"):
+                break
+            deadline.wait(0.05)
+        assert got.is_condition_true("Uploaded")
+        assert got.build_upload.stored_md5_checksum == tb.md5_checksum
+        # kaniko storage job exists; completing it publishes the image
+        kube.patch("batch/v1", "Job", "default", "up1-model-bld", {"status": {
+            "succeeded": 1, "conditions": [{"type": "Complete",
+                                            "status": "True"}]}})
+        for _ in range(100):
+            raw = kube.get("substratus.ai/v1", "Model", "default", "up1")
+            got = Model.from_dict(raw)
+            if got.get_image():
+                break
+            deadline.wait(0.05)
+        assert got.get_image().endswith(":" + tb.md5_checksum)
+    finally:
+        stop.set()
+        t.join(timeout=2)
+        grpc_server.stop(0)
+        httpd.shutdown()
+
+
+def test_cli_manifest_discovery(tmp_path):
+    from runbooks_amd.cli.manifests import find_manifests, next_version_name
+    (tmp_path / "model.yaml").write_text(
+        "apiVersion: substratus.ai/v1\nkind: Model\n"
+        "metadata: {name: m1}\nspec: {image: i}\n---\n"
+        "apiVersion: v1\nkind: ConfigMap\nmetadata: {name: x}\n")
+    (tmp_path / "server.yaml").write_text(
+        "apiVersion: substratus.ai/v1\nkind: Server\n"
+        "metadata: {name: s1}\nspec: {image: i, model: {name: m1}}\n")
+    objs = find_manifests(str(tmp_path))
+    assert [o.kind for o in objs] == ["Model", "Server"]
+    assert find_manifests(str(tmp_path), kind_filter="server")[0].name == "s1"
+
+    kube = MemoryKubeClient()
+    assert next_version_name(kube, "Model", "default", "m") == "m"
+    kube.create(Model(name="m").to_dict())
+    assert next_version_name(kube, "Model", "default", "m") == "m-1"
+    kube.create(Model(name="m-4").to_dict())
+    assert next_version_name(kube, "Model", "default", "m") == "m-5"
+    assert next_version_name(kube, "Model", "default", "m-2") == "m-5"
+
+
+def test_cli_get_delete(monkeypatch, capsys):
+    from click.testing import CliRunner
+    import runbooks_amd.cli.main as cli_main
+
+    kube = MemoryKubeClient()
+    kube.create(Model(name="m1", image="i").to_dict())
+    monkeypatch.setattr(cli_main, "_kube", lambda: kube)
+    r = CliRunner().invoke(cli_main.main, ["get", "models"])
+    assert r.exit_code == 0 and "m1" in r.output
+    r = CliRunner().invoke(cli_main.main, ["delete", "model", "m1"])
+    assert r.exit_code == 0
+    assert kube.get("substratus.ai/v1", "Model", "default", "m1") is None
+
+
+def test_nbwatch_events(tmp_path):
+    from runbooks_amd import nbwatch
+    root = tmp_path / "content"
+    (root / "data").mkdir(parents=True)   # special dir: ignored
+    (root / "nb").mkdir()
+    f = root / "nb" / "train.py"
+    gen = nbwatch.watch(str(root), interval=0.01, once=True)
+    f.write_text("x = 1\n")
+    (root / "data" / "ignored.bin").write_text("z")
+    events = list(gen)
+    assert {(e["op"], os.path.basename(e["path"])) for e in events} == {
+        ("CREATE", "train.py")}
