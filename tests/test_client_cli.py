@@ -129,8 +129,9 @@ def test_cli_manifest_discovery(tmp_path):
 
 
 def test_cli_get_delete(monkeypatch, capsys):
+    import importlib
     from click.testing import CliRunner
-    import runbooks_amd.cli.main as cli_main
+    cli_main = importlib.import_module("runbooks_amd.cli.main")
 
     kube = MemoryKubeClient()
     kube.create(Model(name="m1", image="i").to_dict())
