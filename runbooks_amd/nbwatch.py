@@ -45,8 +45,14 @@ def _scan(root: str) -> dict[str, float]:
 
 
 def watch(root: str = "/content", interval: float = 0.5, once: bool = False):
-    index = 0
+    """The baseline scan happens eagerly at call time (like registering an
+    fsnotify watcher); the returned generator yields change events."""
     prev = _scan(root)
+    return _watch_iter(root, prev, interval, once)
+
+
+def _watch_iter(root, prev, interval, once):
+    index = 0
     while True:
         time.sleep(interval)
         cur = _scan(root)
