@@ -26,7 +26,8 @@ class ModelConfig:
     act: str = "silu_glu"            # silu_glu | gelu | relu
     pos: str = "rope"                # rope | learned
     rope_theta: float = 10000.0
-    parallel_residual: bool = False  # falcon-style single-norm parallel block
+    parallel_residual: bool = False  # falcon-style attn+mlp parallel block
+    single_norm: bool = False        # falcon-7b: one shared ln for attn+mlp
     tie_embeddings: bool = False
     mlp_bias: bool = False
     attn_bias: bool = False
@@ -81,7 +82,7 @@ register(ModelConfig("falcon-7b", vocab_size=65024, hidden_size=4544,
                      num_layers=32, num_heads=71, num_kv_heads=1,
                      intermediate_size=4 * 4544, head_dim=64,
                      norm="layernorm", act="gelu", parallel_residual=True,
-                     tie_embeddings=True))
+                     single_norm=True, tie_embeddings=True))
 register(ModelConfig("falcon-40b", vocab_size=65024, hidden_size=8192,
                      num_layers=60, num_heads=128, num_kv_heads=8,
                      intermediate_size=4 * 8192, head_dim=64,
@@ -108,7 +109,7 @@ register(ModelConfig("tiny-falcon", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=8, num_kv_heads=1,
                      intermediate_size=128, head_dim=8, max_seq_len=128,
                      norm="layernorm", act="gelu", parallel_residual=True,
-                     tie_embeddings=True))
+                     single_norm=True, tie_embeddings=True))
 register(ModelConfig("tiny-opt", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=4,
                      intermediate_size=128, max_seq_len=128,

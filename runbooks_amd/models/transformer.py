@@ -138,12 +138,16 @@ class Block(nn.Module):
         self.attn = Attention(cfg, dtype, tp)
         self.mlp = MLP(cfg, dtype, tp)
         self.norm1 = _norm_module(cfg, dtype)
-        self.norm2 = _norm_module(cfg, dtype)
+        self.norm2 = None if cfg.single_norm else _norm_module(cfg, dtype)
 
     def forward(self, x, attn_fn):
         """attn_fn(h) runs the right attention mode on normed input."""
         if self.cfg.parallel_residual:
-            # falcon: x + attn(ln1(x)) + mlp(ln2(x))
+            if self.cfg.single_norm:
+                # falcon-7b: one shared ln feeds both branches
+                h = self.norm1(x)
+                return x + attn_fn(h) + self.mlp(h)
+            # falcon-40b: x + attn(ln_attn(x)) + mlp(ln_mlp(x))
             return x + attn_fn(self.norm1(x)) + self.mlp(self.norm2(x))
         x = x + attn_fn(self.norm1(x))
         return x + self.mlp(self.norm2(x))

@@ -1,17 +1,23 @@
 """HTTP serving layer satisfying the reference's container contract:
-port 8080, 200 OK on "/" when ready, POST /v1/completions
-(reference docs/container-contract.md "Server"; exercised by the
-reference system test test/system.sh:70-77).
+port 8080, 200 OK on "/" when ready, POST /v1/completions with optional
+SSE streaming (reference docs/container-contract.md "Server"; exercised
+by the reference system test test/system.sh:70-77; the reference's
+basaran image exposes the same OpenAI-style surface).
+
+Requests are submitted to the Engine's continuous-batching scheduler; a
+single background thread drives engine.step() so concurrent HTTP
+requests decode together in one batch.
 """
 from __future__ import annotations
 
-import asyncio
+import json
+import queue
 import threading
 import time
 import uuid
 
 from fastapi import FastAPI
-from fastapi.responses import JSONResponse
+from fastapi.responses import JSONResponse, StreamingResponse
 from pydantic import BaseModel
 
 from .engine import Engine
@@ -26,10 +32,72 @@ class CompletionRequest(BaseModel):
     stream: bool = False
 
 
-def build_app(engine: Engine, tokenizer=None, model_name: str = "model") -> FastAPI:
+class EngineLoop:
+    """Background thread driving the continuous-batching engine."""
+
+    def __init__(self, engine: Engine):
+        self.engine = engine
+        self._lock = threading.Lock()
+        self._wake = threading.Event()
+        self._watchers: dict[int, queue.Queue] = {}  # request_id -> token q
+        self._stop = False
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+
+    def submit(self, prompt_ids, max_new_tokens, temperature) -> queue.Queue:
+        """Returns a queue yielding (token_id | None); None = finished."""
+        q: queue.Queue = queue.Queue()
+        with self._lock:
+            req = self.engine.submit(prompt_ids, max_new_tokens, temperature)
+            self._watchers[req.request_id] = q
+            req._watch_sent = 0
+        self._wake.set()
+        return q
+
+    def _run(self):
+        while not self._stop:
+            if not self.engine.has_work():
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+                continue
+            with self._lock:
+                try:
+                    finished = self.engine.step()
+                except RuntimeError:
+                    # oversize prompt rejected by the scheduler: notify all
+                    # watchers whose request vanished from the queues
+                    alive = {r.request_id for r in
+                             self.engine.running + self.engine.waiting}
+                    for rid in list(self._watchers):
+                        if rid not in alive:
+                            self._watchers.pop(rid).put(None)
+                    continue
+                live = list(self.engine.running) + finished
+                for r in live:
+                    q = self._watchers.get(r.request_id)
+                    if q is None:
+                        continue
+                    sent = getattr(r, "_watch_sent", 0)
+                    for t in r.output_ids[sent:]:
+                        q.put(t)
+                    r._watch_sent = len(r.output_ids)
+                for r in finished:
+                    q = self._watchers.pop(r.request_id, None)
+                    if q is not None:
+                        q.put(None)
+
+    def shutdown(self):
+        self._stop = True
+        self._wake.set()
+        self._thread.join(timeout=2)
+
+
+def build_app(engine: Engine, tokenizer=None,
+              model_name: str = "model") -> FastAPI:
     app = FastAPI(title="runbooks-amd-server")
     tok = tokenizer or load_tokenizer(None)
-    lock = threading.Lock()
+    loop = EngineLoop(engine)
+    app.state.engine_loop = loop
 
     @app.get("/")
     def ready():
@@ -39,28 +107,51 @@ def build_app(engine: Engine, tokenizer=None, model_name: str = "model") -> Fast
     def healthz():
         return {"status": "ok"}
 
-    @app.post("/v1/completions")
-    async def completions(req: CompletionRequest):
-        ids = tok.encode(req.prompt)[-engine.cfg.max_seq_len + req.max_tokens + 1:]
-        loop = asyncio.get_event_loop()
+    def _usage(n_prompt, n_out):
+        return {"prompt_tokens": n_prompt, "completion_tokens": n_out,
+                "total_tokens": n_prompt + n_out}
 
-        def run():
-            with lock:
-                return engine.generate(ids, max_new_tokens=req.max_tokens,
-                                       temperature=req.temperature)
-        t0 = time.time()
-        out_ids = await loop.run_in_executor(None, run)
-        text = tok.decode(out_ids)
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        ids = tok.encode(req.prompt)[-engine.cfg.max_seq_len +
+                                     req.max_tokens + 1:]
+        cid = f"cmpl-{uuid.uuid4().hex[:12]}"
+        t0 = int(time.time())
+        q = loop.submit(ids, req.max_tokens, req.temperature)
+
+        if req.stream:
+            def gen():
+                out = []
+                while True:
+                    t = q.get()
+                    if t is None:
+                        break
+                    out.append(t)
+                    piece = tok.decode(out)
+                    chunk = {"id": cid, "object": "text_completion",
+                             "created": t0,
+                             "model": req.model or model_name,
+                             "choices": [{"text": piece[len(tok.decode(out[:-1])):],
+                                          "index": 0, "logprobs": None,
+                                          "finish_reason": None}]}
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                yield "data: [DONE]\n\n"
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        out = []
+        while True:
+            t = q.get()
+            if t is None:
+                break
+            out.append(t)
         return JSONResponse({
-            "id": f"cmpl-{uuid.uuid4().hex[:12]}",
+            "id": cid,
             "object": "text_completion",
-            "created": int(t0),
+            "created": t0,
             "model": req.model or model_name,
-            "choices": [{"text": text, "index": 0, "logprobs": None,
-                         "finish_reason": "length"}],
-            "usage": {"prompt_tokens": len(ids),
-                      "completion_tokens": len(out_ids),
-                      "total_tokens": len(ids) + len(out_ids)},
+            "choices": [{"text": tok.decode(out), "index": 0,
+                         "logprobs": None, "finish_reason": "length"}],
+            "usage": _usage(len(ids), len(out)),
         })
 
     return app
