@@ -81,7 +81,11 @@ class Engine:
         self.max_blocks_per_seq = min(
             kv_blocks, (self.cfg.max_seq_len + self.bs - 1) // self.bs)
         self._graphed = None
-        self.use_graphs = torch.cuda.is_available()
+        # hipGraphs drive the single-GPU decode; the TP>1 path runs eager
+        # so the worker-follow protocol (serve/tp_worker.py) sees every
+        # collective.
+        self.tp = comm.world_size()
+        self.use_graphs = torch.cuda.is_available() and self.tp == 1
 
     def _auto_kv_blocks(self, mem_fraction: float) -> int:
         bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
@@ -120,6 +124,9 @@ class Engine:
         slots = torch.tensor(
             [req.blocks[i // self.bs] * self.bs + i % self.bs for i in range(S)],
             dtype=torch.int32, device=self.device)
+        if self.tp > 1:
+            from .tp_worker import broadcast_prefill
+            broadcast_prefill(tokens, positions, slots)
         logits = self.model.prefill(tokens, positions, self.caches, slots)
         tok = ops.sample_tokens(logits, req.temperature,
                                 seed=self.seed + req.request_id * 65537 + S)
@@ -158,8 +165,12 @@ class Engine:
             slot_t = torch.tensor(slots, dtype=torch.int32, device=dev)
             seq_lens = torch.tensor([p + 1 for p in pos], dtype=torch.int32,
                                     device=dev)
+            bt = bt.to(dev)
+            if self.tp > 1:
+                from .tp_worker import broadcast_decode
+                broadcast_decode(tokens, positions, slot_t, bt, seq_lens)
             logits = self.model.decode(tokens, positions, self.caches, slot_t,
-                                       bt.to(dev), seq_lens)
+                                       bt, seq_lens)
         toks = ops.sample_tokens(
             logits, reqs[0].temperature,
             seed=self.seed + 1_000_003 * reqs[0].seq_len)

@@ -20,7 +20,22 @@ CSRC = ROOT / "runbooks_amd" / "ops" / "csrc"
 sources = sorted(str(p) for p in CSRC.glob("*.hip")) + [str(CSRC / "bindings.cpp")]
 
 setup(
-    name="runbooks_amd_kernels",
+    name="runbooks_amd",
+    version="0.1.0",
+    packages=["runbooks_amd"] + [
+        f"runbooks_amd.{p}" for p in
+        ("api", "cli", "client", "cloud", "controller", "k8s", "models",
+         "ops", "parallel", "sci", "serve", "train", "workloads")],
+    entry_points={"console_scripts": [
+        # parity: reference cmd/ — sub CLI, controllermanager, sci servers,
+        # nbwatch (goreleaser targets .goreleaser.yaml:8-37)
+        "sub = runbooks_amd.cli.main:main",
+        "runbooks-controller-manager = runbooks_amd.controller.manager:run_manager",
+        "sci-kind = runbooks_amd.sci.kind_server:main",
+        "sci-gcp = runbooks_amd.sci.gcp_server:main",
+        "sci-aws = runbooks_amd.sci.aws_server:main",
+        "nbwatch = runbooks_amd.nbwatch:main",
+    ]},
     ext_modules=[
         cpp_extension.CUDAExtension(
             name="runbooks_amd.ops._hip",

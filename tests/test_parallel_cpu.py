@@ -136,6 +136,42 @@ def _bucket_overlap_many_buckets(rank, world):
     assert torch.allclose(g, m3[0].weight.grad, atol=1e-6)
 
 
+def _tp_engine_serving(rank, world):
+    """TP=2 serving through the worker-follow protocol
+    (serve/tp_worker.py): rank 0 drives the engine, rank 1 follows;
+    output must equal a TP=1 engine run."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.serve import Engine
+    from runbooks_amd.serve.tp_worker import broadcast_shutdown, worker_loop
+
+    torch.manual_seed(0)
+    tp_model = build_model("tiny-llama", dtype=torch.float32, tp=world, seed=3)
+    single = build_model("tiny-llama", dtype=torch.float32, tp=1, seed=3)
+    sd = single.state_dict()
+    tsd = tp_model.state_dict()
+    for name, t in tsd.items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:
+            t.copy_(full[rank * t.shape[0]:(rank + 1) * t.shape[0]])
+        else:
+            t.copy_(full[:, rank * t.shape[1]:(rank + 1) * t.shape[1]])
+
+    eng = Engine(tp_model, device="cpu", kv_blocks=64, seed=11)
+    prompt = [3, 1, 4, 1, 5, 9, 2, 6]
+    if rank == 0:
+        out_tp = eng.generate(prompt, max_new_tokens=6)
+        broadcast_shutdown(torch.device("cpu"))
+        ref = Engine(single, device="cpu", kv_blocks=64, seed=11)
+        # reference engine must not broadcast: it thinks tp==world; force 1
+        ref.tp = 1
+        out_1 = ref.generate(prompt, max_new_tokens=6)
+        assert out_tp == out_1, (out_tp, out_1)
+    else:
+        worker_loop(eng)
+
+
 # --- test entries -----------------------------------------------------------
 
 def test_dp_gradient_allreduce():
@@ -152,3 +188,7 @@ def test_tp_model_full():
 
 def test_ddp_many_buckets():
     _run_dist(_bucket_overlap_many_buckets, port=PORT + 3)
+
+
+def test_tp_engine_serving():
+    _run_dist(_tp_engine_serving, port=PORT + 4)
