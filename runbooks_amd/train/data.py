@@ -30,7 +30,8 @@ class JsonlTextDataset(torch.utils.data.Dataset):
     """Reads {"text": ...} or {"prompt","completion"} jsonl, tokenizes with a
     byte-level fallback tokenizer when none is supplied."""
 
-    def __init__(self, path: str | Path, seq_len: int, tokenizer=None):
+    def __init__(self, path: str | Path, seq_len: int, tokenizer=None,
+                 vocab_size: int | None = None):
         self.rows = []
         with open(path) as f:
             for line in f:
@@ -39,6 +40,9 @@ class JsonlTextDataset(torch.utils.data.Dataset):
                     self.rows.append(json.loads(line))
         self.seq = seq_len
         self.tok = tokenizer
+        # clamp ids into the model's vocab (a byte-fallback tokenizer can
+        # emit ids past a tiny model's embedding table)
+        self.vocab = vocab_size
 
     def __len__(self):
         return len(self.rows)
@@ -52,6 +56,8 @@ class JsonlTextDataset(torch.utils.data.Dataset):
         row = self.rows[i]
         text = row.get("text") or (row.get("prompt", "") + row.get("completion", ""))
         ids = self._encode(text)[: self.seq]
+        if self.vocab is not None:
+            ids = [i % self.vocab for i in ids]
         ids = ids + [0] * (self.seq - len(ids))
         return torch.tensor(ids, dtype=torch.long)
 

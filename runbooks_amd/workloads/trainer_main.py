@@ -94,7 +94,8 @@ def main():
     jsonl = sorted(data_dir.glob("*.jsonl")) if data_dir.exists() else []
     if jsonl:
         tok = load_tokenizer(model_dir if model_dir.exists() else None)
-        dataset = JsonlTextDataset(jsonl[0], cfg.seq_len + 1, tokenizer=tok)
+        dataset = JsonlTextDataset(jsonl[0], cfg.seq_len + 1, tokenizer=tok,
+                                   vocab_size=trainer.model.cfg.vocab_size)
         epochs = _param("num_train_epochs", None, float)
         if epochs is not None:
             steps = int(epochs * len(dataset)

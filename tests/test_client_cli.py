@@ -155,3 +155,60 @@ def test_nbwatch_events(tmp_path):
     events = list(gen)
     assert {(e["op"], os.path.basename(e["path"])) for e in events} == {
         ("CREATE", "train.py")}
+
+
+def test_entrypoint_params_to_env(tmp_path, monkeypatch):
+    """The contract's params.json -> PARAM_* conversion
+    (docs/container-contract.md; reference container-contract.md:34-48)."""
+    import json
+    from runbooks_amd.workloads.entrypoint import params_to_env
+    assert params_to_env({"epochs": 1, "use-lora": True, "name": "x/y"}) == {
+        "PARAM_EPOCHS": "1", "PARAM_USE_LORA": "true", "PARAM_NAME": "x/y"}
+
+
+def test_dataset_loader_synthetic(tmp_path, monkeypatch):
+    from runbooks_amd.workloads import dataset_loader
+    monkeypatch.setenv("ARTIFACTS_DIR", str(tmp_path))
+    monkeypatch.setenv("PARAM_SYNTHETIC", "true")
+    assert dataset_loader.main() == 0
+    lines = (tmp_path / "data.jsonl").read_text().strip().splitlines()
+    assert len(lines) == 256
+
+
+def test_model_loader_synthetic(tmp_path, monkeypatch):
+    from runbooks_amd.workloads import model_loader
+    monkeypatch.setenv("ARTIFACTS_DIR", str(tmp_path))
+    monkeypatch.setenv("PARAM_NAME", "tiny-llama")
+    monkeypatch.setenv("PARAM_SYNTHETIC", "true")
+    assert model_loader.main() == 0
+    assert (tmp_path / "model.safetensors").exists()
+    assert "tiny-llama" in (tmp_path / "config.json").read_text()
+
+
+def test_trainer_main_end_to_end(tmp_path, monkeypatch):
+    """The trainer image main on CPU: loader artifacts -> fine-tune ->
+    checkpoints in /content/artifacts (contract round trip)."""
+    from runbooks_amd.workloads import dataset_loader, model_loader, trainer_main
+    model_dir = tmp_path / "model"
+    data_dir = tmp_path / "data"
+    out_dir = tmp_path / "artifacts"
+    model_dir.mkdir(), data_dir.mkdir()
+    monkeypatch.setenv("ARTIFACTS_DIR", str(model_dir))
+    monkeypatch.setenv("PARAM_NAME", "tiny-llama")
+    monkeypatch.setenv("PARAM_SYNTHETIC", "true")
+    assert model_loader.main() == 0
+    monkeypatch.setenv("ARTIFACTS_DIR", str(data_dir))
+    assert dataset_loader.main() == 0
+
+    monkeypatch.setenv("MODEL_DIR", str(model_dir))
+    monkeypatch.setenv("DATA_DIR", str(data_dir))
+    monkeypatch.setenv("ARTIFACTS_DIR", str(out_dir))
+    monkeypatch.setenv("PARAM_NUM_TRAIN_STEPS", "2")
+    monkeypatch.setenv("PARAM_SAVE_STEPS", "2")
+    monkeypatch.setenv("PARAM_SEQ_LEN", "32")
+    monkeypatch.setenv("PARAM_PER_DEVICE_TRAIN_BATCH_SIZE", "2")
+    monkeypatch.delenv("PARAM_SYNTHETIC")
+    assert trainer_main.main() == 0
+    ckpts = list(out_dir.glob("checkpoint-*"))
+    assert ckpts, "trainer wrote no checkpoint"
+    assert (ckpts[0] / "model.safetensors").exists()
