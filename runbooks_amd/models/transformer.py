@@ -63,9 +63,9 @@ class Attention(nn.Module):
                                                bias=cfg.attn_bias, tp_size=tp,
                                                dtype=dtype, gather_input=False)
         else:  # replicate KV (MQA with tp > num_kv_heads)
-            self.k_proj = nn.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
+            self.k_proj = ops.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
                                     dtype=dtype)
-            self.v_proj = nn.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
+            self.v_proj = ops.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
                                     dtype=dtype)
         self.o_proj = RowParallelLinear(cfg.num_heads * self.dh, cfg.hidden_size,
                                         bias=cfg.attn_bias, tp_size=tp, dtype=dtype)
@@ -166,7 +166,7 @@ class Transformer(nn.Module):
         self.blocks = nn.ModuleList(Block(cfg, dtype, self.tp)
                                     for _ in range(cfg.num_layers))
         self.norm_f = _norm_module(cfg, dtype)
-        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False, dtype=dtype)
+        self.lm_head = ops.Linear(cfg.hidden_size, cfg.vocab_size, bias=False, dtype=dtype)
         if cfg.tie_embeddings:
             self.lm_head.weight = self.embed.weight
         if cfg.pos == "rope":
@@ -181,7 +181,7 @@ class Transformer(nn.Module):
 
     def reset_parameters(self):
         for m in self.modules():
-            if isinstance(m, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):
+            if isinstance(m, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):  # ops.Linear subclasses nn.Linear
                 nn.init.normal_(m.weight, std=0.02)
                 if getattr(m, "bias", None) is not None:
                     nn.init.zeros_(m.bias)

@@ -35,6 +35,8 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                              double gscale, int64_t ignore_index);
 at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
+at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
+int64_t skinny_gemm_mmax();
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "runbooks_amd gfx950 HIP kernels";
@@ -61,4 +63,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward -> dlogits");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "layout probe: one 32x32x16 bf16 MFMA as a plain matmul");
+  m.def("skinny_gemm", &skinny_gemm,
+        "HBM-rate decode GEMM: y[M<=32,N] = x @ W^T (bf16)");
+  m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
 }

@@ -16,6 +16,7 @@ import torch.distributed as dist
 from torch import nn
 
 from . import comm
+from ..ops.linear import fast_linear
 
 
 class _AllReduceFn(torch.autograd.Function):
@@ -75,7 +76,7 @@ class ColumnParallelLinear(nn.Module):
     def forward(self, x):
         if self.tp > 1 and self.gather_input:
             x = copy_to_tp(x)
-        return torch.nn.functional.linear(x, self.weight, self.bias)
+        return fast_linear(x, self.weight, self.bias)
 
 
 class RowParallelLinear(nn.Module):
@@ -94,7 +95,7 @@ class RowParallelLinear(nn.Module):
         self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
 
     def forward(self, x):
-        y = torch.nn.functional.linear(x, self.weight)
+        y = fast_linear(x, self.weight)
         if self.tp > 1:
             y = reduce_from_tp(y)
         if self.bias is not None:
