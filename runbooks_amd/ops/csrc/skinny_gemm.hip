@@ -2,25 +2,26 @@
 // M <= 32 (decode batch), bf16 in / bf16 out, fp32 accumulate.
 //
 // Why: batch-<=32 decode GEMMs are pure weight streams (W is ~99.9% of
-// the traffic) and hipBLASLt's tiles measured only 25-45% of HBM
-// bandwidth on these shapes (profiles/). Design:
+// the traffic); hipBLASLt measured ~2 TB/s in the real decode loop
+// (profiles/). The one thing that matters here is shaping the W read as
+// long coalesced bursts and keeping enough of them in flight:
 //
-//  * grid = (N/128 n-blocks) x KSPLIT k-slices -> >= 512 WGs on every
-//    llama/falcon shape. WG = 4 waves; wave owns 32 W rows.
-//  * x^T slice staged once to LDS in B-fragment-ready layout.
-//  * main loop unrolled 8x: 8 independent nontemporal A-fragment loads
-//    (16 B/lane, perfectly coalesced, nt keeps the one-pass W stream out
-//    of L2 — MI355X_MICROARCH.md nt-weights) issued back-to-back, then 8
-//    ds_read_b128 + 8 v_mfma_f32_32x32x16_bf16. ~8 loads in flight per
-//    wave x 8+ waves/CU covers the ~900-cycle HBM latency
-//    (cdna_hip_programming.md Guideline 7).
-//  * split-K partials go to fp32 slabs with PLAIN stores; the kernel
-//    boundary is the release, and a small combine kernel reduces
-//    KSPLIT slabs -> bf16 y (no in-kernel cross-WG sync: the
-//    boundary costs ~1.4 us, an agent-scope fence storm costs far more
-//    — MI355X_MICROARCH.md boundary vs splitk-seam rows).
-//
-// mfma_f32_32x32x16_bf16 layouts as in attention_prefill.hip.
+//  * grid = (N/64) x (K/256): each WG owns a [64 rows x 256 k] W tile
+//    (32 KB) -> >= 1024 WGs on every llama shape, 3 WGs/CU by LDS, and
+//    split-K slab traffic stays ~25% of the W stream (fp32 partials:
+//    M*4 bytes per row-slice vs KT*2 of W).
+//  * W tile is staged global->LDS with nontemporal 16 B/lane loads that
+//    walk the tile row-major (256 B contiguous per row, rows adjacent in
+//    the k-slice) — proper bursts, unlike direct MFMA A-fragment loads
+//    whose 32 B/row at 8 KB stride waste the DRAM granule (that design
+//    measured 1.5 TB/s; this file's history).
+//  * x^T slice staged once in B-fragment-ready layout (tiny).
+//  * main loop is pure LDS + MFMA: per 16-k step one ds_read_b128 (A,
+//    row stride 272 B ≡ 4 mod 64 dwords -> conflict-free 16-lane groups,
+//    same padding trick as attention_prefill.hip) + one ds_read_b128 (B)
+//    + one v_mfma_f32_32x32x16_bf16.
+//  * split-K partials: plain fp32 slab stores; the kernel boundary is
+//    the release and skinny_combine reduces slabs -> bf16 y.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -29,25 +30,26 @@
 
 namespace {
 
-constexpr int BLOCK = 256;        // 4 waves
-constexpr int NB = 128;           // W rows per WG (32 per wave)
+constexpr int BLOCK = 128;        // 2 waves; wave owns 32 W rows
+constexpr int NB = 64;            // W rows per WG
+constexpr int KT = 256;           // k per WG
 constexpr int MMAX = 32;
-constexpr int UNROLL = 8;         // 16-k steps in flight
+constexpr int W_STRIDE = KT * 2 + 16;  // LDS row bytes (+16: bank offset 4)
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 typedef __attribute__((ext_vector_type(16))) float f32x16v;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4v;
 
-__device__ __forceinline__ bf16x8v nt_load_frag(const uint16_t *p) {
-  union { u32x4v u; bf16x8v v; } c;
+__device__ __forceinline__ rb::bf16x8 nt_load8(const uint16_t *p) {
+  union { u32x4v u; rb::bf16x8 v; } c;
   c.u = __builtin_nontemporal_load(reinterpret_cast<const u32x4v *>(p));
   return c.v;
 }
 
 // slabs layout: [nblk][kslice][NB][MMAX] f32
-__global__ __launch_bounds__(BLOCK, 4) void skinny_gemm_kernel(
+__global__ __launch_bounds__(BLOCK, 2) void skinny_gemm_kernel(
     const uint16_t *__restrict__ xp, const uint16_t *__restrict__ wp,
-    float *__restrict__ slabs, int M, int N, int K, int ksplit) {
+    float *__restrict__ slabs, int M, int N, int K) {
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
@@ -56,15 +58,30 @@ __global__ __launch_bounds__(BLOCK, 4) void skinny_gemm_kernel(
 
   const int nblk = blockIdx.x;
   const int kslice = blockIdx.y;
-  const int kper = K / ksplit;
-  const int k0 = kslice * kper;
+  const int k0 = kslice * KT;
+  const int ksplit = gridDim.y;
 
-  // ---- stage x^T slice in B-fragment-ready layout -----------------------
-  // slot s = (16-k step)*64 + lane: 8 contiguous k of x row (lane&31).
-  extern __shared__ __attribute__((aligned(16))) uint16_t xfrag[];
+  __shared__ __attribute__((aligned(16))) char w_img[NB * W_STRIDE];
+  __shared__ __attribute__((aligned(16))) uint16_t xfrag[KT * MMAX];
+
+  // ---- stage W tile, coalesced + nontemporal (one-pass stream) ----------
   {
-    const int slots = (kper / 16) * 64;
-    for (int s = tid; s < slots; s += BLOCK) {
+    constexpr int PER_ROW = KT / 8;         // 16B slots per row (16)
+    constexpr int SLOTS = NB * PER_ROW;     // 2048
+#pragma unroll
+    for (int p = 0; p < SLOTS / BLOCK; ++p) {
+      const int s = p * BLOCK + tid;
+      const int r = s / PER_ROW;
+      const int c8 = (s % PER_ROW) * 8;
+      rb::bf16x8 v = nt_load8(
+          wp + (int64_t)(nblk * NB + r) * K + k0 + c8);
+      *reinterpret_cast<rb::bf16x8 *>(w_img + r * W_STRIDE + c8 * 2) = v;
+    }
+  }
+  // ---- stage x^T slice in B-fragment-ready layout -----------------------
+  {
+    constexpr int SLOTS = (KT / 16) * 64;   // 512
+    for (int s = tid; s < SLOTS; s += BLOCK) {
       const int ks = s >> 6;
       const int l = s & 63;
       const int m = l & 31;
@@ -81,33 +98,18 @@ __global__ __launch_bounds__(BLOCK, 4) void skinny_gemm_kernel(
   }
   __syncthreads();
 
-  // ---- main loop: UNROLL A loads in flight, then the MFMAs --------------
-  const uint16_t *wrow =
-      wp + (int64_t)(nblk * NB + wid * 32 + col) * K + k0 + hi * 8;
+  // ---- LDS -> MFMA ------------------------------------------------------
   f32x16v acc = (f32x16v)(0.0f);
-  const int steps = kper / 16;
-  int s = 0;
-  for (; s + UNROLL <= steps; s += UNROLL) {
-    bf16x8v a[UNROLL];
 #pragma unroll
-    for (int u = 0; u < UNROLL; ++u)
-      a[u] = nt_load_frag(wrow + (s + u) * 16);
-#pragma unroll
-    for (int u = 0; u < UNROLL; ++u) {
-      const bf16x8v b = *reinterpret_cast<const bf16x8v *>(
-          xfrag + (s + u) * 512 + (lane << 3));
-      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[u], b, acc, 0, 0, 0);
-    }
-  }
-  for (; s < steps; ++s) {
-    const bf16x8v a = nt_load_frag(wrow + s * 16);
+  for (int s = 0; s < KT / 16; ++s) {
+    const bf16x8v a = *reinterpret_cast<const bf16x8v *>(
+        w_img + (wid * 32 + col) * W_STRIDE + (s * 16 + hi * 8) * 2);
     const bf16x8v b = *reinterpret_cast<const bf16x8v *>(
         xfrag + s * 512 + (lane << 3));
     acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
   }
 
-  // ---- write this slice's partial tile (plain stores; the kernel
-  // boundary orders them before the combine kernel) -----------------------
+  // ---- partial tile -> slab (plain stores; boundary = release) ----------
   float *slab = slabs + (((int64_t)nblk * ksplit + kslice) * NB) * MMAX;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
@@ -153,36 +155,33 @@ SkinnyWorkspace &ws_for(const at::Tensor &ref, int64_t need) {
 
 int64_t skinny_gemm_mmax() { return MMAX; }
 
+bool skinny_gemm_supported(int64_t M, int64_t N, int64_t K) {
+  return M <= MMAX && N % NB == 0 && K % KT == 0;
+}
+
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous(),
               "skinny_gemm: contiguous GPU tensors");
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               w.scalar_type() == at::kBFloat16, "skinny_gemm: bf16 only");
   const int M = (int)x.size(0), K = (int)x.size(1), N = (int)w.size(0);
-  TORCH_CHECK(M <= MMAX, "skinny_gemm: M must be <= ", MMAX);
   TORCH_CHECK((int)w.size(1) == K, "skinny_gemm: K mismatch");
-  TORCH_CHECK(N % NB == 0, "skinny_gemm: N % 128");
-  TORCH_CHECK(K % 16 == 0, "skinny_gemm: K % 16");
+  TORCH_CHECK(skinny_gemm_supported(M, N, K),
+              "skinny_gemm: unsupported shape ", M, "x", N, "x", K);
 
   const int nblocks = N / NB;
-  int ksplit = 1;
-  while (nblocks * ksplit < 512 && ksplit < 16 &&
-         (K / (ksplit * 2)) % 16 == 0 && K / (ksplit * 2) >= 128)
-    ksplit *= 2;
-
+  const int ksplit = K / KT;
   auto y = at::empty({M, N}, x.options());
   auto &ws = ws_for(x, (int64_t)nblocks * ksplit * NB * MMAX);
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t shmem = (size_t)(K / ksplit) * MMAX * sizeof(uint16_t);
-  TORCH_CHECK(shmem <= 160 * 1024, "skinny_gemm: K/ksplit too large");
   hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, ksplit), dim3(BLOCK),
-                     shmem, stream,
+                     0, stream,
                      (const uint16_t *)x.data_ptr(),
                      (const uint16_t *)w.data_ptr(),
-                     ws.slabs.data_ptr<float>(), M, N, K, ksplit);
+                     ws.slabs.data_ptr<float>(), M, N, K);
   const int cgrid = rb::rb_grid_1d((int64_t)N * M, 256);
   hipLaunchKernelGGL(skinny_combine_kernel, dim3(cgrid), dim3(256), 0,
                      stream, ws.slabs.data_ptr<float>(),
-                     (uint16_t *)y.data_ptr(), M, N, ksplit);
+                     (uint16_t *)y.data_ptr(), M, N, K / KT);
   return y;
 }

@@ -21,7 +21,7 @@ def fast_linear(x: torch.Tensor, weight: torch.Tensor,
         k = x.shape[-1]
         m = x.numel() // k
         n = weight.shape[0]
-        if m <= 32 and n % 128 == 0 and k % 16 == 0:
+        if m <= 32 and n % 64 == 0 and k % 256 == 0:
             y = _backend.ext().skinny_gemm(x.reshape(m, k).contiguous(),
                                            weight)
             return y.view(*x.shape[:-1], n)

@@ -235,7 +235,8 @@ def test_train_step_gpu():
     (32, 11008, 4096),     # mlp up
     (32, 4096, 11008),     # mlp down (odd K split)
     (17, 32000, 4096),     # lm_head, ragged M
-    (8, 128, 64),          # minimal single block, ksplit=1 path
+    (8, 128, 256),         # minimal grid
+    (32, 4096, 512),       # falcon-7b-ish head shard
 ])
 def test_skinny_gemm(M, N, K):
     """csrc/skinny_gemm.hip vs fp32 matmul."""
