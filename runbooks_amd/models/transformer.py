@@ -73,9 +73,20 @@ class Attention(nn.Module):
     def _qkv(self, x, cos, sin, positions):
         # x: [T, hidden] token-major
         T = x.shape[0]
-        q = self.q_proj(x).view(T, self.hq, self.dh)
-        k = self.k_proj(x).view(T, self.hkv, self.dh)
-        v = self.v_proj(x).view(T, self.hkv, self.dh)
+        if getattr(self, "_qkv_w", None) is not None and \
+                not torch.is_grad_enabled():
+            # fused inference path (fuse_for_inference): one wide GEMM
+            # instead of three skinny ones
+            qo = self.hq * self.dh
+            kvo = self.hkv * self.dh
+            y = ops.fast_linear(x, self._qkv_w)
+            q = y[:, :qo].contiguous().view(T, self.hq, self.dh)
+            k = y[:, qo:qo + kvo].contiguous().view(T, self.hkv, self.dh)
+            v = y[:, qo + kvo:].contiguous().view(T, self.hkv, self.dh)
+        else:
+            q = self.q_proj(x).view(T, self.hq, self.dh)
+            k = self.k_proj(x).view(T, self.hkv, self.dh)
+            v = self.v_proj(x).view(T, self.hkv, self.dh)
         if self.cfg.pos == "rope":
             q = ops.rope(q, cos, sin, positions)
             k = ops.rope(k, cos, sin, positions)
@@ -125,6 +136,12 @@ class MLP(nn.Module):
 
     def forward(self, x):
         if self.act == "silu_glu":
+            if getattr(self, "_gateup_w", None) is not None and \
+                    not torch.is_grad_enabled():
+                y = ops.fast_linear(x, self._gateup_w)
+                half = y.shape[-1] // 2
+                return self.down_proj(ops.swiglu(
+                    y[:, :half].contiguous(), y[:, half:].contiguous()))
             return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
         h = self.fc1(x)
         h = torch.nn.functional.gelu(h) if self.act == "gelu" else torch.relu(h)
@@ -247,6 +264,45 @@ class Transformer(nn.Module):
         return [ops.alloc_kv_cache(num_blocks, self.local_kv_heads(),
                                    self.cfg.head_dim, device, dtype=self.dtype)
                 for _ in range(self.cfg.num_layers)]
+
+
+def fuse_for_inference(model: "Transformer") -> "Transformer":
+    """Fuse each block's QKV and gate/up weights into single tensors for
+    serving: one wide GEMM per group instead of 2-3 skinny ones (the
+    batch-<=32 decode GEMMs are weight-bandwidth bound, and hipBLASLt is
+    markedly more efficient at the fused N — profiles/). The original
+    parameters become views into the fused tensor, so no extra HBM and
+    the separate-projection training path keeps working.
+    """
+    for blk in model.blocks:
+        attn = blk.attn
+        if (attn.q_proj.bias is None and
+                getattr(attn.k_proj, "bias", None) is None and
+                attn.q_proj.weight.shape[1] == attn.k_proj.weight.shape[1]):
+            fused = torch.cat([attn.q_proj.weight.data,
+                               attn.k_proj.weight.data,
+                               attn.v_proj.weight.data], dim=0).contiguous()
+            qo = attn.q_proj.weight.shape[0]
+            kvo = attn.k_proj.weight.shape[0]
+            attn._qkv_w = fused
+            attn.q_proj.weight = nn.Parameter(fused[:qo],
+                                              requires_grad=False)
+            attn.k_proj.weight = nn.Parameter(fused[qo:qo + kvo],
+                                              requires_grad=False)
+            attn.v_proj.weight = nn.Parameter(fused[qo + kvo:],
+                                              requires_grad=False)
+        mlp = blk.mlp
+        if model.cfg.act == "silu_glu" and mlp.gate_proj.bias is None:
+            fused = torch.cat([mlp.gate_proj.weight.data,
+                               mlp.up_proj.weight.data],
+                              dim=0).contiguous()
+            half = mlp.gate_proj.weight.shape[0]
+            mlp._gateup_w = fused
+            mlp.gate_proj.weight = nn.Parameter(fused[:half],
+                                                requires_grad=False)
+            mlp.up_proj.weight = nn.Parameter(fused[half:],
+                                              requires_grad=False)
+    return model
 
 
 def build_model(name_or_cfg, dtype=torch.bfloat16, tp: int | None = None,

@@ -7,15 +7,21 @@ on hipBLASLt via F.linear — the library is the right tool for big GEMMs.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn.functional as F
 
 from . import _backend
 
+# The custom kernel currently trails hipBLASLt on most decode shapes
+# (see profiles/README.md measurements); keep it opt-in until it wins.
+_USE_SKINNY = os.environ.get("RB_SKINNY_GEMM", "0") == "1"
+
 
 def fast_linear(x: torch.Tensor, weight: torch.Tensor,
                 bias: torch.Tensor | None = None) -> torch.Tensor:
-    if (bias is None and not torch.is_grad_enabled()
+    if (_USE_SKINNY and bias is None and not torch.is_grad_enabled()
             and x.dtype == torch.bfloat16 and _backend.use_hip(x)
             and weight.is_contiguous()):
         k = x.shape[-1]

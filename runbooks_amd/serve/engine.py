@@ -64,6 +64,9 @@ class Engine:
         if isinstance(model, str):
             model = build_model(model, dtype=dtype, seed=seed)
         self.model = model.to(self.device).eval()
+        if torch.cuda.is_available() and model.tp == 1:
+            from ..models.transformer import fuse_for_inference
+            fuse_for_inference(self.model)
         self.cfg = model.cfg
         self.bs = ops.BLOCK_SIZE
         self.max_batch = max_batch

@@ -90,3 +90,22 @@ def test_block_allocator_release():
     free0 = len(eng.allocator.free)
     eng.generate([1, 2, 3], max_new_tokens=3)
     assert len(eng.allocator.free) == free0, "blocks leaked"
+
+
+def test_fuse_for_inference_matches_unfused():
+    """Fused QKV / gate-up serving path == separate projections."""
+    import torch
+    from runbooks_amd.models import build_model
+    from runbooks_amd.models.transformer import fuse_for_inference
+
+    m = build_model("tiny-llama", dtype=torch.float32, tp=1, seed=5)
+    tokens = torch.randint(0, 256, (2, 9))
+    with torch.no_grad():
+        ref = m(tokens)
+    fuse_for_inference(m)
+    with torch.no_grad():
+        got = m(tokens)
+    assert torch.allclose(got, ref, atol=1e-6)
+    # parameters are views into the fused tensors (no weight duplication)
+    blk = m.blocks[0]
+    assert blk.attn.q_proj.weight.data_ptr() == blk.attn._qkv_w.data_ptr()
