@@ -110,11 +110,20 @@ class Attention(nn.Module):
 
     def forward_decode(self, x, cos, sin, positions, kc, vc, slot_mapping,
                        block_tables, seq_lens, nsplit=None):
-        q, k, v = self._qkv(x, cos, sin, positions)
-        ops.kv_append(k, v, kc, vc, slot_mapping)
+        B = x.shape[0]
+        if getattr(self, "_qkv_w", None) is not None and \
+                self.cfg.pos == "rope" and ops.use_hip(x) and \
+                x.dtype == torch.bfloat16:
+            # fused decode hot path: one wide GEMM, then one kernel doing
+            # rope(q), rope(k)->cache, v->cache (csrc/qkv_fused.hip)
+            y = ops.fast_linear(x, self._qkv_w)
+            q = ops.ext().qkv_rope_append(y, cos, sin, positions, kc, vc,
+                                          slot_mapping, self.hq)
+        else:
+            q, k, v = self._qkv(x, cos, sin, positions)
+            ops.kv_append(k, v, kc, vc, slot_mapping)
         o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale,
                              nsplit=nsplit)
-        B = x.shape[0]
         return self.o_proj(o.reshape(B, self.hq * self.dh))
 
 
@@ -139,6 +148,8 @@ class MLP(nn.Module):
             if getattr(self, "_gateup_w", None) is not None and \
                     not torch.is_grad_enabled():
                 y = ops.fast_linear(x, self._gateup_w)
+                if ops.use_hip(y):
+                    return self.down_proj(ops.ext().swiglu_packed(y))
                 half = y.shape[-1] // 2
                 return self.down_proj(ops.swiglu(
                     y[:, :half].contiguous(), y[:, half:].contiguous()))

@@ -37,6 +37,11 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
 at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
 int64_t skinny_gemm_mmax();
+at::Tensor qkv_rope_append(at::Tensor y, at::Tensor cos, at::Tensor sin,
+                           at::Tensor positions, at::Tensor k_cache,
+                           at::Tensor v_cache, at::Tensor slot_mapping,
+                           int64_t hq);
+at::Tensor swiglu_packed(at::Tensor y);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "runbooks_amd gfx950 HIP kernels";
@@ -66,4 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm,
         "HBM-rate decode GEMM: y[M<=32,N] = x @ W^T (bf16)");
   m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
+  m.def("qkv_rope_append", &qkv_rope_append,
+        "packed qkv -> rope'd q + rope'd k / v appended to the paged cache");
+  m.def("swiglu_packed", &swiglu_packed,
+        "silu(y[:, :I]) * y[:, I:] from the fused gate/up GEMM output");
 }

@@ -267,6 +267,45 @@ def test_fast_linear_dispatch():
     assert xg.grad is not None
 
 
+@pytest.mark.parametrize("hq,hkv,dh", [(8, 8, 128), (8, 2, 128), (16, 1, 64)])
+def test_qkv_rope_append(hq, hkv, dh):
+    """Fused packed-qkv epilogue vs the composed rope + kv_append ops."""
+    _assert_hip()
+    torch.manual_seed(hq)
+    T, BS, nblocks = 7, 16, 8
+    y = torch.randn(T, (hq + 2 * hkv) * dh, dtype=torch.bfloat16, device=DEV)
+    cos, sin = ops.rope_tables(dh, 64, device=DEV)
+    pos = torch.randint(0, 64, (T,), dtype=torch.int32, device=DEV)
+    slots = torch.randperm(nblocks * BS, device=DEV)[:T].to(torch.int32)
+
+    kc = torch.zeros(nblocks, hkv, BS, dh, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    q = ops.ext().qkv_rope_append(y, cos, sin, pos, kc, vc, slots, hq)
+
+    qo, kvo = hq * dh, hkv * dh
+    q_ref = ops.rope(y[:, :qo].contiguous().view(T, hq, dh), cos, sin, pos)
+    k_ref = ops.rope(y[:, qo:qo + kvo].contiguous().view(T, hkv, dh),
+                     cos, sin, pos)
+    v_ref = y[:, qo + kvo:].contiguous().view(T, hkv, dh)
+    kc_ref = torch.zeros_like(kc)
+    vc_ref = torch.zeros_like(vc)
+    ops.kv_append(k_ref, v_ref, kc_ref, vc_ref, slots)
+
+    assert torch.allclose(q.float(), q_ref.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(kc.float(), kc_ref.float(), atol=2e-2, rtol=2e-2)
+    assert torch.equal(vc, vc_ref)
+
+
+def test_swiglu_packed():
+    _assert_hip()
+    torch.manual_seed(0)
+    y = torch.randn(33, 2 * 1024, dtype=torch.bfloat16, device=DEV)
+    out = ops.ext().swiglu_packed(y)
+    g, u = y[:, :1024].contiguous(), y[:, 1024:].contiguous()
+    ref = torch.nn.functional.silu(g.float()) * u.float()
+    assert torch.allclose(out.float(), ref, atol=2e-2, rtol=2e-2)
+
+
 def test_mfma_32x32x16_layout_probe():
     """One MFMA vs torch matmul — pinpoints a wrong fragment map."""
     _assert_hip()

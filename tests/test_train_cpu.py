@@ -41,3 +41,31 @@ def test_trainer_save_resume(tmp_path):
     tr2 = Trainer(cfg)
     assert tr2.resume()
     assert tr2.step_num == 3
+
+
+def test_lora_fused_backward_matches_eager():
+    """_LoRAFused (GEMM-only backward) vs plain autograd over the same math."""
+    import torch
+    from runbooks_amd.train.lora import _LoRAFused
+
+    torch.manual_seed(3)
+    T, IN, OUT, r, s = 9, 32, 24, 4, 2.0
+    x = torch.randn(T, IN, requires_grad=True)
+    w = torch.randn(OUT, IN)
+    a = torch.randn(r, IN, requires_grad=True)
+    b = torch.randn(OUT, r, requires_grad=True)
+    dy = torch.randn(T, OUT)
+
+    y = _LoRAFused.apply(x, w, a, b, s)
+    y.backward(dy)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    a2 = a.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = x2 @ w.t() + s * (x2 @ a2.t()) @ b2.t()
+    y2.backward(dy)
+
+    assert torch.allclose(y, y2, atol=1e-5)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+    assert torch.allclose(a.grad, a2.grad, atol=1e-5)
+    assert torch.allclose(b.grad, b2.grad, atol=1e-5)
