@@ -39,7 +39,9 @@
 
 namespace {
 
-constexpr int BLOCK = 512;  // 8 waves
+constexpr int BLOCK = 256;  // 4 waves (128-row WGs: finer grid
+                            // halves the causal-triangle imbalance and
+                            // doubles WG count vs 8-wave WGs)
 constexpr int TILE = 64;    // rows staged to LDS per loop step
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
@@ -133,7 +135,7 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dq_kernel(
   const int hi = lane >> 5;
   const int col = lane & 31;
 
-  const int q0 = blockIdx.x * 256;
+  const int q0 = blockIdx.x * 128;
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int h_kv = h / (Hq / Hkv);
@@ -167,7 +169,7 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dq_kernel(
 #pragma unroll
   for (int dt = 0; dt < DTILES; ++dt) acc_dq[dt] = (f32x16v)(0.0f);
 
-  const int kv_end = min(S, q0 + 256);
+  const int kv_end = min(S, q0 + 128);
   const int n_tiles = (kv_end + TILE - 1) / TILE;
   const int wave_kmax = q0 + wid * 32 + 31;
 
@@ -312,7 +314,7 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dkdv_kernel(
   const int hi = lane >> 5;
   const int col = lane & 31;
 
-  const int k0 = blockIdx.x * 256;
+  const int k0 = blockIdx.x * 128;
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int h_kv = h / (Hq / Hkv);
@@ -535,7 +537,7 @@ std::vector<at::Tensor> fa_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                          dvec.data_ptr<float>(), B, S, Hq);
   }
 
-  const dim3 grid((S + 255) / 256, Hq, B);
+  const dim3 grid((S + 127) / 128, Hq, B);
 #define RB_LAUNCH_BWD(DHV)                                                    \
   do {                                                                        \
     constexpr int ROW_STRIDE = DHV * 2 + 16;                                  \
