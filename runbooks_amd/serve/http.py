@@ -63,9 +63,22 @@ class EngineLoop:
             with self._lock:
                 try:
                     finished = self.engine.step()
-                except RuntimeError:
-                    # oversize prompt rejected by the scheduler: notify all
-                    # watchers whose request vanished from the queues
+                    self._failures = 0
+                except Exception:
+                    # a rejected request is dropped from the queues —
+                    # notify its watcher; on repeated failures (a request
+                    # that keeps crashing the step) fail everything rather
+                    # than spin
+                    self._failures = getattr(self, "_failures", 0) + 1
+                    if self._failures >= 3:
+                        for r in (self.engine.running +
+                                  self.engine.waiting):
+                            r.finished = True
+                            if r.blocks:
+                                self.engine.allocator.release(r.blocks)
+                                r.blocks = []
+                        self.engine.running.clear()
+                        self.engine.waiting.clear()
                     alive = {r.request_id for r in
                              self.engine.running + self.engine.waiting}
                     for rid in list(self._watchers):
@@ -99,6 +112,36 @@ def build_app(engine: Engine, tokenizer=None,
     loop = EngineLoop(engine)
     app.state.engine_loop = loop
 
+    # Prometheus metrics (the reference exposes controller metrics behind
+    # kube-rbac-proxy; the serving runtime gets request/token counters and
+    # latency histograms — SURVEY.md §5 observability)
+    try:
+        from prometheus_client import (
+            CONTENT_TYPE_LATEST,
+            Counter,
+            Gauge,
+            Histogram,
+            generate_latest,
+        )
+        m_reqs = Counter("rb_requests_total", "completion requests")
+        m_tokens = Counter("rb_generated_tokens_total", "generated tokens")
+        m_lat = Histogram("rb_request_seconds", "request latency",
+                          buckets=(.05, .1, .25, .5, 1, 2.5, 5, 10, 30, 60))
+        m_running = Gauge("rb_running_requests", "requests decoding")
+        m_kv_free = Gauge("rb_kv_blocks_free", "free KV cache blocks")
+
+        @app.get("/metrics")
+        def metrics():
+            m_running.set(len(engine.running))
+            m_kv_free.set(len(engine.allocator.free))
+            from fastapi import Response
+            return Response(generate_latest(),
+                            media_type=CONTENT_TYPE_LATEST)
+    except ImportError:
+        m_reqs = m_tokens = m_lat = None
+
+    app.state.metrics = (m_reqs, m_tokens, m_lat)
+
     @app.get("/")
     def ready():
         return {"status": "ok", "model": model_name}
@@ -115,6 +158,10 @@ def build_app(engine: Engine, tokenizer=None,
     def completions(req: CompletionRequest):
         ids = tok.encode(req.prompt)[-engine.cfg.max_seq_len +
                                      req.max_tokens + 1:]
+        # guard: a fallback tokenizer may emit ids past a small model's
+        # vocab (identity for any properly paired tokenizer)
+        vocab = engine.cfg.vocab_size
+        ids = [i % vocab for i in ids]
         cid = f"cmpl-{uuid.uuid4().hex[:12]}"
         t0 = int(time.time())
         q = loop.submit(ids, req.max_tokens, req.temperature)
@@ -144,6 +191,10 @@ def build_app(engine: Engine, tokenizer=None,
             if t is None:
                 break
             out.append(t)
+        if m_reqs is not None:
+            m_reqs.inc()
+            m_tokens.inc(len(out))
+            m_lat.observe(time.time() - t0)
         return JSONResponse({
             "id": cid,
             "object": "text_completion",
