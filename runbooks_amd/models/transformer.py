@@ -277,7 +277,8 @@ class Transformer(nn.Module):
                 for _ in range(self.cfg.num_layers)]
 
 
-def fuse_for_inference(model: "Transformer") -> "Transformer":
+def fuse_for_inference(model: "Transformer",
+                       load_in_8bit: bool = False) -> "Transformer":
     """Fuse each block's QKV and gate/up weights into single tensors for
     serving: one wide GEMM per group instead of 2-3 skinny ones (the
     batch-<=32 decode GEMMs are weight-bandwidth bound, and hipBLASLt is
@@ -313,6 +314,25 @@ def fuse_for_inference(model: "Transformer") -> "Transformer":
                                                 requires_grad=False)
             mlp.up_proj.weight = nn.Parameter(fused[half:],
                                               requires_grad=False)
+
+    if load_in_8bit and torch.cuda.is_available():
+        # MODEL_LOAD_IN_8BIT: decode weights additionally stored as OCP
+        # e4m3 + per-channel scales — the decode GEMM streams half the
+        # bytes (ops/csrc/skinny_gemm.hip fp8 path). bf16 masters stay for
+        # prefill/big-M GEMMs: 288 GB HBM3E makes both-copies the right
+        # trade on MI355X.
+        from ..ops.linear import quantize_fp8
+        for blk in model.blocks:
+            for w in (getattr(blk.attn, "_qkv_w", None),
+                      getattr(blk.mlp, "_gateup_w", None),
+                      blk.attn.o_proj.weight.data,
+                      blk.mlp.down_proj.weight.data):
+                if w is not None and w.shape[0] % 64 == 0 and \
+                        w.shape[1] % 256 == 0:
+                    quantize_fp8(w)
+        if model.lm_head.weight.shape[0] % 64 == 0 and \
+                model.lm_head.weight.shape[1] % 256 == 0:
+            quantize_fp8(model.lm_head.weight.data)
     return model
 
 

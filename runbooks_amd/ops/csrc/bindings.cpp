@@ -36,6 +36,7 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor l
                              double gscale, int64_t ignore_index);
 at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
+at::Tensor skinny_gemm_fp8(at::Tensor x, at::Tensor w8, at::Tensor scale);
 int64_t skinny_gemm_mmax();
 at::Tensor qkv_rope_append(at::Tensor y, at::Tensor cos, at::Tensor sin,
                            at::Tensor positions, at::Tensor k_cache,
@@ -71,6 +72,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm,
         "HBM-rate decode GEMM: y[M<=32,N] = x @ W^T (bf16)");
   m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
+  m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
+        "fp8-e4m3 weight-only decode GEMM with per-channel scales");
   m.def("qkv_rope_append", &qkv_rope_append,
         "packed qkv -> rope'd q + rope'd k / v appended to the paged cache");
   m.def("swiglu_packed", &swiglu_packed,

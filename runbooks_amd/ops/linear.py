@@ -19,18 +19,48 @@ from . import _backend
 _USE_SKINNY = os.environ.get("RB_SKINNY_GEMM", "0") == "1"
 
 
+# data_ptr(weight) -> (w8 uint8 view, f32 per-channel scale). Populated by
+# quantize_fp8 (serving with MODEL_LOAD_IN_8BIT); looked up on the decode
+# fast path.
+_FP8_REGISTRY: dict[int, tuple[torch.Tensor, torch.Tensor]] = {}
+
+E4M3_MAX = 448.0
+
+
+def quantize_fp8(weight: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Row-wise (per-output-channel) OCP e4m3 quantization; registers the
+    pair so fast_linear dispatches to the fp8 decode GEMM."""
+    scale = weight.float().abs().amax(dim=1).clamp(min=1e-8) / E4M3_MAX
+    w8 = (weight.float() / scale[:, None]).to(torch.float8_e4m3fn)
+    w8_bytes = w8.view(torch.uint8).contiguous()
+    _FP8_REGISTRY[weight.data_ptr()] = (w8_bytes, scale.contiguous())
+    return w8_bytes, scale
+
+
+def dequantize_fp8(w8_bytes: torch.Tensor, scale: torch.Tensor,
+                   dtype=torch.bfloat16) -> torch.Tensor:
+    return (w8_bytes.view(torch.float8_e4m3fn).float() *
+            scale[:, None]).to(dtype)
+
+
 def fast_linear(x: torch.Tensor, weight: torch.Tensor,
                 bias: torch.Tensor | None = None) -> torch.Tensor:
-    if (_USE_SKINNY and bias is None and not torch.is_grad_enabled()
+    if (bias is None and not torch.is_grad_enabled()
             and x.dtype == torch.bfloat16 and _backend.use_hip(x)
             and weight.is_contiguous()):
         k = x.shape[-1]
         m = x.numel() // k
         n = weight.shape[0]
         if m <= 32 and n % 64 == 0 and k % 256 == 0:
-            y = _backend.ext().skinny_gemm(x.reshape(m, k).contiguous(),
-                                           weight)
-            return y.view(*x.shape[:-1], n)
+            q = _FP8_REGISTRY.get(weight.data_ptr())
+            if q is not None:
+                y = _backend.ext().skinny_gemm_fp8(
+                    x.reshape(m, k).contiguous(), q[0], q[1])
+                return y.view(*x.shape[:-1], n)
+            if _USE_SKINNY:
+                y = _backend.ext().skinny_gemm(x.reshape(m, k).contiguous(),
+                                               weight)
+                return y.view(*x.shape[:-1], n)
     return F.linear(x, weight, bias)
 
 

@@ -58,7 +58,8 @@ class Request:
 class Engine:
     def __init__(self, model: Transformer | str, device=None,
                  dtype=torch.bfloat16, kv_blocks: int | None = None,
-                 max_batch: int = 64, mem_fraction: float = 0.85, seed: int = 0):
+                 max_batch: int = 64, mem_fraction: float = 0.85, seed: int = 0,
+                 load_in_8bit: bool = False):
         self.device = device if device is not None else (
             f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
         if isinstance(model, str):
@@ -66,7 +67,7 @@ class Engine:
         self.model = model.to(self.device).eval()
         if torch.cuda.is_available() and model.tp == 1:
             from ..models.transformer import fuse_for_inference
-            fuse_for_inference(self.model)
+            fuse_for_inference(self.model, load_in_8bit=load_in_8bit)
         self.cfg = model.cfg
         self.bs = ops.BLOCK_SIZE
         self.max_batch = max_batch

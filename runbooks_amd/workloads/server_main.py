@@ -68,7 +68,8 @@ def main():
                         tp=comm.world_size(), strict=False)
         print(f"server: loaded weights from {model_dir}")
 
-    engine = Engine(model)
+    in_8bit = os.environ.get("MODEL_LOAD_IN_8BIT", "").lower() in ("1", "true")
+    engine = Engine(model, load_in_8bit=in_8bit)
     tok = load_tokenizer(model_dir if model_dir.exists() else None)
     if comm.rank() == 0:
         serve_forever(engine, tok, port=int(os.environ.get("PORT", "8080")),

@@ -411,3 +411,24 @@ def test_cross_entropy_fwd_bwd(V):
     assert abs(float(loss) - float(ref)) < 2e-2, (float(loss), float(ref))
     assert torch.allclose(logits.grad.cpu().float(), lc.grad, atol=2e-3,
                           rtol=2e-2), (logits.grad.cpu().float() - lc.grad).abs().max()
+
+
+def test_skinny_gemm_fp8():
+    """fp8 weight-only decode GEMM vs matmul on the dequantized weights
+    (exact dequant path) and vs the original bf16 weights (e4m3 bound)."""
+    _assert_hip()
+    from runbooks_amd.ops.linear import dequantize_fp8, quantize_fp8, _FP8_REGISTRY
+    torch.manual_seed(4)
+    M, N, K = 32, 4096, 4096
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+    w8, scale = quantize_fp8(w)
+    y = ops.ext().skinny_gemm_fp8(x, w8, scale)
+    deq = dequantize_fp8(w8, scale).to(DEV)
+    ref = (x.float() @ deq.float().t())
+    d = (y.float() - ref).abs().max() / ref.abs().max()
+    assert d < 2e-2, f"vs dequant rel {d}"
+    full = (x.float() @ w.float().t())
+    d2 = (y.float() - full).abs().max() / full.abs().max()
+    assert d2 < 0.08, f"vs bf16 rel {d2}"
+    _FP8_REGISTRY.clear()

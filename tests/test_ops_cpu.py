@@ -117,3 +117,19 @@ def test_causal_attention_matches_naive():
             s = s.masked_fill(mask, float("-inf"))
             o = torch.softmax(s, -1) @ vv[b, :, h]
             assert torch.allclose(out[b, :, h], o, atol=1e-5)
+
+
+def test_fp8_rowwise_quantization_roundtrip():
+    """quantize_fp8 / dequantize_fp8: per-channel e4m3 error bound."""
+    import torch
+    from runbooks_amd.ops.linear import dequantize_fp8, quantize_fp8, _FP8_REGISTRY
+
+    torch.manual_seed(0)
+    w = torch.randn(64, 256, dtype=torch.bfloat16) * 3.0
+    w8, scale = quantize_fp8(w)
+    assert w8.dtype == torch.uint8 and scale.shape == (64,)
+    back = dequantize_fp8(w8, scale)
+    rel = (back.float() - w.float()).abs().max() / w.float().abs().max()
+    assert rel < 0.08, rel  # e4m3 has ~2 mantissa bits at full scale
+    assert _FP8_REGISTRY.get(w.data_ptr()) is not None
+    _FP8_REGISTRY.clear()
