@@ -153,6 +153,84 @@ class ControllerManager:
                 self._stop.wait(1.0)
 
 
+class LeaderElector:
+    """Lease-based leader election (parity: the reference manager's
+    --leader-elect flag, reference cmd/controllermanager/main.go:67-68,
+    backed by controller-runtime's coordination.k8s.io Lease lock)."""
+
+    def __init__(self, kube: KubeClient, identity: str,
+                 namespace: str = "substratus",
+                 name: str = "runbooks-amd-controller-manager",
+                 lease_seconds: int = 15):
+        self.kube = kube
+        self.identity = identity
+        self.namespace = namespace
+        self.name = name
+        self.lease_seconds = lease_seconds
+
+    def _now(self) -> str:
+        import datetime
+        return datetime.datetime.now(datetime.timezone.utc).strftime(
+            "%Y-%m-%dT%H:%M:%S.%f")[:-3] + "Z"
+
+    def _lease_obj(self) -> dict:
+        return {
+            "apiVersion": "coordination.k8s.io/v1", "kind": "Lease",
+            "metadata": {"name": self.name, "namespace": self.namespace},
+            "spec": {"holderIdentity": self.identity,
+                     "leaseDurationSeconds": self.lease_seconds,
+                     "renewTime": self._now()},
+        }
+
+    def try_acquire(self) -> bool:
+        import datetime
+        cur = self.kube.get("coordination.k8s.io/v1", "Lease",
+                            self.namespace, self.name)
+        if cur is None:
+            try:
+                self.kube.create(self._lease_obj())
+                return True
+            except Exception:
+                return False
+        spec = cur.get("spec") or {}
+        holder = spec.get("holderIdentity")
+        if holder == self.identity:
+            self.renew()
+            return True
+        renew = spec.get("renewTime", "")
+        try:
+            t = datetime.datetime.strptime(renew[:19], "%Y-%m-%dT%H:%M:%S")
+            age = (datetime.datetime.utcnow() - t).total_seconds()
+        except ValueError:
+            age = 1e9
+        if age > spec.get("leaseDurationSeconds", self.lease_seconds) * 2:
+            self.kube.patch("coordination.k8s.io/v1", "Lease",
+                            self.namespace, self.name,
+                            {"spec": self._lease_obj()["spec"]})
+            return True
+        return False
+
+    def renew(self) -> None:
+        self.kube.patch("coordination.k8s.io/v1", "Lease", self.namespace,
+                        self.name, {"spec": {"renewTime": self._now()}})
+
+    def run(self, stop: threading.Event) -> None:
+        """Block until leadership is acquired; keep renewing in the
+        background."""
+        while not stop.is_set() and not self.try_acquire():
+            stop.wait(2.0)
+
+        def _renew_loop():
+            while not stop.is_set():
+                stop.wait(self.lease_seconds / 3)
+                try:
+                    self.renew()
+                except Exception:
+                    log.exception("lease renew failed")
+
+        threading.Thread(target=_renew_loop, daemon=True).start()
+
+
 def run_manager(argv: Optional[list[str]] = None) -> None:
     """controllermanager entrypoint (reference cmd/controllermanager/main.go).
     """
@@ -163,12 +241,19 @@ def run_manager(argv: Optional[list[str]] = None) -> None:
     p.add_argument("--sci-address", default="sci.substratus.svc.cluster.local:10080")
     p.add_argument("--health-probe-bind-address", default=":8081")
     p.add_argument("--namespace", default="")
+    p.add_argument("--leader-elect", action="store_true")
     args = p.parse_args(argv)
 
     logging.basicConfig(level=logging.INFO)
     kube = HTTPKubeClient()
     cloud = new_cloud()
     sci_client = ControllerClient(args.sci_address)
+
+    if args.leader_elect:
+        import os as _os
+        import socket
+        stop = threading.Event()
+        LeaderElector(kube, f"{socket.gethostname()}_{_os.getpid()}").run(stop)
 
     # Associate the SCI server's own service account principal
     # (reference main.go:117-127).

@@ -43,6 +43,7 @@ PLURALS = {
     "ConfigMap": "configmaps", "Secret": "secrets", "Namespace": "namespaces",
     "Job": "jobs", "Deployment": "deployments", "Event": "events",
     "CustomResourceDefinition": "customresourcedefinitions",
+    "Lease": "leases",
 }
 
 CLUSTER_SCOPED = {"Namespace", "CustomResourceDefinition"}

@@ -306,3 +306,20 @@ def test_notebook_flow_and_suspend(env):
     nb = Notebook.from_dict(kube.get(API, "Notebook", "default", "nb"))
     assert not nb.ready
     assert nb.get_condition("Serving")["reason"] == "Suspended"
+
+
+def test_leader_election(env):
+    """Lease-based leader election: first holder wins, a second identity
+    waits until the lease goes stale, same identity renews."""
+    kube, cloud, sci, mgr = env
+    from runbooks_amd.controller.manager import LeaderElector
+
+    a = LeaderElector(kube, "mgr-a", lease_seconds=15)
+    b = LeaderElector(kube, "mgr-b", lease_seconds=15)
+    assert a.try_acquire()
+    assert not b.try_acquire()
+    assert a.try_acquire()          # renewal by the holder
+    # stale lease (old renewTime) is taken over
+    kube.patch("coordination.k8s.io/v1", "Lease", "substratus",
+               a.name, {"spec": {"renewTime": "2000-01-01T00:00:00.000Z"}})
+    assert b.try_acquire()
