@@ -43,6 +43,7 @@ class Request:
     prompt_ids: list[int]
     max_new_tokens: int = 64
     temperature: float = 0.0
+    top_p: float = 1.0
     stop_ids: tuple[int, ...] = ()
     # state
     output_ids: list[int] = field(default_factory=list)
@@ -103,8 +104,9 @@ class Engine:
 
     # -- request API ------------------------------------------------------------
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 64,
-               temperature: float = 0.0) -> Request:
-        req = Request(self._next_id, list(prompt_ids), max_new_tokens, temperature)
+               temperature: float = 0.0, top_p: float = 1.0) -> Request:
+        req = Request(self._next_id, list(prompt_ids), max_new_tokens,
+                      temperature, top_p)
         self._next_id += 1
         self.waiting.append(req)
         return req
@@ -132,7 +134,7 @@ class Engine:
             from .tp_worker import broadcast_prefill
             broadcast_prefill(tokens, positions, slots)
         logits = self.model.prefill(tokens, positions, self.caches, slots)
-        tok = ops.sample_tokens(logits, req.temperature,
+        tok = ops.sample_tokens(logits, req.temperature, top_p=req.top_p,
                                 seed=self.seed + req.request_id * 65537 + S)
         return int(tok[0])
 
@@ -175,10 +177,37 @@ class Engine:
                 broadcast_decode(tokens, positions, slot_t, bt, seq_lens)
             logits = self.model.decode(tokens, positions, self.caches, slot_t,
                                        bt, seq_lens)
-        toks = ops.sample_tokens(
-            logits, reqs[0].temperature,
-            seed=self.seed + 1_000_003 * reqs[0].seq_len)
-        return [int(t) for t in toks]
+        return self._sample_batch(logits, reqs)
+
+    def _sample_batch(self, logits, reqs: list[Request]) -> list[int]:
+        seed = self.seed + 1_000_003 * reqs[0].seq_len
+        params = {(r.temperature, r.top_p) for r in reqs}
+        if len(params) == 1:
+            t, p = params.pop()
+            toks = ops.sample_tokens(logits, t, top_p=p, seed=seed)
+            return [int(x) for x in toks]
+        # heterogeneous sampling params: sample row-by-row
+        out = []
+        for i, r in enumerate(reqs):
+            tok = ops.sample_tokens(logits[i:i + 1], r.temperature,
+                                    top_p=r.top_p,
+                                    seed=seed + r.request_id)
+            out.append(int(tok[0]))
+        return out
+
+    def cancel(self, request_id: int) -> bool:
+        """Stop a request: waiting -> dropped now; running -> finishes at
+        the next scheduler sweep."""
+        for r in self.waiting:
+            if r.request_id == request_id:
+                self.waiting.remove(r)
+                r.finished = True
+                return True
+        for r in self.running:
+            if r.request_id == request_id:
+                r.max_new_tokens = max(1, len(r.output_ids))
+                return True
+        return False
 
     # -- scheduler step ----------------------------------------------------------
     def step(self) -> list[Request]:

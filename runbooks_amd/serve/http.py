@@ -28,8 +28,15 @@ class CompletionRequest(BaseModel):
     prompt: str = ""
     max_tokens: int = 16
     temperature: float = 0.0
+    top_p: float = 1.0
+    stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
+
+    def stop_list(self) -> list[str]:
+        if self.stop is None:
+            return []
+        return [self.stop] if isinstance(self.stop, str) else list(self.stop)
 
 
 class EngineLoop:
@@ -44,15 +51,21 @@ class EngineLoop:
         self._thread = threading.Thread(target=self._run, daemon=True)
         self._thread.start()
 
-    def submit(self, prompt_ids, max_new_tokens, temperature) -> queue.Queue:
-        """Returns a queue yielding (token_id | None); None = finished."""
+    def submit(self, prompt_ids, max_new_tokens, temperature,
+               top_p=1.0) -> tuple[queue.Queue, int]:
+        """Returns (queue yielding token_id | None, request_id)."""
         q: queue.Queue = queue.Queue()
         with self._lock:
-            req = self.engine.submit(prompt_ids, max_new_tokens, temperature)
+            req = self.engine.submit(prompt_ids, max_new_tokens, temperature,
+                                     top_p=top_p)
             self._watchers[req.request_id] = q
             req._watch_sent = 0
         self._wake.set()
-        return q
+        return q, req.request_id
+
+    def cancel(self, request_id: int) -> None:
+        with self._lock:
+            self.engine.cancel(request_id)
 
     def _run(self):
         while not self._stop:
@@ -164,33 +177,68 @@ def build_app(engine: Engine, tokenizer=None,
         ids = [i % vocab for i in ids]
         cid = f"cmpl-{uuid.uuid4().hex[:12]}"
         t0 = int(time.time())
-        q = loop.submit(ids, req.max_tokens, req.temperature)
+        stops = req.stop_list()
+        q, rid = loop.submit(ids, req.max_tokens, req.temperature,
+                             top_p=req.top_p)
+
+        def _stop_hit(text: str):
+            for s in stops:
+                i = text.find(s)
+                if i >= 0:
+                    return text[:i]
+            return None
 
         if req.stream:
             def gen():
                 out = []
+                sent = ""
+                finish = "length"
                 while True:
                     t = q.get()
                     if t is None:
                         break
                     out.append(t)
-                    piece = tok.decode(out)
-                    chunk = {"id": cid, "object": "text_completion",
-                             "created": t0,
-                             "model": req.model or model_name,
-                             "choices": [{"text": piece[len(tok.decode(out[:-1])):],
-                                          "index": 0, "logprobs": None,
-                                          "finish_reason": None}]}
-                    yield f"data: {json.dumps(chunk)}\n\n"
+                    text = tok.decode(out)
+                    trunc = _stop_hit(text)
+                    if trunc is not None:
+                        loop.cancel(rid)
+                        piece, finish = trunc[len(sent):], "stop"
+                    else:
+                        piece = text[len(sent):]
+                    sent += piece
+                    if piece:
+                        chunk = {"id": cid, "object": "text_completion",
+                                 "created": t0,
+                                 "model": req.model or model_name,
+                                 "choices": [{"text": piece, "index": 0,
+                                              "logprobs": None,
+                                              "finish_reason": None}]}
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                    if finish == "stop":
+                        while q.get() is not None:
+                            pass
+                        break
                 yield "data: [DONE]\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
         out = []
+        finish = "length"
+        text = ""
         while True:
             t = q.get()
             if t is None:
                 break
             out.append(t)
+            if stops:
+                trunc = _stop_hit(tok.decode(out))
+                if trunc is not None:
+                    loop.cancel(rid)
+                    text, finish = trunc, "stop"
+                    while q.get() is not None:
+                        pass
+                    break
+        if finish != "stop":
+            text = tok.decode(out)
         if m_reqs is not None:
             m_reqs.inc()
             m_tokens.inc(len(out))
@@ -200,8 +248,8 @@ def build_app(engine: Engine, tokenizer=None,
             "object": "text_completion",
             "created": t0,
             "model": req.model or model_name,
-            "choices": [{"text": tok.decode(out), "index": 0,
-                         "logprobs": None, "finish_reason": "length"}],
+            "choices": [{"text": text, "index": 0,
+                         "logprobs": None, "finish_reason": finish}],
             "usage": _usage(len(ids), len(out)),
         })
 

@@ -110,13 +110,26 @@ class Trainer:
             return
         state = (self.model.state_dict() if self.cfg.full_finetune
                  else lora_state_dict(self.model))
-        ckpt_mod.save_checkpoint(self.cfg.output_dir, self.step_num, state)
+        ckpt_mod.save_checkpoint(self.cfg.output_dir, self.step_num, state,
+                                 optim_state=self.optimizer.state_dict())
 
     def resume(self) -> bool:
+        """Resume model weights, optimizer moments, and the step counter
+        from the latest checkpoint in output_dir (the bucket-mounted
+        /content/artifacts in-cluster — artifacts survive cluster
+        re-creation via the deterministic bucket path)."""
         latest = ckpt_mod.latest_checkpoint(self.cfg.output_dir)
         if latest is None:
             return False
-        state, _, step = ckpt_mod.load_checkpoint(latest)
+        state, optim, step = ckpt_mod.load_checkpoint(latest)
         self.model.load_state_dict(state, strict=False)
+        if optim is not None:
+            try:
+                self.optimizer.load_state_dict(optim)
+            except (ValueError, KeyError):
+                pass  # param set changed; train on with fresh moments
         self.step_num = step
+        if comm.is_dist():
+            # ranks > 0 load the same weights via rank-0 broadcast
+            self.ddp._broadcast_params()
         return True

@@ -69,3 +69,35 @@ def test_lora_fused_backward_matches_eager():
     assert torch.allclose(x.grad, x2.grad, atol=1e-5)
     assert torch.allclose(a.grad, a2.grad, atol=1e-5)
     assert torch.allclose(b.grad, b2.grad, atol=1e-5)
+
+
+def test_checkpoint_resume_with_optimizer(tmp_path):
+    """save -> fresh Trainer -> resume restores step, weights, AND the
+    AdamW moments (training continues instead of restarting)."""
+    import torch
+    from runbooks_amd.train import TrainConfig, Trainer
+
+    cfg = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                      num_train_steps=3, save_steps=0, dtype="float32",
+                      output_dir=str(tmp_path), seed=1)
+    t1 = Trainer(cfg)
+    batch = torch.randint(0, 256, (2, 17))
+    for _ in range(3):
+        t1.train_step(batch)
+    t1.save()
+
+    t2 = Trainer(cfg)
+    assert t2.resume()
+    assert t2.step_num == 3
+    # lora weights restored
+    sd1 = {k: v for k, v in t1.model.state_dict().items() if "lora" in k}
+    sd2 = {k: v for k, v in t2.model.state_dict().items() if "lora" in k}
+    for k in sd1:
+        assert torch.equal(sd1[k], sd2[k]), k
+    # optimizer moments restored (non-zero after 3 steps)
+    states = list(t2.optimizer.state.values())
+    assert states and any(s["exp_avg"].abs().sum() > 0 for s in states)
+    assert all(s["step"] == 3 for s in states)
+    # training continues
+    t2.train_step(batch)
+    assert t2.step_num == 4
