@@ -77,6 +77,22 @@ register(ModelConfig("llama2-70b", vocab_size=32000, hidden_size=8192,
                      num_layers=80, num_heads=64, num_kv_heads=8,
                      intermediate_size=28672))
 
+register(ModelConfig("llama3-8b", vocab_size=128256, hidden_size=4096,
+                     num_layers=32, num_heads=32, num_kv_heads=8,
+                     intermediate_size=14336, max_seq_len=8192,
+                     rope_theta=500000.0))
+register(ModelConfig("llama3-70b", vocab_size=128256, hidden_size=8192,
+                     num_layers=80, num_heads=64, num_kv_heads=8,
+                     intermediate_size=28672, max_seq_len=8192,
+                     rope_theta=500000.0))
+# mistral-7b is llama-shaped with GQA-8 (its 4k sliding window is not
+# applied — the paged cache holds the full context, which is strictly
+# more general and exact for seq <= 4096)
+register(ModelConfig("mistral-7b", vocab_size=32000, hidden_size=4096,
+                     num_layers=32, num_heads=32, num_kv_heads=8,
+                     intermediate_size=14336, max_seq_len=8192,
+                     rope_theta=1000000.0))
+
 # --- falcon family (LayerNorm, RoPE, GELU, parallel residual, MQA/GQA) -----
 register(ModelConfig("falcon-7b", vocab_size=65024, hidden_size=4544,
                      num_layers=32, num_heads=71, num_kv_heads=1,

@@ -161,7 +161,7 @@ def config_from_hf_json(path: str | Path) -> ModelConfig:
     registry)."""
     d = json.loads(Path(path).read_text())
     mt = d.get("model_type", "llama")
-    if mt == "llama":
+    if mt in ("llama", "mistral"):
         return ModelConfig(
             d.get("_name_or_path", "hf-llama"),
             vocab_size=d["vocab_size"], hidden_size=d["hidden_size"],

@@ -96,3 +96,28 @@ def test_falcon_hf_parity():
         theirs = hf(tokens).logits
         got = ours(tokens)
     _logits_close(got, theirs)
+
+
+def test_mistral_hf_parity():
+    """mistral-7b-class (llama-shaped GQA, theta 1e6) through the llama
+    mapping; sliding window unused below 4k context."""
+    from transformers import MistralConfig, MistralForCausalLM
+    hf_cfg = MistralConfig(vocab_size=128, hidden_size=64,
+                           intermediate_size=128, num_hidden_layers=2,
+                           num_attention_heads=4, num_key_value_heads=2,
+                           max_position_embeddings=64, rope_theta=1e6,
+                           sliding_window=None, tie_word_embeddings=False)
+    torch.manual_seed(6)
+    hf = MistralForCausalLM(hf_cfg).eval()
+    cfg = ModelConfig("t-mistral", vocab_size=128, hidden_size=64,
+                      num_layers=2, num_heads=4, num_kv_heads=2,
+                      intermediate_size=128, max_seq_len=64,
+                      rope_theta=1e6)
+    ours = build_model(cfg, dtype=torch.float32)
+    state = convert_hf_state_dict(hf.state_dict(), cfg)
+    missing, _ = ours.load_state_dict(state, strict=False)
+    assert not [m for m in missing if not m.startswith("rope_")], missing
+    tokens = torch.randint(0, 128, (2, 15))
+    with torch.no_grad():
+        # HF Mistral's eager attention upcasts differently; ~7e-3 drift
+        _logits_close(ours(tokens), hf(tokens).logits, tol=1e-2)
