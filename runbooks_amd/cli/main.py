@@ -84,13 +84,30 @@ def apply(filename, namespace, build_path, wait):
         _apply_with_optional_upload(kube, obj, build_path, wait)
 
 
+def _stream_pod_logs(namespace: str, selector: str):
+    """Background `kubectl logs -f` for the workload pods (the reference
+    TUI streams pod logs per role, reference internal/tui/pods.go:222+).
+    Returns the Popen or None when kubectl is unavailable."""
+    import shutil
+    import subprocess
+    k = shutil.which("kubectl")
+    if k is None:
+        return None
+    return subprocess.Popen(
+        [k, "logs", "-f", "-n", namespace, "-l", selector, "--all-containers",
+         "--prefix", "--ignore-errors"],
+        stderr=subprocess.DEVNULL)
+
+
 @main.command()
 @click.argument("path", default=".")
 @click.option("-n", "--namespace", default="default")
 @click.option("-i", "--increment", is_flag=True,
               help="create a new auto-versioned name-N object")
 @click.option("-r", "--replace", is_flag=True, help="replace existing object")
-def run(path, namespace, increment, replace):
+@click.option("--logs/--no-logs", default=True,
+              help="stream workload pod logs while waiting")
+def run(path, namespace, increment, replace, logs):
     """Upload PATH (with Dockerfile) and run it as a Model build+train."""
     kube = _kube()
     objs = find_manifests(path, kind_filter="Model") or \
@@ -105,7 +122,13 @@ def run(path, namespace, increment, replace):
         _say(f"auto-versioned name: {obj.name}")
     elif replace:
         kube.delete("substratus.ai/v1", obj.kind, namespace, obj.name)
-    _apply_with_optional_upload(kube, obj, path, wait=True)
+    log_proc = _stream_pod_logs(namespace, f"model={obj.name}") if logs \
+        else None
+    try:
+        _apply_with_optional_upload(kube, obj, path, wait=True)
+    finally:
+        if log_proc is not None:
+            log_proc.terminate()
 
 
 @main.command()
