@@ -48,6 +48,10 @@ class DataParallel(nn.Module):
         self._buckets: list[_Bucket] = []
         self._param_bucket: dict[int, tuple[_Bucket, int]] = {}
         self._hooks = []
+        # gradient accumulation: when False, completed buckets do NOT
+        # launch their all-reduce (grads keep accumulating locally);
+        # the final micro-batch sets it back to True.
+        self.sync = True
         self._build_buckets()
         if comm.is_dist():
             self._broadcast_params()
@@ -99,8 +103,15 @@ class DataParallel(nn.Module):
         b.views[i].add_(p.grad)
         p.grad = None
         b.pending -= 1
-        if b.pending == 0 and comm.is_dist():
+        if b.pending == 0 and comm.is_dist() and self.sync:
             b.work = dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM, async_op=True)
+
+    def start_microbatch(self, sync: bool = True):
+        """Arm the hooks for one backward pass. sync=False accumulates
+        into the flat buckets without reducing (grad-accumulation
+        micro-batches); the last micro-batch passes sync=True."""
+        self.sync = sync
+        self._reset_pending()
 
     def finish_backward(self):
         """Wait for in-flight reductions, average, and expose ``.grad``.

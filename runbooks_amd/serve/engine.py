@@ -195,6 +195,26 @@ class Engine:
             out.append(int(tok[0]))
         return out
 
+    def _preempt(self, r: Request) -> None:
+        """Release a running request's cache and requeue it: its generated
+        tokens become part of the prompt for the re-prefill."""
+        self.allocator.release(r.blocks)
+        r.blocks = []
+        r.prompt_ids = r.prompt_ids + r.output_ids
+        r.max_new_tokens -= len(r.output_ids)
+        r.output_ids = []
+        if hasattr(r, "_watch_sent"):
+            r._watch_sent = 0  # post-preempt tokens are all new to watchers
+        self.running.remove(r)
+        self.waiting.insert(0, r)
+
+    def _preempt_newest(self, exclude: Request) -> bool:
+        for r in reversed(self.running):
+            if r is not exclude:
+                self._preempt(r)
+                return True
+        return False
+
     def cancel(self, request_id: int) -> bool:
         """Stop a request: waiting -> dropped now; running -> finishes at
         the next scheduler sweep."""
@@ -227,13 +247,24 @@ class Engine:
             req.output_ids.append(first)
             self.running.append(req)
         elif self.running:
-            # ensure every running seq has a block for the next position
-            for r in self.running:
-                if r.seq_len >= len(r.blocks) * self.bs:
-                    r.blocks.extend(self.allocator.alloc(1))
-            toks = self._decode_batch(self.running)
-            for r, t in zip(self.running, toks):
-                r.output_ids.append(t)
+            # ensure every running seq has a block for the next position;
+            # when the pool is exhausted, preempt the newest sequence
+            # (its blocks are freed and it requeues for a fresh prefill)
+            # instead of failing the whole batch.
+            for r in list(self.running):
+                if r not in self.running:
+                    continue
+                while r.seq_len >= len(r.blocks) * self.bs:
+                    if self.allocator.free:
+                        r.blocks.extend(self.allocator.alloc(1))
+                    elif not self._preempt_newest(exclude=r):
+                        # nothing left to preempt: preempt r itself
+                        self._preempt(r)
+                        break
+            if self.running:
+                toks = self._decode_batch(self.running)
+                for r, t in zip(self.running, toks):
+                    r.output_ids.append(t)
         for r in list(self.running):
             if (len(r.output_ids) >= r.max_new_tokens
                     or (r.stop_ids and r.output_ids[-1] in r.stop_ids)

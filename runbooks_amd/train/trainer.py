@@ -30,6 +30,7 @@ class TrainConfig:
     weight_decay: float = 0.0
     num_train_steps: int = 100
     save_steps: int = 0                 # 0 = no checkpoints
+    grad_accum_steps: int = 1
     lora_r: int = 16
     lora_alpha: int = 32
     full_finetune: bool = False
@@ -57,13 +58,19 @@ class Trainer:
         self.step_num = 0
 
     # -- core step -------------------------------------------------------------
-    def train_step(self, tokens: torch.Tensor) -> float:
-        """tokens [B, S+1] (inputs + shifted labels). Returns loss."""
+    def train_step(self, tokens: torch.Tensor, sync: bool = True) -> float:
+        """tokens [B, S+1] (inputs + shifted labels). Returns loss.
+        sync=False runs a gradient-accumulation micro-batch: grads pile
+        into the flat buckets, no all-reduce, no optimizer step."""
         inputs = tokens[:, :-1].to(self.device)
         labels = tokens[:, 1:].to(self.device)
+        self.ddp.start_microbatch(sync=sync)
         logits = self.ddp(inputs)
+        scale = 1.0 / max(1, self.cfg.grad_accum_steps)
         loss = ops.cross_entropy(logits, labels)
-        loss.backward()
+        (loss * scale if scale != 1.0 else loss).backward()
+        if not sync:
+            return float(loss.detach())
         self.ddp.finish_backward()
         if self.cfg.grad_clip > 0:
             # grads are views into the DDP flat buckets — clip the few flat
@@ -94,6 +101,13 @@ class Trainer:
             except StopIteration:
                 it = iter(loader)
                 batch = next(it)
+            for _ in range(self.cfg.grad_accum_steps - 1):
+                self.train_step(batch, sync=False)
+                try:
+                    batch = next(it)
+                except StopIteration:
+                    it = iter(loader)
+                    batch = next(it)
             loss = self.train_step(batch)
             if cfg.save_steps and self.step_num % cfg.save_steps == 0:
                 self.save()

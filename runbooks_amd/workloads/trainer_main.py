@@ -80,6 +80,7 @@ def main():
         lr=_param("learning_rate", 2e-4, float),
         num_train_steps=_param("num_train_steps", 100, int),
         save_steps=_param("save_steps", 50, int),
+        grad_accum_steps=_param("gradient_accumulation_steps", 1, int),
         lora_r=_param("lora_r", 16, int),
         full_finetune=_param("full_finetune", False, bool),
         dtype="bfloat16" if torch.cuda.is_available() else "float32",

@@ -101,3 +101,33 @@ def test_checkpoint_resume_with_optimizer(tmp_path):
     # training continues
     t2.train_step(batch)
     assert t2.step_num == 4
+
+
+def test_gradient_accumulation_matches_big_batch():
+    """2 micro-batches with grad_accum_steps=2 == one batch of 2x size."""
+    import torch
+    from runbooks_amd.train import TrainConfig, Trainer
+
+    torch.manual_seed(0)
+    b1 = torch.randint(0, 256, (2, 17))
+    b2 = torch.randint(0, 256, (2, 17))
+    big = torch.cat([b1, b2])
+
+    cfg_a = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                        num_train_steps=1, grad_accum_steps=2,
+                        dtype="float32", grad_clip=0, seed=2)
+    ta = Trainer(cfg_a)
+    ta.train_step(b1, sync=False)
+    ta.train_step(b2, sync=True)
+    assert ta.step_num == 1
+
+    cfg_b = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=4,
+                        num_train_steps=1, grad_accum_steps=1,
+                        dtype="float32", grad_clip=0, seed=2)
+    tb = Trainer(cfg_b)
+    tb.train_step(big)
+
+    sa = {k: v for k, v in ta.model.state_dict().items() if "lora" in k}
+    sb = {k: v for k, v in tb.model.state_dict().items() if "lora" in k}
+    for k in sa:
+        assert torch.allclose(sa[k], sb[k], atol=1e-5), k
