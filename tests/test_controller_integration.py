@@ -323,3 +323,40 @@ def test_leader_election(env):
     kube.patch("coordination.k8s.io/v1", "Lease", "substratus",
                a.name, {"spec": {"renewTime": "2000-01-01T00:00:00.000Z"}})
     assert b.try_acquire()
+
+
+def test_manager_watch_loop(env):
+    """The production watch-driven loop: events in the in-memory API
+    server flow through watches -> workqueue -> reconcilers, including
+    the dependency fan-out (Model ready -> dependent Server reconciled)."""
+    import threading
+    import time
+
+    kube, cloud, sci, mgr = env
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    try:
+        kube.create(Model(name="wm", image="img:1").to_dict())
+
+        def wait_for(fn, timeout=10.0):
+            end = time.time() + timeout
+            while time.time() < end:
+                if fn():
+                    return True
+                time.sleep(0.02)
+            return False
+
+        assert wait_for(lambda: kube.get("batch/v1", "Job", "default",
+                                         "wm-modeller") is not None)
+        kube.create(Server(name="ws", image="img:s",
+                           model=ObjectRef("wm")).to_dict())
+        assert wait_for(lambda: (kube.get(API, "Server", "default", "ws")
+                                 .get("status") or {}).get("conditions"))
+        # completing the Job flips the Model; the watch fan-out then
+        # reconciles the Server into creating its Deployment
+        fake_job_complete(kube, "default", "wm-modeller")
+        assert wait_for(lambda: kube.get("apps/v1", "Deployment", "default",
+                                         "ws-server") is not None)
+    finally:
+        mgr.stop()
+        t.join(timeout=3)
