@@ -189,7 +189,9 @@ def serve(path, namespace):
 @click.argument("kind")
 @click.argument("name", required=False)
 @click.option("-n", "--namespace", default="default")
-def get(kind, name, namespace):
+@click.option("-o", "--output", default="table",
+              type=click.Choice(["table", "json"]))
+def get(kind, name, namespace, output):
     kube = _kube()
     kind = {k.lower(): k for k in KINDS}.get(kind.rstrip("s").lower(), kind)
     if name:
@@ -201,9 +203,50 @@ def get(kind, name, namespace):
         print(json.dumps(raw, indent=2))
         return
     rows = kube.list("substratus.ai/v1", kind, namespace)
+    if output == "json":
+        import json
+        print(json.dumps(rows, indent=2))
+        return
     for o in rows:
-        ready = (o.get("status") or {}).get("ready", False)
-        print(f"{o['metadata']['name']}\tready={ready}")
+        status = o.get("status") or {}
+        ready = status.get("ready", False)
+        conds = status.get("conditions") or []
+        last = conds[-1] if conds else {}
+        cond = f"{last.get('type', '-')}:{last.get('reason', '-')}" \
+            if last else "-"
+        print(f"{o['metadata']['name']}\tready={ready}\t{cond}")
+
+
+@main.command()
+@click.option("-m", "--model", "server_name", required=True,
+              help="Server object name")
+@click.option("-p", "--prompt", default="Hello")
+@click.option("-n", "--namespace", default="default")
+@click.option("--max-tokens", default=32)
+@click.option("--port", default=18080)
+def infer(server_name, prompt, namespace, max_tokens, port):
+    """Send a completion request to a Server through a port-forward
+    (the reference's `sub infer` exists but is disabled,
+    reference internal/cli/root.go:19 — here it works)."""
+    import json
+    import time
+    import urllib.request
+
+    from .. import client as sclient
+    fwd = sclient.sync.port_forward(namespace, f"{server_name}-server",
+                                    port, 8080, resource="service")
+    try:
+        time.sleep(2)
+        body = json.dumps({"prompt": prompt,
+                           "max_tokens": max_tokens}).encode()
+        req = urllib.request.Request(
+            f"http://localhost:{port}/v1/completions", data=body,
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=300) as r:
+            out = json.loads(r.read())
+        print(out["choices"][0]["text"])
+    finally:
+        fwd.terminate()
 
 
 @main.command()

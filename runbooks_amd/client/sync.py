@@ -80,10 +80,11 @@ def sync_files_from_notebook(namespace: str, pod: str, local_dir: str,
         proc.terminate()
 
 
-def port_forward(namespace: str, pod: str, local_port: int, pod_port: int
-                 ) -> subprocess.Popen:
-    """Start a kubectl port-forward; caller terminates the returned proc."""
+def port_forward(namespace: str, name: str, local_port: int, pod_port: int,
+                 resource: str = "pod") -> subprocess.Popen:
+    """Start a kubectl port-forward (pod or service); caller terminates
+    the returned proc."""
     return subprocess.Popen(
-        [_kubectl(), "port-forward", "-n", namespace, f"pod/{pod}",
+        [_kubectl(), "port-forward", "-n", namespace, f"{resource}/{name}",
          f"{local_port}:{pod_port}"],
         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)

@@ -212,3 +212,32 @@ def test_trainer_main_end_to_end(tmp_path, monkeypatch):
     ckpts = list(out_dir.glob("checkpoint-*"))
     assert ckpts, "trainer wrote no checkpoint"
     assert (ckpts[0] / "model.safetensors").exists()
+
+
+def test_bench_contract_cpu(tmp_path):
+    """bench.py emits exactly one driver-contract JSON line for both
+    modes on CPU (tiny model)."""
+    import json as _json
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--micro-batch", "2", "--seq-len", "16"],
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-800:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = _json.loads(line)
+    assert d["metric"] == "finetune_samples_per_sec"
+    assert d["n_gpus"] == 1 and d["steps"] == 2 and d["scaling"] == "weak"
+    assert set(d) >= {"value", "unit", "ms_per_step", "higher_is_better",
+                      "vs_baseline", "dtype", "data", "config"}
+
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--mode", "serve", "--steps", "3",
+         "--warmup", "1", "--serve-batch", "2", "--prompt-len", "8"],
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-800:]
+    d = _json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["metric"] == "serve_tokens_per_sec"
+    assert d["value"] > 0
