@@ -44,21 +44,23 @@ def _report_conditions(raw):
 
 
 def _apply_with_optional_upload(kube, obj, build_path=None, wait=False):
+    from ..tui import ReadinessChecklist, UploadProgress
     if build_path is not None:
-        tb = sclient.prepare_image_tarball(build_path,
-                                           progress=lambda p: None)
-        sclient.set_upload_container_spec(obj, tb, uuid.uuid4().hex)
-        sclient.clear_image(obj)
-        kube.apply(obj.to_dict())
-        _say(f"[bold]{obj.kind}/{obj.name}[/bold]: uploading build context "
-             f"(md5 {tb.md5_checksum[:12]}…)")
-        sclient.upload(kube, obj, tb)
+        with UploadProgress(f"{obj.kind}/{obj.name}") as up:
+            tb = sclient.prepare_image_tarball(build_path,
+                                               progress=up.on_file)
+            sclient.set_upload_container_spec(obj, tb, uuid.uuid4().hex)
+            sclient.clear_image(obj)
+            kube.apply(obj.to_dict())
+            _say(f"[bold]{obj.kind}/{obj.name}[/bold]: uploading build "
+                 f"context (md5 {tb.md5_checksum[:12]}…)")
+            sclient.upload(kube, obj, tb, progress=up.on_fraction)
     else:
         kube.apply(obj.to_dict())
         _say(f"[bold]{obj.kind}/{obj.name}[/bold]: applied")
     if wait:
-        _say(f"waiting for {obj.kind}/{obj.name} to become ready…")
-        raw = sclient.wait_ready(kube, obj)
+        with ReadinessChecklist(f"{obj.kind}/{obj.name}") as view:
+            raw = sclient.wait_ready(kube, obj, callback=view.update)
         _report_conditions(raw)
 
 
@@ -115,7 +117,8 @@ def run(path, namespace, increment, replace, logs):
     if not objs:
         _say("[red]no substratus manifests found[/red]")
         sys.exit(1)
-    obj = objs[0]
+    from ..tui import select_manifest
+    obj = select_manifest(objs)
     obj.namespace = namespace
     if increment:
         obj.name = next_version_name(kube, obj.kind, namespace, obj.name)
@@ -146,7 +149,8 @@ def notebook(path, namespace, sync_dir, port, no_sync):
     if not objs:
         _say("[red]no substratus manifests found[/red]")
         sys.exit(1)
-    nb = sclient.notebook_for_object(objs[0])
+    from ..tui import select_manifest
+    nb = sclient.notebook_for_object(select_manifest(objs))
     nb.namespace = namespace
     _apply_with_optional_upload(kube, nb, path, wait=True)
 

@@ -241,3 +241,23 @@ def test_bench_contract_cpu(tmp_path):
     d = _json.loads(out.stdout.strip().splitlines()[-1])
     assert d["metric"] == "serve_tokens_per_sec"
     assert d["value"] > 0
+
+
+def test_tui_components():
+    from runbooks_amd.tui import format_conditions, select_manifest
+
+    raw = {"status": {"ready": False, "conditions": [
+        {"type": "Built", "status": "True", "reason": "JobComplete"},
+        {"type": "Complete", "status": "False", "reason": "JobNotComplete"},
+    ]}}
+    rows = format_conditions(raw)
+    assert rows == [("✓", "Built", "JobComplete"),
+                    ("…", "Complete", "JobNotComplete"),
+                    ("…", "Ready", "")]
+    raw["status"]["ready"] = True
+    assert format_conditions(raw)[-1] == ("✓", "Ready", "")
+
+    m1, m2 = Model(name="a"), Model(name="b")
+    # non-TTY path picks the first deterministically
+    assert select_manifest([m1, m2]) is m1
+    assert select_manifest([m2]) is m2
