@@ -341,4 +341,10 @@ def build_model(name_or_cfg, dtype=torch.bfloat16, tp: int | None = None,
     from .config import get_config
     cfg = name_or_cfg if isinstance(name_or_cfg, ModelConfig) else get_config(name_or_cfg)
     torch.manual_seed(seed)
-    return Transformer(cfg, dtype=dtype, tp=tp, device=device)
+    if device is not None:
+        # construct directly on the target device — a 7B bf16 random init
+        # takes minutes on CPU but is instant on the GPU, and 8 DP ranks
+        # would otherwise all churn host RAM at once
+        with torch.device(device):
+            return Transformer(cfg, dtype=dtype, tp=tp)
+    return Transformer(cfg, dtype=dtype, tp=tp)

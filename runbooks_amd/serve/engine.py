@@ -64,7 +64,8 @@ class Engine:
         self.device = device if device is not None else (
             f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
         if isinstance(model, str):
-            model = build_model(model, dtype=dtype, seed=seed)
+            model = build_model(model, dtype=dtype, seed=seed,
+                                device=self.device)
         self.model = model.to(self.device).eval()
         if torch.cuda.is_available() and model.tp == 1:
             from ..models.transformer import fuse_for_inference

@@ -46,10 +46,10 @@ class Trainer:
         self.device = device if device is not None else (
             f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
         dtype = getattr(torch, cfg.dtype)
-        model = build_model(cfg.model, dtype=dtype, tp=1, seed=cfg.seed)
+        model = build_model(cfg.model, dtype=dtype, tp=1, seed=cfg.seed,
+                            device=self.device)
         if not cfg.full_finetune:
             apply_lora(model, r=cfg.lora_r, alpha=cfg.lora_alpha)
-        model = model.to(self.device)
         self.ddp = DataParallel(model)
         self.model = model
         trainable = [p for p in model.parameters() if p.requires_grad]

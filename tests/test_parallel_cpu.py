@@ -192,3 +192,75 @@ def test_ddp_many_buckets():
 
 def test_tp_engine_serving():
     _run_dist(_tp_engine_serving, port=PORT + 4)
+
+
+def _dp_grad_accum(rank, world):
+    """grad accumulation across ranks: 2 micro-batches/rank, all-reduce
+    only on the final one; result equals per-rank big-batch averaging."""
+    from runbooks_amd.train import TrainConfig, Trainer
+
+    torch.manual_seed(0)
+    cfg = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                      num_train_steps=1, grad_accum_steps=2,
+                      dtype="float32", grad_clip=0, seed=4)
+    t = Trainer(cfg, device="cpu")
+    torch.manual_seed(500 + rank)
+    b1 = torch.randint(0, 256, (2, 17))
+    b2 = torch.randint(0, 256, (2, 17))
+    t.train_step(b1, sync=False)
+    t.train_step(b2, sync=True)
+    assert t.step_num == 1
+    # every rank must end with identical (all-reduced) LoRA weights
+    for n, p in t.model.named_parameters():
+        if not p.requires_grad:
+            continue
+        flat = p.detach().clone()
+        gathered = [torch.empty_like(flat) for _ in range(world)]
+        dist.all_gather(gathered, flat)
+        for g in gathered:
+            assert torch.allclose(g, flat, atol=1e-6), n
+
+
+def _tp_falcon_mqa_replicated_kv(rank, world):
+    """falcon-style MQA at tp>num_kv_heads: KV projection replicated,
+    each rank keeps a full KV cache; engine output == single GPU."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.serve import Engine
+    from runbooks_amd.serve.tp_worker import broadcast_shutdown, worker_loop
+
+    torch.manual_seed(0)
+    tp_model = build_model("tiny-falcon", dtype=torch.float32, tp=world,
+                           seed=8)
+    single = build_model("tiny-falcon", dtype=torch.float32, tp=1, seed=8)
+    sd = single.state_dict()
+    for name, t in tp_model.state_dict().items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:
+            t.copy_(full[rank * t.shape[0]:(rank + 1) * t.shape[0]])
+        else:
+            t.copy_(full[:, rank * t.shape[1]:(rank + 1) * t.shape[1]])
+    # MQA: k/v projections replicated (full size on every rank)
+    assert tp_model.blocks[0].attn.k_proj.weight.shape == \
+        single.blocks[0].attn.k_proj.weight.shape
+
+    eng = Engine(tp_model, device="cpu", kv_blocks=64, seed=21)
+    prompt = [2, 7, 1, 8]
+    if rank == 0:
+        out_tp = eng.generate(prompt, max_new_tokens=5)
+        broadcast_shutdown(torch.device("cpu"))
+        ref = Engine(single, device="cpu", kv_blocks=64, seed=21)
+        ref.tp = 1
+        out_1 = ref.generate(prompt, max_new_tokens=5)
+        assert out_tp == out_1, (out_tp, out_1)
+    else:
+        worker_loop(eng)
+
+
+def test_dp_grad_accumulation():
+    _run_dist(_dp_grad_accum, port=PORT + 5)
+
+
+def test_tp_falcon_mqa():
+    _run_dist(_tp_falcon_mqa_replicated_kv, port=PORT + 6)
