@@ -101,6 +101,13 @@ def bench_serve(args) -> dict:
     eng = Engine(args.model, dtype=dtype,
                  kv_blocks=None if torch.cuda.is_available() else 4096,
                  max_batch=args.serve_batch, seed=17)
+    if comm.is_dist() and comm.world_size() > 1:
+        # TP serving: rank 0 drives the engine + measures; workers follow
+        # the broadcast protocol until shutdown (serve/tp_worker.py).
+        if comm.rank() != 0:
+            from runbooks_amd.serve.tp_worker import worker_loop
+            worker_loop(eng)
+            return {}
     vocab = eng.cfg.vocab_size
     g = torch.Generator().manual_seed(17)
 
@@ -124,6 +131,9 @@ def bench_serve(args) -> dict:
         tokens += after - before
     _sync()
     dt = time.perf_counter() - t0
+    if comm.is_dist() and comm.world_size() > 1:
+        from runbooks_amd.serve.tp_worker import broadcast_shutdown
+        broadcast_shutdown(torch.device(eng.device))
     return {
         "metric": "serve_tokens_per_sec",
         "value": tokens / dt,
@@ -133,7 +143,8 @@ def bench_serve(args) -> dict:
         "warmup": args.warmup,
         "ms_per_step": dt / args.steps * 1e3,
         "higher_is_better": True,
-        "scaling": "weak",
+        # TP serving keeps total work fixed as ranks grow
+        "scaling": "strong" if comm.world_size() > 1 else "weak",
         "vs_baseline": None,
         "dtype": "bf16" if torch.cuda.is_available() else "float32",
         "data": "synthetic",
