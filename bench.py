@@ -13,9 +13,9 @@ Launch (driver):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
 
-Weak scaling: per-GPU work is fixed (micro batch x seq per rank for
-train; the serve bench is single-GPU TP=1 per rank... serve mode is
-rank-0 only). Rank 0 prints ONE JSON line.
+Scaling: train is weak (per-GPU micro-batch fixed, DP over RCCL);
+serve under torchrun is strong (TP: rank 0 drives the engine, worker
+ranks follow the broadcast protocol). Rank 0 prints ONE JSON line.
 """
 from __future__ import annotations
 

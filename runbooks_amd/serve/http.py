@@ -47,6 +47,7 @@ class EngineLoop:
         self._lock = threading.Lock()
         self._wake = threading.Event()
         self._watchers: dict[int, queue.Queue] = {}  # request_id -> token q
+        self._failures = 0
         self._stop = False
         self._thread = threading.Thread(target=self._run, daemon=True)
         self._thread.start()
@@ -82,7 +83,7 @@ class EngineLoop:
                     # notify its watcher; on repeated failures (a request
                     # that keeps crashing the step) fail everything rather
                     # than spin
-                    self._failures = getattr(self, "_failures", 0) + 1
+                    self._failures += 1
                     if self._failures >= 3:
                         for r in (self.engine.running +
                                   self.engine.waiting):
