@@ -35,6 +35,8 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits, at::Tensor targets,
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                              double gscale, int64_t ignore_index);
 at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
+at::Tensor mfma_probe_16x16x32(at::Tensor a, at::Tensor b);
+at::Tensor train_gemm_nt(at::Tensor a, at::Tensor b);
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
 at::Tensor skinny_gemm_fp8(at::Tensor x, at::Tensor w8, at::Tensor scale);
 int64_t skinny_gemm_mmax();
@@ -72,6 +74,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm,
         "HBM-rate decode GEMM: y[M<=32,N] = x @ W^T (bf16)");
   m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
+  m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32,
+        "layout probe: one 16x16x32 bf16 MFMA as a plain matmul");
+  m.def("train_gemm_nt", &train_gemm_nt,
+        "EXPERIMENTAL 256^2 8-phase bf16 GEMM (C = A @ B^T)");
   m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
         "fp8-e4m3 weight-only decode GEMM with per-channel scales");
   m.def("qkv_rope_append", &qkv_rope_append,

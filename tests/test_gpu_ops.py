@@ -432,3 +432,38 @@ def test_skinny_gemm_fp8():
     d2 = (y.float() - full).abs().max() / full.abs().max()
     assert d2 < 0.08, f"vs bf16 rel {d2}"
     _FP8_REGISTRY.clear()
+
+
+_EXPERIMENTAL = __import__("os").environ.get("RB_EXPERIMENTAL") == "1"
+
+
+@pytest.mark.skipif(not _EXPERIMENTAL, reason="RB_EXPERIMENTAL=1 only "
+                    "(drafted offline; validate before enabling)")
+def test_mfma_16x16x32_layout_probe():
+    _assert_hip()
+    torch.manual_seed(0)
+    a = torch.randn(16, 32, dtype=torch.bfloat16, device=DEV)
+    b = torch.arange(32 * 16, dtype=torch.float32, device=DEV).reshape(32, 16)
+    b = ((b % 11) / 5.0 - 1.0).to(torch.bfloat16)   # asymmetric
+    c = ops.ext().mfma_probe_16x16x32(a, b)
+    ref = a.float() @ b.float()
+    assert torch.allclose(c, ref, atol=2e-2, rtol=2e-2), (c - ref).abs().max()
+
+
+@pytest.mark.skipif(not _EXPERIMENTAL, reason="RB_EXPERIMENTAL=1 only")
+@pytest.mark.parametrize("M,N,K", [
+    (256, 256, 128),      # single tile, 2 K-tiles (minimum)
+    (256, 256, 512),
+    (512, 768, 1024),     # multi-tile, XCD remap with nwg%8 != 0
+    (2048, 4096, 4096),   # the training shape
+])
+def test_train_gemm_nt(M, N, K):
+    _assert_hip()
+    torch.manual_seed(M + N)
+    a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+    c = ops.ext().train_gemm_nt(a, b)
+    ref = (a.float() @ b.float().t())
+    d = (c.float() - ref).abs().max().item()
+    rel = d / ref.abs().max().item()
+    assert rel < 2e-2, f"max abs {d} rel {rel}"
