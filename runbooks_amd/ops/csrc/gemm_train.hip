@@ -10,8 +10,16 @@
 // Status: drafted offline at the end of round 1 (no GPU budget left to
 // validate); NOT wired into any model path. Gated behind
 // RB_EXPERIMENTAL: tests skip unless it is set. Round 2: run
-// `RB_EXPERIMENTAL=1 pytest tests/test_gpu_ops.py -k train_gemm`first,
+// `RB_EXPERIMENTAL=1 pytest tests/test_gpu_ops.py -k train_gemm` first,
 // then benchmarks/kernels.py.
+//
+// KNOWN LIMIT (audit): at M=2048, N=4096 the 256^2 tiling yields only
+// 128 WGs -> half the CUs idle; expect ~660 TF there, BELOW hipBLASLt.
+// The template wins where (M/256)*(N/256) >= 256: the N=11008 MLP
+// shapes (344 WGs) and any fused-projection N. To cover N=4096 either
+// add split-K=2 with fp32 slabs + combine (skinny_gemm.hip pattern) or
+// fuse q/k/v for the training forward (frozen base weights need no
+// wgrad under LoRA).
 //
 // Template geometry (guide table):
 //   tile BM x BN = 256 x 256, BK = 64, 8 waves (2M x 4N), 512 threads
