@@ -12,7 +12,6 @@ Parity: reference internal/client/upload.go —
 from __future__ import annotations
 
 import base64
-import gzip
 import hashlib
 import os
 import tarfile

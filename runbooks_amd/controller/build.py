@@ -10,7 +10,7 @@ from __future__ import annotations
 import datetime
 from typing import Callable
 
-from ..api.types import UploadStatus, now_rfc3339
+from ..api.types import UploadStatus
 from ..api import conditions as cond
 from ..k8s import Conflict, KubeClient
 from ..resources import container_builder_resources

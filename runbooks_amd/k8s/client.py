@@ -13,7 +13,7 @@ import ssl
 import threading
 import urllib.parse
 import urllib.request
-from typing import Any, Callable, Iterator, Optional
+from typing import Any, Iterator, Optional
 
 
 class NotFound(Exception):

@@ -11,7 +11,7 @@ from __future__ import annotations
 import copy
 import queue
 import threading
-from typing import Iterator, Optional
+from typing import Iterator
 
 from .client import Conflict, KubeClient, NotFound
 

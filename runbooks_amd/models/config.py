@@ -7,7 +7,7 @@ configs of our own runtime.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass

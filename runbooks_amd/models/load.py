@@ -15,7 +15,7 @@ from pathlib import Path
 
 import torch
 
-from .config import ModelConfig, get_config
+from .config import ModelConfig
 from .transformer import Transformer
 
 

@@ -17,7 +17,6 @@ import argparse
 import datetime
 import hashlib
 import hmac
-import json
 import os
 import urllib.parse
 import urllib.request

@@ -15,8 +15,6 @@ from __future__ import annotations
 import torch
 import torch.distributed as dist
 
-from ..parallel import comm
-
 OP_SHUTDOWN = 0
 OP_PREFILL = 1
 OP_DECODE = 2

@@ -51,7 +51,7 @@ def main():
         return 0
     import torch
 
-    from ..models import get_config, list_configs
+    from ..models import list_configs
     from ..models.load import load_pretrained
     from ..parallel import comm
     from ..serve.tokenizer import load_tokenizer
