@@ -222,6 +222,28 @@ def get(kind, name, namespace, output):
 
 
 @main.command()
+@click.argument("kind")
+@click.argument("name")
+@click.option("-n", "--namespace", default="default")
+@click.option("-f", "--follow", is_flag=True)
+def logs(kind, name, namespace, follow):
+    """Stream the workload pod logs of a substratus object."""
+    import shutil
+    import subprocess
+    k = shutil.which("kubectl")
+    if k is None:
+        _say("[red]kubectl not found[/red]")
+        sys.exit(1)
+    kind = {kk.lower(): kk for kk in KINDS}.get(kind.rstrip("s").lower(),
+                                                kind).lower()
+    args = [k, "logs", "-n", namespace, "-l", f"{kind}={name}",
+            "--all-containers", "--prefix", "--ignore-errors"]
+    if follow:
+        args.append("-f")
+    subprocess.run(args)
+
+
+@main.command()
 @click.option("-m", "--model", "server_name", required=True,
               help="Server object name")
 @click.option("-p", "--prompt", default="Hello")

@@ -164,6 +164,12 @@ def build_app(engine: Engine, tokenizer=None,
     def healthz():
         return {"status": "ok"}
 
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "owned_by": "runbooks-amd"}]}
+
     def _usage(n_prompt, n_out):
         return {"prompt_tokens": n_prompt, "completion_tokens": n_out,
                 "total_tokens": n_prompt + n_out}

@@ -27,6 +27,8 @@ class TrainConfig:
     seq_len: int = 512
     micro_batch: int = 4
     lr: float = 2e-4
+    lr_scheduler: str = "constant"      # constant | cosine | linear
+    warmup_steps: int = 0
     weight_decay: float = 0.0
     num_train_steps: int = 100
     save_steps: int = 0                 # 0 = no checkpoints
@@ -81,10 +83,28 @@ class Trainer:
             scale = self.cfg.grad_clip / (total + 1e-6)
             if float(scale) < 1.0:
                 torch._foreach_mul_(bufs, scale)
+        self._apply_lr_schedule()
         self.optimizer.step()
         self.ddp.zero_grad()
         self.step_num += 1
         return float(loss.detach())
+
+    def _apply_lr_schedule(self):
+        """HF-TrainingArguments-style warmup + decay on the base lr."""
+        import math
+        cfg = self.cfg
+        step = self.step_num + 1
+        scale = 1.0
+        if cfg.warmup_steps > 0 and step <= cfg.warmup_steps:
+            scale = step / cfg.warmup_steps
+        elif cfg.lr_scheduler in ("cosine", "linear"):
+            total = max(1, cfg.num_train_steps - cfg.warmup_steps)
+            done = min(total, step - cfg.warmup_steps)
+            frac = done / total
+            scale = (0.5 * (1.0 + math.cos(math.pi * frac))
+                     if cfg.lr_scheduler == "cosine" else 1.0 - frac)
+        for g in self.optimizer.param_groups:
+            g["lr"] = cfg.lr * scale
 
     # -- full loop -------------------------------------------------------------
     def fit(self, dataset=None, log_every: int = 10):

@@ -78,6 +78,8 @@ def main():
         seq_len=_param("seq_len", 512, int),
         micro_batch=_param("per_device_train_batch_size", 4, int),
         lr=_param("learning_rate", 2e-4, float),
+        lr_scheduler=_param("lr_scheduler_type", "constant"),
+        warmup_steps=_param("warmup_steps", 0, int),
         num_train_steps=_param("num_train_steps", 100, int),
         save_steps=_param("save_steps", 50, int),
         grad_accum_steps=_param("gradient_accumulation_steps", 1, int),

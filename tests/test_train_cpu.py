@@ -131,3 +131,41 @@ def test_gradient_accumulation_matches_big_batch():
     sb = {k: v for k, v in tb.model.state_dict().items() if "lora" in k}
     for k in sa:
         assert torch.allclose(sa[k], sb[k], atol=1e-5), k
+
+
+def test_lr_schedule_warmup_cosine():
+    import pytest
+    import torch
+    from runbooks_amd.train import TrainConfig, Trainer
+
+    cfg = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                      num_train_steps=10, lr=1e-2, lr_scheduler="cosine",
+                      warmup_steps=2, dtype="float32", seed=0)
+    t = Trainer(cfg)
+    b = torch.randint(0, 256, (2, 17))
+    lrs = []
+    for _ in range(10):
+        t.train_step(b)
+        lrs.append(t.optimizer.param_groups[0]["lr"])
+    assert lrs[0] == pytest.approx(1e-2 * 0.5)   # warmup step 1/2
+    assert lrs[1] == pytest.approx(1e-2)         # warmup done
+    assert lrs[2] < lrs[1]                       # cosine decays
+    assert lrs[-1] == pytest.approx(0.0, abs=1e-6)
+
+
+def test_full_finetune_step():
+    """full_finetune=True trains every parameter (no LoRA wrap)."""
+    import torch
+    from runbooks_amd.train import TrainConfig, Trainer
+
+    cfg = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                      num_train_steps=1, full_finetune=True,
+                      dtype="float32", seed=0)
+    t = Trainer(cfg)
+    n_trainable = sum(p.numel() for p in t.model.parameters()
+                      if p.requires_grad)
+    n_total = sum(p.numel() for p in t.model.parameters())
+    assert n_trainable == n_total
+    before = t.model.embed.weight.detach().clone()
+    t.train_step(torch.randint(0, 256, (2, 17)))
+    assert not torch.equal(before, t.model.embed.weight.detach())
