@@ -126,6 +126,10 @@ def build_app(engine: Engine, tokenizer=None,
     loop = EngineLoop(engine)
     app.state.engine_loop = loop
 
+    @app.on_event("shutdown")
+    def _shutdown():
+        loop.shutdown()
+
     # Prometheus metrics (the reference exposes controller metrics behind
     # kube-rbac-proxy; the serving runtime gets request/token counters and
     # latency histograms — SURVEY.md §5 observability)
@@ -176,6 +180,9 @@ def build_app(engine: Engine, tokenizer=None,
 
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
+        # clamp generation to what the context window can hold
+        req.max_tokens = max(1, min(req.max_tokens,
+                                    engine.cfg.max_seq_len - 1))
         ids = tok.encode(req.prompt)[-engine.cfg.max_seq_len +
                                      req.max_tokens + 1:]
         # guard: a fallback tokenizer may emit ids past a small model's

@@ -16,7 +16,8 @@ def test_forward_shapes(name):
 
 
 def test_registry_has_flagship_configs():
-    for name in ("llama2-7b", "llama2-70b", "falcon-40b", "falcon-7b", "opt-125m"):
+    for name in ("llama2-7b", "llama2-70b", "llama3-8b", "llama3-70b",
+                 "mistral-7b", "falcon-40b", "falcon-7b", "opt-125m"):
         cfg = get_config(name)
         assert cfg.num_heads % cfg.num_kv_heads == 0
     # llama2-7b parameter count sanity (±10%)
@@ -150,3 +151,21 @@ def test_preempted_output_preserved():
     assert len(combined) == 10
     # greedy continuation after re-prefill matches the uninterrupted run
     assert combined == ref, (combined, ref)
+
+
+@pytest.mark.parametrize("name", ["tiny-opt", "tiny-falcon"])
+def test_engine_generates_all_families(name):
+    """Decode path per family: OPT (learned positions, no rope, biases)
+    and falcon (parallel residual, MQA) — the BASELINE config families."""
+    m = build_model(name, dtype=torch.float32, seed=6)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=9)
+    out = eng.generate([5, 4, 3], max_new_tokens=6)
+    assert len(out) == 6
+    assert all(0 <= t < m.cfg.vocab_size for t in out)
+    # decode must agree with the full-forward argmax continuation
+    full = [5, 4, 3]
+    for _ in range(6):
+        with torch.no_grad():
+            logits = m(torch.tensor([full]))
+        full.append(int(logits[0, -1].argmax()))
+    assert out == full[3:], (out, full[3:])
