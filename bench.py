@@ -100,7 +100,8 @@ def bench_serve(args) -> dict:
     dtype = torch.bfloat16 if torch.cuda.is_available() else torch.float32
     eng = Engine(args.model, dtype=dtype,
                  kv_blocks=None if torch.cuda.is_available() else 4096,
-                 max_batch=args.serve_batch, seed=17)
+                 max_batch=args.serve_batch, seed=17,
+                 load_in_8bit=args.serve_8bit)
     if comm.is_dist() and comm.world_size() > 1:
         # TP serving: rank 0 drives the engine + measures; workers follow
         # the broadcast protocol until shutdown (serve/tp_worker.py).
@@ -146,7 +147,8 @@ def bench_serve(args) -> dict:
         # TP serving keeps total work fixed as ranks grow
         "scaling": "strong" if comm.world_size() > 1 else "weak",
         "vs_baseline": None,
-        "dtype": "bf16" if torch.cuda.is_available() else "float32",
+        "dtype": ("bf16-act+fp8-w" if args.serve_8bit else "bf16")
+                 if torch.cuda.is_available() else "float32",
         "data": "synthetic",
         "config": {"model": args.model, "global_batch": args.serve_batch,
                    "seq_len": args.prompt_len,
@@ -165,6 +167,9 @@ def main():
     p.add_argument("--micro-batch", type=int, default=4)
     p.add_argument("--serve-batch", type=int, default=32)
     p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--serve-8bit", action="store_true",
+                   help="fp8-e4m3 weight-only decode (NOT the headline "
+                        "config; reported with dtype=bf16-act+fp8-w)")
     args = p.parse_args()
     if args.model is None:
         args.model = "llama2-7b" if torch.cuda.is_available() else "tiny-llama"
