@@ -28,6 +28,19 @@ log = logging.getLogger("runbooks_amd.controller")
 API = "substratus.ai/v1"
 
 
+try:
+    from prometheus_client import Counter, Histogram
+    _M_RECONCILES = Counter("rb_reconciles_total", "reconcile calls",
+                            ["kind"])
+    _M_ERRORS = Counter("rb_reconcile_errors_total", "reconcile errors",
+                        ["kind"])
+    _M_LAT = Histogram("rb_reconcile_seconds", "reconcile latency",
+                       ["kind"],
+                       buckets=(.001, .005, .01, .05, .1, .5, 1, 5))
+except ImportError:  # pragma: no cover
+    _M_RECONCILES = _M_ERRORS = _M_LAT = None
+
+
 class ControllerManager:
     def __init__(self, kube: KubeClient, cloud, sci_client):
         self.kube = kube
@@ -49,6 +62,20 @@ class ControllerManager:
 
     # -- single-object reconcile -------------------------------------------
     def reconcile_object(self, kind: str, namespace: str, name: str) -> None:
+        import time as _time
+        t0 = _time.perf_counter()
+        try:
+            self._reconcile_object(kind, namespace, name)
+        except Exception:
+            if _M_ERRORS is not None:
+                _M_ERRORS.labels(kind).inc()
+            raise
+        finally:
+            if _M_RECONCILES is not None:
+                _M_RECONCILES.labels(kind).inc()
+                _M_LAT.labels(kind).observe(_time.perf_counter() - t0)
+
+    def _reconcile_object(self, kind: str, namespace: str, name: str) -> None:
         raw = self.kube.get(API, kind, namespace, name)
         if raw is None:
             return
@@ -275,6 +302,21 @@ def _serve_health(addr: str) -> None:
 
     class H(http.server.BaseHTTPRequestHandler):
         def do_GET(self):
+            if self.path == "/metrics":
+                try:
+                    from prometheus_client import (
+                        CONTENT_TYPE_LATEST,
+                        generate_latest,
+                    )
+                    body = generate_latest()
+                    self.send_response(200)
+                    self.send_header("Content-Type", CONTENT_TYPE_LATEST)
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                    return
+                except ImportError:
+                    pass
             code = 200 if self.path in ("/healthz", "/readyz") else 404
             self.send_response(code)
             self.end_headers()

@@ -360,3 +360,16 @@ def test_manager_watch_loop(env):
     finally:
         mgr.stop()
         t.join(timeout=3)
+
+
+def test_manager_metrics(env):
+    """Reconcile counters/latency exported for Prometheus (the reference
+    exposes controller-runtime metrics behind kube-rbac-proxy)."""
+    kube, cloud, sci, mgr = env
+    from prometheus_client import REGISTRY
+
+    kube.create(Model(name="metr", image="img:1").to_dict())
+    mgr.reconcile_object("Model", "default", "metr")
+    v = REGISTRY.get_sample_value("rb_reconciles_total",
+                                  {"kind": "Model"})
+    assert v and v >= 1
