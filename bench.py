@@ -24,6 +24,16 @@ import json
 import os
 import time
 
+# Optional hipBLASLt TunableOp: point RB_TUNABLEOP_FILE at a previously
+# tuned CSV to replay the tuned GEMM selections (tuning itself is done
+# offline: PYTORCH_TUNABLEOP_TUNING=1; see NOTES-ROUND2.md). Must be set
+# before torch initializes its GEMM backends.
+if os.environ.get("RB_TUNABLEOP_FILE"):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          os.environ["RB_TUNABLEOP_FILE"])
+
 import torch
 
 from runbooks_amd.parallel import comm
