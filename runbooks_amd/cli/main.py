@@ -276,6 +276,37 @@ def infer(server_name, prompt, namespace, max_tokens, port):
 
 
 @main.command()
+@click.argument("name")
+@click.option("-n", "--namespace", default="default")
+def suspend(name, namespace):
+    """Suspend a Notebook (deletes its pod, keeps artifacts) — the
+    reference TUI's suspend command (reference tui/common.go:271-301)."""
+    kube = _kube()
+    try:
+        kube.patch("substratus.ai/v1", "Notebook", namespace, name,
+                   {"spec": {"suspend": True}})
+        _say(f"notebook {name} suspended")
+    except Exception:
+        _say(f"[red]Notebook/{name} not found[/red]")
+        sys.exit(1)
+
+
+@main.command()
+@click.argument("name")
+@click.option("-n", "--namespace", default="default")
+def resume(name, namespace):
+    """Resume a suspended Notebook."""
+    kube = _kube()
+    try:
+        kube.patch("substratus.ai/v1", "Notebook", namespace, name,
+                   {"spec": {"suspend": False}})
+        _say(f"notebook {name} resuming")
+    except Exception:
+        _say(f"[red]Notebook/{name} not found[/red]")
+        sys.exit(1)
+
+
+@main.command()
 @click.argument("kind")
 @click.argument("name")
 @click.option("-n", "--namespace", default="default")

@@ -261,3 +261,21 @@ def test_tui_components():
     # non-TTY path picks the first deterministically
     assert select_manifest([m1, m2]) is m1
     assert select_manifest([m2]) is m2
+
+
+def test_cli_suspend_resume(monkeypatch):
+    import importlib
+    from click.testing import CliRunner
+    cli_main = importlib.import_module("runbooks_amd.cli.main")
+
+    kube = MemoryKubeClient()
+    kube.create(Notebook(name="nb1", image="i").to_dict())
+    monkeypatch.setattr(cli_main, "_kube", lambda: kube)
+    r = CliRunner().invoke(cli_main.main, ["suspend", "nb1"])
+    assert r.exit_code == 0
+    raw = kube.get("substratus.ai/v1", "Notebook", "default", "nb1")
+    assert raw["spec"]["suspend"] is True
+    r = CliRunner().invoke(cli_main.main, ["resume", "nb1"])
+    assert r.exit_code == 0
+    raw = kube.get("substratus.ai/v1", "Notebook", "default", "nb1")
+    assert raw["spec"]["suspend"] is False
