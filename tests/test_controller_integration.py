@@ -373,3 +373,25 @@ def test_manager_metrics(env):
     v = REGISTRY.get_sample_value("rb_reconciles_total",
                                   {"kind": "Model"})
     assert v and v >= 1
+
+
+def test_build_job_recreated_when_out_of_date(env):
+    """Changing the build spec (new branch -> new image URL) deletes the
+    stale kaniko Job and creates a fresh one
+    (reference build_reconciler.go:128-136)."""
+    kube, cloud, sci, mgr = env
+    m = Model(name="m-ood",
+              build=Build(git=BuildGit(url="https://x", branch="v1")))
+    kube.create(m.to_dict())
+    mgr.reconcile_all(rounds=1)
+    job1 = kube.get("batch/v1", "Job", "default", "m-ood-model-bld")
+    assert job1["metadata"]["annotations"]["image"].endswith(":v1")
+
+    kube.patch("substratus.ai/v1", "Model", "default", "m-ood",
+               {"spec": {"build": {"git": {"url": "https://x",
+                                           "branch": "v2"}}}})
+    mgr.reconcile_all(rounds=1)
+    job2 = kube.get("batch/v1", "Job", "default", "m-ood-model-bld")
+    assert job2["metadata"]["annotations"]["image"].endswith(":v2")
+    # fresh object (recreated, not patched)
+    assert job2["metadata"]["uid"] != job1["metadata"]["uid"]
