@@ -10,6 +10,7 @@ the skinny adapter GEMMs running through hipBLASLt-backed torch.matmul
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 from torch import nn
@@ -18,6 +19,35 @@ from ..parallel.tp import ColumnParallelLinear, RowParallelLinear
 
 DEFAULT_TARGETS = ("q_proj", "k_proj", "v_proj", "o_proj",
                    "gate_proj", "up_proj", "down_proj", "fc1", "fc2")
+
+
+# RB_EXPERIMENTAL_GEMM=1 routes the big frozen-weight GEMMs through the
+# 8-phase 256^2 kernel (ops/csrc/gemm_train.hip) instead of hipBLASLt.
+# The dgrad (dy @ W) is NN-shaped, so frozen weights get a one-time
+# transposed copy (cheap next to 288 GB HBM) and run as NT(dy, W^T).
+_USE_CUSTOM_GEMM = os.environ.get("RB_EXPERIMENTAL_GEMM", "0") == "1"
+_WT_CACHE: dict[int, torch.Tensor] = {}
+
+
+def _nt_ok(m: int, n: int, k: int) -> bool:
+    return m % 256 == 0 and n % 256 == 0 and k % 128 == 0
+
+
+def _nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """a[M,K] @ b[N,K]^T — custom kernel when eligible, else hipBLASLt."""
+    if (_USE_CUSTOM_GEMM and a.is_cuda and a.dtype == torch.bfloat16
+            and _nt_ok(a.shape[0], b.shape[0], a.shape[1])):
+        from ..ops import _backend
+        return _backend.ext().train_gemm_nt(a.contiguous(), b.contiguous())
+    return a @ b.t()
+
+
+def _wt(w: torch.Tensor) -> torch.Tensor:
+    wt = _WT_CACHE.get(w.data_ptr())
+    if wt is None:
+        wt = w.t().contiguous()
+        _WT_CACHE[w.data_ptr()] = wt
+    return wt
 
 
 class _LoRAFused(torch.autograd.Function):
@@ -34,7 +64,7 @@ class _LoRAFused(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, a, b, scale):
         t = x @ a.t()                       # [T, r]
-        y = torch.addmm(x @ w.t(), t, b.t(), beta=1.0, alpha=scale)
+        y = torch.addmm(_nt(x, w), t, b.t(), beta=1.0, alpha=scale)
         ctx.save_for_backward(x, w, a, b, t)
         ctx.scale = scale
         return y
@@ -44,7 +74,11 @@ class _LoRAFused(torch.autograd.Function):
         x, w, a, b, t = ctx.saved_tensors
         s = ctx.scale
         t2 = dy @ b                         # [T, r]
-        dx = torch.addmm(dy @ w, t2, a, beta=1.0, alpha=s)
+        if (_USE_CUSTOM_GEMM and dy.is_cuda and dy.dtype == torch.bfloat16
+                and _nt_ok(dy.shape[0], w.shape[1], dy.shape[1])):
+            dx = torch.addmm(_nt(dy, _wt(w)), t2, a, beta=1.0, alpha=s)
+        else:
+            dx = torch.addmm(dy @ w, t2, a, beta=1.0, alpha=s)
         da = torch.mm(t2.t(), x).mul_(s)
         db = torch.mm(dy.t(), t).mul_(s)
         return dx, None, da, db, None
