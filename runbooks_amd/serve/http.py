@@ -23,6 +23,26 @@ from pydantic import BaseModel
 from .engine import Engine
 from .tokenizer import load_tokenizer
 
+# Prometheus metrics are process-global: created once here so a second
+# build_app() in the same process (tests, uvicorn reload) cannot trip
+# "Duplicated timeseries".
+try:
+    from prometheus_client import (
+        CONTENT_TYPE_LATEST,
+        Counter,
+        Gauge,
+        Histogram,
+        generate_latest,
+    )
+    M_REQS = Counter("rb_requests_total", "completion requests")
+    M_TOKENS = Counter("rb_generated_tokens_total", "generated tokens")
+    M_LAT = Histogram("rb_request_seconds", "request latency",
+                      buckets=(.05, .1, .25, .5, 1, 2.5, 5, 10, 30, 60))
+    M_RUNNING = Gauge("rb_running_requests", "requests decoding")
+    M_KV_FREE = Gauge("rb_kv_blocks_free", "free KV cache blocks")
+except ImportError:  # pragma: no cover
+    M_REQS = M_TOKENS = M_LAT = M_RUNNING = M_KV_FREE = None
+
 
 class CompletionRequest(BaseModel):
     prompt: str = ""
@@ -133,30 +153,15 @@ def build_app(engine: Engine, tokenizer=None,
     # Prometheus metrics (the reference exposes controller metrics behind
     # kube-rbac-proxy; the serving runtime gets request/token counters and
     # latency histograms — SURVEY.md §5 observability)
-    try:
-        from prometheus_client import (
-            CONTENT_TYPE_LATEST,
-            Counter,
-            Gauge,
-            Histogram,
-            generate_latest,
-        )
-        m_reqs = Counter("rb_requests_total", "completion requests")
-        m_tokens = Counter("rb_generated_tokens_total", "generated tokens")
-        m_lat = Histogram("rb_request_seconds", "request latency",
-                          buckets=(.05, .1, .25, .5, 1, 2.5, 5, 10, 30, 60))
-        m_running = Gauge("rb_running_requests", "requests decoding")
-        m_kv_free = Gauge("rb_kv_blocks_free", "free KV cache blocks")
-
+    m_reqs, m_tokens, m_lat = M_REQS, M_TOKENS, M_LAT
+    if M_REQS is not None:
         @app.get("/metrics")
         def metrics():
-            m_running.set(len(engine.running))
-            m_kv_free.set(len(engine.allocator.free))
+            M_RUNNING.set(len(engine.running))
+            M_KV_FREE.set(len(engine.allocator.free))
             from fastapi import Response
             return Response(generate_latest(),
                             media_type=CONTENT_TYPE_LATEST)
-    except ImportError:
-        m_reqs = m_tokens = m_lat = None
 
     app.state.metrics = (m_reqs, m_tokens, m_lat)
 

@@ -89,3 +89,12 @@ def test_top_p_filter_math():
     # 0.5 + 0.3 crosses 0.7 -> first two kept, rest -inf
     assert torch.isfinite(f[0, 0]) and torch.isfinite(f[0, 1])
     assert torch.isinf(f[0, 2]) and torch.isinf(f[0, 3])
+
+
+def test_build_app_twice_no_metric_collision():
+    eng = Engine("tiny-llama", device="cpu", dtype=torch.float32,
+                 kv_blocks=64, seed=1)
+    a1 = build_app(eng, model_name="m1")
+    a2 = build_app(eng, model_name="m2")
+    a1.state.engine_loop.shutdown()
+    a2.state.engine_loop.shutdown()
