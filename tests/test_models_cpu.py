@@ -169,3 +169,25 @@ def test_engine_generates_all_families(name):
             logits = m(torch.tensor([full]))
         full.append(int(logits[0, -1].argmax()))
     assert out == full[3:], (out, full[3:])
+
+
+def test_graphed_decoder_staging():
+    """GraphedDecoder._stage fills the pinned buffer correctly, including
+    dummy-padding of rows beyond the live batch (GPU-independent part of
+    serve/graph.py)."""
+    from runbooks_amd.serve.graph import GraphedDecoder
+
+    gd = GraphedDecoder(model=None, caches=None, max_batch=8, max_blocks=4,
+                        dummy_block=99, device="cpu")
+    bt = gd._stage(4, tokens=[7, 8], positions=[3, 5], slots=[12, 20],
+                   block_rows=[[1, 2], [4]], seq_lens=[4, 6])
+    h = gd.h_staging
+    assert h[0:4].tolist() == [7, 8, 0, 0]          # tokens + pad
+    assert h[4:8].tolist() == [3, 5, 0, 0]          # positions + pad
+    assert h[8:10].tolist() == [12, 20]             # live slots
+    assert (h[10:12] == 99 * 16).all()              # dummy slots
+    assert h[12:16].tolist() == [4, 6, 1, 1]        # seq_lens + pad
+    assert bt.shape == (4, 4)
+    assert bt[0].tolist() == [1, 2, 99, 99]
+    assert bt[1].tolist() == [4, 99, 99, 99]
+    assert (bt[2:] == 99).all()
