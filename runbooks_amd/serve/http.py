@@ -141,14 +141,18 @@ class EngineLoop:
 
 def build_app(engine: Engine, tokenizer=None,
               model_name: str = "model") -> FastAPI:
-    app = FastAPI(title="runbooks-amd-server")
-    tok = tokenizer or load_tokenizer(None)
-    loop = EngineLoop(engine)
-    app.state.engine_loop = loop
+    import contextlib
 
-    @app.on_event("shutdown")
-    def _shutdown():
+    loop = EngineLoop(engine)
+
+    @contextlib.asynccontextmanager
+    async def _lifespan(app):
+        yield
         loop.shutdown()
+
+    app = FastAPI(title="runbooks-amd-server", lifespan=_lifespan)
+    tok = tokenizer or load_tokenizer(None)
+    app.state.engine_loop = loop
 
     # Prometheus metrics (the reference exposes controller metrics behind
     # kube-rbac-proxy; the serving runtime gets request/token counters and
