@@ -3,9 +3,11 @@
 The reference platform's fine-tune path is LoRA via the external HF
 trainer image (reference examples/llama2-7b/finetuned-model.yaml,
 SURVEY.md §2b "trainer image"); here it is native: frozen base weights,
-trainable low-rank A/B pairs on the attention + MLP projections, with
-the skinny adapter GEMMs running through hipBLASLt-backed torch.matmul
-(a fused gfx950 LoRA kernel is the planned upgrade on this path).
+trainable low-rank A/B pairs on the attention + MLP projections. The
+fused fwd+bwd path (_LoRAFused) keeps every product a single hipBLASLt
+call and rides both gradient merges on addmm epilogues — no separate
+mul/add kernels (the eager formulation measured ~450 extra elementwise
+launches per llama2-7b step; profiles/).
 """
 from __future__ import annotations
 
