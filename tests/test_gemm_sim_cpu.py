@@ -1,0 +1,181 @@
+"""CPU simulation of ops/csrc/gemm_train.hip's index dataflow.
+
+The 8-phase GEMM was drafted without GPU budget (round 1 end). This
+test mirrors its exact address arithmetic in numpy — glds staging with
+the pre-swizzled source, st_16x32-swizzled LDS images, per-wave fragment
+reads, phase-by-phase MFMA accumulation with the documented
+mfma_f32_16x16x32_bf16 lane maps, and the C epilogue — and checks the
+result against A @ B^T. It validates every LOGIC decision in the kernel;
+what it cannot check is hardware timing (vmcnt/barriers) and the
+intrinsic's true lane maps (covered on-GPU by mfma_probe_16x16x32).
+
+Keep in sync with gemm_train.hip when editing either.
+"""
+import numpy as np
+import pytest
+
+BM = BN = 256
+BK = 64
+THREADS = 512
+HALF_ELEMS = 128 * BK            # one half-tile image, in bf16 elements
+
+
+def swz_e(e):
+    """st_16x32 swizzle in ELEMENT units (2 B each): byte bit9->bit5 is
+    element bit8->bit4."""
+    return e ^ (((e >> 8) & 1) << 4)
+
+
+def stage_half(src, lds_img):
+    """src: [128, BK] float array (a half-tile slice, already row-major).
+    Mirrors stage_half(): lane-linear LDS write, pre-swizzled source."""
+    for tid in range(THREADS):
+        for p in range(2):
+            e0 = p * (THREADS * 8) + tid * 8    # element offset this lane
+            q = swz_e(e0)
+            row, col = q // BK, q % BK
+            lds_img[e0:e0 + 8] = src[row, col:col + 8]
+
+
+def lds_frag(lds_img, logical_elem):
+    e = swz_e(logical_elem)
+    return lds_img[e:e + 8]
+
+
+def mfma_16x16x32(a_frags, b_frags, acc):
+    """a_frags/b_frags: per-lane 8-val fragments, lists indexed by lane.
+    Reconstruct A[16,32], B[32,16] via the documented maps and do the
+    matmul; acc [16,16] += A @ B."""
+    A = np.zeros((16, 32))
+    B = np.zeros((32, 16))
+    for lane in range(64):
+        i = lane & 15
+        kq = lane >> 4
+        for e in range(8):
+            A[i, kq * 8 + e] = a_frags[lane][e]
+            B[kq * 8 + e, i] = b_frags[lane][e]
+    acc += A @ B
+    return acc
+
+
+def xcd_remap(orig, nwg):
+    q8, r8 = nwg // 8, nwg % 8
+    xcd, pos = orig % 8, orig // 8
+    return (xcd * (q8 + 1) if xcd < r8 else
+            r8 * (q8 + 1) + (xcd - r8) * q8) + pos
+
+
+def a_byte_elem(lane, fm, ks):
+    return (fm * 16 + (lane & 15)) * BK + ks * 32 + ((lane >> 4) * 8)
+
+
+def b_byte_elem(lane, wn, fn, ks):
+    return (((wn * 64 + fn * 16 + (lane & 15)) & 127) * BK + ks * 32 +
+            ((lane >> 4) * 8))
+
+
+def simulate(A, B):
+    """Run the kernel's dataflow. A [M,K], B [N,K] float64."""
+    M, K = A.shape
+    N = B.shape[0]
+    C = np.zeros((M, N))
+    ktiles = K // BK
+    nwg = (M // BM) * (N // BN)
+    ntiles_n = N // BN
+
+    for orig in range(nwg):
+        wg = xcd_remap(orig, nwg)
+        tile_m, tile_n = wg // ntiles_n, wg % ntiles_n
+        a_tile = A[tile_m * BM:(tile_m + 1) * BM]
+        b_tile = B[tile_n * BN:(tile_n + 1) * BN]
+
+        # LDS: [db][op][half] images
+        lds = np.zeros((2, 2, 2, HALF_ELEMS))
+
+        def stage(kt, op, half):
+            src = (a_tile if op == 0 else b_tile)[
+                half * 128:(half + 1) * 128, kt * BK:(kt + 1) * BK]
+            stage_half(src, lds[kt & 1, op, half])
+
+        # the kernel stages in prologue order then one half per phase; the
+        # DATA outcome is order-independent in this simulation, so stage
+        # each K-tile fully before its compute (the timing correctness is
+        # the vmcnt discipline, not simulated here)
+        for wid in range(8):
+            pass  # staging is workgroup-wide; done below per K-tile
+
+        acc = np.zeros((8, 2, 8, 4, 16, 16))  # [wid][..] -> use dict
+        acc = {wid: np.zeros((8, 4, 16, 16)) for wid in range(8)}
+
+        for kt in range(ktiles):
+            db = kt & 1
+            for op in range(2):
+                for half in range(2):
+                    stage(kt, op, half)
+            for wid in range(8):
+                wm, wn = wid >> 2, wid & 3
+                A_img = lds[db, 0, wm]
+                B_img = lds[db, 1, wn >> 1]
+                # phases 1-4 = quadrants (fmh, fnh) in kernel order
+                for fmh, fnh in ((0, 0), (0, 1), (1, 1), (1, 0)):
+                    for fm in range(4):
+                        for fn in range(2):
+                            FM, FN = fmh * 4 + fm, fnh * 2 + fn
+                            for ks in range(2):
+                                af = [lds_frag(A_img,
+                                               a_byte_elem(l, FM, ks))
+                                      for l in range(64)]
+                                bf = [lds_frag(B_img,
+                                               b_byte_elem(l, wn, FN, ks))
+                                      for l in range(64)]
+                                mfma_16x16x32(af, bf, acc[wid][FM, FN])
+
+        # epilogue: C[i=(lane>>4)*4+e][j=lane&15] per fragment
+        for wid in range(8):
+            wm, wn = wid >> 2, wid & 3
+            crow0 = tile_m * BM + wm * 128
+            ccol0 = tile_n * BN + wn * 64
+            for FM in range(8):
+                for FN in range(4):
+                    C[crow0 + FM * 16:crow0 + FM * 16 + 16,
+                      ccol0 + FN * 16:ccol0 + FN * 16 + 16] += \
+                        acc[wid][FM, FN]
+    return C
+
+
+def test_xcd_remap_bijective():
+    for nwg in (8, 16, 128, 768 // 256 * 3, 13, 100):
+        seen = {xcd_remap(i, nwg) for i in range(nwg)}
+        assert seen == set(range(nwg)), nwg
+
+
+def test_swizzle_involution_and_16B_chunks():
+    for e in range(0, HALF_ELEMS, 8):
+        assert swz_e(swz_e(e)) == e
+        # a 16-B chunk maps contiguously
+        assert swz_e(e + 7) == swz_e(e) + 7
+
+
+def test_staging_roundtrip():
+    """stage_half + swizzled reads reproduce the source exactly."""
+    rng = np.random.default_rng(0)
+    src = rng.standard_normal((128, BK))
+    img = np.zeros(HALF_ELEMS)
+    stage_half(src, img)
+    for row in range(0, 128, 7):
+        for col in range(0, BK, 8):
+            got = lds_frag(img, row * BK + col)
+            assert np.array_equal(got, src[row, col:col + 8]), (row, col)
+
+
+@pytest.mark.parametrize("M,N,K", [
+    (256, 256, 128),
+    (512, 512, 128),   # multi-tile + XCD remap (nwg=4, nwg%8 != 0)
+])
+def test_gemm_dataflow_matches_matmul(M, N, K):
+    rng = np.random.default_rng(M + K)
+    A = rng.standard_normal((M, K))
+    B = rng.standard_normal((N, K))
+    C = simulate(A, B)
+    ref = A @ B.T
+    assert np.allclose(C, ref, atol=1e-9), np.abs(C - ref).max()
