@@ -13,13 +13,12 @@
 // `RB_EXPERIMENTAL=1 pytest tests/test_gpu_ops.py -k train_gemm` first,
 // then benchmarks/kernels.py.
 //
-// KNOWN LIMIT (audit): at M=2048, N=4096 the 256^2 tiling yields only
-// 128 WGs -> half the CUs idle; expect ~660 TF there, BELOW hipBLASLt.
-// The template wins where (M/256)*(N/256) >= 256: the N=11008 MLP
-// shapes (344 WGs) and any fused-projection N. To cover N=4096 either
-// add split-K=2 with fp32 slabs + combine (skinny_gemm.hip pattern) or
-// fuse q/k/v for the training forward (frozen base weights need no
-// wgrad under LoRA).
+// Grid fill: when (M/256)*(N/256) < 256 the host splits K (fp32 slabs
+// + a combine kernel, the skinny_gemm.hip pattern) — M=2048 N=4096 runs
+// as 128 tiles x ksplit 2 = 256 WGs. The whole index dataflow (staging
+// swizzle, fragment maps, phase accumulation, split-K slabs + combine,
+// epilogue) is verified by the CPU simulation in
+// tests/test_gemm_sim_cpu.py; keep the two in sync.
 //
 // Template geometry (guide table):
 //   tile BM x BN = 256 x 256, BK = 64, 8 waves (2M x 4N), 512 threads
@@ -84,9 +83,13 @@ __device__ __forceinline__ bf16x8v lds_read_frag(const char *lds,
   return *reinterpret_cast<const bf16x8v *>(lds + swz(logical_byte));
 }
 
+// cp: bf16 output when gridDim.z == 1; otherwise fp32 partial slabs
+// [tile][kslice][BM][BN] (combined by gemm_nt_combine_kernel — the
+// kernel boundary is the release, as in skinny_gemm.hip).
 __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
     const uint16_t *__restrict__ ap, const uint16_t *__restrict__ bp,
-    uint16_t *__restrict__ cp, int M, int N, int K) {
+    uint16_t *__restrict__ cp, float *__restrict__ slabs,
+    int M, int N, int K) {
   extern __shared__ __attribute__((aligned(16))) char lds[];
 
   const int tid = threadIdx.x;
@@ -105,8 +108,11 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
   const int tile_m = wg / ntiles_n;
   const int tile_n = wg % ntiles_n;
 
-  const uint16_t *a_tile = ap + (int64_t)tile_m * BM * K;
-  const uint16_t *b_tile = bp + (int64_t)tile_n * BN * K;
+  const int ksplit = gridDim.z;
+  const int kslice = blockIdx.z;
+  const int kper = K / ksplit;             // this slice's contraction depth
+  const uint16_t *a_tile = ap + (int64_t)tile_m * BM * K + kslice * kper;
+  const uint16_t *b_tile = bp + (int64_t)tile_n * BN * K + kslice * kper;
 
   // LDS layout: [db][op][half] images of 16 KiB each
   auto lds_img = [&](int db, int op, int half) -> uint32_t {
@@ -115,7 +121,7 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
 
   // ---- prologue: stage K-tiles 0 and half of 1 --------------------------
   // order: A0h0 A0h1 B0h0 B0h1 | A1h0 A1h1 B1h0 (7 half-tiles, 14 glds)
-  const int ktiles = K / BK;
+  const int ktiles = kper / BK;
   {
     for (int p = 0; p < 2; ++p) stage_half(lds, a_tile + 0 * K + 0, K,
                                            lds_img(0, 0, 0), tid, p);
@@ -281,19 +287,57 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
     __builtin_amdgcn_s_barrier();
   }
 
-  // ---- epilogue: acc -> C (bf16). C[i=(lane>>4)*4+e][j=lane&15] --------
-  const int crow0 = tile_m * BM + wm * 128;
-  const int ccol0 = tile_n * BN + wn * 64;
+  // ---- epilogue: acc -> C (bf16) or fp32 slab ---------------------------
+  // C[i=(lane>>4)*4+e][j=lane&15] per 16x16 fragment.
+  if (ksplit == 1) {
+    const int crow0 = tile_m * BM + wm * 128;
+    const int ccol0 = tile_n * BN + wn * 64;
 #pragma unroll
-  for (int fm = 0; fm < 8; ++fm)
+    for (int fm = 0; fm < 8; ++fm)
 #pragma unroll
-    for (int fn = 0; fn < 4; ++fn) {
-      const int r0 = crow0 + fm * 16 + ((lane >> 4) * 4);
-      const int c = ccol0 + fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 4; ++fn) {
+        const int r0 = crow0 + fm * 16 + ((lane >> 4) * 4);
+        const int c = ccol0 + fn * 16 + (lane & 15);
 #pragma unroll
-      for (int e = 0; e < 4; ++e)
-        cp[(int64_t)(r0 + e) * N + c] = rb::f32_to_bf16(acc[fm][fn][e]);
-    }
+        for (int e = 0; e < 4; ++e)
+          cp[(int64_t)(r0 + e) * N + c] = rb::f32_to_bf16(acc[fm][fn][e]);
+      }
+  } else {
+    float *slab = slabs +
+        (((int64_t)wg * ksplit + kslice) * BM) * BN;  // [tile][slice] image
+    const int srow0 = wm * 128;
+    const int scol0 = wn * 64;
+#pragma unroll
+    for (int fm = 0; fm < 8; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        const int r0 = srow0 + fm * 16 + ((lane >> 4) * 4);
+        const int c = scol0 + fn * 16 + (lane & 15);
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          slab[(int64_t)(r0 + e) * BN + c] = acc[fm][fn][e];
+      }
+  }
+}
+
+// Combine ksplit fp32 slabs -> bf16 C. wg/tile mapping mirrors the main
+// kernel's (post-XCD-remap wg index owns slab block wg).
+__global__ void gemm_nt_combine_kernel(const float *__restrict__ slabs,
+                                       uint16_t *__restrict__ cp,
+                                       int M, int N, int ksplit) {
+  const int ntiles_n = N / BN;
+  const int64_t total = (int64_t)M * N;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int r = (int)(i / N), c = (int)(i % N);
+    const int wg = (r / BM) * ntiles_n + (c / BN);
+    const float *base = slabs + (((int64_t)wg * ksplit) * BM) * BN +
+        (int64_t)(r % BM) * BN + (c % BN);
+    float v = 0.0f;
+    for (int ks = 0; ks < ksplit; ++ks) v += base[(int64_t)ks * BM * BN];
+    cp[i] = rb::f32_to_bf16(v);
+  }
 }
 
 }  // namespace
@@ -307,9 +351,17 @@ at::Tensor train_gemm_nt(at::Tensor a, at::Tensor b) {
   TORCH_CHECK((int)b.size(1) == K, "train_gemm_nt: K mismatch");
   TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % (2 * BK) == 0,
               "train_gemm_nt: M%256, N%256, K%128 required (experimental)");
+  // split K until the grid covers the chip (fixes the M=2048 N=4096
+  // underfill: 128 tiles -> ksplit 2 -> 256 WGs)
+  const int tiles = (M / BM) * (N / BN);
+  int ksplit = 1;
+  while (tiles * ksplit < 256 && ksplit < 8 &&
+         (K / (ksplit * 2)) % (2 * BK) == 0)
+    ksplit *= 2;
+
   auto c = at::empty({M, N}, a.options());
   auto stream = at::hip::getCurrentHIPStream();
-  const dim3 grid(M / BM, N / BN);
+  const dim3 grid(M / BM, N / BN, ksplit);
   constexpr size_t shmem = 2 * DB_BYTES;   // 128 KiB
   static bool cfg_done = false;
   if (!cfg_done) {
@@ -318,10 +370,27 @@ at::Tensor train_gemm_nt(at::Tensor a, at::Tensor b) {
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)shmem);
     cfg_done = true;
   }
+  float *slabs = nullptr;
+  at::Tensor ws;
+  if (ksplit > 1) {
+    static at::Tensor ws_cache;
+    const int64_t need = (int64_t)tiles * ksplit * BM * BN;
+    if (!ws_cache.defined() || ws_cache.numel() < need ||
+        ws_cache.device() != a.device())
+      ws_cache = at::empty({need}, a.options().dtype(at::kFloat));
+    ws = ws_cache;
+    slabs = ws.data_ptr<float>();
+  }
   hipLaunchKernelGGL(gemm_nt_8phase_kernel, grid, dim3(THREADS), shmem,
                      stream, (const uint16_t *)a.data_ptr(),
                      (const uint16_t *)b.data_ptr(),
-                     (uint16_t *)c.data_ptr(), M, N, K);
+                     (uint16_t *)c.data_ptr(), slabs, M, N, K);
+  if (ksplit > 1) {
+    const int cgrid = rb::rb_grid_1d((int64_t)M * N, 256);
+    hipLaunchKernelGGL(gemm_nt_combine_kernel, dim3(cgrid), dim3(256), 0,
+                       stream, slabs, (uint16_t *)c.data_ptr(), M, N,
+                       ksplit);
+  }
   return c;
 }
 

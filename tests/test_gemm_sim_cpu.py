@@ -74,10 +74,42 @@ def b_byte_elem(lane, wn, fn, ks):
             ((lane >> 4) * 8))
 
 
-def simulate(A, B):
-    """Run the kernel's dataflow. A [M,K], B [N,K] float64."""
+def host_ksplit(tiles, K):
+    """Mirrors train_gemm_nt's split-K selection."""
+    ksplit = 1
+    while tiles * ksplit < 256 and ksplit < 8 and \
+            (K // (ksplit * 2)) % (2 * BK) == 0:
+        ksplit *= 2
+    return ksplit
+
+
+def simulate(A, B, force_ksplit=None):
+    """Run the kernel's dataflow (incl. split-K slabs + combine).
+    A [M,K], B [N,K] float64."""
     M, K = A.shape
     N = B.shape[0]
+    tiles = (M // BM) * (N // BN)
+    ksplit = force_ksplit or host_ksplit(tiles, K)
+    if ksplit > 1:
+        # main kernel: per-slice partials into slabs indexed by logical wg
+        slabs = np.zeros((tiles, ksplit, BM, BN))
+        kper = K // ksplit
+        for ks in range(ksplit):
+            part = simulate(A[:, ks * kper:(ks + 1) * kper],
+                            B[:, ks * kper:(ks + 1) * kper], force_ksplit=1)
+            ntn = N // BN
+            for wg in range(tiles):
+                tm, tn = wg // ntn, wg % ntn
+                slabs[wg, ks] = part[tm * BM:(tm + 1) * BM,
+                                     tn * BN:(tn + 1) * BN]
+        # combine kernel mapping
+        C = np.zeros((M, N))
+        ntn = N // BN
+        for r in range(0, M, BM):
+            for c in range(0, N, BN):
+                wg = (r // BM) * ntn + (c // BN)
+                C[r:r + BM, c:c + BN] = slabs[wg].sum(axis=0)
+        return C
     C = np.zeros((M, N))
     ktiles = K // BK
     nwg = (M // BM) * (N // BN)
@@ -171,6 +203,7 @@ def test_staging_roundtrip():
 @pytest.mark.parametrize("M,N,K", [
     (256, 256, 128),
     (512, 512, 128),   # multi-tile + XCD remap (nwg=4, nwg%8 != 0)
+    (256, 512, 512),   # 2 tiles -> host picks split-K; slab + combine path
 ])
 def test_gemm_dataflow_matches_matmul(M, N, K):
     rng = np.random.default_rng(M + K)
