@@ -395,3 +395,45 @@ def test_build_job_recreated_when_out_of_date(env):
     assert job2["metadata"]["annotations"]["image"].endswith(":v2")
     # fresh object (recreated, not patched)
     assert job2["metadata"]["uid"] != job1["metadata"]["uid"]
+
+
+def test_controller_fuzz_random_object_graph(env):
+    """Random Models/Datasets/Servers/Notebooks with random (possibly
+    dangling) references: reconciliation must never wedge or corrupt —
+    every object ends with a coherent status once its chain is unblocked."""
+    import random
+
+    kube, cloud, sci, mgr = env
+    rng = random.Random(7)
+    names = [f"o{i}" for i in range(12)]
+    for i, n in enumerate(names):
+        kind = rng.choice(["Model", "Dataset", "Server", "Notebook"])
+        ref = rng.choice(names + ["dangling"])
+        if kind == "Model":
+            o = Model(name=n, image="img:x",
+                      model=ObjectRef(ref) if rng.random() < 0.4 else None,
+                      dataset=ObjectRef(ref) if rng.random() < 0.3 else None)
+        elif kind == "Dataset":
+            o = Dataset(name=n, image="img:x")
+        elif kind == "Server":
+            o = Server(name=n, image="img:x", model=ObjectRef(ref))
+        else:
+            o = Notebook(name=n, image="img:x",
+                         model=ObjectRef(ref) if rng.random() < 0.5 else None)
+        kube.create(o.to_dict())
+
+    for _ in range(4):
+        mgr.reconcile_all(rounds=2)
+        # complete any Jobs that appeared, unblocking chains
+        for j in kube.list("batch/v1", "Job"):
+            fake_job_complete(kube, "default", j["metadata"]["name"])
+
+    # invariant: every object has a status and either ready or a
+    # condition explaining why not
+    for kind in ("Model", "Dataset", "Server", "Notebook"):
+        for raw in kube.list(API, kind):
+            st = raw.get("status") or {}
+            assert "ready" in st, raw["metadata"]["name"]
+            if not st["ready"]:
+                assert st.get("conditions"), \
+                    f"{kind}/{raw['metadata']['name']} stuck with no reason"
