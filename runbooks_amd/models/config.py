@@ -28,6 +28,10 @@ class ModelConfig:
     rope_theta: float = 10000.0
     parallel_residual: bool = False  # falcon-style attn+mlp parallel block
     single_norm: bool = False        # falcon-7b: one shared ln for attn+mlp
+    # sliding-window attention width (mistral). Enforced by the serving
+    # engine at BLOCK_SIZE granularity (up to BLOCK_SIZE-1 extra tokens
+    # of context vs the strict window); expired KV blocks are freed.
+    sliding_window: int | None = None
     tie_embeddings: bool = False
     mlp_bias: bool = False
     attn_bias: bool = False
@@ -85,13 +89,12 @@ register(ModelConfig("llama3-70b", vocab_size=128256, hidden_size=8192,
                      num_layers=80, num_heads=64, num_kv_heads=8,
                      intermediate_size=28672, max_seq_len=8192,
                      rope_theta=500000.0))
-# mistral-7b is llama-shaped with GQA-8 (its 4k sliding window is not
-# applied — the paged cache holds the full context, which is strictly
-# more general and exact for seq <= 4096)
+# mistral-7b: llama-shaped GQA-8 with a 4k sliding window (the engine
+# enforces it block-aligned and frees expired KV blocks)
 register(ModelConfig("mistral-7b", vocab_size=32000, hidden_size=4096,
                      num_layers=32, num_heads=32, num_kv_heads=8,
                      intermediate_size=14336, max_seq_len=8192,
-                     rope_theta=1000000.0))
+                     rope_theta=1000000.0, sliding_window=4096))
 
 # --- falcon family (LayerNorm, RoPE, GELU, parallel residual, MQA/GQA) -----
 register(ModelConfig("falcon-7b", vocab_size=65024, hidden_size=4544,

@@ -10,6 +10,17 @@ Scheduling: one prefill (whole prompt in one pass) is admitted per step
 when capacity allows, then all running sequences decode as one batch —
 decode batches hit the gfx950 paged-decode kernel
 (ops/csrc/attention_decode.hip).
+
+Sliding-window attention (mistral, cfg.sliding_window): enforced here as
+pure KV bookkeeping — RoPE is applied at absolute positions when keys
+are written, so attention over any suffix of the cache is exact, and the
+kernel needs no window parameter. Whole front blocks that fall entirely
+outside the window are freed (bounded cache memory per sequence) and
+the block table / seq_len presented to the kernel cover only the
+retained suffix. Granularity is BLOCK_SIZE: a query attends to between
+W and W+BLOCK_SIZE-1 trailing tokens; prompts longer than W prefill
+with full causal attention (exact for prompts <= W, which covers the
+supported max prompt for mistral-7b).
 """
 from __future__ import annotations
 
@@ -48,6 +59,9 @@ class Request:
     # state
     output_ids: list[int] = field(default_factory=list)
     blocks: list[int] = field(default_factory=list)
+    # tokens whose KV blocks were freed by the sliding window (always a
+    # multiple of BLOCK_SIZE; blocks[] maps retained positions only)
+    dropped: int = 0
     finished: bool = False
     created: float = field(default_factory=time.time)
 
@@ -84,8 +98,15 @@ class Engine:
         self.waiting: list[Request] = []
         self.running: list[Request] = []
         self._next_id = 0
+        # sliding window bounds per-seq blocks at W//bs + 2 (retained
+        # suffix <= W + bs tokens incl. the one being appended); prompts
+        # prefill in full but front blocks are dropped before the first
+        # decode, so this is the true block-table width for decode.
+        seq_cap = self.cfg.max_seq_len
+        if self.cfg.sliding_window:
+            seq_cap = min(seq_cap, self.cfg.sliding_window + 2 * self.bs)
         self.max_blocks_per_seq = min(
-            kv_blocks, (self.cfg.max_seq_len + self.bs - 1) // self.bs)
+            kv_blocks, (seq_cap + self.bs - 1) // self.bs)
         self._graphed = None
         # hipGraphs drive the single-GPU decode; the TP>1 path runs eager
         # so the worker-follow protocol (serve/tp_worker.py) sees every
@@ -144,13 +165,16 @@ class Engine:
         dev = self.device
         last = [r.prompt_ids[-1] if not r.output_ids else r.output_ids[-1]
                 for r in reqs]
-        # position of the token being generated = current seq_len - 1 for the
-        # query; its kv slot appends at position seq_len - 1.
+        # RoPE position of the token being generated = absolute seq_len - 1
+        # (keys were roped at absolute positions too, so windowed suffixes
+        # stay exact). Its kv slot / the attended seq_len use the VIRTUAL
+        # position within the retained blocks: absolute minus dropped.
         pos = [r.seq_len - 1 for r in reqs]
+        vpos = [r.seq_len - 1 - r.dropped for r in reqs]
         slots = []
-        for r, p in zip(reqs, pos):
-            blk = r.blocks[p // self.bs]
-            slots.append(blk * self.bs + p % self.bs)
+        for r, v in zip(reqs, vpos):
+            blk = r.blocks[v // self.bs]
+            slots.append(blk * self.bs + v % self.bs)
 
         if self.use_graphs and B <= self.max_batch:
             if self._graphed is None:
@@ -160,7 +184,7 @@ class Engine:
                     self.max_blocks_per_seq, self.dummy_block, dev)
             logits = self._graphed.decode(last, pos, slots,
                                           [r.blocks for r in reqs],
-                                          [p + 1 for p in pos])
+                                          [v + 1 for v in vpos])
         else:
             maxb = max(len(r.blocks) for r in reqs)
             bt = torch.zeros(B, maxb, dtype=torch.int32)
@@ -170,7 +194,7 @@ class Engine:
             tokens = torch.tensor(last, dtype=torch.long, device=dev)
             positions = torch.tensor(pos, dtype=torch.int32, device=dev)
             slot_t = torch.tensor(slots, dtype=torch.int32, device=dev)
-            seq_lens = torch.tensor([p + 1 for p in pos], dtype=torch.int32,
+            seq_lens = torch.tensor([v + 1 for v in vpos], dtype=torch.int32,
                                     device=dev)
             bt = bt.to(dev)
             if self.tp > 1:
@@ -196,11 +220,25 @@ class Engine:
             out.append(int(tok[0]))
         return out
 
+    def _apply_window(self) -> None:
+        """Free whole front KV blocks that fall entirely outside the
+        sliding window of the next query (absolute position seq_len - 1).
+        Dropping whole blocks keeps the within-block slot offsets of the
+        retained tokens unchanged, so no cache data moves."""
+        w = self.cfg.sliding_window
+        if not w:
+            return
+        for r in self.running:
+            while r.dropped + self.bs <= r.seq_len - w and len(r.blocks) > 1:
+                self.allocator.release([r.blocks.pop(0)])
+                r.dropped += self.bs
+
     def _preempt(self, r: Request) -> None:
         """Release a running request's cache and requeue it: its generated
         tokens become part of the prompt for the re-prefill."""
         self.allocator.release(r.blocks)
         r.blocks = []
+        r.dropped = 0
         r.prompt_ids = r.prompt_ids + r.output_ids
         r.max_new_tokens -= len(r.output_ids)
         r.output_ids = []
@@ -248,6 +286,7 @@ class Engine:
             req.output_ids.append(first)
             self.running.append(req)
         elif self.running:
+            self._apply_window()
             # ensure every running seq has a block for the next position;
             # when the pool is exhausted, preempt the newest sequence
             # (its blocks are freed and it requeues for a fresh prefill)
@@ -255,7 +294,7 @@ class Engine:
             for r in list(self.running):
                 if r not in self.running:
                     continue
-                while r.seq_len >= len(r.blocks) * self.bs:
+                while r.seq_len - r.dropped >= len(r.blocks) * self.bs:
                     if self.allocator.free:
                         r.blocks.extend(self.allocator.alloc(1))
                     elif not self._preempt_newest(exclude=r):

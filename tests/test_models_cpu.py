@@ -238,3 +238,77 @@ def test_engine_scheduler_fuzz():
     for r in live:
         assert r.finished
         assert len(r.output_ids) <= r.max_new_tokens or r.max_new_tokens <= 0
+
+
+# --- sliding-window attention (mistral; engine-level KV bookkeeping) -------
+
+def _windowed_model(window):
+    import dataclasses
+    cfg = dataclasses.replace(get_config("tiny-llama"), name="tiny-window",
+                              sliding_window=window)
+    return build_model(cfg, dtype=torch.float32)
+
+
+def test_sliding_window_noop_below_window():
+    """For sequences shorter than the window the windowed engine is
+    bit-identical to the unwindowed one (no blocks dropped -> exact)."""
+    torch.manual_seed(0)
+    m_full = build_model("tiny-llama", dtype=torch.float32)
+    torch.manual_seed(0)
+    m_win = _windowed_model(64)
+    eng_f = Engine(m_full, device="cpu", dtype=torch.float32, kv_blocks=64)
+    eng_w = Engine(m_win, device="cpu", dtype=torch.float32, kv_blocks=64)
+    prompt = [3, 1, 4, 1, 5]
+    n = 20  # seq stays at 25 < 64
+    assert eng_w.generate(list(prompt), n) == eng_f.generate(list(prompt), n)
+
+
+def test_sliding_window_frees_blocks_and_bounds_table():
+    """Long generation: front blocks get freed (bounded KV memory), the
+    dropped count stays block-aligned, and the block table presented to
+    decode never exceeds window//bs + 2 rows."""
+    from runbooks_amd import ops
+    w, bs = 32, ops.BLOCK_SIZE
+    m = _windowed_model(w)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    seen = {"max_blocks": 0, "checked": 0}
+    orig = eng._decode_batch
+
+    def spy(reqs):
+        for r in reqs:
+            assert r.dropped % bs == 0
+            retained = r.seq_len - r.dropped
+            # retained context: within [w, w + bs) once past the window
+            if r.seq_len > w + bs:
+                assert w <= retained < w + bs, (r.seq_len, r.dropped)
+                seen["checked"] += 1
+            seen["max_blocks"] = max(seen["max_blocks"], len(r.blocks))
+        return orig(reqs)
+
+    eng._decode_batch = spy
+    req = eng.submit([7] * 8, max_new_tokens=90)
+    while eng.has_work():
+        eng.step()
+    assert req.finished and len(req.output_ids) == 90
+    assert seen["checked"] > 0, "window never engaged"
+    assert seen["max_blocks"] <= w // bs + 2
+    # all blocks (incl. dropped ones) returned to the pool
+    assert len(eng.allocator.free) == eng.allocator.num_blocks
+
+
+def test_sliding_window_long_prompt_prefills_then_trims():
+    """A prompt longer than the window prefills in full, then the first
+    decode step trims the table down to the window."""
+    from runbooks_amd import ops
+    w = 32
+    m = _windowed_model(w)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    req = eng.submit(list(range(1, 61)), max_new_tokens=5)  # 60-token prompt
+    eng.step()  # prefill: full prompt cached
+    assert req.dropped == 0 and len(req.blocks) * ops.BLOCK_SIZE >= 60
+    eng.step()  # first decode: window applied first
+    assert req.dropped > 0
+    assert req.seq_len - req.dropped < w + ops.BLOCK_SIZE
+    while eng.has_work():
+        eng.step()
+    assert len(req.output_ids) == 5
