@@ -279,3 +279,31 @@ def test_cli_suspend_resume(monkeypatch):
     assert r.exit_code == 0
     raw = kube.get("substratus.ai/v1", "Notebook", "default", "nb1")
     assert raw["spec"]["suspend"] is False
+
+
+def test_server_main_builds_engine(tmp_path, monkeypatch):
+    """The server image main on CPU: config marker -> engine + tokenizer,
+    handed to serve_forever (patched out; binding :8080 is the only part
+    skipped). Covers the contract env parsing incl. MODEL_LOAD_IN_8BIT."""
+    import json as _json
+    from runbooks_amd.workloads import server_main
+    model_dir = tmp_path / "model"
+    model_dir.mkdir()
+    (model_dir / "config.json").write_text(
+        _json.dumps({"runbooks_amd_config": "tiny-llama"}))
+    monkeypatch.setenv("MODEL_DIR", str(model_dir))
+    monkeypatch.setenv("MODEL_LOAD_IN_8BIT", "true")  # bf16 fallback on CPU
+    monkeypatch.delenv("TP", raising=False)
+
+    got = {}
+
+    def fake_serve(engine, tok, port, model_name):
+        got.update(engine=engine, tok=tok, port=port, model=model_name)
+
+    import runbooks_amd.serve.http as http_mod
+    monkeypatch.setattr(http_mod, "serve_forever", fake_serve)
+    assert server_main.main() == 0
+    assert got["model"] == "tiny-llama" and got["port"] == 8080
+    assert got["engine"].cfg.name == "tiny-llama"
+    out = got["engine"].generate([1, 2, 3], max_new_tokens=2)
+    assert len(out) == 2
