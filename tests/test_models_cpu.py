@@ -312,3 +312,51 @@ def test_sliding_window_long_prompt_prefills_then_trims():
     while eng.has_work():
         eng.step()
     assert len(req.output_ids) == 5
+
+
+def test_sliding_window_matches_masked_forward():
+    """Exactness of the window semantics: greedy windowed decode must
+    equal a from-scratch forward under the equivalent banded attention
+    mask. With a rolling KV cache, each token's K/V are computed ONCE
+    under its own generation-time window (so deeper layers see a growing
+    receptive field -- mistral's actual semantics; a fresh prefill of the
+    retained suffix is NOT equivalent). The mask reference reproduces
+    that exactly: prompt rows are fully causal (prefill), generated row
+    i attends [block_drop(i), i] per the engine's block-aligned rule."""
+    import math
+    from runbooks_amd import ops
+    w, bs = 32, ops.BLOCK_SIZE
+    m = _windowed_model(w)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    prompt = [5, 9, 2, 7, 1, 8, 3, 6]
+    req = eng.submit(list(prompt), max_new_tokens=56, temperature=0.0)
+    while eng.has_work():
+        eng.step()
+    all_ids = req.prompt_ids + req.output_ids
+    L, P = len(all_ids), len(prompt)
+
+    mask = torch.zeros(L, L, dtype=torch.bool)
+    for i in range(L):
+        start = 0 if i < P else max(0, bs * ((i + 1 - w) // bs))
+        mask[i, start:i + 1] = True
+
+    def masked_attn(q, k, v, scale=None, q_block=256):
+        B, S, Hq, Dh = q.shape
+        g = Hq // k.shape[2]
+        if g > 1:
+            k, v = (t.repeat_interleave(g, dim=2) for t in (k, v))
+        s = torch.einsum("bihd,bjhd->bhij", q.float(), k.float())
+        s = (s * (scale or 1 / math.sqrt(Dh))).masked_fill(
+            ~mask[:S, :S], float("-inf"))
+        return torch.einsum("bhij,bjhd->bihd", s.softmax(-1),
+                            v.float()).to(q.dtype)
+
+    orig = ops.causal_attention
+    ops.causal_attention = masked_attn
+    try:
+        with torch.no_grad():
+            logits = m(torch.tensor([all_ids]))
+    finally:
+        ops.causal_attention = orig
+    ref = logits[0, P - 1:L - 1].argmax(-1).tolist()
+    assert req.output_ids == ref, (req.output_ids, ref)
