@@ -49,6 +49,8 @@ class CompletionRequest(BaseModel):
     max_tokens: int = 16
     temperature: float = 0.0
     top_p: float = 1.0
+    n: int = 1              # choices per prompt (non-stream)
+    echo: bool = False      # prepend the prompt to each choice
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -201,8 +203,15 @@ def build_app(engine: Engine, tokenizer=None,
         cid = f"cmpl-{uuid.uuid4().hex[:12]}"
         t0 = int(time.time())
         stops = req.stop_list()
-        q, rid = loop.submit(ids, req.max_tokens, req.temperature,
-                             top_p=req.top_p)
+        if req.stream and req.n > 1:
+            return JSONResponse(status_code=400, content={"error": {
+                "message": "n > 1 is not supported with stream=true"}})
+        # all n choices submitted up front so they decode as one batch;
+        # per-request sampling noise is row-independent (ops/sampling.py)
+        n = max(1, min(req.n, 16))
+        subs = [loop.submit(list(ids), req.max_tokens, req.temperature,
+                            top_p=req.top_p) for _ in range(n)]
+        q, rid = subs[0]
 
         def _stop_hit(text: str):
             for s in stops:
@@ -216,6 +225,14 @@ def build_app(engine: Engine, tokenizer=None,
                 out = []
                 sent = ""
                 finish = "length"
+                if req.echo and req.prompt:
+                    chunk = {"id": cid, "object": "text_completion",
+                             "created": t0,
+                             "model": req.model or model_name,
+                             "choices": [{"text": req.prompt, "index": 0,
+                                          "logprobs": None,
+                                          "finish_reason": None}]}
+                    yield f"data: {json.dumps(chunk)}\n\n"
                 while True:
                     t = q.get()
                     if t is None:
@@ -244,36 +261,46 @@ def build_app(engine: Engine, tokenizer=None,
                 yield "data: [DONE]\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
-        out = []
-        finish = "length"
-        text = ""
-        while True:
-            t = q.get()
-            if t is None:
-                break
-            out.append(t)
-            if stops:
-                trunc = _stop_hit(tok.decode(out))
-                if trunc is not None:
-                    loop.cancel(rid)
-                    text, finish = trunc, "stop"
-                    while q.get() is not None:
-                        pass
+        def _collect(q, rid):
+            out = []
+            finish = "length"
+            text = ""
+            while True:
+                t = q.get()
+                if t is None:
                     break
-        if finish != "stop":
-            text = tok.decode(out)
+                out.append(t)
+                if stops:
+                    trunc = _stop_hit(tok.decode(out))
+                    if trunc is not None:
+                        loop.cancel(rid)
+                        text, finish = trunc, "stop"
+                        while q.get() is not None:
+                            pass
+                        break
+            if finish != "stop":
+                text = tok.decode(out)
+            return text, finish, len(out)
+
+        choices, n_out = [], 0
+        for i, (cq, crid) in enumerate(subs):
+            text, finish, produced = _collect(cq, crid)
+            if req.echo:
+                text = req.prompt + text
+            choices.append({"text": text, "index": i,
+                            "logprobs": None, "finish_reason": finish})
+            n_out += produced
         if m_reqs is not None:
             m_reqs.inc()
-            m_tokens.inc(len(out))
+            m_tokens.inc(n_out)
             m_lat.observe(time.time() - t0)
         return JSONResponse({
             "id": cid,
             "object": "text_completion",
             "created": t0,
             "model": req.model or model_name,
-            "choices": [{"text": text, "index": 0,
-                         "logprobs": None, "finish_reason": finish}],
-            "usage": _usage(len(ids), len(out)),
+            "choices": choices,
+            "usage": _usage(len(ids), n_out),
         })
 
     return app

@@ -98,3 +98,35 @@ def test_build_app_twice_no_metric_collision():
     a2 = build_app(eng, model_name="m2")
     a1.state.engine_loop.shutdown()
     a2.state.engine_loop.shutdown()
+
+
+def test_n_choices_and_echo(client):
+    """OpenAI-parity knobs the reference's basaran server exposes: n
+    parallel choices (decoded as one batch) and echo."""
+    r = client.post("/v1/completions",
+                    json={"prompt": "ab", "max_tokens": 4, "n": 3,
+                          "temperature": 0.9, "echo": True})
+    assert r.status_code == 200
+    body = r.json()
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+    assert all(c["text"].startswith("ab") for c in body["choices"])
+    assert body["usage"]["completion_tokens"] == 12
+    # sampled choices are row-independent: not all three identical
+    texts = {c["text"] for c in body["choices"]}
+    assert len(texts) >= 2, texts
+
+
+def test_stream_rejects_multi_choice(client):
+    r = client.post("/v1/completions",
+                    json={"prompt": "x", "n": 2, "stream": True})
+    assert r.status_code == 400
+
+
+def test_stream_echo_first_chunk(client):
+    with client.stream("POST", "/v1/completions",
+                       json={"prompt": "hi", "max_tokens": 2,
+                             "echo": True, "stream": True}) as r:
+        lines = [ln for ln in r.iter_lines() if ln.startswith("data: ")]
+    first = json.loads(lines[0][len("data: "):])
+    assert first["choices"][0]["text"] == "hi"
+    assert lines[-1] == "data: [DONE]"
