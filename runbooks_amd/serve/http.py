@@ -61,6 +61,22 @@ class CompletionRequest(BaseModel):
         return [self.stop] if isinstance(self.stop, str) else list(self.stop)
 
 
+class ChatMessage(BaseModel):
+    role: str
+    content: str
+
+
+class ChatRequest(BaseModel):
+    messages: list[ChatMessage]
+    max_tokens: int = 16
+    temperature: float = 0.0
+    top_p: float = 1.0
+    n: int = 1
+    stop: list[str] | str | None = None
+    model: str = ""
+    stream: bool = False
+
+
 class EngineLoop:
     """Background thread driving the continuous-batching engine."""
 
@@ -189,17 +205,70 @@ def build_app(engine: Engine, tokenizer=None,
         return {"prompt_tokens": n_prompt, "completion_tokens": n_out,
                 "total_tokens": n_prompt + n_out}
 
+    def _stop_hit(text: str, stops):
+        for s in stops:
+            i = text.find(s)
+            if i >= 0:
+                return text[:i]
+        return None
+
+    def _encode(prompt: str, max_tokens: int) -> list[int]:
+        ids = tok.encode(prompt)[-engine.cfg.max_seq_len + max_tokens + 1:]
+        # guard: a fallback tokenizer may emit ids past a small model's
+        # vocab (identity for any properly paired tokenizer)
+        return [i % engine.cfg.vocab_size for i in ids]
+
+    def _pieces(q, rid, stops):
+        """Incremental decoded text pieces for one request; drains the
+        queue after a stop-sequence hit so cancel cleanup stays local."""
+        out, sent = [], ""
+        while True:
+            t = q.get()
+            if t is None:
+                return
+            out.append(t)
+            text = tok.decode(out)
+            trunc = _stop_hit(text, stops) if stops else None
+            if trunc is not None:
+                loop.cancel(rid)
+                piece = trunc[len(sent):]
+                if piece:
+                    yield piece
+                while q.get() is not None:
+                    pass
+                return
+            piece = text[len(sent):]
+            sent += piece
+            if piece:
+                yield piece
+
+    def _collect(q, rid, stops):
+        out = []
+        finish = "length"
+        text = ""
+        while True:
+            t = q.get()
+            if t is None:
+                break
+            out.append(t)
+            if stops:
+                trunc = _stop_hit(tok.decode(out), stops)
+                if trunc is not None:
+                    loop.cancel(rid)
+                    text, finish = trunc, "stop"
+                    while q.get() is not None:
+                        pass
+                    break
+        if finish != "stop":
+            text = tok.decode(out)
+        return text, finish, len(out)
+
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
         # clamp generation to what the context window can hold
         req.max_tokens = max(1, min(req.max_tokens,
                                     engine.cfg.max_seq_len - 1))
-        ids = tok.encode(req.prompt)[-engine.cfg.max_seq_len +
-                                     req.max_tokens + 1:]
-        # guard: a fallback tokenizer may emit ids past a small model's
-        # vocab (identity for any properly paired tokenizer)
-        vocab = engine.cfg.vocab_size
-        ids = [i % vocab for i in ids]
+        ids = _encode(req.prompt, req.max_tokens)
         cid = f"cmpl-{uuid.uuid4().hex[:12]}"
         t0 = int(time.time())
         stops = req.stop_list()
@@ -213,78 +282,25 @@ def build_app(engine: Engine, tokenizer=None,
                             top_p=req.top_p) for _ in range(n)]
         q, rid = subs[0]
 
-        def _stop_hit(text: str):
-            for s in stops:
-                i = text.find(s)
-                if i >= 0:
-                    return text[:i]
-            return None
-
         if req.stream:
             def gen():
-                out = []
-                sent = ""
-                finish = "length"
+                def chunk(text):
+                    c = {"id": cid, "object": "text_completion",
+                         "created": t0, "model": req.model or model_name,
+                         "choices": [{"text": text, "index": 0,
+                                      "logprobs": None,
+                                      "finish_reason": None}]}
+                    return f"data: {json.dumps(c)}\n\n"
                 if req.echo and req.prompt:
-                    chunk = {"id": cid, "object": "text_completion",
-                             "created": t0,
-                             "model": req.model or model_name,
-                             "choices": [{"text": req.prompt, "index": 0,
-                                          "logprobs": None,
-                                          "finish_reason": None}]}
-                    yield f"data: {json.dumps(chunk)}\n\n"
-                while True:
-                    t = q.get()
-                    if t is None:
-                        break
-                    out.append(t)
-                    text = tok.decode(out)
-                    trunc = _stop_hit(text)
-                    if trunc is not None:
-                        loop.cancel(rid)
-                        piece, finish = trunc[len(sent):], "stop"
-                    else:
-                        piece = text[len(sent):]
-                    sent += piece
-                    if piece:
-                        chunk = {"id": cid, "object": "text_completion",
-                                 "created": t0,
-                                 "model": req.model or model_name,
-                                 "choices": [{"text": piece, "index": 0,
-                                              "logprobs": None,
-                                              "finish_reason": None}]}
-                        yield f"data: {json.dumps(chunk)}\n\n"
-                    if finish == "stop":
-                        while q.get() is not None:
-                            pass
-                        break
+                    yield chunk(req.prompt)
+                for piece in _pieces(q, rid, stops):
+                    yield chunk(piece)
                 yield "data: [DONE]\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
-        def _collect(q, rid):
-            out = []
-            finish = "length"
-            text = ""
-            while True:
-                t = q.get()
-                if t is None:
-                    break
-                out.append(t)
-                if stops:
-                    trunc = _stop_hit(tok.decode(out))
-                    if trunc is not None:
-                        loop.cancel(rid)
-                        text, finish = trunc, "stop"
-                        while q.get() is not None:
-                            pass
-                        break
-            if finish != "stop":
-                text = tok.decode(out)
-            return text, finish, len(out)
-
         choices, n_out = [], 0
         for i, (cq, crid) in enumerate(subs):
-            text, finish, produced = _collect(cq, crid)
+            text, finish, produced = _collect(cq, crid, stops)
             if req.echo:
                 text = req.prompt + text
             choices.append({"text": text, "index": i,
@@ -301,6 +317,64 @@ def build_app(engine: Engine, tokenizer=None,
             "model": req.model or model_name,
             "choices": choices,
             "usage": _usage(len(ids), n_out),
+        })
+
+    @app.post("/v1/chat/completions")
+    def chat(req: ChatRequest):
+        """Chat surface over the same engine: messages are flattened with
+        role prefixes (no chat-template metadata in tokenizer.json /
+        tokenizer.model; models loaded from /content/model use whatever
+        plain-text convention they were tuned on). Reuses the tested
+        completions path and reshapes the payload."""
+        prompt = "\n".join(f"{m.role}: {m.content}" for m in req.messages)
+        prompt += "\nassistant:"
+        cid = f"chatcmpl-{uuid.uuid4().hex[:12]}"
+        t0 = int(time.time())
+        mdl = req.model or model_name
+
+        if req.stream:
+            if req.n > 1:
+                return JSONResponse(status_code=400, content={"error": {
+                    "message": "n > 1 is not supported with stream=true"}})
+            max_tokens = max(1, min(req.max_tokens,
+                                    engine.cfg.max_seq_len - 1))
+            ids = _encode(prompt, max_tokens)
+            stops = CompletionRequest(stop=req.stop).stop_list()
+            q, rid = loop.submit(ids, max_tokens, req.temperature,
+                                 top_p=req.top_p)
+
+            def gen():
+                def chunk(delta):
+                    c = {"id": cid, "object": "chat.completion.chunk",
+                         "created": t0, "model": mdl,
+                         "choices": [{"index": 0, "delta": delta,
+                                      "finish_reason": None}]}
+                    return f"data: {json.dumps(c)}\n\n"
+                yield chunk({"role": "assistant"})
+                for piece in _pieces(q, rid, stops):
+                    yield chunk({"content": piece})
+                yield "data: [DONE]\n\n"
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        creq = CompletionRequest(
+            prompt=prompt, max_tokens=req.max_tokens,
+            temperature=req.temperature, top_p=req.top_p, n=req.n,
+            stop=req.stop, model=req.model, stream=False)
+        resp = completions(creq)
+        if resp.status_code != 200:
+            return resp
+        body = json.loads(bytes(resp.body))
+        return JSONResponse({
+            "id": cid,
+            "object": "chat.completion",
+            "created": t0,
+            "model": mdl,
+            "choices": [{"index": c["index"],
+                         "message": {"role": "assistant",
+                                     "content": c["text"]},
+                         "finish_reason": c["finish_reason"]}
+                        for c in body["choices"]],
+            "usage": body["usage"],
         })
 
     return app

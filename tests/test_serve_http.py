@@ -130,3 +130,31 @@ def test_stream_echo_first_chunk(client):
     first = json.loads(lines[0][len("data: "):])
     assert first["choices"][0]["text"] == "hi"
     assert lines[-1] == "data: [DONE]"
+
+
+def test_chat_completions(client):
+    r = client.post("/v1/chat/completions",
+                    json={"messages": [{"role": "system", "content": "be brief"},
+                                       {"role": "user", "content": "hi"}],
+                          "max_tokens": 4, "n": 2, "temperature": 0.8})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert len(body["choices"]) == 2
+    for c in body["choices"]:
+        assert c["message"]["role"] == "assistant"
+        assert isinstance(c["message"]["content"], str)
+    assert body["usage"]["completion_tokens"] == 8
+
+
+def test_chat_streaming_deltas(client):
+    with client.stream("POST", "/v1/chat/completions",
+                       json={"messages": [{"role": "user", "content": "go"}],
+                             "max_tokens": 3, "stream": True}) as r:
+        lines = [ln for ln in r.iter_lines() if ln.startswith("data: ")]
+    first = json.loads(lines[0][len("data: "):])
+    assert first["choices"][0]["delta"] == {"role": "assistant"}
+    assert lines[-1] == "data: [DONE]"
+    mids = [json.loads(ln[len("data: "):]) for ln in lines[1:-1]]
+    assert all(m["object"] == "chat.completion.chunk" for m in mids)
+    assert all("content" in m["choices"][0]["delta"] for m in mids)
