@@ -26,18 +26,49 @@ class SyntheticTokens(torch.utils.data.Dataset):
         return torch.randint(0, self.vocab, (self.seq,), generator=g)
 
 
-class JsonlTextDataset(torch.utils.data.Dataset):
-    """Reads {"text": ...} or {"prompt","completion"} jsonl, tokenizes with a
-    byte-level fallback tokenizer when none is supplied."""
+TEXT_SUFFIXES = (".jsonl", ".parquet", ".csv", ".txt")
 
-    def __init__(self, path: str | Path, seq_len: int, tokenizer=None,
-                 vocab_size: int | None = None):
-        self.rows = []
+
+def _load_rows(path: Path) -> list[dict]:
+    suf = path.suffix.lower()
+    if suf == ".jsonl":
+        rows = []
         with open(path) as f:
             for line in f:
                 line = line.strip()
                 if line:
-                    self.rows.append(json.loads(line))
+                    rows.append(json.loads(line))
+        return rows
+    if suf == ".parquet":
+        import pyarrow.parquet as pq
+        return pq.read_table(path).to_pylist()
+    if suf == ".csv":
+        import csv
+        with open(path, newline="") as f:
+            return list(csv.DictReader(f))
+    if suf == ".txt":
+        with open(path) as f:
+            return [{"text": ln.rstrip("\n")} for ln in f if ln.strip()]
+    raise ValueError(f"unsupported dataset file {path} "
+                     f"(supported: {TEXT_SUFFIXES})")
+
+
+class TextDataset(torch.utils.data.Dataset):
+    """Instruction/text rows from jsonl / parquet / csv / txt files —
+    the formats dataset-loader jobs commonly produce ({"text": ...} or
+    {"prompt","completion"} rows). Accepts one file or a directory (all
+    supported files, sorted). Tokenizes with a byte-level fallback when
+    no tokenizer is supplied."""
+
+    def __init__(self, path: str | Path, seq_len: int, tokenizer=None,
+                 vocab_size: int | None = None):
+        p = Path(path)
+        files = (sorted(f for f in p.iterdir()
+                        if f.suffix.lower() in TEXT_SUFFIXES)
+                 if p.is_dir() else [p])
+        if not files:
+            raise FileNotFoundError(f"no dataset files under {p}")
+        self.rows = [row for f in files for row in _load_rows(f)]
         self.seq = seq_len
         self.tok = tokenizer
         # clamp ids into the model's vocab (a byte-fallback tokenizer can
@@ -60,6 +91,10 @@ class JsonlTextDataset(torch.utils.data.Dataset):
             ids = [i % self.vocab for i in ids]
         ids = ids + [0] * (self.seq - len(ids))
         return torch.tensor(ids, dtype=torch.long)
+
+
+# legacy name (the class now reads all supported formats)
+JsonlTextDataset = TextDataset
 
 
 def data_loader(dataset, batch_size: int, rank: int = 0, world: int = 1,

@@ -56,7 +56,7 @@ def main():
     from ..parallel import comm
     from ..serve.tokenizer import load_tokenizer
     from ..train import TrainConfig, Trainer
-    from ..train.data import JsonlTextDataset, SyntheticTokens
+    from ..train.data import TEXT_SUFFIXES, SyntheticTokens, TextDataset
 
     comm.init_from_env()
     model_dir = Path(os.environ.get("MODEL_DIR", "/content/model"))
@@ -94,11 +94,13 @@ def main():
         load_pretrained(trainer.model, model_dir, rank=0, tp=1, strict=False)
         print(f"trainer: loaded base weights from {model_dir}")
 
-    jsonl = sorted(data_dir.glob("*.jsonl")) if data_dir.exists() else []
-    if jsonl:
+    files = (sorted(p for p in data_dir.iterdir()
+                    if p.suffix.lower() in TEXT_SUFFIXES)
+             if data_dir.exists() else [])
+    if files:
         tok = load_tokenizer(model_dir if model_dir.exists() else None)
-        dataset = JsonlTextDataset(jsonl[0], cfg.seq_len + 1, tokenizer=tok,
-                                   vocab_size=trainer.model.cfg.vocab_size)
+        dataset = TextDataset(data_dir, cfg.seq_len + 1, tokenizer=tok,
+                              vocab_size=trainer.model.cfg.vocab_size)
         epochs = _param("num_train_epochs", None, float)
         if epochs is not None:
             steps = int(epochs * len(dataset)

@@ -169,3 +169,26 @@ def test_full_finetune_step():
     before = t.model.embed.weight.detach().clone()
     t.train_step(torch.randint(0, 256, (2, 17)))
     assert not torch.equal(before, t.model.embed.weight.detach())
+
+
+def test_text_dataset_formats(tmp_path):
+    """TextDataset reads jsonl / parquet / csv / txt (the formats
+    dataset-loader jobs produce) and concatenates a directory."""
+    import json as _json
+    from runbooks_amd.train import TextDataset
+
+    (tmp_path / "a.jsonl").write_text(
+        _json.dumps({"text": "alpha"}) + "\n" +
+        _json.dumps({"prompt": "b", "completion": "eta"}) + "\n")
+    (tmp_path / "b.txt").write_text("gamma\ndelta\n")
+    (tmp_path / "c.csv").write_text("text\nepsilon\n")
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    pq.write_table(pa.table({"text": ["zeta", "eta"]}), tmp_path / "d.parquet")
+
+    ds = TextDataset(tmp_path, seq_len=16, vocab_size=256)
+    assert len(ds) == 7
+    row = ds[0]
+    assert row.shape == (16,) and row.dtype == torch.long
+    single = TextDataset(tmp_path / "d.parquet", seq_len=8, vocab_size=256)
+    assert len(single) == 2
