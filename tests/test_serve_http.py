@@ -158,3 +158,53 @@ def test_chat_streaming_deltas(client):
     mids = [json.loads(ln[len("data: "):]) for ln in lines[1:-1]]
     assert all(m["object"] == "chat.completion.chunk" for m in mids)
     assert all("content" in m["choices"][0]["delta"] for m in mids)
+
+
+def test_concurrent_mixed_requests_stress(client):
+    """Thread-stress the EngineLoop: mixed n/echo/stop/stream requests in
+    flight together must all complete coherently (the watchers dict and
+    allocator are shared across HTTP threads)."""
+    import concurrent.futures as cf
+
+    def completion(i):
+        r = client.post("/v1/completions",
+                        json={"prompt": f"p{i}", "max_tokens": 3 + i % 3,
+                              "n": 1 + i % 3, "temperature": 0.5,
+                              "echo": i % 2 == 0})
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["choices"]) == 1 + i % 3
+        return body["usage"]["completion_tokens"]
+
+    def chat(i):
+        r = client.post("/v1/chat/completions",
+                        json={"messages": [{"role": "user",
+                                            "content": f"m{i}"}],
+                              "max_tokens": 3})
+        assert r.status_code == 200
+        return r.json()["usage"]["completion_tokens"]
+
+    def stream(i):
+        with client.stream("POST", "/v1/completions",
+                           json={"prompt": f"s{i}", "max_tokens": 4,
+                                 "stream": True}) as r:
+            lines = [ln for ln in r.iter_lines()
+                     if ln.startswith("data: ")]
+        assert lines[-1] == "data: [DONE]"
+        return 1
+
+    with cf.ThreadPoolExecutor(max_workers=8) as ex:
+        futs = [ex.submit((completion, chat, stream)[i % 3], i)
+                for i in range(15)]
+        results = [f.result(timeout=60) for f in futs]
+    assert len(results) == 15
+    # engine fully drained, no leaked blocks or watchers
+    eng = client.app.state.engine_loop.engine
+    for _ in range(200):
+        if not eng.has_work():
+            break
+        import time as _t
+        _t.sleep(0.02)
+    assert not eng.has_work()
+    assert len(eng.allocator.free) == eng.allocator.num_blocks
+    assert not client.app.state.engine_loop._watchers
