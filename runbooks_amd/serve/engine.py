@@ -55,9 +55,12 @@ class Request:
     max_new_tokens: int = 64
     temperature: float = 0.0
     top_p: float = 1.0
+    logprobs: int | None = None     # record top-k logprobs per token
     stop_ids: tuple[int, ...] = ()
     # state
     output_ids: list[int] = field(default_factory=list)
+    # per output token, when logprobs is set: {"logprob": f, "top": [(id, f)]}
+    logprob_data: list = field(default_factory=list)
     blocks: list[int] = field(default_factory=list)
     # tokens whose KV blocks were freed by the sliding window (always a
     # multiple of BLOCK_SIZE; blocks[] maps retained positions only)
@@ -126,12 +129,26 @@ class Engine:
 
     # -- request API ------------------------------------------------------------
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 64,
-               temperature: float = 0.0, top_p: float = 1.0) -> Request:
+               temperature: float = 0.0, top_p: float = 1.0,
+               logprobs: int | None = None) -> Request:
         req = Request(self._next_id, list(prompt_ids), max_new_tokens,
-                      temperature, top_p)
+                      temperature, top_p, logprobs)
         self._next_id += 1
         self.waiting.append(req)
         return req
+
+    @staticmethod
+    def _record_logprobs(req: Request, logits_row, tok_id: int) -> None:
+        """Top-k logprob bookkeeping for one sampled token (host sync;
+        only runs for requests that asked — sampling already lives
+        outside the hipGraph, so the graphed decode path is untouched)."""
+        lp = torch.log_softmax(logits_row.float(), dim=-1)
+        entry = {"logprob": float(lp[tok_id])}
+        k = req.logprobs or 0
+        if k > 0:
+            v, idx = lp.topk(min(k, lp.numel()))
+            entry["top"] = list(zip(idx.tolist(), v.tolist()))
+        req.logprob_data.append(entry)
 
     def _admit(self) -> Request | None:
         if not self.waiting or len(self.running) >= self.max_batch:
@@ -158,7 +175,10 @@ class Engine:
         logits = self.model.prefill(tokens, positions, self.caches, slots)
         tok = ops.sample_tokens(logits, req.temperature, top_p=req.top_p,
                                 seed=self.seed + req.request_id * 65537 + S)
-        return int(tok[0])
+        t = int(tok[0])
+        if req.logprobs is not None:
+            self._record_logprobs(req, logits[0], t)
+        return t
 
     def _decode_batch(self, reqs: list[Request]) -> list[int]:
         B = len(reqs)
@@ -210,14 +230,18 @@ class Engine:
         if len(params) == 1:
             t, p = params.pop()
             toks = ops.sample_tokens(logits, t, top_p=p, seed=seed)
-            return [int(x) for x in toks]
-        # heterogeneous sampling params: sample row-by-row
-        out = []
+            out = [int(x) for x in toks]
+        else:
+            # heterogeneous sampling params: sample row-by-row
+            out = []
+            for i, r in enumerate(reqs):
+                tok = ops.sample_tokens(logits[i:i + 1], r.temperature,
+                                        top_p=r.top_p,
+                                        seed=seed + r.request_id)
+                out.append(int(tok[0]))
         for i, r in enumerate(reqs):
-            tok = ops.sample_tokens(logits[i:i + 1], r.temperature,
-                                    top_p=r.top_p,
-                                    seed=seed + r.request_id)
-            out.append(int(tok[0]))
+            if r.logprobs is not None:
+                self._record_logprobs(r, logits[i], out[i])
         return out
 
     def _apply_window(self) -> None:

@@ -51,6 +51,7 @@ class CompletionRequest(BaseModel):
     top_p: float = 1.0
     n: int = 1              # choices per prompt (non-stream)
     echo: bool = False      # prepend the prompt to each choice
+    logprobs: int | None = None  # top-k logprobs per generated token
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -91,16 +92,16 @@ class EngineLoop:
         self._thread.start()
 
     def submit(self, prompt_ids, max_new_tokens, temperature,
-               top_p=1.0) -> tuple[queue.Queue, int]:
-        """Returns (queue yielding token_id | None, request_id)."""
+               top_p=1.0, logprobs=None) -> tuple[queue.Queue, "object"]:
+        """Returns (queue yielding token_id | None, engine Request)."""
         q: queue.Queue = queue.Queue()
         with self._lock:
             req = self.engine.submit(prompt_ids, max_new_tokens, temperature,
-                                     top_p=top_p)
+                                     top_p=top_p, logprobs=logprobs)
             self._watchers[req.request_id] = q
             req._watch_sent = 0
         self._wake.set()
-        return q, req.request_id
+        return q, req
 
     def cancel(self, request_id: int) -> None:
         with self._lock:
@@ -263,6 +264,25 @@ def build_app(engine: Engine, tokenizer=None,
             text = tok.decode(out)
         return text, finish, len(out)
 
+    def _fmt_logprobs(ereq, produced: int):
+        """OpenAI-style logprobs block for one finished choice. `produced`
+        caps at the tokens actually delivered (stop-truncation)."""
+        ids = ereq.output_ids[:produced]
+        data = ereq.logprob_data[:produced]
+        toks = [tok.decode([t]) for t in ids]
+        offs, pos = [], 0
+        for t in toks:
+            offs.append(pos)
+            pos += len(t)
+        top = None
+        if data and "top" in data[0]:
+            top = [{tok.decode([i]): lp for i, lp in d["top"]}
+                   for d in data]
+        return {"tokens": toks,
+                "token_logprobs": [d["logprob"] for d in data],
+                "top_logprobs": top,
+                "text_offset": offs}
+
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
         # clamp generation to what the context window can hold
@@ -279,8 +299,9 @@ def build_app(engine: Engine, tokenizer=None,
         # per-request sampling noise is row-independent (ops/sampling.py)
         n = max(1, min(req.n, 16))
         subs = [loop.submit(list(ids), req.max_tokens, req.temperature,
-                            top_p=req.top_p) for _ in range(n)]
-        q, rid = subs[0]
+                            top_p=req.top_p, logprobs=req.logprobs)
+                for _ in range(n)]
+        q, ereq = subs[0]
 
         if req.stream:
             def gen():
@@ -293,18 +314,20 @@ def build_app(engine: Engine, tokenizer=None,
                     return f"data: {json.dumps(c)}\n\n"
                 if req.echo and req.prompt:
                     yield chunk(req.prompt)
-                for piece in _pieces(q, rid, stops):
+                for piece in _pieces(q, ereq.request_id, stops):
                     yield chunk(piece)
                 yield "data: [DONE]\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
         choices, n_out = [], 0
-        for i, (cq, crid) in enumerate(subs):
-            text, finish, produced = _collect(cq, crid, stops)
+        for i, (cq, creq_) in enumerate(subs):
+            text, finish, produced = _collect(cq, creq_.request_id, stops)
             if req.echo:
                 text = req.prompt + text
+            lp = (_fmt_logprobs(creq_, produced)
+                  if req.logprobs is not None else None)
             choices.append({"text": text, "index": i,
-                            "logprobs": None, "finish_reason": finish})
+                            "logprobs": lp, "finish_reason": finish})
             n_out += produced
         if m_reqs is not None:
             m_reqs.inc()
@@ -340,8 +363,8 @@ def build_app(engine: Engine, tokenizer=None,
                                     engine.cfg.max_seq_len - 1))
             ids = _encode(prompt, max_tokens)
             stops = CompletionRequest(stop=req.stop).stop_list()
-            q, rid = loop.submit(ids, max_tokens, req.temperature,
-                                 top_p=req.top_p)
+            q, ereq = loop.submit(ids, max_tokens, req.temperature,
+                                  top_p=req.top_p)
 
             def gen():
                 def chunk(delta):
@@ -351,7 +374,7 @@ def build_app(engine: Engine, tokenizer=None,
                                       "finish_reason": None}]}
                     return f"data: {json.dumps(c)}\n\n"
                 yield chunk({"role": "assistant"})
-                for piece in _pieces(q, rid, stops):
+                for piece in _pieces(q, ereq.request_id, stops):
                     yield chunk({"content": piece})
                 yield "data: [DONE]\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")

@@ -208,3 +208,26 @@ def test_concurrent_mixed_requests_stress(client):
     assert not eng.has_work()
     assert len(eng.allocator.free) == eng.allocator.num_blocks
     assert not client.app.state.engine_loop._watchers
+
+
+def test_logprobs(client):
+    """OpenAI logprobs surface: per-token logprob + top-k alternatives +
+    text offsets; greedy sampling means the chosen token is the top-1."""
+    import math
+    r = client.post("/v1/completions",
+                    json={"prompt": "lp", "max_tokens": 4, "logprobs": 2,
+                          "temperature": 0.0})
+    assert r.status_code == 200
+    lp = r.json()["choices"][0]["logprobs"]
+    assert lp is not None
+    assert len(lp["tokens"]) == len(lp["token_logprobs"]) == 4
+    assert len(lp["top_logprobs"]) == 4 and len(lp["text_offset"]) == 4
+    for tl, top in zip(lp["token_logprobs"], lp["top_logprobs"]):
+        # dict keys are decoded token strings; the byte fallback can
+        # collide distinct ids onto one replacement char, so 1 <= k <= 2
+        assert tl <= 0 and 1 <= len(top) <= 2
+        # greedy: the chosen token's logprob is the max of the top-k
+        assert tl >= max(top.values()) - 1e-5
+    # offsets are cumulative over the decoded pieces
+    assert lp["text_offset"][0] == 0
+    assert lp["text_offset"] == sorted(lp["text_offset"])
