@@ -172,6 +172,48 @@ def _tp_engine_serving(rank, world):
         worker_loop(eng)
 
 
+def _tp_engine_sliding_window(rank, world):
+    """TP=2 + sliding window (mistral deployment shape): the window is
+    rank-0 bookkeeping, workers follow broadcast tensors; a long
+    generation (window engaged, front blocks freed) must match the
+    TP=1 windowed engine bit-for-bit."""
+    import dataclasses
+
+    from runbooks_amd.models import build_model, get_config
+    from runbooks_amd.serve import Engine
+    from runbooks_amd.serve.tp_worker import broadcast_shutdown, worker_loop
+
+    cfg = dataclasses.replace(get_config("tiny-llama"), name="tiny-window",
+                              sliding_window=32)
+    torch.manual_seed(0)
+    tp_model = build_model(cfg, dtype=torch.float32, tp=world, seed=3)
+    single = build_model(cfg, dtype=torch.float32, tp=1, seed=3)
+    sd = single.state_dict()
+    for name, t in tp_model.state_dict().items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:
+            t.copy_(full[rank * t.shape[0]:(rank + 1) * t.shape[0]])
+        else:
+            t.copy_(full[:, rank * t.shape[1]:(rank + 1) * t.shape[1]])
+
+    eng = Engine(tp_model, device="cpu", kv_blocks=64, seed=11)
+    prompt = [3, 1, 4, 1, 5, 9, 2, 6]
+    if rank == 0:
+        req = eng.submit(list(prompt), max_new_tokens=60)
+        while eng.has_work():
+            eng.step()
+        broadcast_shutdown(torch.device("cpu"))
+        assert req.dropped > 0, "window never engaged"
+        ref = Engine(single, device="cpu", kv_blocks=64, seed=11)
+        ref.tp = 1
+        out_1 = ref.generate(list(prompt), max_new_tokens=60)
+        assert req.output_ids == out_1
+    else:
+        worker_loop(eng)
+
+
 # --- test entries -----------------------------------------------------------
 
 def test_dp_gradient_allreduce():
@@ -264,3 +306,7 @@ def test_dp_grad_accumulation():
 
 def test_tp_falcon_mqa():
     _run_dist(_tp_falcon_mqa_replicated_kv, port=PORT + 6)
+
+
+def test_tp_engine_sliding_window():
+    _run_dist(_tp_engine_sliding_window, port=PORT + 11)
