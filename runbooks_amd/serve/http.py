@@ -11,6 +11,7 @@ requests decode together in one batch.
 from __future__ import annotations
 
 import json
+import logging
 import queue
 import threading
 import time
@@ -22,6 +23,10 @@ from pydantic import BaseModel
 
 from .engine import Engine
 from .tokenizer import load_tokenizer
+
+# per-request latency logging (SURVEY.md §5 tracing: the reference has
+# none; the serving runtime logs one structured line per request)
+logger = logging.getLogger("runbooks_amd.serve")
 
 # Prometheus metrics are process-global: created once here so a second
 # build_app() in the same process (tests, uvicorn reload) cannot trip
@@ -333,6 +338,10 @@ def build_app(engine: Engine, tokenizer=None,
             m_reqs.inc()
             m_tokens.inc(n_out)
             m_lat.observe(time.time() - t0)
+        logger.info(
+            "completion id=%s prompt_tokens=%d completion_tokens=%d "
+            "choices=%d latency_ms=%.1f", cid, len(ids), n_out, len(choices),
+            (time.time() - t0) * 1000.0)
         return JSONResponse({
             "id": cid,
             "object": "text_completion",

@@ -231,3 +231,11 @@ def test_logprobs(client):
     # offsets are cumulative over the decoded pieces
     assert lp["text_offset"][0] == 0
     assert lp["text_offset"] == sorted(lp["text_offset"])
+
+
+def test_per_request_latency_log(client, caplog):
+    import logging as _logging
+    with caplog.at_level(_logging.INFO, logger="runbooks_amd.serve"):
+        client.post("/v1/completions", json={"prompt": "log", "max_tokens": 2})
+    recs = [r for r in caplog.records if "completion id=" in r.getMessage()]
+    assert recs and "latency_ms=" in recs[-1].getMessage()
