@@ -35,17 +35,33 @@ from ..parallel import comm
 
 
 class BlockAllocator:
+    """Free-list allocator with reference counts: prefix-cached blocks
+    are shared across requests (and held by the cache itself); a block
+    returns to the free list when its last reference is released."""
+
     def __init__(self, num_blocks: int):
         self.free = list(range(num_blocks - 1, -1, -1))
         self.num_blocks = num_blocks
+        self.refs = [0] * num_blocks
 
     def alloc(self, n: int) -> list[int]:
         if n > len(self.free):
             raise RuntimeError("KV cache exhausted")
-        return [self.free.pop() for _ in range(n)]
+        out = [self.free.pop() for _ in range(n)]
+        for b in out:
+            self.refs[b] = 1
+        return out
+
+    def share(self, block: int) -> None:
+        assert self.refs[block] > 0, "sharing a free block"
+        self.refs[block] += 1
 
     def release(self, blocks: list[int]):
-        self.free.extend(blocks)
+        for b in blocks:
+            self.refs[b] -= 1
+            assert self.refs[b] >= 0, "double release"
+            if self.refs[b] == 0:
+                self.free.append(b)
 
 
 @dataclass
@@ -77,7 +93,7 @@ class Engine:
     def __init__(self, model: Transformer | str, device=None,
                  dtype=torch.bfloat16, kv_blocks: int | None = None,
                  max_batch: int = 64, mem_fraction: float = 0.85, seed: int = 0,
-                 load_in_8bit: bool = False):
+                 load_in_8bit: bool = False, prefix_cache: bool | None = None):
         self.device = device if device is not None else (
             f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
         if isinstance(model, str):
@@ -116,6 +132,18 @@ class Engine:
         # collective.
         self.tp = comm.world_size()
         self.use_graphs = torch.cuda.is_available() and self.tp == 1
+        # Prefix caching (opt-in, RB_PREFIX_CACHE=1 or prefix_cache=True):
+        # full 16-token prompt-prefix chunks are shared across requests by
+        # refcount — a hit skips the KV writes for those chunks (the
+        # prefill still computes the full prompt, so attention inputs are
+        # identical; the kernels are untouched, only the block table
+        # changes). Bounds cache memory for common-system-prompt serving.
+        import os as _os
+        self.prefix_cache_enabled = (
+            prefix_cache if prefix_cache is not None
+            else _os.environ.get("RB_PREFIX_CACHE", "0") == "1")
+        from collections import OrderedDict
+        self._pc: "OrderedDict[tuple, int]" = OrderedDict()
 
     def _auto_kv_blocks(self, mem_fraction: float) -> int:
         bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
@@ -154,20 +182,73 @@ class Engine:
         if not self.waiting or len(self.running) >= self.max_batch:
             return None
         req = self.waiting[0]
-        need = (len(req.prompt_ids) + self.bs - 1) // self.bs + 1
+        S = len(req.prompt_ids)
+        hits: list[tuple[tuple, int]] = []
+        if self.prefix_cache_enabled:
+            for k in range(1, S // self.bs + 1):
+                key = tuple(req.prompt_ids[:k * self.bs])
+                b = self._pc.get(key)
+                if b is None:
+                    break
+                hits.append((key, b))
+        need = (S + self.bs - 1) // self.bs + 1 - len(hits)
+        if need > len(self.allocator.free) and self.prefix_cache_enabled:
+            self._evict_prefix(need - len(self.allocator.free),
+                               protect={b for _, b in hits})
         if need > len(self.allocator.free):
             return None
         self.waiting.pop(0)
-        req.blocks = self.allocator.alloc(need)
+        for key, b in hits:
+            self.allocator.share(b)
+            self._pc.move_to_end(key)
+        req.blocks = [b for _, b in hits] + self.allocator.alloc(need)
+        req._shared_chunks = len(hits)
         return req
+
+    def _evict_prefix(self, n_blocks: int, protect: set = frozenset()):
+        """Drop LRU prefix-cache entries until ~n_blocks come free (an
+        entry only frees its block if no live request still shares it)."""
+        for key in list(self._pc):
+            if n_blocks <= 0:
+                return
+            b = self._pc[key]
+            if b in protect:
+                continue
+            del self._pc[key]
+            before = len(self.allocator.free)
+            self.allocator.release([b])
+            n_blocks -= len(self.allocator.free) - before
+
+    def _pc_insert(self, req: Request) -> None:
+        """After a successful prefill, publish the request's full prompt
+        chunks (cache holds its own reference per block)."""
+        S = len(req.prompt_ids)
+        for k in range(1, S // self.bs + 1):
+            key = tuple(req.prompt_ids[:k * self.bs])
+            if key not in self._pc:
+                b = req.blocks[k - 1]
+                self._pc[key] = b
+                self.allocator.share(b)
+            self._pc.move_to_end(key)
+
+    def flush_prefix_cache(self) -> None:
+        for b in self._pc.values():
+            self.allocator.release([b])
+        self._pc.clear()
 
     # -- model invocations ------------------------------------------------------
     def _prefill(self, req: Request) -> int:
         S = len(req.prompt_ids)
         tokens = torch.tensor([req.prompt_ids], dtype=torch.long, device=self.device)
         positions = torch.arange(S, dtype=torch.int32, device=self.device)
+        # prefix-cache hits: those chunks' KV already sit in the shared
+        # blocks, so their writes are redirected to the dummy block
+        # (trash); everything else lands in the request's own blocks.
+        shared_tokens = getattr(req, "_shared_chunks", 0) * self.bs
         slots = torch.tensor(
-            [req.blocks[i // self.bs] * self.bs + i % self.bs for i in range(S)],
+            [(self.dummy_block * self.bs + i % self.bs) if i < shared_tokens
+             else req.blocks[i // self.bs] * self.bs + i % self.bs
+             for i in range(S)],
             dtype=torch.int32, device=self.device)
         if self.tp > 1:
             from .tp_worker import broadcast_prefill
@@ -298,7 +379,14 @@ class Engine:
         finished = []
         req = self._admit()
         if req is None and not self.running and self.waiting:
-            need = (len(self.waiting[0].prompt_ids) + self.bs - 1) // self.bs + 1
+            head = self.waiting[0]
+            need = (len(head.prompt_ids) + self.bs - 1) // self.bs + 1
+            if self.prefix_cache_enabled:
+                for k in range(1, len(head.prompt_ids) // self.bs + 1):
+                    if tuple(head.prompt_ids[:k * self.bs]) in self._pc:
+                        need -= 1
+                    else:
+                        break
             if need > self.allocator.num_blocks:
                 r = self.waiting.pop(0)
                 r.finished = True
@@ -307,6 +395,8 @@ class Engine:
                     f"cache ({self.allocator.num_blocks} blocks)")
         if req is not None:
             first = self._prefill(req)
+            if self.prefix_cache_enabled:
+                self._pc_insert(req)
             req.output_ids.append(first)
             self.running.append(req)
         elif self.running:

@@ -360,3 +360,68 @@ def test_sliding_window_matches_masked_forward():
         ops.causal_attention = orig
     ref = logits[0, P - 1:L - 1].argmax(-1).tolist()
     assert req.output_ids == ref, (req.output_ids, ref)
+
+
+# --- prefix caching (refcounted shared prompt blocks) ----------------------
+
+def test_prefix_cache_shares_blocks_and_matches_uncached():
+    """Identical prompt prefixes share KV blocks; outputs are identical
+    to an uncached engine (the shared blocks hold bitwise-identical KV,
+    written once by the first prefill)."""
+    m = build_model("tiny-llama", dtype=torch.float32)
+    base = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=3)
+    pc = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=3,
+                prefix_cache=True)
+    sys_prompt = list(range(1, 33))  # 2 full chunks (bs=16)
+    prompts = [sys_prompt + [40 + i] for i in range(3)]
+    want = [base.generate(list(p), max_new_tokens=5) for p in prompts]
+
+    reqs = [pc.submit(list(p), max_new_tokens=5) for p in prompts]
+    shared = []
+    while pc.has_work():
+        pc.step()
+        shared = [getattr(r, "_shared_chunks", 0) for r in reqs]
+    assert [r.output_ids for r in reqs] == want
+    # later requests hit the 2-chunk prefix
+    assert any(s == 2 for s in shared[1:]), shared
+    # both chunks cached exactly once; cache holds one ref each
+    assert len(pc._pc) == 2
+    free0 = len(pc.allocator.free)
+    pc.flush_prefix_cache()
+    assert len(pc.allocator.free) == free0 + 2
+    assert len(pc.allocator.free) == pc.allocator.num_blocks
+
+
+def test_prefix_cache_eviction_under_pressure():
+    """Distinct prompts churn a small pool: LRU cache entries get
+    evicted so admission never wedges, and accounting stays exact."""
+    m = build_model("tiny-llama", dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=10, seed=1,
+                 prefix_cache=True)
+    for i in range(6):
+        out = eng.generate([i * 7 % 250 + 1] * 20, max_new_tokens=3)
+        assert len(out) == 3
+    # pool of 9 usable blocks, each prompt caches 1 chunk: evictions
+    # must have kept cached+free == total
+    cached = len(eng._pc)
+    assert cached >= 1
+    assert len(eng.allocator.free) + cached == eng.allocator.num_blocks
+    eng.flush_prefix_cache()
+    assert len(eng.allocator.free) == eng.allocator.num_blocks
+
+
+def test_prefix_cache_with_sliding_window():
+    """Prefix sharing composes with the rolling window: dropped shared
+    blocks just decrement the refcount; outputs still match the
+    uncached windowed engine."""
+    m = _windowed_model(32)
+    base = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=2)
+    pc = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=2,
+                prefix_cache=True)
+    prompt = list(range(1, 33))
+    want = base.generate(list(prompt), max_new_tokens=40)
+    got1 = pc.generate(list(prompt), max_new_tokens=40)
+    got2 = pc.generate(list(prompt), max_new_tokens=40)  # cache hit run
+    assert got1 == want and got2 == want
+    pc.flush_prefix_cache()
+    assert len(pc.allocator.free) == pc.allocator.num_blocks
