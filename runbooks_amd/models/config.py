@@ -34,7 +34,8 @@ class ModelConfig:
     sliding_window: int | None = None
     tie_embeddings: bool = False
     mlp_bias: bool = False
-    attn_bias: bool = False
+    attn_bias: bool = False   # bias on q/k/v AND o_proj (OPT)
+    qkv_bias: bool = False    # bias on q/k/v only (qwen2)
 
     def __post_init__(self):
         if self.head_dim is None:
@@ -96,6 +97,12 @@ register(ModelConfig("mistral-7b", vocab_size=32000, hidden_size=4096,
                      intermediate_size=14336, max_seq_len=8192,
                      rope_theta=1000000.0, sliding_window=4096))
 
+# --- qwen2 family (llama-shaped + q/k/v bias, GQA, huge vocab) -------------
+register(ModelConfig("qwen2-7b", vocab_size=152064, hidden_size=3584,
+                     num_layers=28, num_heads=28, num_kv_heads=4,
+                     intermediate_size=18944, max_seq_len=32768,
+                     norm_eps=1e-6, rope_theta=1000000.0, qkv_bias=True))
+
 # --- falcon family (LayerNorm, RoPE, GELU, parallel residual, MQA/GQA) -----
 register(ModelConfig("falcon-7b", vocab_size=65024, hidden_size=4544,
                      num_layers=32, num_heads=71, num_kv_heads=1,
@@ -129,6 +136,10 @@ register(ModelConfig("tiny-falcon", vocab_size=256, hidden_size=64,
                      intermediate_size=128, head_dim=8, max_seq_len=128,
                      norm="layernorm", act="gelu", parallel_residual=True,
                      single_norm=True, tie_embeddings=True))
+register(ModelConfig("tiny-qwen", vocab_size=256, hidden_size=64,
+                     num_layers=2, num_heads=4, num_kv_heads=2,
+                     intermediate_size=128, max_seq_len=128,
+                     norm_eps=1e-6, qkv_bias=True))
 register(ModelConfig("tiny-opt", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=4,
                      intermediate_size=128, max_seq_len=128,

@@ -161,6 +161,20 @@ def config_from_hf_json(path: str | Path) -> ModelConfig:
     registry)."""
     d = json.loads(Path(path).read_text())
     mt = d.get("model_type", "llama")
+    if mt == "qwen2":
+        return ModelConfig(
+            d.get("_name_or_path", "hf-qwen2"),
+            vocab_size=d["vocab_size"], hidden_size=d["hidden_size"],
+            num_layers=d["num_hidden_layers"],
+            num_heads=d["num_attention_heads"],
+            num_kv_heads=d.get("num_key_value_heads",
+                               d["num_attention_heads"]),
+            intermediate_size=d["intermediate_size"],
+            max_seq_len=d.get("max_position_embeddings", 32768),
+            norm_eps=d.get("rms_norm_eps", 1e-6),
+            rope_theta=d.get("rope_theta", 1000000.0),
+            qkv_bias=True,
+            tie_embeddings=d.get("tie_word_embeddings", False))
     if mt in ("llama", "mistral"):
         return ModelConfig(
             d.get("_name_or_path", "hf-llama"),

@@ -53,19 +53,21 @@ class Attention(nn.Module):
 
         q_out = cfg.num_heads * self.dh
         kv_out = cfg.num_kv_heads * self.dh
+        # qwen2-style models carry bias on q/k/v but not on o_proj
+        in_bias = cfg.attn_bias or cfg.qkv_bias
         self.q_proj = ColumnParallelLinear(cfg.hidden_size, q_out,
-                                           bias=cfg.attn_bias, tp_size=tp, dtype=dtype)
+                                           bias=in_bias, tp_size=tp, dtype=dtype)
         if self.kv_sharded:
             self.k_proj = ColumnParallelLinear(cfg.hidden_size, kv_out,
-                                               bias=cfg.attn_bias, tp_size=tp,
+                                               bias=in_bias, tp_size=tp,
                                                dtype=dtype, gather_input=False)
             self.v_proj = ColumnParallelLinear(cfg.hidden_size, kv_out,
-                                               bias=cfg.attn_bias, tp_size=tp,
+                                               bias=in_bias, tp_size=tp,
                                                dtype=dtype, gather_input=False)
         else:  # replicate KV (MQA with tp > num_kv_heads)
-            self.k_proj = ops.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
+            self.k_proj = ops.Linear(cfg.hidden_size, kv_out, bias=in_bias,
                                     dtype=dtype)
-            self.v_proj = ops.Linear(cfg.hidden_size, kv_out, bias=cfg.attn_bias,
+            self.v_proj = ops.Linear(cfg.hidden_size, kv_out, bias=in_bias,
                                     dtype=dtype)
         self.o_proj = RowParallelLinear(cfg.num_heads * self.dh, cfg.hidden_size,
                                         bias=cfg.attn_bias, tp_size=tp, dtype=dtype)

@@ -174,3 +174,43 @@ def test_load_pretrained_tp_sharding(tmp_path):
                            dw[:, rank * s2:(rank + 1) * s2])
         # replicated norm
         assert torch.equal(shard.norm_f.weight, full.norm_f.weight)
+
+
+def test_qwen2_hf_parity():
+    """qwen2: llama-shaped with bias on q/k/v only (cfg.qkv_bias)."""
+    from transformers import Qwen2Config, Qwen2ForCausalLM
+    hf_cfg = Qwen2Config(vocab_size=128, hidden_size=64,
+                         intermediate_size=128, num_hidden_layers=2,
+                         num_attention_heads=4, num_key_value_heads=2,
+                         max_position_embeddings=64, rms_norm_eps=1e-6,
+                         rope_theta=10000.0, tie_word_embeddings=False)
+    torch.manual_seed(2)
+    hf = Qwen2ForCausalLM(hf_cfg).eval()
+
+    cfg = ModelConfig("t-qwen", vocab_size=128, hidden_size=64, num_layers=2,
+                      num_heads=4, num_kv_heads=2, intermediate_size=128,
+                      max_seq_len=64, norm_eps=1e-6, qkv_bias=True)
+    ours = build_model(cfg, dtype=torch.float32)
+    state = convert_hf_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(state, strict=False)
+    assert not [m for m in missing if not m.startswith("rope_")], missing
+    assert not unexpected, unexpected
+
+    tokens = torch.randint(0, 128, (2, 17))
+    with torch.no_grad():
+        theirs = hf(tokens).logits
+        got = ours(tokens)
+    _logits_close(got, theirs)
+
+
+def test_qwen2_config_from_hf_json(tmp_path):
+    import json as _json
+    (tmp_path / "config.json").write_text(_json.dumps({
+        "model_type": "qwen2", "vocab_size": 1024, "hidden_size": 64,
+        "num_hidden_layers": 2, "num_attention_heads": 4,
+        "num_key_value_heads": 2, "intermediate_size": 128,
+        "rope_theta": 1000000.0, "tie_word_embeddings": True}))
+    from runbooks_amd.models.load import config_from_hf_json
+    cfg = config_from_hf_json(tmp_path / "config.json")
+    assert cfg.qkv_bias and not cfg.attn_bias
+    assert cfg.tie_embeddings and cfg.rope_theta == 1000000.0
