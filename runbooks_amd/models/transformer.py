@@ -232,12 +232,25 @@ class Transformer(nn.Module):
         positions = torch.arange(S, device=tokens.device, dtype=torch.int32)
         positions = positions.repeat(B)
         x = self._embed(tokens, positions)
+        ckpt = getattr(self, "grad_checkpointing", False) and \
+            torch.is_grad_enabled()
         for blk in self.blocks:
             fn = lambda h, b=blk: b.attn.forward_train(  # noqa: E731
                 h, self.rope_cos, self.rope_sin, positions, B, S)
-            x = blk(x, fn)
+            if ckpt:
+                # recompute this block's activations in backward: peak
+                # activation memory drops from O(layers) to O(1) blocks
+                # (full fine-tunes of 70B-class models at long seq; LoRA
+                # at default shapes fits 288 GB without it)
+                x = torch.utils.checkpoint.checkpoint(
+                    blk, x, fn, use_reentrant=False)
+            else:
+                x = blk(x, fn)
         x = self.norm_f(x)
         return self.lm_head(x).view(B, S, -1)
+
+    def enable_grad_checkpointing(self, enabled: bool = True) -> None:
+        self.grad_checkpointing = enabled
 
     # -- serving ---------------------------------------------------------------
     @torch.no_grad()

@@ -36,6 +36,7 @@ class TrainConfig:
     lora_r: int = 16
     lora_alpha: int = 32
     full_finetune: bool = False
+    grad_checkpointing: bool = False    # recompute block activations
     grad_clip: float = 1.0
     dtype: str = "bfloat16"
     seed: int = 0
@@ -52,6 +53,8 @@ class Trainer:
                             device=self.device)
         if not cfg.full_finetune:
             apply_lora(model, r=cfg.lora_r, alpha=cfg.lora_alpha)
+        if cfg.grad_checkpointing:
+            model.enable_grad_checkpointing()
         self.ddp = DataParallel(model)
         self.model = model
         trainable = [p for p in model.parameters() if p.requires_grad]

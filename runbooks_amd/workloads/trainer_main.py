@@ -85,6 +85,7 @@ def main():
         grad_accum_steps=_param("gradient_accumulation_steps", 1, int),
         lora_r=_param("lora_r", 16, int),
         full_finetune=_param("full_finetune", False, bool),
+        grad_checkpointing=_param("gradient_checkpointing", False, bool),
         dtype="bfloat16" if torch.cuda.is_available() else "float32",
         output_dir=out_dir)
     trainer = Trainer(cfg)

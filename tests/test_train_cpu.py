@@ -192,3 +192,31 @@ def test_text_dataset_formats(tmp_path):
     assert row.shape == (16,) and row.dtype == torch.long
     single = TextDataset(tmp_path / "d.parquet", seq_len=8, vocab_size=256)
     assert len(single) == 2
+
+
+def test_grad_checkpointing_matches_plain():
+    """Activation checkpointing recomputes forward in backward: loss and
+    gradients must be bitwise-equal to the plain path (fp32 CPU)."""
+    from runbooks_amd.models import build_model
+
+    tokens = torch.randint(0, 256, (2, 17))
+
+    def run(ckpt):
+        torch.manual_seed(0)
+        m = build_model("tiny-llama", dtype=torch.float32, seed=7)
+        if ckpt:
+            m.enable_grad_checkpointing()
+        logits = m(tokens[:, :-1])
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]), tokens[:, 1:].reshape(-1))
+        loss.backward()
+        grads = {n: p.grad.clone() for n, p in m.named_parameters()
+                 if p.grad is not None}
+        return float(loss), grads
+
+    l0, g0 = run(False)
+    l1, g1 = run(True)
+    assert l0 == l1
+    assert g0.keys() == g1.keys() and len(g0) > 0
+    for n in g0:
+        assert torch.equal(g0[n], g1[n]), n
