@@ -36,6 +36,7 @@ class ModelConfig:
     mlp_bias: bool = False
     attn_bias: bool = False   # bias on q/k/v AND o_proj (OPT)
     qkv_bias: bool = False    # bias on q/k/v only (qwen2)
+    embed_scale: float = 1.0  # gemma multiplies embeddings by sqrt(hidden)
 
     def __post_init__(self):
         if self.head_dim is None:
@@ -103,6 +104,13 @@ register(ModelConfig("qwen2-7b", vocab_size=152064, hidden_size=3584,
                      intermediate_size=18944, max_seq_len=32768,
                      norm_eps=1e-6, rope_theta=1000000.0, qkv_bias=True))
 
+# --- gemma family (GeGLU, RMSNorm(1+w) folded at load, scaled embed) -------
+register(ModelConfig("gemma-7b", vocab_size=256000, hidden_size=3072,
+                     num_layers=28, num_heads=16, num_kv_heads=16,
+                     intermediate_size=24576, max_seq_len=8192, head_dim=256,
+                     norm_eps=1e-6, act="gelu_glu", tie_embeddings=True,
+                     embed_scale=3072 ** 0.5))
+
 # --- falcon family (LayerNorm, RoPE, GELU, parallel residual, MQA/GQA) -----
 register(ModelConfig("falcon-7b", vocab_size=65024, hidden_size=4544,
                      num_layers=32, num_heads=71, num_kv_heads=1,
@@ -140,6 +148,11 @@ register(ModelConfig("tiny-qwen", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=2,
                      intermediate_size=128, max_seq_len=128,
                      norm_eps=1e-6, qkv_bias=True))
+register(ModelConfig("tiny-gemma", vocab_size=256, hidden_size=64,
+                     num_layers=2, num_heads=4, num_kv_heads=2,
+                     intermediate_size=128, head_dim=16, max_seq_len=128,
+                     norm_eps=1e-6, act="gelu_glu", tie_embeddings=True,
+                     embed_scale=8.0))
 register(ModelConfig("tiny-opt", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=4,
                      intermediate_size=128, max_seq_len=128,

@@ -105,6 +105,12 @@ def convert_hf_state_dict(hf_state: dict, cfg: ModelConfig) -> dict:
             # OPTLearnedPositionalEmbedding); drop the two pad rows.
             w = w[2:]
         out[n] = w
+    # gemma stores RMSNorm weights zero-centered (x * (1 + w)); fold the
+    # +1 at import so the fused rmsnorm kernel stays family-agnostic
+    if cfg.act == "gelu_glu" and cfg.norm == "rmsnorm":
+        for k in list(out):
+            if k.endswith((".norm1.weight", ".norm2.weight")) or                     k == "norm_f.weight":
+                out[k] = out[k].float() + 1.0
     # tied embeddings: derive lm_head when absent
     if "lm_head.weight" not in out and "embed.weight" in out:
         out["lm_head.weight"] = out["embed.weight"]
@@ -161,6 +167,20 @@ def config_from_hf_json(path: str | Path) -> ModelConfig:
     registry)."""
     d = json.loads(Path(path).read_text())
     mt = d.get("model_type", "llama")
+    if mt == "gemma":
+        heads = d["num_attention_heads"]
+        return ModelConfig(
+            d.get("_name_or_path", "hf-gemma"),
+            vocab_size=d["vocab_size"], hidden_size=d["hidden_size"],
+            num_layers=d["num_hidden_layers"], num_heads=heads,
+            num_kv_heads=d.get("num_key_value_heads", heads),
+            intermediate_size=d["intermediate_size"],
+            max_seq_len=d.get("max_position_embeddings", 8192),
+            head_dim=d.get("head_dim", d["hidden_size"] // heads),
+            norm_eps=d.get("rms_norm_eps", 1e-6),
+            rope_theta=d.get("rope_theta", 10000.0),
+            act="gelu_glu", tie_embeddings=True,
+            embed_scale=d["hidden_size"] ** 0.5)
     if mt == "qwen2":
         return ModelConfig(
             d.get("_name_or_path", "hf-qwen2"),

@@ -133,7 +133,7 @@ class MLP(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype, tp: int):
         super().__init__()
         self.act = cfg.act
-        if cfg.act == "silu_glu":
+        if cfg.act in ("silu_glu", "gelu_glu"):
             self.gate_proj = ColumnParallelLinear(cfg.hidden_size, cfg.intermediate_size,
                                                   bias=cfg.mlp_bias, tp_size=tp, dtype=dtype)
             self.up_proj = ColumnParallelLinear(cfg.hidden_size, cfg.intermediate_size,
@@ -156,6 +156,12 @@ class MLP(nn.Module):
                 return self.down_proj(ops.swiglu(
                     y[:, :half].contiguous(), y[:, half:].contiguous()))
             return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+        if self.act == "gelu_glu":
+            # gemma GeGLU (tanh approximation); eager torch on GPU — a
+            # fused kernel like swiglu_packed is a round-2 item
+            return self.down_proj(
+                torch.nn.functional.gelu(self.gate_proj(x),
+                                         approximate="tanh") * self.up_proj(x))
         h = self.fc1(x)
         h = torch.nn.functional.gelu(h) if self.act == "gelu" else torch.relu(h)
         return self.down_proj(h)
@@ -220,7 +226,11 @@ class Transformer(nn.Module):
 
     # -- shared ----------------------------------------------------------------
     def _embed(self, tokens, positions):
+        """Token (+learned position) embedding; gemma scales by
+        sqrt(hidden) (cfg.embed_scale, 1.0 elsewhere)."""
         x = self.embed(tokens.reshape(-1))
+        if self.cfg.embed_scale != 1.0:
+            x = x * self.cfg.embed_scale
         if self.cfg.pos == "learned":
             x = x + self.embed_pos(positions.long())
         return x

@@ -214,3 +214,34 @@ def test_qwen2_config_from_hf_json(tmp_path):
     cfg = config_from_hf_json(tmp_path / "config.json")
     assert cfg.qkv_bias and not cfg.attn_bias
     assert cfg.tie_embeddings and cfg.rope_theta == 1000000.0
+
+
+def test_gemma_hf_parity():
+    """gemma: GeGLU (tanh), RMSNorm(1+w) (folded at import), embeddings
+    scaled by sqrt(hidden), tied lm_head, wide head_dim."""
+    from transformers import GemmaConfig, GemmaForCausalLM
+    hf_cfg = GemmaConfig(vocab_size=128, hidden_size=64,
+                         intermediate_size=128, num_hidden_layers=2,
+                         num_attention_heads=4, num_key_value_heads=2,
+                         head_dim=16, max_position_embeddings=64,
+                         rms_norm_eps=1e-6, rope_theta=10000.0,
+                         hidden_act="gelu_pytorch_tanh",
+                         tie_word_embeddings=True)
+    torch.manual_seed(3)
+    hf = GemmaForCausalLM(hf_cfg).eval()
+
+    cfg = ModelConfig("t-gemma", vocab_size=128, hidden_size=64, num_layers=2,
+                      num_heads=4, num_kv_heads=2, intermediate_size=128,
+                      head_dim=16, max_seq_len=64, norm_eps=1e-6,
+                      act="gelu_glu", tie_embeddings=True,
+                      embed_scale=64 ** 0.5)
+    ours = build_model(cfg, dtype=torch.float32)
+    state = convert_hf_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(state, strict=False)
+    assert not [m for m in missing if not m.startswith("rope_")], missing
+
+    tokens = torch.randint(0, 128, (2, 17))
+    with torch.no_grad():
+        theirs = hf(tokens).logits
+        got = ours(tokens)
+    _logits_close(got, theirs)

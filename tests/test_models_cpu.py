@@ -153,7 +153,8 @@ def test_preempted_output_preserved():
     assert combined == ref, (combined, ref)
 
 
-@pytest.mark.parametrize("name", ["tiny-opt", "tiny-falcon"])
+@pytest.mark.parametrize("name", ["tiny-opt", "tiny-falcon", "tiny-gemma",
+                                  "tiny-qwen"])
 def test_engine_generates_all_families(name):
     """Decode path per family: OPT (learned positions, no rope, biases)
     and falcon (parallel residual, MQA) — the BASELINE config families."""
