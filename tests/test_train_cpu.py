@@ -212,7 +212,7 @@ def test_grad_checkpointing_matches_plain():
         loss.backward()
         grads = {n: p.grad.clone() for n, p in m.named_parameters()
                  if p.grad is not None}
-        return float(loss), grads
+        return float(loss.detach()), grads
 
     l0, g0 = run(False)
     l1, g1 = run(True)
