@@ -267,13 +267,14 @@ def build_app(engine: Engine, tokenizer=None,
                     break
         if finish != "stop":
             text = tok.decode(out)
-        return text, finish, len(out)
+        return text, finish, out
 
-    def _fmt_logprobs(ereq, produced: int):
-        """OpenAI-style logprobs block for one finished choice. `produced`
-        caps at the tokens actually delivered (stop-truncation)."""
-        ids = ereq.output_ids[:produced]
-        data = ereq.logprob_data[:produced]
+    def _fmt_logprobs(ereq, ids: list[int]):
+        """OpenAI-style logprobs block for one finished choice. `ids` are
+        the tokens actually delivered (stop-truncation; preemptions move
+        early output into prompt_ids, so ereq.output_ids can be shorter —
+        logprob_data spans the full delivered stream)."""
+        data = ereq.logprob_data[:len(ids)]
         toks = [tok.decode([t]) for t in ids]
         offs, pos = [], 0
         for t in toks:
@@ -326,14 +327,14 @@ def build_app(engine: Engine, tokenizer=None,
 
         choices, n_out = [], 0
         for i, (cq, creq_) in enumerate(subs):
-            text, finish, produced = _collect(cq, creq_.request_id, stops)
+            text, finish, out_ids = _collect(cq, creq_.request_id, stops)
             if req.echo:
                 text = req.prompt + text
-            lp = (_fmt_logprobs(creq_, produced)
+            lp = (_fmt_logprobs(creq_, out_ids)
                   if req.logprobs is not None else None)
             choices.append({"text": text, "index": i,
                             "logprobs": lp, "finish_reason": finish})
-            n_out += produced
+            n_out += len(out_ids)
         if m_reqs is not None:
             m_reqs.inc()
             m_tokens.inc(n_out)
