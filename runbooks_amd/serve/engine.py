@@ -144,6 +144,9 @@ class Engine:
             else _os.environ.get("RB_PREFIX_CACHE", "0") == "1")
         from collections import OrderedDict
         self._pc: "OrderedDict[tuple, int]" = OrderedDict()
+        self.stats = {"steps": 0, "prefills": 0, "decode_tokens": 0,
+                      "preemptions": 0, "prefix_hits": 0,
+                      "prefix_hit_blocks": 0, "window_dropped_blocks": 0}
 
     def _auto_kv_blocks(self, mem_fraction: float) -> int:
         bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
@@ -198,6 +201,9 @@ class Engine:
         if need > len(self.allocator.free):
             return None
         self.waiting.pop(0)
+        if hits:
+            self.stats["prefix_hits"] += 1
+            self.stats["prefix_hit_blocks"] += len(hits)
         for key, b in hits:
             self.allocator.share(b)
             self._pc.move_to_end(key)
@@ -337,6 +343,7 @@ class Engine:
             while r.dropped + self.bs <= r.seq_len - w and len(r.blocks) > 1:
                 self.allocator.release([r.blocks.pop(0)])
                 r.dropped += self.bs
+                self.stats["window_dropped_blocks"] += 1
 
     def _preempt(self, r: Request) -> None:
         """Release a running request's cache and requeue it: its generated
@@ -347,6 +354,7 @@ class Engine:
         r.prompt_ids = r.prompt_ids + r.output_ids
         r.max_new_tokens -= len(r.output_ids)
         r.output_ids = []
+        self.stats["preemptions"] += 1
         if hasattr(r, "_watch_sent"):
             r._watch_sent = 0  # post-preempt tokens are all new to watchers
         self.running.remove(r)
@@ -376,6 +384,7 @@ class Engine:
     # -- scheduler step ----------------------------------------------------------
     def step(self) -> list[Request]:
         """One engine iteration. Returns requests finished this step."""
+        self.stats["steps"] += 1
         finished = []
         req = self._admit()
         if req is None and not self.running and self.waiting:
@@ -395,6 +404,7 @@ class Engine:
                     f"cache ({self.allocator.num_blocks} blocks)")
         if req is not None:
             first = self._prefill(req)
+            self.stats["prefills"] += 1
             if self.prefix_cache_enabled:
                 self._pc_insert(req)
             req.output_ids.append(first)
@@ -417,6 +427,7 @@ class Engine:
                         break
             if self.running:
                 toks = self._decode_batch(self.running)
+                self.stats["decode_tokens"] += len(toks)
                 for r, t in zip(self.running, toks):
                     r.output_ids.append(t)
         for r in list(self.running):

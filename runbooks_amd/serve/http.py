@@ -45,8 +45,9 @@ try:
                       buckets=(.05, .1, .25, .5, 1, 2.5, 5, 10, 30, 60))
     M_RUNNING = Gauge("rb_running_requests", "requests decoding")
     M_KV_FREE = Gauge("rb_kv_blocks_free", "free KV cache blocks")
+    M_ENGINE = Gauge("rb_engine_stat", "engine counters", ["stat"])
 except ImportError:  # pragma: no cover
-    M_REQS = M_TOKENS = M_LAT = M_RUNNING = M_KV_FREE = None
+    M_REQS = M_TOKENS = M_LAT = M_RUNNING = M_KV_FREE = M_ENGINE = None
 
 
 class CompletionRequest(BaseModel):
@@ -187,6 +188,8 @@ def build_app(engine: Engine, tokenizer=None,
         def metrics():
             M_RUNNING.set(len(engine.running))
             M_KV_FREE.set(len(engine.allocator.free))
+            for k, v in engine.stats.items():
+                M_ENGINE.labels(stat=k).set(v)
             from fastapi import Response
             return Response(generate_latest(),
                             media_type=CONTENT_TYPE_LATEST)

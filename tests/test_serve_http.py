@@ -239,3 +239,10 @@ def test_per_request_latency_log(client, caplog):
         client.post("/v1/completions", json={"prompt": "log", "max_tokens": 2})
     recs = [r for r in caplog.records if "completion id=" in r.getMessage()]
     assert recs and "latency_ms=" in recs[-1].getMessage()
+
+
+def test_engine_stats_in_metrics(client):
+    client.post("/v1/completions", json={"prompt": "st", "max_tokens": 2})
+    r = client.get("/metrics")
+    assert 'rb_engine_stat{stat="prefills"}' in r.text
+    assert 'rb_engine_stat{stat="decode_tokens"}' in r.text
