@@ -440,3 +440,54 @@ def test_engine_generates_qwen_family():
             logits = m(torch.tensor([full]))
         full.append(int(logits[0, -1].argmax()))
     assert out == full[3:], (out, full[3:])
+
+
+def test_engine_fuzz_prefix_cache_and_window():
+    """Randomized traffic with prefix caching ON and a sliding window:
+    per-block refcounts must equal (live request holders) + (cache
+    holds), the free list must be exactly the zero-ref blocks, and
+    everything terminates with a clean pool after a cache flush."""
+    import random
+
+    rng = random.Random(99)
+    m = _windowed_model(32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=28,
+                 max_batch=4, seed=5, prefix_cache=True)
+    # few shared prompt templates -> lots of prefix hits
+    templates = [[i + 1] * 24 for i in range(3)]
+    live = []
+    submitted = 0
+    for step in range(400):
+        if rng.random() < 0.3 and submitted < 25:
+            p = list(rng.choice(templates)) + \
+                [rng.randrange(256) for _ in range(rng.randint(0, 8))]
+            live.append(eng.submit(p, max_new_tokens=rng.randint(1, 30)))
+            submitted += 1
+        if rng.random() < 0.05 and live:
+            eng.cancel(rng.choice(live).request_id)
+        if eng.has_work():
+            eng.step()
+        # refcount invariants
+        counts = [0] * eng.allocator.num_blocks
+        for r in eng.running:
+            assert r.dropped % eng.bs == 0
+            for b in r.blocks:
+                counts[b] += 1
+        for b in eng._pc.values():
+            counts[b] += 1
+        assert counts == eng.allocator.refs, (step, counts,
+                                              eng.allocator.refs)
+        free = sorted(eng.allocator.free)
+        assert free == sorted(set(free)), "duplicate free blocks"
+        assert free == [b for b in range(eng.allocator.num_blocks)
+                        if counts[b] == 0]
+    for _ in range(3000):
+        if not eng.has_work():
+            break
+        eng.step()
+    assert not eng.has_work()
+    assert eng.stats["prefix_hits"] > 0, "fuzz never hit the cache"
+    eng.flush_prefix_cache()
+    assert len(eng.allocator.free) == eng.allocator.num_blocks
+    for r in live:
+        assert r.finished
