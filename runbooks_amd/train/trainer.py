@@ -37,6 +37,7 @@ class TrainConfig:
     lora_alpha: int = 32
     full_finetune: bool = False
     grad_checkpointing: bool = False    # recompute block activations
+    eval_steps: int = 0                 # 0 = no evaluation
     grad_clip: float = 1.0
     dtype: str = "bfloat16"
     seed: int = 0
@@ -110,7 +111,30 @@ class Trainer:
             g["lr"] = cfg.lr * scale
 
     # -- full loop -------------------------------------------------------------
-    def fit(self, dataset=None, log_every: int = 10):
+    @torch.no_grad()
+    def evaluate(self, dataset, max_batches: int = 16) -> float:
+        """Mean cross-entropy over (up to) max_batches of an eval split —
+        the eval_loss the reference's HF trainer images log."""
+        self.model.eval()
+        loader = data_loader(dataset, self.cfg.micro_batch,
+                             rank=comm.rank(), world=comm.world_size(),
+                             seed=self.cfg.seed + 1)
+        total, n = 0.0, 0
+        for batch in loader:
+            inputs = batch[:, :-1].to(self.device)
+            labels = batch[:, 1:].to(self.device)
+            total += float(ops.cross_entropy(self.model(inputs), labels))
+            n += 1
+            if n >= max_batches:
+                break
+        self.model.train()
+        if comm.world_size() > 1:
+            t = torch.tensor([total, float(n)])
+            torch.distributed.all_reduce(t)
+            total, n = float(t[0]), int(t[1])
+        return total / max(1, n)
+
+    def fit(self, dataset=None, eval_dataset=None, log_every: int = 10):
         cfg = self.cfg
         if dataset is None:
             dataset = SyntheticTokens(self.model.cfg.vocab_size, cfg.seq_len + 1)
@@ -132,6 +156,12 @@ class Trainer:
                     it = iter(loader)
                     batch = next(it)
             loss = self.train_step(batch)
+            if cfg.eval_steps and eval_dataset is not None and \
+                    self.step_num % cfg.eval_steps == 0:
+                ev = self.evaluate(eval_dataset)
+                if comm.rank() == 0:
+                    print(f"step {self.step_num} eval_loss {ev:.4f}",
+                          flush=True)
             if cfg.save_steps and self.step_num % cfg.save_steps == 0:
                 self.save()
             if comm.rank() == 0 and self.step_num % log_every == 0:

@@ -86,6 +86,7 @@ def main():
         lora_r=_param("lora_r", 16, int),
         full_finetune=_param("full_finetune", False, bool),
         grad_checkpointing=_param("gradient_checkpointing", False, bool),
+        eval_steps=_param("eval_steps", 0, int),
         dtype="bfloat16" if torch.cuda.is_available() else "float32",
         output_dir=out_dir)
     trainer = Trainer(cfg)
@@ -111,8 +112,17 @@ def main():
         dataset = SyntheticTokens(trainer.model.cfg.vocab_size,
                                   cfg.seq_len + 1)
 
+    eval_dataset = None
+    if trainer.cfg.eval_steps and files:
+        # hold out ~5% of rows (HF-style eval split) for eval_loss
+        n_eval = max(1, len(dataset) // 20)
+        eval_dataset = torch.utils.data.Subset(
+            dataset, range(len(dataset) - n_eval, len(dataset)))
+        dataset = torch.utils.data.Subset(
+            dataset, range(len(dataset) - n_eval))
+
     trainer.resume()
-    trainer.fit(dataset)
+    trainer.fit(dataset, eval_dataset=eval_dataset)
     if comm.rank() == 0:
         print(f"trainer: done at step {trainer.step_num}; "
               f"artifacts in {out_dir}")

@@ -220,3 +220,21 @@ def test_grad_checkpointing_matches_plain():
     assert g0.keys() == g1.keys() and len(g0) > 0
     for n in g0:
         assert torch.equal(g0[n], g1[n]), n
+
+
+def test_trainer_evaluate_and_eval_steps(capsys):
+    """evaluate() returns a finite mean CE; fit() logs eval_loss at the
+    configured cadence."""
+    from runbooks_amd.train import SyntheticTokens, TrainConfig, Trainer
+
+    cfg = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                      num_train_steps=4, eval_steps=2, dtype="float32",
+                      save_steps=0, output_dir="/tmp/rb-eval-test")
+    tr = Trainer(cfg, device="cpu")
+    ev = SyntheticTokens(tr.model.cfg.vocab_size, 17, n=8, seed=9)
+    loss = tr.evaluate(ev, max_batches=2)
+    assert loss == loss and loss > 0  # finite, positive
+    tr.fit(SyntheticTokens(tr.model.cfg.vocab_size, 17, n=16), ev,
+           log_every=100)
+    out = capsys.readouterr().out
+    assert out.count("eval_loss") == 2, out
