@@ -103,6 +103,10 @@ class Engine:
         if torch.cuda.is_available() and model.tp == 1:
             from ..models.transformer import fuse_for_inference
             fuse_for_inference(self.model, load_in_8bit=load_in_8bit)
+        elif load_in_8bit and model.tp > 1:
+            print("engine: MODEL_LOAD_IN_8BIT ignored at TP>1 "
+                  "(fp8 decode path is single-GPU; 288 GB/GPU rarely "
+                  "needs 8-bit at TP>1)", flush=True)
         self.cfg = model.cfg
         self.bs = ops.BLOCK_SIZE
         self.max_batch = max_batch
