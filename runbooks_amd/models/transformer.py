@@ -157,8 +157,20 @@ class MLP(nn.Module):
                     y[:, :half].contiguous(), y[:, half:].contiguous()))
             return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
         if self.act == "gelu_glu":
-            # gemma GeGLU (tanh approximation); eager torch on GPU — a
-            # fused kernel like swiglu_packed is a round-2 item
+            # gemma GeGLU (tanh approximation). The fused gate/up path
+            # (csrc geglu_packed) exists but fuse_for_inference only
+            # enables it under RB_FUSED_GEGLU=1 until a GPU validation
+            # pass; eager torch is the default.
+            if getattr(self, "_gateup_w", None) is not None and \
+                    not torch.is_grad_enabled():
+                y = ops.fast_linear(x, self._gateup_w)
+                if ops.use_hip(y):
+                    return self.down_proj(ops.ext().geglu_packed(y))
+                half = y.shape[-1] // 2
+                return self.down_proj(
+                    torch.nn.functional.gelu(y[:, :half],
+                                             approximate="tanh") *
+                    y[:, half:])
             return self.down_proj(
                 torch.nn.functional.gelu(self.gate_proj(x),
                                          approximate="tanh") * self.up_proj(x))
@@ -329,7 +341,11 @@ def fuse_for_inference(model: "Transformer",
             attn.v_proj.weight = nn.Parameter(fused[qo + kvo:],
                                               requires_grad=False)
         mlp = blk.mlp
-        if model.cfg.act == "silu_glu" and mlp.gate_proj.bias is None:
+        import os as _os
+        glu_fusable = model.cfg.act == "silu_glu" or (
+            model.cfg.act == "gelu_glu" and
+            _os.environ.get("RB_FUSED_GEGLU", "0") == "1")
+        if glu_fusable and mlp.gate_proj.bias is None:
             fused = torch.cat([mlp.gate_proj.weight.data,
                                mlp.up_proj.weight.data],
                               dim=0).contiguous()

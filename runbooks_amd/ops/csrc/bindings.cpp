@@ -45,6 +45,7 @@ at::Tensor qkv_rope_append(at::Tensor y, at::Tensor cos, at::Tensor sin,
                            at::Tensor v_cache, at::Tensor slot_mapping,
                            int64_t hq);
 at::Tensor swiglu_packed(at::Tensor y);
+at::Tensor geglu_packed(at::Tensor y);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "runbooks_amd gfx950 HIP kernels";
@@ -84,4 +85,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "packed qkv -> rope'd q + rope'd k / v appended to the paged cache");
   m.def("swiglu_packed", &swiglu_packed,
         "silu(y[:, :I]) * y[:, I:] from the fused gate/up GEMM output");
+  m.def("geglu_packed", &geglu_packed,
+        "packed GeGLU (tanh) epilogue [T,2I]->[T,I]");
 }

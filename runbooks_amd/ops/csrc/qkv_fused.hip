@@ -120,6 +120,36 @@ __global__ void swiglu_packed_kernel(const T *__restrict__ y,
   }
 }
 
+// gemma GeGLU: gelu_tanh(y[:, :I]) * y[:, I:]. Same memory shape as
+// swiglu_packed; only the activation differs. Gated in python behind
+// RB_FUSED_GEGLU until a GPU validation pass (round 2) - the eager
+// torch path is the default for gelu_glu models.
+template <typename T>
+__global__ void geglu_packed_kernel(const T *__restrict__ y,
+                                    T *__restrict__ out,
+                                    int64_t n_rows, int inter) {
+  constexpr int W = rb::VIO<T>::W;
+  const int nvec = inter / W;
+  const int64_t total = n_rows * nvec;
+  const int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < total;
+       i += stride) {
+    const int64_t row = i / nvec;
+    const int v = (int)(i % nvec);
+    float g[W], u[W];
+    rb::VIO<T>::load(y + (row * 2 * inter) + v * W, g);
+    rb::VIO<T>::load(y + (row * 2 * inter) + inter + v * W, u);
+#pragma unroll
+    for (int k = 0; k < W; ++k) {
+      // tanh approximation (HF gelu_pytorch_tanh)
+      const float x = g[k];
+      const float t = tanhf(0.7978845608028654f * (x + 0.044715f * x * x * x));
+      g[k] = 0.5f * x * (1.0f + t) * u[k];
+    }
+    rb::VIO<T>::store(out + row * inter + v * W, g);
+  }
+}
+
 }  // namespace
 
 at::Tensor qkv_rope_append(at::Tensor y, at::Tensor cos, at::Tensor sin,
@@ -175,6 +205,32 @@ at::Tensor swiglu_packed(at::Tensor y) {
                        (float *)out.data_ptr(), n_rows, inter);
   } else {
     TORCH_CHECK(false, "swiglu_packed: dtype");
+  }
+  return out;
+}
+
+at::Tensor geglu_packed(at::Tensor y) {
+  TORCH_CHECK(y.is_cuda() && y.is_contiguous(), "geglu_packed: y");
+  const int64_t inter2 = y.size(-1);
+  TORCH_CHECK(inter2 % 2 == 0, "geglu_packed: last dim even");
+  const int inter = (int)(inter2 / 2);
+  const int64_t n_rows = y.numel() / inter2;
+  auto out = at::empty({n_rows, (int64_t)inter}, y.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = rb::rb_grid_1d(n_rows * (inter / 8), BLOCK);
+  if (y.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(inter % 8 == 0, "geglu_packed bf16: I % 8");
+    hipLaunchKernelGGL(geglu_packed_kernel<uint16_t>, dim3(grid),
+                       dim3(BLOCK), 0, stream,
+                       (const uint16_t *)y.data_ptr(),
+                       (uint16_t *)out.data_ptr(), n_rows, inter);
+  } else if (y.scalar_type() == at::kFloat) {
+    TORCH_CHECK(inter % 4 == 0, "geglu_packed f32: I % 4");
+    hipLaunchKernelGGL(geglu_packed_kernel<float>, dim3(grid), dim3(BLOCK),
+                       0, stream, (const float *)y.data_ptr(),
+                       (float *)out.data_ptr(), n_rows, inter);
+  } else {
+    TORCH_CHECK(false, "geglu_packed: dtype");
   }
   return out;
 }

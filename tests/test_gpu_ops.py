@@ -467,3 +467,17 @@ def test_train_gemm_nt(M, N, K):
     d = (c.float() - ref).abs().max().item()
     rel = d / ref.abs().max().item()
     assert rel < 2e-2, f"max abs {d} rel {rel}"
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not _EXPERIMENTAL, reason="RB_EXPERIMENTAL=1 only "
+                    "(geglu_packed awaits its first on-GPU validation)")
+def test_geglu_packed():
+    """csrc geglu_packed vs fp32 tanh-gelu reference."""
+    from runbooks_amd import ops
+    torch.manual_seed(0)
+    y = torch.randn(64, 2 * 256, device="cuda", dtype=torch.bfloat16)
+    got = ops.ext().geglu_packed(y)
+    g, u = y[:, :256].float(), y[:, 256:].float()
+    ref = torch.nn.functional.gelu(g, approximate="tanh") * u
+    assert torch.allclose(got.float(), ref, atol=3e-2, rtol=3e-2)

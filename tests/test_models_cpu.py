@@ -491,3 +491,21 @@ def test_engine_fuzz_prefix_cache_and_window():
     assert len(eng.allocator.free) == eng.allocator.num_blocks
     for r in live:
         assert r.finished
+
+
+def test_geglu_fused_path_matches_eager(monkeypatch):
+    """RB_FUSED_GEGLU=1: gemma's gate/up fuse into one GEMM; the packed
+    GeGLU epilogue (torch fallback on CPU; csrc geglu_packed on GPU)
+    must equal the eager projections."""
+    from runbooks_amd.models.transformer import fuse_for_inference
+
+    m = build_model("tiny-gemma", dtype=torch.float32, seed=8)
+    tokens = torch.randint(0, 256, (2, 9))
+    with torch.no_grad():
+        ref = m(tokens)
+    monkeypatch.setenv("RB_FUSED_GEGLU", "1")
+    fuse_for_inference(m)
+    assert getattr(m.blocks[0].mlp, "_gateup_w", None) is not None
+    with torch.no_grad():
+        got = m(tokens)
+    assert torch.allclose(got, ref, atol=1e-6)
