@@ -25,3 +25,11 @@ RB_TUNABLEOP_FILE=gpurun_out/tunableop_train.csv \
 timeout 300 python bench.py --steps 10 --warmup 3
 timeout 300 python bench.py --mode serve --steps 60 --warmup 5
 timeout 300 python bench.py --mode serve --steps 60 --warmup 5 --serve-8bit
+
+# 6. Late-round-1 gated features: validate then flip defaults (~2 min)
+RB_EXPERIMENTAL=1 python -m pytest tests/test_gpu_ops.py -q -k geglu_packed
+# prefix caching on GPU (engine bookkeeping only; kernels see a block table)
+RB_PREFIX_CACHE=1 python -m pytest tests/test_gpu_engine.py -q -m gpu
+# then: flip RB_PREFIX_CACHE default in serve/engine.py, enable
+# RB_FUSED_GEGLU, and add seq_start to attention_decode.hip for
+# strict-W sliding windows (see NOTES-ROUND2.md).
