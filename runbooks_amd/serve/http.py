@@ -283,10 +283,15 @@ def build_app(engine: Engine, tokenizer=None,
         for t in toks:
             offs.append(pos)
             pos += len(t)
+        def _tok_str(i):
+            t = tok.decode([i])
+            # undecodable bytes all render as U+FFFD and would collide
+            # as dict keys; use an explicit byte form instead
+            return t if t and "\ufffd" not in t else f"bytes:0x{i:x}"
+
         top = None
         if data and "top" in data[0]:
-            top = [{tok.decode([i]): lp for i, lp in d["top"]}
-                   for d in data]
+            top = [{_tok_str(i): lp for i, lp in d["top"]} for d in data]
         return {"tokens": toks,
                 "token_logprobs": [d["logprob"] for d in data],
                 "top_logprobs": top,
