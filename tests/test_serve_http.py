@@ -246,3 +246,48 @@ def test_engine_stats_in_metrics(client):
     r = client.get("/metrics")
     assert 'rb_engine_stat{stat="prefills"}' in r.text
     assert 'rb_engine_stat{stat="decode_tokens"}' in r.text
+
+
+def test_tokenizer_loading_real_files(tmp_path):
+    """load_tokenizer picks tokenizer.json (tokenizers lib) or
+    tokenizer.model (sentencepiece) from a model dir — the formats
+    shipped in real HF checkpoints — and both round-trip text."""
+    from runbooks_amd.serve.tokenizer import (ByteTokenizer, HFTokenizer,
+                                              SPTokenizer, load_tokenizer)
+
+    # tokenizers-lib json
+    from tokenizers import Tokenizer
+    from tokenizers.models import BPE
+    from tokenizers.pre_tokenizers import Whitespace
+    from tokenizers.trainers import BpeTrainer
+    t = Tokenizer(BPE(unk_token="[UNK]"))
+    t.pre_tokenizer = Whitespace()
+    t.train_from_iterator(["hello world", "hello there world"] * 8,
+                          BpeTrainer(special_tokens=["[UNK]"],
+                                     vocab_size=64))
+    d1 = tmp_path / "hf"
+    d1.mkdir()
+    t.save(str(d1 / "tokenizer.json"))
+    tok = load_tokenizer(d1)
+    assert isinstance(tok, HFTokenizer)
+    ids = tok.encode("hello world")
+    assert ids and "hello" in tok.decode(ids)
+
+    # sentencepiece model
+    import sentencepiece as spm
+    corpus = tmp_path / "corpus.txt"
+    corpus.write_text("hello world\n" * 64 + "the quick brown fox\n" * 64)
+    spm.SentencePieceTrainer.train(
+        input=str(corpus), model_prefix=str(tmp_path / "sp"),
+        vocab_size=26)
+    d2 = tmp_path / "sp_dir"
+    d2.mkdir()
+    (d2 / "tokenizer.model").write_bytes(
+        (tmp_path / "sp.model").read_bytes())
+    tok2 = load_tokenizer(d2)
+    assert isinstance(tok2, SPTokenizer)
+    ids2 = tok2.encode("hello world")
+    assert ids2 and "hello" in tok2.decode(ids2)
+
+    assert isinstance(load_tokenizer(None), ByteTokenizer)
+    assert isinstance(load_tokenizer(tmp_path), ByteTokenizer)  # no files
