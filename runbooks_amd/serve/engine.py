@@ -72,6 +72,7 @@ class Request:
     temperature: float = 0.0
     top_p: float = 1.0
     logprobs: int | None = None     # record top-k logprobs per token
+    seed: int | None = None         # per-request sampling seed override
     stop_ids: tuple[int, ...] = ()
     # state
     output_ids: list[int] = field(default_factory=list)
@@ -165,9 +166,10 @@ class Engine:
     # -- request API ------------------------------------------------------------
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 64,
                temperature: float = 0.0, top_p: float = 1.0,
-               logprobs: int | None = None) -> Request:
+               logprobs: int | None = None,
+               seed: int | None = None) -> Request:
         req = Request(self._next_id, list(prompt_ids), max_new_tokens,
-                      temperature, top_p, logprobs)
+                      temperature, top_p, logprobs, seed)
         self._next_id += 1
         self.waiting.append(req)
         return req
@@ -264,8 +266,12 @@ class Engine:
             from .tp_worker import broadcast_prefill
             broadcast_prefill(tokens, positions, slots)
         logits = self.model.prefill(tokens, positions, self.caches, slots)
+        # a per-request seed makes sampling reproducible across runs and
+        # independent of the request_id the scheduler happened to assign
+        base = (req.seed * 2654435761 if req.seed is not None
+                else self.seed + req.request_id * 65537)
         tok = ops.sample_tokens(logits, req.temperature, top_p=req.top_p,
-                                seed=self.seed + req.request_id * 65537 + S)
+                                seed=base + S)
         t = int(tok[0])
         if req.logprobs is not None:
             self._record_logprobs(req, logits[0], t)
@@ -318,17 +324,18 @@ class Engine:
     def _sample_batch(self, logits, reqs: list[Request]) -> list[int]:
         seed = self.seed + 1_000_003 * reqs[0].seq_len
         params = {(r.temperature, r.top_p) for r in reqs}
-        if len(params) == 1:
+        if len(params) == 1 and all(r.seed is None for r in reqs):
             t, p = params.pop()
             toks = ops.sample_tokens(logits, t, top_p=p, seed=seed)
             out = [int(x) for x in toks]
         else:
-            # heterogeneous sampling params: sample row-by-row
+            # heterogeneous sampling params / custom seeds: row-by-row
             out = []
             for i, r in enumerate(reqs):
+                rs = (r.seed * 2654435761 + r.seq_len
+                      if r.seed is not None else seed + r.request_id)
                 tok = ops.sample_tokens(logits[i:i + 1], r.temperature,
-                                        top_p=r.top_p,
-                                        seed=seed + r.request_id)
+                                        top_p=r.top_p, seed=rs)
                 out.append(int(tok[0]))
         for i, r in enumerate(reqs):
             if r.logprobs is not None:

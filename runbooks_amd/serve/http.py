@@ -58,6 +58,7 @@ class CompletionRequest(BaseModel):
     n: int = 1              # choices per prompt (non-stream)
     echo: bool = False      # prepend the prompt to each choice
     logprobs: int | None = None  # top-k logprobs per generated token
+    seed: int | None = None      # reproducible sampling
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -98,12 +99,14 @@ class EngineLoop:
         self._thread.start()
 
     def submit(self, prompt_ids, max_new_tokens, temperature,
-               top_p=1.0, logprobs=None) -> tuple[queue.Queue, "object"]:
+               top_p=1.0, logprobs=None,
+               seed=None) -> tuple[queue.Queue, "object"]:
         """Returns (queue yielding token_id | None, engine Request)."""
         q: queue.Queue = queue.Queue()
         with self._lock:
             req = self.engine.submit(prompt_ids, max_new_tokens, temperature,
-                                     top_p=top_p, logprobs=logprobs)
+                                     top_p=top_p, logprobs=logprobs,
+                                     seed=seed)
             self._watchers[req.request_id] = q
             req._watch_sent = 0
         self._wake.set()
@@ -313,8 +316,10 @@ def build_app(engine: Engine, tokenizer=None,
         # per-request sampling noise is row-independent (ops/sampling.py)
         n = max(1, min(req.n, 16))
         subs = [loop.submit(list(ids), req.max_tokens, req.temperature,
-                            top_p=req.top_p, logprobs=req.logprobs)
-                for _ in range(n)]
+                            top_p=req.top_p, logprobs=req.logprobs,
+                            seed=(None if req.seed is None
+                                  else req.seed + i))
+                for i in range(n)]
         q, ereq = subs[0]
 
         if req.stream:

@@ -291,3 +291,16 @@ def test_tokenizer_loading_real_files(tmp_path):
 
     assert isinstance(load_tokenizer(None), ByteTokenizer)
     assert isinstance(load_tokenizer(tmp_path), ByteTokenizer)  # no files
+
+
+def test_seed_reproducible_sampling(client):
+    """Same prompt + same seed + temperature>0 gives identical text
+    across separate requests (request_id no longer enters the noise)."""
+    body = {"prompt": "seed", "max_tokens": 8, "temperature": 1.0,
+            "seed": 42}
+    r1 = client.post("/v1/completions", json=body).json()
+    r2 = client.post("/v1/completions", json=body).json()
+    assert r1["choices"][0]["text"] == r2["choices"][0]["text"]
+    body2 = dict(body, seed=43)
+    r3 = client.post("/v1/completions", json=body2).json()
+    assert r3["choices"][0]["text"] != r1["choices"][0]["text"]
