@@ -56,6 +56,7 @@ class CompletionRequest(BaseModel):
     temperature: float = 0.0
     top_p: float = 1.0
     n: int = 1              # choices per prompt (non-stream)
+    best_of: int | None = None  # sample best_of, keep the n most likely
     echo: bool = False      # prepend the prompt to each choice
     logprobs: int | None = None  # top-k logprobs per generated token
     seed: int | None = None      # reproducible sampling
@@ -312,14 +313,19 @@ def build_app(engine: Engine, tokenizer=None,
         if req.stream and req.n > 1:
             return JSONResponse(status_code=400, content={"error": {
                 "message": "n > 1 is not supported with stream=true"}})
-        # all n choices submitted up front so they decode as one batch;
-        # per-request sampling noise is row-independent (ops/sampling.py)
+        # all candidates submitted up front so they decode as one batch;
+        # per-request sampling noise is row-independent (ops/sampling.py).
+        # best_of > n: sample best_of candidates with per-token logprobs
+        # and keep the n with the highest mean logprob (OpenAI semantics).
         n = max(1, min(req.n, 16))
+        n_sample = max(n, min(req.best_of or n, 16))
+        want_lp = req.logprobs if req.logprobs is not None else (
+            0 if n_sample > n else None)
         subs = [loop.submit(list(ids), req.max_tokens, req.temperature,
-                            top_p=req.top_p, logprobs=req.logprobs,
+                            top_p=req.top_p, logprobs=want_lp,
                             seed=(None if req.seed is None
                                   else req.seed + i))
-                for i in range(n)]
+                for i in range(n_sample)]
         q, ereq = subs[0]
 
         if req.stream:
@@ -338,16 +344,26 @@ def build_app(engine: Engine, tokenizer=None,
                 yield "data: [DONE]\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
-        choices, n_out = [], 0
-        for i, (cq, creq_) in enumerate(subs):
+        collected = []
+        n_out = 0
+        for cq, creq_ in subs:
             text, finish, out_ids = _collect(cq, creq_.request_id, stops)
             if req.echo:
                 text = req.prompt + text
             lp = (_fmt_logprobs(creq_, out_ids)
                   if req.logprobs is not None else None)
-            choices.append({"text": text, "index": i,
-                            "logprobs": lp, "finish_reason": finish})
+            mean_lp = (sum(d["logprob"] for d in
+                           creq_.logprob_data[:len(out_ids)])
+                       / max(1, len(out_ids))
+                       if creq_.logprob_data else 0.0)
+            collected.append((mean_lp, text, finish, lp, len(out_ids)))
             n_out += len(out_ids)
+        if n_sample > n:
+            collected.sort(key=lambda c: c[0], reverse=True)
+            collected = collected[:n]
+        choices = [{"text": c[1], "index": i, "logprobs": c[3],
+                    "finish_reason": c[2]}
+                   for i, c in enumerate(collected)]
         if m_reqs is not None:
             m_reqs.inc()
             m_tokens.inc(n_out)

@@ -304,3 +304,16 @@ def test_seed_reproducible_sampling(client):
     body2 = dict(body, seed=43)
     r3 = client.post("/v1/completions", json=body2).json()
     assert r3["choices"][0]["text"] != r1["choices"][0]["text"]
+
+
+def test_best_of_picks_most_likely(client):
+    """best_of=4, n=1: four candidates decode as one batch; the returned
+    choice is the one with the highest mean token logprob."""
+    r = client.post("/v1/completions",
+                    json={"prompt": "bo", "max_tokens": 5, "n": 1,
+                          "best_of": 4, "temperature": 1.0, "seed": 7})
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["choices"]) == 1
+    # all 4 candidates' tokens are billed
+    assert body["usage"]["completion_tokens"] == 20
