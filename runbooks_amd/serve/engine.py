@@ -73,6 +73,8 @@ class Request:
     top_p: float = 1.0
     logprobs: int | None = None     # record top-k logprobs per token
     seed: int | None = None         # per-request sampling seed override
+    presence_penalty: float = 0.0   # subtract once per seen token
+    frequency_penalty: float = 0.0  # subtract per occurrence
     stop_ids: tuple[int, ...] = ()
     # state
     output_ids: list[int] = field(default_factory=list)
@@ -166,13 +168,33 @@ class Engine:
     # -- request API ------------------------------------------------------------
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 64,
                temperature: float = 0.0, top_p: float = 1.0,
-               logprobs: int | None = None,
-               seed: int | None = None) -> Request:
+               logprobs: int | None = None, seed: int | None = None,
+               presence_penalty: float = 0.0,
+               frequency_penalty: float = 0.0) -> Request:
         req = Request(self._next_id, list(prompt_ids), max_new_tokens,
-                      temperature, top_p, logprobs, seed)
+                      temperature, top_p, logprobs, seed,
+                      presence_penalty, frequency_penalty)
         self._next_id += 1
         self.waiting.append(req)
         return req
+
+    @staticmethod
+    def _apply_penalties(row, r: Request):
+        """OpenAI presence/frequency penalties over the text so far
+        (prompt + generated). Torch ops on the logits row, outside the
+        decode hipGraph — zero cost for requests that don't ask."""
+        if not (r.presence_penalty or r.frequency_penalty):
+            return row
+        from collections import Counter
+        cnt = Counter(r.prompt_ids)
+        cnt.update(r.output_ids)
+        ids = torch.tensor(list(cnt.keys()), dtype=torch.long,
+                           device=row.device)
+        c = torch.tensor([float(v) for v in cnt.values()],
+                         device=row.device)
+        row = row.clone().float()
+        row[ids] -= r.frequency_penalty * c + r.presence_penalty
+        return row
 
     @staticmethod
     def _record_logprobs(req: Request, logits_row, tok_id: int) -> None:
@@ -270,11 +292,12 @@ class Engine:
         # independent of the request_id the scheduler happened to assign
         base = (req.seed * 2654435761 if req.seed is not None
                 else self.seed + req.request_id * 65537)
-        tok = ops.sample_tokens(logits, req.temperature, top_p=req.top_p,
-                                seed=base + S)
+        logits0 = self._apply_penalties(logits[0], req)
+        tok = ops.sample_tokens(logits0.unsqueeze(0), req.temperature,
+                                top_p=req.top_p, seed=base + S)
         t = int(tok[0])
         if req.logprobs is not None:
-            self._record_logprobs(req, logits[0], t)
+            self._record_logprobs(req, logits0, t)
         return t
 
     def _decode_batch(self, reqs: list[Request]) -> list[int]:
@@ -324,7 +347,9 @@ class Engine:
     def _sample_batch(self, logits, reqs: list[Request]) -> list[int]:
         seed = self.seed + 1_000_003 * reqs[0].seq_len
         params = {(r.temperature, r.top_p) for r in reqs}
-        if len(params) == 1 and all(r.seed is None for r in reqs):
+        plain = all(r.seed is None and not r.presence_penalty and
+                    not r.frequency_penalty for r in reqs)
+        if len(params) == 1 and plain:
             t, p = params.pop()
             toks = ops.sample_tokens(logits, t, top_p=p, seed=seed)
             out = [int(x) for x in toks]
@@ -334,7 +359,8 @@ class Engine:
             for i, r in enumerate(reqs):
                 rs = (r.seed * 2654435761 + r.seq_len
                       if r.seed is not None else seed + r.request_id)
-                tok = ops.sample_tokens(logits[i:i + 1], r.temperature,
+                row = self._apply_penalties(logits[i], r)
+                tok = ops.sample_tokens(row.unsqueeze(0), r.temperature,
                                         top_p=r.top_p, seed=rs)
                 out.append(int(tok[0]))
         for i, r in enumerate(reqs):

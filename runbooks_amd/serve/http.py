@@ -60,6 +60,8 @@ class CompletionRequest(BaseModel):
     echo: bool = False      # prepend the prompt to each choice
     logprobs: int | None = None  # top-k logprobs per generated token
     seed: int | None = None      # reproducible sampling
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -100,14 +102,17 @@ class EngineLoop:
         self._thread.start()
 
     def submit(self, prompt_ids, max_new_tokens, temperature,
-               top_p=1.0, logprobs=None,
-               seed=None) -> tuple[queue.Queue, "object"]:
+               top_p=1.0, logprobs=None, seed=None,
+               presence_penalty=0.0,
+               frequency_penalty=0.0) -> tuple[queue.Queue, "object"]:
         """Returns (queue yielding token_id | None, engine Request)."""
         q: queue.Queue = queue.Queue()
         with self._lock:
             req = self.engine.submit(prompt_ids, max_new_tokens, temperature,
                                      top_p=top_p, logprobs=logprobs,
-                                     seed=seed)
+                                     seed=seed,
+                                     presence_penalty=presence_penalty,
+                                     frequency_penalty=frequency_penalty)
             self._watchers[req.request_id] = q
             req._watch_sent = 0
         self._wake.set()
@@ -324,7 +329,9 @@ def build_app(engine: Engine, tokenizer=None,
         subs = [loop.submit(list(ids), req.max_tokens, req.temperature,
                             top_p=req.top_p, logprobs=want_lp,
                             seed=(None if req.seed is None
-                                  else req.seed + i))
+                                  else req.seed + i),
+                            presence_penalty=req.presence_penalty,
+                            frequency_penalty=req.frequency_penalty)
                 for i in range(n_sample)]
         q, ereq = subs[0]
 

@@ -509,3 +509,19 @@ def test_geglu_fused_path_matches_eager(monkeypatch):
     with torch.no_grad():
         got = m(tokens)
     assert torch.allclose(got, ref, atol=1e-6)
+
+
+def test_frequency_penalty_prevents_repetition():
+    """A huge frequency penalty (greedy) forces every generated token to
+    be distinct from all prior text; without it the tiny random model
+    repeats (greedy cycles)."""
+    m = build_model("tiny-llama", dtype=torch.float32, seed=1)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    plain = eng.generate([5, 5, 5], max_new_tokens=20)
+    assert len(set(plain)) < 20, "expected the unpenalized run to repeat"
+    r = eng.submit([5, 5, 5], max_new_tokens=20, frequency_penalty=100.0)
+    while not r.finished:
+        eng.step()
+    out = r.output_ids
+    assert len(set(out)) == 20, out
+    assert 5 not in out  # prompt tokens penalized too (OpenAI semantics)
