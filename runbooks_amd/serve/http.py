@@ -83,6 +83,9 @@ class ChatRequest(BaseModel):
     temperature: float = 0.0
     top_p: float = 1.0
     n: int = 1
+    seed: int | None = None
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -410,7 +413,9 @@ def build_app(engine: Engine, tokenizer=None,
             ids = _encode(prompt, max_tokens)
             stops = CompletionRequest(stop=req.stop).stop_list()
             q, ereq = loop.submit(ids, max_tokens, req.temperature,
-                                  top_p=req.top_p)
+                                  top_p=req.top_p, seed=req.seed,
+                                  presence_penalty=req.presence_penalty,
+                                  frequency_penalty=req.frequency_penalty)
 
             def gen():
                 def chunk(delta):
@@ -428,6 +433,8 @@ def build_app(engine: Engine, tokenizer=None,
         creq = CompletionRequest(
             prompt=prompt, max_tokens=req.max_tokens,
             temperature=req.temperature, top_p=req.top_p, n=req.n,
+            seed=req.seed, presence_penalty=req.presence_penalty,
+            frequency_penalty=req.frequency_penalty,
             stop=req.stop, model=req.model, stream=False)
         resp = completions(creq)
         if resp.status_code != 200:
