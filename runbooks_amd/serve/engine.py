@@ -32,6 +32,7 @@ import torch
 from .. import ops
 from ..models import Transformer, build_model
 from ..parallel import comm
+from ..utils.trace import get_tracer
 
 
 class BlockAllocator:
@@ -439,8 +440,14 @@ class Engine:
                 raise RuntimeError(
                     f"prompt of {len(r.prompt_ids)} tokens cannot fit the KV "
                     f"cache ({self.allocator.num_blocks} blocks)")
+        tr = get_tracer()
         if req is not None:
-            first = self._prefill(req)
+            if tr:
+                with tr.span("prefill", tokens=len(req.prompt_ids),
+                             request=req.request_id):
+                    first = self._prefill(req)
+            else:
+                first = self._prefill(req)
             self.stats["prefills"] += 1
             if self.prefix_cache_enabled:
                 self._pc_insert(req)
@@ -463,7 +470,11 @@ class Engine:
                         self._preempt(r)
                         break
             if self.running:
-                toks = self._decode_batch(self.running)
+                if tr:
+                    with tr.span("decode", batch=len(self.running)):
+                        toks = self._decode_batch(self.running)
+                else:
+                    toks = self._decode_batch(self.running)
                 self.stats["decode_tokens"] += len(toks)
                 for r, t in zip(self.running, toks):
                     r.output_ids.append(t)

@@ -174,6 +174,10 @@ class EngineLoop:
         self._stop = True
         self._wake.set()
         self._thread.join(timeout=2)
+        from ..utils.trace import dump_global
+        p = dump_global()
+        if p:
+            logger.info("trace written to %s", p)
 
 
 def build_app(engine: Engine, tokenizer=None,

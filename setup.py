@@ -25,7 +25,8 @@ setup(
     packages=["runbooks_amd"] + [
         f"runbooks_amd.{p}" for p in
         ("api", "cli", "client", "cloud", "controller", "k8s", "models",
-         "ops", "parallel", "sci", "serve", "train", "workloads")],
+         "ops", "parallel", "sci", "serve", "train", "tui", "utils",
+         "workloads")],
     entry_points={"console_scripts": [
         # parity: reference cmd/ — sub CLI, controllermanager, sci servers,
         # nbwatch (goreleaser targets .goreleaser.yaml:8-37)

@@ -317,3 +317,24 @@ def test_best_of_picks_most_likely(client):
     assert len(body["choices"]) == 1
     # all 4 candidates' tokens are billed
     assert body["usage"]["completion_tokens"] == 20
+
+
+def test_rb_trace_engine_spans(tmp_path, monkeypatch):
+    """RB_TRACE=path: the engine emits prefill/decode spans; shutdown
+    writes a chrome-trace JSON."""
+    import json as _json
+
+    import runbooks_amd.utils.trace as trace_mod
+    out = tmp_path / "trace.json"
+    monkeypatch.setenv("RB_TRACE", str(out))
+    monkeypatch.setattr(trace_mod, "_GLOBAL", None)
+    eng = Engine("tiny-llama", device="cpu", dtype=torch.float32,
+                 kv_blocks=64, seed=2)
+    app = build_app(eng, model_name="tiny-llama")
+    with TestClient(app) as c:
+        c.post("/v1/completions", json={"prompt": "tr", "max_tokens": 3})
+    data = _json.loads(out.read_text())
+    names = {e["name"] for e in data["traceEvents"]}
+    assert "prefill" in names and "decode" in names
+    assert all("ts" in e and "dur" in e for e in data["traceEvents"]
+               if e["ph"] == "X")
