@@ -351,11 +351,16 @@ def build_app(engine: Engine, tokenizer=None,
                                       "logprobs": None,
                                       "finish_reason": None}]}
                     return f"data: {json.dumps(c)}\n\n"
-                if req.echo and req.prompt:
-                    yield chunk(req.prompt)
-                for piece in _pieces(q, ereq.request_id, stops):
-                    yield chunk(piece)
-                yield "data: [DONE]\n\n"
+                try:
+                    if req.echo and req.prompt:
+                        yield chunk(req.prompt)
+                    for piece in _pieces(q, ereq.request_id, stops):
+                        yield chunk(piece)
+                    yield "data: [DONE]\n\n"
+                finally:
+                    # client disconnect abandons the generator mid-stream:
+                    # stop generating for it (no-op after normal finish)
+                    loop.cancel(ereq.request_id)
             return StreamingResponse(gen(), media_type="text/event-stream")
 
         collected = []
@@ -428,10 +433,13 @@ def build_app(engine: Engine, tokenizer=None,
                          "choices": [{"index": 0, "delta": delta,
                                       "finish_reason": None}]}
                     return f"data: {json.dumps(c)}\n\n"
-                yield chunk({"role": "assistant"})
-                for piece in _pieces(q, ereq.request_id, stops):
-                    yield chunk({"content": piece})
-                yield "data: [DONE]\n\n"
+                try:
+                    yield chunk({"role": "assistant"})
+                    for piece in _pieces(q, ereq.request_id, stops):
+                        yield chunk({"content": piece})
+                    yield "data: [DONE]\n\n"
+                finally:
+                    loop.cancel(ereq.request_id)
             return StreamingResponse(gen(), media_type="text/event-stream")
 
         creq = CompletionRequest(

@@ -155,7 +155,13 @@ class Trainer:
                 except StopIteration:
                     it = iter(loader)
                     batch = next(it)
-            loss = self.train_step(batch)
+            from ..utils.trace import get_tracer
+            tr = get_tracer()
+            if tr:
+                with tr.span("train_step", step=self.step_num):
+                    loss = self.train_step(batch)
+            else:
+                loss = self.train_step(batch)
             if cfg.eval_steps and eval_dataset is not None and \
                     self.step_num % cfg.eval_steps == 0:
                 ev = self.evaluate(eval_dataset)
@@ -171,6 +177,8 @@ class Trainer:
                       f"samples/sec {sps:.2f}", flush=True)
         if cfg.save_steps:
             self.save()
+        from ..utils.trace import dump_global
+        dump_global()
 
     def save(self):
         if comm.rank() != 0:

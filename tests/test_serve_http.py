@@ -338,3 +338,25 @@ def test_rb_trace_engine_spans(tmp_path, monkeypatch):
     assert "prefill" in names and "decode" in names
     assert all("ts" in e and "dur" in e for e in data["traceEvents"]
                if e["ph"] == "X")
+
+
+def test_stream_disconnect_cancels_request():
+    """Abandoning an SSE stream mid-generation cancels the engine
+    request instead of generating to max_tokens."""
+    eng = Engine("tiny-llama", device="cpu", dtype=torch.float32,
+                 kv_blocks=64, seed=3)
+    app = build_app(eng, model_name="tiny-llama")
+    with TestClient(app) as c:
+        with c.stream("POST", "/v1/completions",
+                      json={"prompt": "dc", "max_tokens": 10_000,
+                            "stream": True}) as r:
+            it = r.iter_lines()
+            next(it)  # read one chunk, then drop the connection
+        # the request must wind down long before 10k tokens
+        import time as _t
+        loop = app.state.engine_loop
+        for _ in range(400):
+            if not loop.engine.has_work():
+                break
+            _t.sleep(0.01)
+        assert not loop.engine.has_work(), "request kept generating"
