@@ -214,6 +214,48 @@ def _tp_engine_sliding_window(rank, world):
         worker_loop(eng)
 
 
+def _dp_save_resume(rank, world):
+    """Checkpoint round-trip under DP: rank 0 writes, resume() loads on
+    rank 0 and BROADCASTS to the other ranks; weights and step count
+    must agree everywhere afterwards."""
+    import shutil
+    import tempfile
+    from pathlib import Path
+
+    from runbooks_amd.train import TrainConfig, Trainer
+
+    out = Path(tempfile.gettempdir()) / "rb-dp-resume-test"
+    if rank == 0:
+        shutil.rmtree(out, ignore_errors=True)
+        out.mkdir(parents=True, exist_ok=True)
+    dist.barrier()
+    cfg = TrainConfig(model="tiny-llama", seq_len=16, micro_batch=2,
+                      num_train_steps=3, save_steps=3, dtype="float32",
+                      lora_r=4, output_dir=str(out))
+    t1 = Trainer(cfg, device="cpu")
+    t1.fit(log_every=100)
+    dist.barrier()
+
+    t2 = Trainer(cfg, device="cpu")
+    assert t2.resume(), "no checkpoint found"
+    assert t2.step_num == 3
+    # every rank must hold identical resumed weights
+    for n, p in t2.model.named_parameters():
+        if not p.requires_grad:
+            continue
+        ref = p.detach().clone()
+        dist.broadcast(ref, src=0)
+        assert torch.equal(ref, p.detach()), f"rank {rank} diverged on {n}"
+    # and match what rank 0 trained to
+    for (n1, p1), (n2, p2) in zip(t1.model.named_parameters(),
+                                  t2.model.named_parameters()):
+        if p1.requires_grad:
+            assert torch.allclose(p1.detach(), p2.detach(), atol=1e-7), n1
+    dist.barrier()
+    if rank == 0:
+        shutil.rmtree(out, ignore_errors=True)
+
+
 # --- test entries -----------------------------------------------------------
 
 def test_dp_gradient_allreduce():
@@ -310,3 +352,7 @@ def test_tp_falcon_mqa():
 
 def test_tp_engine_sliding_window():
     _run_dist(_tp_engine_sliding_window, port=PORT + 11)
+
+
+def test_dp_save_resume():
+    _run_dist(_dp_save_resume, port=PORT + 12)
