@@ -50,6 +50,36 @@ class _ReduceFn(torch.autograd.Function):
         return dy, None
 
 
+class _GatherFn(torch.autograd.Function):
+    """All-gather along the last dim fwd; slice own shard bwd. Used to
+    assemble full-vocab logits from a vocab-parallel lm_head."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        if not comm.is_dist():
+            return x
+        ctx.group = group
+        ws = dist.get_world_size(group)
+        ctx.rank = dist.get_rank(group)
+        ctx.shard = x.shape[-1]
+        x = x.contiguous()
+        parts = [torch.empty_like(x) for _ in range(ws)]
+        dist.all_gather(parts, x, group=group)
+        return torch.cat(parts, dim=-1)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if not comm.is_dist():
+            return dy, None
+        s = ctx.shard
+        lo = ctx.rank * s
+        return dy[..., lo:lo + s].contiguous(), None
+
+
+def gather_from_tp(x, group=None):
+    return _GatherFn.apply(x, group)
+
+
 def copy_to_tp(x, group=None):
     return _AllReduceFn.apply(x, group)
 
