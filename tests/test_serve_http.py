@@ -360,3 +360,33 @@ def test_stream_disconnect_cancels_request():
                 break
             _t.sleep(0.01)
         assert not loop.engine.has_work(), "request kept generating"
+
+
+def test_top_p_filter_property():
+    """Property check over random distributions: the kept set is exactly
+    the smallest descending-probability prefix reaching top_p, and
+    renormalized sampling mass is preserved for kept entries."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from runbooks_amd.ops.sampling import top_p_filter
+
+    @settings(max_examples=50, deadline=None)
+    @given(st.lists(st.floats(-8, 8), min_size=2, max_size=32),
+           st.floats(0.05, 0.95))
+    def check(vals, p):
+        logits = torch.tensor([vals], dtype=torch.float32)
+        probs = torch.softmax(logits, dim=-1)[0]
+        out = torch.isfinite(top_p_filter(logits, p)[0])
+        order = torch.argsort(probs, descending=True, stable=True)
+        cum = torch.cumsum(probs[order], 0)
+        k = int((cum >= p - 1e-6).nonzero()[0]) + 1
+        expect = torch.zeros_like(out)
+        expect[order[:k]] = True
+        # ties at the cut can include either member; sizes must agree
+        # and every strictly-heavier token must be kept
+        assert out.sum() == expect.sum()
+        strict = probs > probs[order[k - 1]] + 1e-9
+        assert bool(out[strict].all())
+
+    check()
