@@ -525,3 +525,29 @@ def test_frequency_penalty_prevents_repetition():
     out = r.output_ids
     assert len(set(out)) == 20, out
     assert 5 not in out  # prompt tokens penalized too (OpenAI semantics)
+
+
+def test_rope_rotation_invariants():
+    """RoPE reference invariants on CPU: per-pair norms are preserved
+    (it's a rotation) and q·k depends only on the relative position —
+    the property windowed/suffix attention exactness relies on."""
+    from runbooks_amd import ops
+    from runbooks_amd.models.transformer import Transformer  # noqa: F401
+
+    m = build_model("tiny-llama", dtype=torch.float32)
+    cos, sin = m.rope_cos, m.rope_sin
+    torch.manual_seed(0)
+    q = torch.randn(1, 4, 16)
+    k = torch.randn(1, 4, 16)
+
+    def rot(x, p):
+        return ops.rope(x, cos, sin, torch.tensor([p], dtype=torch.int32))
+
+    assert torch.allclose(rot(q, 7).norm(), q.norm(), atol=1e-5)
+    # relative property: <R(p1)q, R(p2)k> == <R(p1+d)q, R(p2+d)k>
+    def score(p1, p2):
+        return torch.einsum("bhd,bhd->bh", rot(q, p1), rot(k, p2))
+
+    assert torch.allclose(score(3, 1), score(23, 21), atol=1e-4)
+    assert torch.allclose(score(10, 10), score(50, 50), atol=1e-4)
+    assert not torch.allclose(score(3, 1), score(3, 2), atol=1e-3)
