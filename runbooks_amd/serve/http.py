@@ -322,9 +322,10 @@ def build_app(engine: Engine, tokenizer=None,
         cid = f"cmpl-{uuid.uuid4().hex[:12]}"
         t0 = int(time.time())
         stops = req.stop_list()
-        if req.stream and req.n > 1:
+        if req.stream and (req.n > 1 or (req.best_of or 1) > 1):
             return JSONResponse(status_code=400, content={"error": {
-                "message": "n > 1 is not supported with stream=true"}})
+                "message": "n > 1 / best_of > 1 are not supported with "
+                           "stream=true"}})
         # all candidates submitted up front so they decode as one batch;
         # per-request sampling noise is row-independent (ops/sampling.py).
         # best_of > n: sample best_of candidates with per-token logprobs
