@@ -204,6 +204,7 @@ def test_staging_roundtrip():
     (256, 256, 128),
     (512, 512, 128),   # multi-tile + XCD remap (nwg=4, nwg%8 != 0)
     (256, 512, 512),   # 2 tiles -> host picks split-K; slab + combine path
+    (256, 256, 512),   # 1 tile  -> ksplit=4 (depth-limited by K%(2*BK))
 ])
 def test_gemm_dataflow_matches_matmul(M, N, K):
     rng = np.random.default_rng(M + K)
