@@ -6,7 +6,8 @@ layers (parallel/tp.py) line up. Before each model invocation rank 0
 broadcasts a small header (op, batch, seq, table width) and the input
 tensors over the same process group; workers replay the call against
 their own weight shards and KV caches and discard the logits (the
-lm_head is replicated, so every rank materializes them).
+vocab-parallel lm_head all-gather assembles full logits on every rank,
+so the collective schedule stays symmetric).
 
 Ops: 1 = prefill, 2 = decode, 0 = shutdown.
 """
