@@ -453,7 +453,12 @@ class Engine:
                 self._pc_insert(req)
             req.output_ids.append(first)
             self.running.append(req)
-        elif self.running:
+            # the prefill token may already satisfy the request
+            self._sweep_finished(finished)
+        # decode every step — admissions must not starve running
+        # sequences (a freshly prefilled request decodes its second
+        # token in the same step it was admitted)
+        if self.running:
             self._apply_window()
             # ensure every running seq has a block for the next position;
             # when the pool is exhausted, preempt the newest sequence
@@ -478,6 +483,10 @@ class Engine:
                 self.stats["decode_tokens"] += len(toks)
                 for r, t in zip(self.running, toks):
                     r.output_ids.append(t)
+        self._sweep_finished(finished)
+        return finished
+
+    def _sweep_finished(self, finished: list[Request]) -> None:
         for r in list(self.running):
             if (len(r.output_ids) >= r.max_new_tokens
                     or (r.stop_ids and r.output_ids[-1] in r.stop_ids)
@@ -487,7 +496,6 @@ class Engine:
                 r.blocks = []
                 self.running.remove(r)
                 finished.append(r)
-        return finished
 
     def has_work(self) -> bool:
         return bool(self.waiting or self.running)
