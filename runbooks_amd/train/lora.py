@@ -94,9 +94,10 @@ class LoRALinear(nn.Module):
     """
 
     def __init__(self, base: nn.Module, r: int = 16, alpha: int = 32,
-                 dtype=None):
+                 dropout: float = 0.0, dtype=None):
         super().__init__()
         self.base = base
+        self.dropout = nn.Dropout(dropout) if dropout > 0 else None
         w = base.weight
         out_f, in_f = w.shape
         dtype = dtype or w.dtype
@@ -111,6 +112,15 @@ class LoRALinear(nn.Module):
 
     def forward(self, x):
         base = self.base
+        if self.dropout is not None and self.training:
+            # dropout on the adapter input (PEFT semantics) breaks the
+            # single-addmm fusion; let autograd handle the composite
+            y = self.base(x)
+            t = torch.nn.functional.linear(self.dropout(x), self.lora_a)
+            out_f = y.shape[-1]
+            return torch.addmm(y.reshape(-1, out_f),
+                               t.reshape(-1, self.r), self.lora_b.t(),
+                               beta=1.0, alpha=self.scale).view(y.shape)
         if (isinstance(base, nn.Linear) or
                 (getattr(base, "tp", 1) == 1 and base.bias is None)) and \
                 x.dim() == 2 and getattr(base, "bias", None) is None:
@@ -134,7 +144,7 @@ class LoRALinear(nn.Module):
 
 
 def apply_lora(model: nn.Module, r: int = 16, alpha: int = 32,
-               targets=DEFAULT_TARGETS) -> list[str]:
+               dropout: float = 0.0, targets=DEFAULT_TARGETS) -> list[str]:
     """Freeze the model and wrap target linears with LoRA. Returns wrapped
     module names."""
     for p in model.parameters():
@@ -144,7 +154,8 @@ def apply_lora(model: nn.Module, r: int = 16, alpha: int = 32,
         for child_name, child in list(module.named_children()):
             if child_name in targets and isinstance(
                     child, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):
-                setattr(module, child_name, LoRALinear(child, r=r, alpha=alpha))
+                setattr(module, child_name,
+                        LoRALinear(child, r=r, alpha=alpha, dropout=dropout))
                 wrapped.append(f"{name}.{child_name}" if name else child_name)
     if not wrapped:
         raise ValueError("apply_lora: no target modules found")

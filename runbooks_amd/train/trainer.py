@@ -35,6 +35,7 @@ class TrainConfig:
     grad_accum_steps: int = 1
     lora_r: int = 16
     lora_alpha: int = 32
+    lora_dropout: float = 0.0
     full_finetune: bool = False
     grad_checkpointing: bool = False    # recompute block activations
     eval_steps: int = 0                 # 0 = no evaluation
@@ -53,7 +54,8 @@ class Trainer:
         model = build_model(cfg.model, dtype=dtype, tp=1, seed=cfg.seed,
                             device=self.device)
         if not cfg.full_finetune:
-            apply_lora(model, r=cfg.lora_r, alpha=cfg.lora_alpha)
+            apply_lora(model, r=cfg.lora_r, alpha=cfg.lora_alpha,
+                       dropout=cfg.lora_dropout)
         if cfg.grad_checkpointing:
             model.enable_grad_checkpointing()
         self.ddp = DataParallel(model)

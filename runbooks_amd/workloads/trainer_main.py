@@ -84,6 +84,7 @@ def main():
         save_steps=_param("save_steps", 50, int),
         grad_accum_steps=_param("gradient_accumulation_steps", 1, int),
         lora_r=_param("lora_r", 16, int),
+        lora_dropout=_param("lora_dropout", 0.0, float),
         full_finetune=_param("full_finetune", False, bool),
         grad_checkpointing=_param("gradient_checkpointing", False, bool),
         eval_steps=_param("eval_steps", 0, int),

@@ -238,3 +238,28 @@ def test_trainer_evaluate_and_eval_steps(capsys):
            log_every=100)
     out = capsys.readouterr().out
     assert out.count("eval_loss") == 2, out
+
+
+def test_lora_dropout():
+    """dropout>0: stochastic in train mode, exact (no-dropout) in eval;
+    dropout=0 keeps the fused GEMM-only path."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.train import apply_lora
+
+    torch.manual_seed(0)
+    m = build_model("tiny-llama", dtype=torch.float32, seed=2)
+    apply_lora(m, r=4, alpha=8, dropout=0.5)
+    # make adapters non-zero so dropout actually matters
+    for n, p in m.named_parameters():
+        if "lora_b" in n:
+            with torch.no_grad():
+                p.add_(torch.randn_like(p) * 0.1)
+    x = torch.randint(0, 256, (1, 8))
+    m.train()
+    a = m(x)
+    b = m(x)
+    assert not torch.allclose(a, b), "dropout inactive in train mode"
+    m.eval()
+    with torch.no_grad():
+        c, d = m(x), m(x)
+    assert torch.equal(c, d)
