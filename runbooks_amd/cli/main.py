@@ -76,6 +76,8 @@ def main():
               help="directory with Dockerfile to upload as build context")
 @click.option("--wait/--no-wait", default=False)
 def apply(filename, namespace, build_path, wait):
+    """Server-side apply substratus manifests (with optional -b upload
+    build and readiness wait) — the reference's `sub apply [-b]`."""
     kube = _kube()
     objs = find_manifests(filename)
     if not objs:
@@ -196,6 +198,8 @@ def serve(path, namespace):
 @click.option("-o", "--output", default="table",
               type=click.Choice(["table", "json"]))
 def get(kind, name, namespace, output):
+    """List substratus objects (`sub get models|datasets|servers|
+    notebooks [name]`), table or -o yaml."""
     kube = _kube()
     kind = {k.lower(): k for k in KINDS}.get(kind.rstrip("s").lower(), kind)
     if name:
@@ -311,6 +315,7 @@ def resume(name, namespace):
 @click.argument("name")
 @click.option("-n", "--namespace", default="default")
 def delete(kind, name, namespace):
+    """Delete a substratus object (`sub delete kind/name`)."""
     kube = _kube()
     kind = {k.lower(): k for k in KINDS}.get(kind.rstrip("s").lower(), kind)
     if kube.delete("substratus.ai/v1", kind, namespace, name):
