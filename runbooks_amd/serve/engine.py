@@ -171,10 +171,12 @@ class Engine:
                temperature: float = 0.0, top_p: float = 1.0,
                logprobs: int | None = None, seed: int | None = None,
                presence_penalty: float = 0.0,
-               frequency_penalty: float = 0.0) -> Request:
+               frequency_penalty: float = 0.0,
+               stop_token_ids: tuple[int, ...] = ()) -> Request:
         req = Request(self._next_id, list(prompt_ids), max_new_tokens,
                       temperature, top_p, logprobs, seed,
-                      presence_penalty, frequency_penalty)
+                      presence_penalty, frequency_penalty,
+                      tuple(stop_token_ids))
         self._next_id += 1
         self.waiting.append(req)
         return req

@@ -106,8 +106,8 @@ class EngineLoop:
 
     def submit(self, prompt_ids, max_new_tokens, temperature,
                top_p=1.0, logprobs=None, seed=None,
-               presence_penalty=0.0,
-               frequency_penalty=0.0) -> tuple[queue.Queue, "object"]:
+               presence_penalty=0.0, frequency_penalty=0.0,
+               stop_token_ids=()) -> tuple[queue.Queue, "object"]:
         """Returns (queue yielding token_id | None, engine Request)."""
         q: queue.Queue = queue.Queue()
         with self._lock:
@@ -115,7 +115,8 @@ class EngineLoop:
                                      top_p=top_p, logprobs=logprobs,
                                      seed=seed,
                                      presence_penalty=presence_penalty,
-                                     frequency_penalty=frequency_penalty)
+                                     frequency_penalty=frequency_penalty,
+                                     stop_token_ids=stop_token_ids)
             self._watchers[req.request_id] = q
             req._watch_sent = 0
         self._wake.set()
@@ -193,6 +194,11 @@ def build_app(engine: Engine, tokenizer=None,
 
     app = FastAPI(title="runbooks-amd-server", lifespan=_lifespan)
     tok = tokenizer or load_tokenizer(None)
+    # finish at the tokenizer's EOS (real checkpoints emit it; the byte
+    # fallback has none). The EOS itself is filtered from the text by the
+    # decoded-piece diffing, and the engine marks the request finished.
+    eos = getattr(tok, "eos_id", None)
+    eos_ids = (eos,) if eos is not None else ()
     app.state.engine_loop = loop
 
     # Prometheus metrics (the reference exposes controller metrics behind
@@ -251,6 +257,8 @@ def build_app(engine: Engine, tokenizer=None,
             t = q.get()
             if t is None:
                 return
+            if t in eos_ids:
+                continue  # engine finishes on EOS; don't stream it
             out.append(t)
             text = tok.decode(out)
             trunc = _stop_hit(text, stops) if stops else None
@@ -284,7 +292,12 @@ def build_app(engine: Engine, tokenizer=None,
                     while q.get() is not None:
                         pass
                     break
-        if finish != "stop":
+        if finish != "stop" and out and out[-1] in eos_ids:
+            # engine stopped at the tokenizer's EOS: report "stop" and
+            # keep the EOS itself out of the decoded text
+            finish = "stop"
+            text = tok.decode(out[:-1])
+        elif finish != "stop":
             text = tok.decode(out)
         return text, finish, out
 
@@ -339,7 +352,8 @@ def build_app(engine: Engine, tokenizer=None,
                             seed=(None if req.seed is None
                                   else req.seed + i),
                             presence_penalty=req.presence_penalty,
-                            frequency_penalty=req.frequency_penalty)
+                            frequency_penalty=req.frequency_penalty,
+                            stop_token_ids=eos_ids)
                 for i in range(n_sample)]
         q, ereq = subs[0]
 
@@ -425,7 +439,8 @@ def build_app(engine: Engine, tokenizer=None,
             q, ereq = loop.submit(ids, max_tokens, req.temperature,
                                   top_p=req.top_p, seed=req.seed,
                                   presence_penalty=req.presence_penalty,
-                                  frequency_penalty=req.frequency_penalty)
+                                  frequency_penalty=req.frequency_penalty,
+                                  stop_token_ids=eos_ids)
 
             def gen():
                 def chunk(delta):

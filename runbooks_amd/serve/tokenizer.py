@@ -14,6 +14,7 @@ class ByteTokenizer:
 
     vocab_size = 257
     bos_id = 256
+    eos_id = None
 
     def encode(self, text: str) -> list[int]:
         return [self.bos_id] + list(text.encode("utf-8"))
@@ -26,6 +27,12 @@ class HFTokenizer:
     def __init__(self, path: str):
         from tokenizers import Tokenizer
         self.tok = Tokenizer.from_file(path)
+        # common eos conventions across llama/mistral/qwen/falcon vocabs
+        self.eos_id = next(
+            (i for i in (self.tok.token_to_id(t) for t in
+                         ("</s>", "<|endoftext|>", "<|end_of_text|>",
+                          "<|im_end|>", "<eos>"))
+             if i is not None), None)
 
     @property
     def vocab_size(self):
@@ -42,6 +49,7 @@ class SPTokenizer:
     def __init__(self, path: str):
         import sentencepiece as spm
         self.sp = spm.SentencePieceProcessor(model_file=path)
+        self.eos_id = self.sp.eos_id() if self.sp.eos_id() >= 0 else None
 
     @property
     def vocab_size(self):

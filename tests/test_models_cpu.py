@@ -535,3 +535,16 @@ def test_rope_rotation_invariants():
     assert torch.allclose(score(3, 1), score(23, 21), atol=1e-4)
     assert torch.allclose(score(10, 10), score(50, 50), atol=1e-4)
     assert not torch.allclose(score(3, 1), score(3, 2), atol=1e-3)
+
+
+def test_stop_token_ids_finish_early():
+    """Engine finishes a request the moment it emits a stop token id
+    (the HTTP layer wires the tokenizer EOS here)."""
+    m = build_model("tiny-llama", dtype=torch.float32, seed=3)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    plain = eng.generate([9, 9, 9], max_new_tokens=8)
+    target = plain[2]
+    r = eng.submit([9, 9, 9], max_new_tokens=8, stop_token_ids=(target,))
+    while not r.finished:
+        eng.step()
+    assert r.output_ids == plain[:3], (r.output_ids, plain)

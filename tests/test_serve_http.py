@@ -390,3 +390,30 @@ def test_top_p_filter_property():
         assert bool(out[strict].all())
 
     check()
+
+
+def test_eos_finish_reason():
+    """A tokenizer with an eos_id makes generation finish with
+    finish_reason="stop" and the EOS excluded from the text."""
+    from runbooks_amd.serve.tokenizer import ByteTokenizer
+
+    eng = Engine("tiny-llama", device="cpu", dtype=torch.float32,
+                 kv_blocks=64, seed=4)
+    # same id pipeline the HTTP layer uses (bos folded into the vocab)
+    ids = [i % eng.cfg.vocab_size for i in ByteTokenizer().encode("eo")]
+    probe = eng.generate(ids, max_new_tokens=6)
+
+    class EosTok(ByteTokenizer):
+        eos_id = probe[2]  # greedy path emits this as the 3rd token
+
+    eng2 = Engine(eng.model, device="cpu", dtype=torch.float32,
+                  kv_blocks=64, seed=4)
+    app = build_app(eng2, tokenizer=EosTok(), model_name="t")
+    with TestClient(app) as c:
+        r = c.post("/v1/completions",
+                   json={"prompt": "eo", "max_tokens": 6}).json()
+    choice = r["choices"][0]
+    assert choice["finish_reason"] == "stop"
+    assert r["usage"]["completion_tokens"] == 3  # incl. the eos token
+    # only the two pre-EOS tokens reach the text
+    assert choice["text"] == ByteTokenizer().decode(probe[:2])
