@@ -371,7 +371,7 @@ def test_top_p_filter_property():
 
     from runbooks_amd.ops.sampling import top_p_filter
 
-    @settings(max_examples=50, deadline=None)
+    @settings(max_examples=50, deadline=None, derandomize=True)
     @given(st.lists(st.floats(-8, 8), min_size=2, max_size=32),
            st.floats(0.05, 0.95))
     def check(vals, p):
@@ -381,13 +381,16 @@ def test_top_p_filter_property():
         order = torch.argsort(probs, descending=True, stable=True)
         cum = torch.cumsum(probs[order], 0)
         k = int((cum >= p - 1e-6).nonzero()[0]) + 1
-        expect = torch.zeros_like(out)
-        expect[order[:k]] = True
-        # ties at the cut can include either member; sizes must agree
-        # and every strictly-heavier token must be kept
-        assert out.sum() == expect.sum()
-        strict = probs > probs[order[k - 1]] + 1e-9
-        assert bool(out[strict].all())
+        # tie-robust properties (exact ties at the cut may keep either
+        # member): something is kept, kept mass reaches top_p, every
+        # strictly-heavier-than-cut token is kept, and nothing lighter
+        # than the lightest "must-keep" is kept beyond the tie band
+        assert out.any()
+        assert float(probs[out].sum()) >= p - 1e-5
+        cut = probs[order[k - 1]]
+        assert bool(out[probs > cut + 1e-9].all())
+        # and nothing strictly lighter than the cut survives
+        assert bool((probs[out] >= cut - 1e-9).all())
 
     check()
 
