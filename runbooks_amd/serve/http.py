@@ -249,21 +249,27 @@ def build_app(engine: Engine, tokenizer=None,
         # vocab (identity for any properly paired tokenizer)
         return [i % engine.cfg.vocab_size for i in ids]
 
-    def _pieces(q, rid, stops):
+    def _pieces(q, rid, stops, state=None):
         """Incremental decoded text pieces for one request; drains the
-        queue after a stop-sequence hit so cancel cleanup stays local."""
+        queue after a stop-sequence hit so cancel cleanup stays local.
+        ``state`` (optional dict) gets {"finish": "stop"|"length"}."""
         out, sent = [], ""
+        if state is None:
+            state = {}
+        state["finish"] = "length"
         while True:
             t = q.get()
             if t is None:
                 return
             if t in eos_ids:
+                state["finish"] = "stop"
                 continue  # engine finishes on EOS; don't stream it
             out.append(t)
             text = tok.decode(out)
             trunc = _stop_hit(text, stops) if stops else None
             if trunc is not None:
                 loop.cancel(rid)
+                state["finish"] = "stop"
                 piece = trunc[len(sent):]
                 if piece:
                     yield piece
@@ -369,8 +375,16 @@ def build_app(engine: Engine, tokenizer=None,
                 try:
                     if req.echo and req.prompt:
                         yield chunk(req.prompt)
-                    for piece in _pieces(q, ereq.request_id, stops):
+                    state = {}
+                    for piece in _pieces(q, ereq.request_id, stops, state):
                         yield chunk(piece)
+                    fin = {"id": cid, "object": "text_completion",
+                           "created": t0,
+                           "model": req.model or model_name,
+                           "choices": [{"text": "", "index": 0,
+                                        "logprobs": None,
+                                        "finish_reason": state["finish"]}]}
+                    yield f"data: {json.dumps(fin)}\n\n"
                     yield "data: [DONE]\n\n"
                 finally:
                     # client disconnect abandons the generator mid-stream:
@@ -451,8 +465,14 @@ def build_app(engine: Engine, tokenizer=None,
                     return f"data: {json.dumps(c)}\n\n"
                 try:
                     yield chunk({"role": "assistant"})
-                    for piece in _pieces(q, ereq.request_id, stops):
+                    state = {}
+                    for piece in _pieces(q, ereq.request_id, stops, state):
                         yield chunk({"content": piece})
+                    fin = {"id": cid, "object": "chat.completion.chunk",
+                           "created": t0, "model": mdl,
+                           "choices": [{"index": 0, "delta": {},
+                                        "finish_reason": state["finish"]}]}
+                    yield f"data: {json.dumps(fin)}\n\n"
                     yield "data: [DONE]\n\n"
                 finally:
                     loop.cancel(ereq.request_id)

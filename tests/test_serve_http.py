@@ -57,8 +57,11 @@ def test_streaming(client):
         events = [ln for ln in r.iter_lines() if ln.startswith("data:")]
     assert events[-1] == "data: [DONE]"
     chunks = [json.loads(e[len("data: "):]) for e in events[:-1]]
-    assert len(chunks) == 3
+    assert len(chunks) == 4  # 3 text chunks + final finish_reason chunk
     assert all(c["object"] == "text_completion" for c in chunks)
+    assert chunks[-1]["choices"][0]["finish_reason"] == "length"
+    assert all(c["choices"][0]["finish_reason"] is None
+               for c in chunks[:-1])
 
 
 def test_metrics(client):
@@ -157,7 +160,8 @@ def test_chat_streaming_deltas(client):
     assert lines[-1] == "data: [DONE]"
     mids = [json.loads(ln[len("data: "):]) for ln in lines[1:-1]]
     assert all(m["object"] == "chat.completion.chunk" for m in mids)
-    assert all("content" in m["choices"][0]["delta"] for m in mids)
+    assert all("content" in m["choices"][0]["delta"] for m in mids[:-1])
+    assert mids[-1]["choices"][0]["finish_reason"] == "length"
 
 
 def test_concurrent_mixed_requests_stress(client):
