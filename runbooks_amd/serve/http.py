@@ -182,7 +182,8 @@ class EngineLoop:
 
 
 def build_app(engine: Engine, tokenizer=None,
-              model_name: str = "model") -> FastAPI:
+              model_name: str = "model",
+              max_queue: int = 512) -> FastAPI:
     import contextlib
 
     loop = EngineLoop(engine)
@@ -337,6 +338,9 @@ def build_app(engine: Engine, tokenizer=None,
         # clamp generation to what the context window can hold
         req.max_tokens = max(1, min(req.max_tokens,
                                     engine.cfg.max_seq_len - 1))
+        if len(engine.waiting) >= max_queue:
+            return JSONResponse(status_code=429, content={"error": {
+                "message": "server overloaded: request queue full"}})
         ids = _encode(req.prompt, req.max_tokens)
         cid = f"cmpl-{uuid.uuid4().hex[:12]}"
         t0 = int(time.time())

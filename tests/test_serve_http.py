@@ -424,3 +424,12 @@ def test_eos_finish_reason():
     assert r["usage"]["completion_tokens"] == 3  # incl. the eos token
     # only the two pre-EOS tokens reach the text
     assert choice["text"] == ByteTokenizer().decode(probe[:2])
+
+
+def test_queue_backpressure_429():
+    eng = Engine("tiny-llama", device="cpu", dtype=torch.float32,
+                 kv_blocks=64, seed=5)
+    app = build_app(eng, model_name="t", max_queue=0)
+    with TestClient(app) as c:  # zero queue capacity -> always shedding
+        r = c.post("/v1/completions", json={"prompt": "x"})
+    assert r.status_code == 429
