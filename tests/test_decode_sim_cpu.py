@@ -31,22 +31,31 @@ def _merge(a, b):
     return mn, l1 * e1 + l2 * e2, a1 * e1 + a2 * e2
 
 
-def simulate_decode(q, kc, vc, bt, seq_lens, scale, nsplit=1):
+def simulate_decode(q, kc, vc, bt, seq_lens, scale, nsplit=1,
+                    seq_starts=None):
     """q [B, Hq, Dh]; kc/vc [nblk, Hkv, BS, Dh]; bt [B, max_blocks].
-    float64 numpy throughout (the kernel's fp32 accum, idealized)."""
+    float64 numpy throughout (the kernel's fp32 accum, idealized).
+
+    ``seq_starts`` models the planned round-2 kernel extension for
+    STRICT sliding windows: attend cache positions [start, seq) only.
+    Kernel-side this is two lines — ``len = seq_len - start; chunk =
+    ceil(len/nsplit); t_begin = start + split*chunk`` — validated here
+    against a masked reference before any GPU budget is spent."""
     B, Hq, Dh = q.shape
     hkv, bs = kc.shape[1], kc.shape[2]
     G = Hq // hkv
     out = np.zeros_like(q)
     for b in range(B):
         seq = int(seq_lens[b])
+        start = int(seq_starts[b]) if seq_starts is not None else 0
         for h in range(hkv):
             for g in range(G):
                 qv = q[b, h * G + g] * scale
                 parts = []
                 for split in range(nsplit):
-                    chunk = (seq + nsplit - 1) // nsplit
-                    t0, t1 = split * chunk, min(seq, (split + 1) * chunk)
+                    chunk = (seq - start + nsplit - 1) // nsplit
+                    t0 = start + split * chunk
+                    t1 = min(seq, t0 + chunk)
                     # one online-softmax stream per (wave, group), tokens
                     # t0 + wid*4 + grp, stride 16 — the kernel's layout
                     streams = []
@@ -116,3 +125,33 @@ def test_decode_sim_matches_ref(B, hkv, G, seqs, nsplit):
         torch.tensor(bt, dtype=torch.int32),
         torch.tensor(sl, dtype=torch.int32), scale).numpy()
     assert np.allclose(got, ref, atol=1e-5), np.abs(got - ref).max()
+
+
+def test_decode_sim_seq_start_strict_window():
+    """The seq_start extension (round-2 strict sliding window): the sim
+    attends only [start, seq) and matches a sliced reference — including
+    a start mid-block and start==0 rows mixed in one batch."""
+    from runbooks_amd.ops.attention import paged_decode_ref
+
+    bs, dh = 16, 64
+    q, kc, vc, bt, sl, scale = _case(3, 2, 2, bs, 4, [50, 61, 9], dh, 7)
+    starts = np.array([18, 0, 5])  # mid-block, none, mid-block
+    got = simulate_decode(q, kc, vc, bt, sl, scale, nsplit=4,
+                          seq_starts=starts)
+    # reference: shift each row's window to the front of a fresh cache
+    for b in range(3):
+        n = int(sl[b]) - int(starts[b])
+        kc2 = np.zeros((4, 2, bs, dh))
+        vc2 = np.zeros((4, 2, bs, dh))
+        for i in range(n):
+            t = int(starts[b]) + i
+            blk = bt[b][t // bs]
+            kc2[i // bs, :, i % bs] = kc[blk, :, t % bs]
+            vc2[i // bs, :, i % bs] = vc[blk, :, t % bs]
+        ref = paged_decode_ref(
+            torch.tensor(q[b:b + 1], dtype=torch.float32),
+            torch.tensor(kc2, dtype=torch.float32),
+            torch.tensor(vc2, dtype=torch.float32),
+            torch.tensor([[0, 1, 2, 3]], dtype=torch.int32),
+            torch.tensor([n], dtype=torch.int32), scale).numpy()
+        assert np.allclose(got[b], ref[0], atol=1e-5), b
