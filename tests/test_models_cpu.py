@@ -445,7 +445,13 @@ def test_engine_fuzz_prefix_cache_and_window():
         if rng.random() < 0.3 and submitted < 25:
             p = list(rng.choice(templates)) + \
                 [rng.randrange(256) for _ in range(rng.randint(0, 8))]
-            live.append(eng.submit(p, max_new_tokens=rng.randint(1, 30)))
+            live.append(eng.submit(
+                p, max_new_tokens=rng.randint(1, 30),
+                temperature=rng.choice([0.0, 0.8]),
+                logprobs=rng.choice([None, 0, 2]),
+                seed=rng.choice([None, 7]),
+                presence_penalty=rng.choice([0.0, 0.5]),
+                frequency_penalty=rng.choice([0.0, 1.0])))
             submitted += 1
         if rng.random() < 0.05 and live:
             eng.cancel(rng.choice(live).request_id)
