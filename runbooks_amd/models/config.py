@@ -123,6 +123,18 @@ register(ModelConfig("falcon-40b", vocab_size=65024, hidden_size=8192,
                      norm="layernorm", act="gelu", parallel_residual=True,
                      tie_embeddings=True))
 
+# --- GPT-2 family (LayerNorm, learned positions, tanh-GELU, biases) --------
+register(ModelConfig("gpt2", vocab_size=50257, hidden_size=768,
+                     num_layers=12, num_heads=12, num_kv_heads=12,
+                     intermediate_size=3072, max_seq_len=1024,
+                     norm="layernorm", act="gelu_tanh", pos="learned",
+                     tie_embeddings=True, mlp_bias=True, attn_bias=True))
+register(ModelConfig("gpt2-xl", vocab_size=50257, hidden_size=1600,
+                     num_layers=48, num_heads=25, num_kv_heads=25,
+                     intermediate_size=6400, max_seq_len=1024,
+                     norm="layernorm", act="gelu_tanh", pos="learned",
+                     tie_embeddings=True, mlp_bias=True, attn_bias=True))
+
 # --- OPT family (LayerNorm, learned positions, ReLU) -----------------------
 register(ModelConfig("opt-125m", vocab_size=50272, hidden_size=768,
                      num_layers=12, num_heads=12, num_kv_heads=12,
@@ -153,6 +165,11 @@ register(ModelConfig("tiny-gemma", vocab_size=256, hidden_size=64,
                      intermediate_size=128, head_dim=16, max_seq_len=128,
                      norm_eps=1e-6, act="gelu_glu", tie_embeddings=True,
                      embed_scale=8.0))
+register(ModelConfig("tiny-gpt2", vocab_size=256, hidden_size=64,
+                     num_layers=2, num_heads=4, num_kv_heads=4,
+                     intermediate_size=128, max_seq_len=128,
+                     norm="layernorm", act="gelu_tanh", pos="learned",
+                     tie_embeddings=True, mlp_bias=True, attn_bias=True))
 register(ModelConfig("tiny-opt", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=4,
                      intermediate_size=128, max_seq_len=128,

@@ -21,7 +21,7 @@ from .transformer import Transformer
 
 def detect_family(cfg: ModelConfig) -> str:
     if cfg.pos == "learned":
-        return "opt"
+        return "gpt2" if cfg.act == "gelu_tanh" else "opt"
     if cfg.parallel_residual:
         return "falcon"
     return "llama"
@@ -57,6 +57,22 @@ def _map_opt(name: str):
     return n
 
 
+def _map_gpt2(name: str):
+    n = name
+    if not n.startswith(("transformer.", "lm_head")):
+        n = "transformer." + n          # GPT2Model vs GPT2LMHeadModel
+    n = n.replace("transformer.wte", "embed")
+    n = n.replace("transformer.wpe", "embed_pos")
+    n = n.replace("transformer.h.", "blocks.")
+    n = n.replace(".attn.c_proj.", ".attn.o_proj.")
+    n = n.replace(".mlp.c_fc.", ".mlp.fc1.")
+    n = n.replace(".mlp.c_proj.", ".mlp.down_proj.")
+    n = n.replace(".ln_1.", ".norm1.")
+    n = n.replace(".ln_2.", ".norm2.")
+    n = n.replace("transformer.ln_f", "norm_f")
+    return n
+
+
 def _map_falcon(name: str):
     n = name
     n = n.replace("transformer.word_embeddings", "embed")
@@ -88,10 +104,28 @@ def convert_hf_state_dict(hf_state: dict, cfg: ModelConfig) -> dict:
     family = detect_family(cfg)
     out: dict[str, torch.Tensor] = {}
     mapper = {"llama": _map_llama, "opt": _map_opt,
-              "falcon": _map_falcon}[family]
+              "falcon": _map_falcon, "gpt2": _map_gpt2}[family]
     for name, w in hf_state.items():
         if name.endswith(".rotary_emb.inv_freq"):
             continue
+        if family == "gpt2":
+            if name.endswith((".attn.bias", ".attn.masked_bias")):
+                continue  # causal-mask buffers, not weights
+            if ".attn.c_attn." in name:
+                # Conv1D fused QKV: weight [in, 3*out] -> three [out, in]
+                blk = name.split(".")[name.split(".").index("h") + 1]
+                if name.endswith(".weight"):
+                    qw, kw, vw = w.t().chunk(3, dim=0)
+                else:
+                    qw, kw, vw = w.chunk(3, dim=0)
+                kind = "weight" if name.endswith(".weight") else "bias"
+                out[f"blocks.{blk}.attn.q_proj.{kind}"] = qw.contiguous()
+                out[f"blocks.{blk}.attn.k_proj.{kind}"] = kw.contiguous()
+                out[f"blocks.{blk}.attn.v_proj.{kind}"] = vw.contiguous()
+                continue
+            if name.endswith(".weight") and w.dim() == 2 and any(
+                    k in name for k in (".c_proj.", ".c_fc.")):
+                w = w.t().contiguous()  # Conv1D stores [in, out]
         if family == "falcon" and ".self_attention.query_key_value." in name:
             blk = name.split(".")[2 if name.startswith("transformer") else 1]
             q, k, v = _split_falcon_qkv(w, cfg)
@@ -219,6 +253,16 @@ def config_from_hf_json(path: str | Path) -> ModelConfig:
             head_dim=d["hidden_size"] // heads,
             norm="layernorm", act="gelu", parallel_residual=True,
             tie_embeddings=True)
+    if mt == "gpt2":
+        return ModelConfig(
+            d.get("_name_or_path", "hf-gpt2"),
+            vocab_size=d["vocab_size"], hidden_size=d["n_embd"],
+            num_layers=d["n_layer"], num_heads=d["n_head"],
+            num_kv_heads=d["n_head"],
+            intermediate_size=d.get("n_inner") or 4 * d["n_embd"],
+            max_seq_len=d.get("n_positions", 1024),
+            norm="layernorm", act="gelu_tanh", pos="learned",
+            tie_embeddings=True, mlp_bias=True, attn_bias=True)
     if mt == "opt":
         return ModelConfig(
             d.get("_name_or_path", "hf-opt"),

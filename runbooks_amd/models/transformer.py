@@ -180,7 +180,12 @@ class MLP(nn.Module):
                 torch.nn.functional.gelu(self.gate_proj(x),
                                          approximate="tanh") * self.up_proj(x))
         h = self.fc1(x)
-        h = torch.nn.functional.gelu(h) if self.act == "gelu" else torch.relu(h)
+        if self.act == "gelu":
+            h = torch.nn.functional.gelu(h)
+        elif self.act == "gelu_tanh":      # gpt2 "gelu_new"
+            h = torch.nn.functional.gelu(h, approximate="tanh")
+        else:
+            h = torch.relu(h)
         return self.down_proj(h)
 
 

@@ -245,3 +245,30 @@ def test_gemma_hf_parity():
         theirs = hf(tokens).logits
         got = ours(tokens)
     _logits_close(got, theirs)
+
+
+def test_gpt2_hf_parity():
+    """gpt2: Conv1D-transposed weights, fused c_attn QKV split,
+    tanh-GELU, learned positions without OPT's +2 offset."""
+    from transformers import GPT2Config, GPT2LMHeadModel
+    hf_cfg = GPT2Config(vocab_size=128, n_embd=64, n_layer=2, n_head=4,
+                        n_positions=64, n_inner=None)
+    torch.manual_seed(4)
+    hf = GPT2LMHeadModel(hf_cfg).eval()
+
+    cfg = ModelConfig("t-gpt2", vocab_size=128, hidden_size=64,
+                      num_layers=2, num_heads=4, num_kv_heads=4,
+                      intermediate_size=256, max_seq_len=64,
+                      norm="layernorm", act="gelu_tanh", pos="learned",
+                      tie_embeddings=True, mlp_bias=True, attn_bias=True)
+    ours = build_model(cfg, dtype=torch.float32)
+    state = convert_hf_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(state, strict=False)
+    assert not [m for m in missing if not m.startswith("rope_")], missing
+    assert not unexpected, unexpected
+
+    tokens = torch.randint(0, 128, (2, 17))
+    with torch.no_grad():
+        theirs = hf(tokens).logits
+        got = ours(tokens)
+    _logits_close(got, theirs)

@@ -154,7 +154,7 @@ def test_preempted_output_preserved():
 
 
 @pytest.mark.parametrize("name", ["tiny-opt", "tiny-falcon", "tiny-gemma",
-                                  "tiny-qwen"])
+                                  "tiny-qwen", "tiny-gpt2"])
 def test_engine_generates_all_families(name):
     """Decode path per family: OPT (learned positions, no rope, biases)
     and falcon (parallel residual, MQA) — the BASELINE config families."""
