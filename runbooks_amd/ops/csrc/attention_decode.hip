@@ -285,8 +285,16 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
       case 16: RB_DEC(64, 16); break;
       default: TORCH_CHECK(false, "paged_decode: unsupported G=", G);
     }
+  } else if (dh == 256) {
+    // gemma-7b (wide heads, MHA). VE=16 -> ~70 VGPRs at G=1; first GPU
+    // validation is the RB_EXPERIMENTAL test (round 2).
+    switch (G) {
+      case 1: RB_DEC(256, 1); break;
+      case 2: RB_DEC(256, 2); break;
+      default: TORCH_CHECK(false, "paged_decode: unsupported G=", G);
+    }
   } else {
-    TORCH_CHECK(false, "paged_decode: Dh must be 64 or 128, got ", dh);
+    TORCH_CHECK(false, "paged_decode: Dh must be 64/128/256, got ", dh);
   }
 #undef RB_DEC
   return out;

@@ -112,6 +112,11 @@ class Engine:
                   "(fp8 decode path is single-GPU; 288 GB/GPU rarely "
                   "needs 8-bit at TP>1)", flush=True)
         self.cfg = model.cfg
+        if torch.cuda.is_available() and model.cfg.head_dim not in (64, 128,
+                                                                    256):
+            raise NotImplementedError(
+                f"paged decode supports head_dim 64/128/256 on gfx950; "
+                f"{model.cfg.name} has {model.cfg.head_dim}")
         self.bs = ops.BLOCK_SIZE
         self.max_batch = max_batch
         self.seed = seed

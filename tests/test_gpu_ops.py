@@ -481,3 +481,24 @@ def test_geglu_packed():
     g, u = y[:, :256].float(), y[:, 256:].float()
     ref = torch.nn.functional.gelu(g, approximate="tanh") * u
     assert torch.allclose(got.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not _EXPERIMENTAL, reason="RB_EXPERIMENTAL=1 only "
+                    "(Dh=256 decode instantiation awaits GPU validation)")
+def test_paged_decode_dh256():
+    """gemma-width decode heads: Dh=256 template vs fp32 reference."""
+    from runbooks_amd import ops
+    torch.manual_seed(0)
+    B, hkv, G, bs, nblk, dh = 2, 2, 1, 16, 8, 256
+    q = torch.randn(B, hkv * G, dh, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(nblk, hkv, bs, dh, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn_like(kc)
+    bt = torch.tensor([[0, 1, 2], [3, 4, 5]], device="cuda",
+                      dtype=torch.int32)
+    sl = torch.tensor([40, 23], device="cuda", dtype=torch.int32)
+    got = ops.paged_decode(q, kc, vc, bt, sl)
+    ref = ops.paged_decode_ref(q.float().cpu(), kc.float().cpu(),
+                               vc.float().cpu(), bt.cpu(), sl.cpu(),
+                               scale=1.0 / 16.0)
+    assert torch.allclose(got.float().cpu(), ref, atol=3e-2, rtol=3e-2)
