@@ -147,6 +147,13 @@ register(ModelConfig("smoke-llama", vocab_size=512, hidden_size=256,
                      num_layers=2, num_heads=4, num_kv_heads=2,
                      intermediate_size=512, max_seq_len=256))
 
+# --- small GPU smoke config for the gemma family (Dh=256 kernels) ----------
+register(ModelConfig("smoke-gemma", vocab_size=512, hidden_size=512,
+                     num_layers=2, num_heads=2, num_kv_heads=2,
+                     intermediate_size=1024, max_seq_len=256, head_dim=256,
+                     norm_eps=1e-6, act="gelu_glu", tie_embeddings=True,
+                     embed_scale=512 ** 0.5))
+
 # --- tiny configs for CPU tests --------------------------------------------
 register(ModelConfig("tiny-llama", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=2,

@@ -27,7 +27,8 @@ timeout 300 python bench.py --mode serve --steps 60 --warmup 5
 timeout 300 python bench.py --mode serve --steps 60 --warmup 5 --serve-8bit
 
 # 6. Late-round-1 gated features: validate then flip defaults (~2 min)
-RB_EXPERIMENTAL=1 python -m pytest tests/test_gpu_ops.py -q -k geglu_packed
+RB_EXPERIMENTAL=1 python -m pytest tests/test_gpu_ops.py -q -k 'geglu_packed or dh256'
+RB_EXPERIMENTAL=1 python -m pytest tests/test_gpu_engine.py -q -k gemma -m gpu
 # prefix caching on GPU (engine bookkeeping only; kernels see a block table)
 RB_PREFIX_CACHE=1 python -m pytest tests/test_gpu_engine.py -q -m gpu
 # then: flip RB_PREFIX_CACHE default in serve/engine.py, enable

@@ -60,3 +60,23 @@ def test_fp8_engine_decode_runs():
     out = eng.generate([10, 20, 30], max_new_tokens=4)
     assert len(out) == 4 and all(0 <= t < m.cfg.vocab_size for t in out)
     _FP8_REGISTRY.clear()
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(
+    __import__("os").environ.get("RB_EXPERIMENTAL") != "1",
+    reason="RB_EXPERIMENTAL=1 only (gemma Dh=256 path awaits first "
+           "on-GPU validation: decode kernel + fused qkv epilogue)")
+def test_gemma_engine_decode_dh256():
+    """End-to-end gemma-family decode on GPU (Dh=256 instantiation +
+    fused qkv_rope_append + GeGLU MLP) vs the CPU fp32 engine."""
+    m = build_model("smoke-gemma", dtype=torch.bfloat16, device="cuda:0",
+                    seed=11)
+    eng = Engine(m, device="cuda:0", kv_blocks=64, seed=2)
+    out = eng.generate([5, 9, 2, 7], max_new_tokens=8)
+
+    cpu = build_model("smoke-gemma", dtype=torch.float32, seed=11)
+    ref = Engine(cpu, device="cpu", dtype=torch.float32, kv_blocks=64,
+                 seed=2).generate([5, 9, 2, 7], max_new_tokens=8)
+    # greedy bf16-vs-fp32 may diverge late; the prefix must agree
+    assert out[:4] == ref[:4], (out, ref)
