@@ -76,6 +76,7 @@ class Request:
     seed: int | None = None         # per-request sampling seed override
     presence_penalty: float = 0.0   # subtract once per seen token
     frequency_penalty: float = 0.0  # subtract per occurrence
+    repetition_penalty: float = 1.0  # HF-style multiplicative penalty
     stop_ids: tuple[int, ...] = ()
     # state
     output_ids: list[int] = field(default_factory=list)
@@ -177,11 +178,12 @@ class Engine:
                logprobs: int | None = None, seed: int | None = None,
                presence_penalty: float = 0.0,
                frequency_penalty: float = 0.0,
+               repetition_penalty: float = 1.0,
                stop_token_ids: tuple[int, ...] = ()) -> Request:
         req = Request(self._next_id, list(prompt_ids), max_new_tokens,
                       temperature, top_p, logprobs, seed,
                       presence_penalty, frequency_penalty,
-                      tuple(stop_token_ids))
+                      repetition_penalty, tuple(stop_token_ids))
         self._next_id += 1
         self.waiting.append(req)
         return req
@@ -191,7 +193,8 @@ class Engine:
         """OpenAI presence/frequency penalties over the text so far
         (prompt + generated). Torch ops on the logits row, outside the
         decode hipGraph — zero cost for requests that don't ask."""
-        if not (r.presence_penalty or r.frequency_penalty):
+        if not (r.presence_penalty or r.frequency_penalty or
+                r.repetition_penalty != 1.0):
             return row
         from collections import Counter
         cnt = Counter(r.prompt_ids)
@@ -202,6 +205,10 @@ class Engine:
                          device=row.device)
         row = row.clone().float()
         row[ids] -= r.frequency_penalty * c + r.presence_penalty
+        if r.repetition_penalty != 1.0:
+            seen = row[ids]
+            row[ids] = torch.where(seen > 0, seen / r.repetition_penalty,
+                                   seen * r.repetition_penalty)
         return row
 
     @staticmethod
@@ -356,7 +363,8 @@ class Engine:
         seed = self.seed + 1_000_003 * reqs[0].seq_len
         params = {(r.temperature, r.top_p) for r in reqs}
         plain = all(r.seed is None and not r.presence_penalty and
-                    not r.frequency_penalty for r in reqs)
+                    not r.frequency_penalty and
+                    r.repetition_penalty == 1.0 for r in reqs)
         if len(params) == 1 and plain:
             t, p = params.pop()
             toks = ops.sample_tokens(logits, t, top_p=p, seed=seed)

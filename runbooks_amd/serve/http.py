@@ -62,6 +62,7 @@ class CompletionRequest(BaseModel):
     seed: int | None = None      # reproducible sampling
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -107,6 +108,7 @@ class EngineLoop:
     def submit(self, prompt_ids, max_new_tokens, temperature,
                top_p=1.0, logprobs=None, seed=None,
                presence_penalty=0.0, frequency_penalty=0.0,
+               repetition_penalty=1.0,
                stop_token_ids=()) -> tuple[queue.Queue, "object"]:
         """Returns (queue yielding token_id | None, engine Request)."""
         q: queue.Queue = queue.Queue()
@@ -116,6 +118,7 @@ class EngineLoop:
                                      seed=seed,
                                      presence_penalty=presence_penalty,
                                      frequency_penalty=frequency_penalty,
+                                     repetition_penalty=repetition_penalty,
                                      stop_token_ids=stop_token_ids)
             self._watchers[req.request_id] = q
             req._watch_sent = 0
@@ -363,6 +366,7 @@ def build_app(engine: Engine, tokenizer=None,
                                   else req.seed + i),
                             presence_penalty=req.presence_penalty,
                             frequency_penalty=req.frequency_penalty,
+                            repetition_penalty=req.repetition_penalty,
                             stop_token_ids=eos_ids)
                 for i in range(n_sample)]
         q, ereq = subs[0]

@@ -554,3 +554,18 @@ def test_stop_token_ids_finish_early():
     while not r.finished:
         eng.step()
     assert r.output_ids == plain[:3], (r.output_ids, plain)
+
+
+def test_repetition_penalty():
+    """HF-style multiplicative penalty: a large value forbids repeats
+    among positive-logit tokens (greedy)."""
+    m = build_model("tiny-llama", dtype=torch.float32, seed=1)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    plain = eng.generate([4, 4, 4], max_new_tokens=16)
+    assert len(set(plain)) < 16
+    r = eng.submit([4, 4, 4], max_new_tokens=16, repetition_penalty=1e6)
+    while not r.finished:
+        eng.step()
+    # tokens with positive logits get crushed after first use; repeats
+    # can only come from the (rare) all-negative rows
+    assert len(set(r.output_ids)) > len(set(plain))
