@@ -87,6 +87,7 @@ class ChatRequest(BaseModel):
     seed: int | None = None
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
     stop: list[str] | str | None = None
     model: str = ""
     stream: bool = False
@@ -462,6 +463,7 @@ def build_app(engine: Engine, tokenizer=None,
                                   top_p=req.top_p, seed=req.seed,
                                   presence_penalty=req.presence_penalty,
                                   frequency_penalty=req.frequency_penalty,
+                                  repetition_penalty=req.repetition_penalty,
                                   stop_token_ids=eos_ids)
 
             def gen():
@@ -491,6 +493,7 @@ def build_app(engine: Engine, tokenizer=None,
             temperature=req.temperature, top_p=req.top_p, n=req.n,
             seed=req.seed, presence_penalty=req.presence_penalty,
             frequency_penalty=req.frequency_penalty,
+            repetition_penalty=req.repetition_penalty,
             stop=req.stop, model=req.model, stream=False)
         resp = completions(creq)
         if resp.status_code != 200:
