@@ -256,6 +256,45 @@ def _dp_save_resume(rank, world):
         shutil.rmtree(out, ignore_errors=True)
 
 
+def _tp_engine_prefix_cache(rank, world):
+    """TP=2 + prefix caching: cache bookkeeping lives on rank 0 (slot
+    mappings broadcast to workers already reflect the shared blocks);
+    outputs of two same-prefix requests must match the TP=1 engine."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.serve import Engine
+    from runbooks_amd.serve.tp_worker import broadcast_shutdown, worker_loop
+
+    torch.manual_seed(0)
+    tp_model = build_model("tiny-llama", dtype=torch.float32, tp=world, seed=3)
+    single = build_model("tiny-llama", dtype=torch.float32, tp=1, seed=3)
+    sd = single.state_dict()
+    for name, t in tp_model.state_dict().items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:
+            t.copy_(full[rank * t.shape[0]:(rank + 1) * t.shape[0]])
+        else:
+            t.copy_(full[:, rank * t.shape[1]:(rank + 1) * t.shape[1]])
+
+    eng = Engine(tp_model, device="cpu", kv_blocks=64, seed=11,
+                 prefix_cache=True)
+    shared = list(range(1, 33))
+    prompts = [shared + [77], shared + [99]]
+    if rank == 0:
+        reqs = [eng.submit(list(p), max_new_tokens=5) for p in prompts]
+        while eng.has_work():
+            eng.step()
+        broadcast_shutdown(torch.device("cpu"))
+        assert eng.stats["prefix_hits"] >= 1
+        ref = Engine(single, device="cpu", kv_blocks=64, seed=11)
+        ref.tp = 1
+        for p, r in zip(prompts, reqs):
+            assert r.output_ids == ref.generate(list(p), max_new_tokens=5)
+    else:
+        worker_loop(eng)
+
+
 # --- test entries -----------------------------------------------------------
 
 def test_dp_gradient_allreduce():
@@ -356,3 +395,7 @@ def test_tp_engine_sliding_window():
 
 def test_dp_save_resume():
     _run_dist(_dp_save_resume, port=PORT + 12)
+
+
+def test_tp_engine_prefix_cache():
+    _run_dist(_tp_engine_prefix_cache, port=PORT + 13)
