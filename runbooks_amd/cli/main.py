@@ -253,8 +253,9 @@ def logs(kind, name, namespace, follow):
 @click.option("-p", "--prompt", default="Hello")
 @click.option("-n", "--namespace", default="default")
 @click.option("--max-tokens", default=32)
+@click.option("--temperature", default=0.0)
 @click.option("--port", default=18080)
-def infer(server_name, prompt, namespace, max_tokens, port):
+def infer(server_name, prompt, namespace, max_tokens, temperature, port):
     """Send a completion request to a Server through a port-forward
     (the reference's `sub infer` exists but is disabled,
     reference internal/cli/root.go:19 — here it works)."""
@@ -267,8 +268,8 @@ def infer(server_name, prompt, namespace, max_tokens, port):
                                     port, 8080, resource="service")
     try:
         time.sleep(2)
-        body = json.dumps({"prompt": prompt,
-                           "max_tokens": max_tokens}).encode()
+        body = json.dumps({"prompt": prompt, "max_tokens": max_tokens,
+                           "temperature": temperature}).encode()
         req = urllib.request.Request(
             f"http://localhost:{port}/v1/completions", data=body,
             headers={"Content-Type": "application/json"})
