@@ -569,3 +569,25 @@ def test_repetition_penalty():
     # tokens with positive logits get crushed after first use; repeats
     # can only come from the (rare) all-negative rows
     assert len(set(r.output_ids)) > len(set(plain))
+
+
+def test_fp8_quant_roundtrip_cpu():
+    """OCP e4m3 row-wise quantization on CPU (torch float8 dtype):
+    round-trip error bounded by the format's relative step, scales
+    positive, registry keyed by the master weight."""
+    from runbooks_amd.ops.linear import (_FP8_REGISTRY, dequantize_fp8,
+                                         quantize_fp8)
+
+    torch.manual_seed(0)
+    w = torch.randn(32, 64) * 3.0
+    w8, scale = quantize_fp8(w)
+    assert w8.dtype == torch.uint8 and scale.shape == (32,)
+    assert (scale > 0).all()
+    back = dequantize_fp8(w8, scale, dtype=torch.float32)
+    # e4m3 has a 3-bit mantissa: relative error <= 2^-4 per element
+    # against the row scale's dynamic range
+    err = (back - w).abs()
+    bound = w.abs().amax(dim=1, keepdim=True) / 448.0 + w.abs() * (2 ** -4)
+    assert (err <= bound + 1e-6).all(), float((err - bound).max())
+    assert w.data_ptr() in _FP8_REGISTRY
+    _FP8_REGISTRY.pop(w.data_ptr())
