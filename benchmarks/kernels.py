@@ -78,6 +78,15 @@ def main():
     us = timeit(lambda: ops.swiglu(g, u))
     out.append(row("swiglu_fwd", "2048x11008", us, 3 * g.numel() * 2))
 
+    # packed GLU epilogues [2048, 2*11008] (swiglu validated; geglu gated)
+    yp = torch.randn(2048, 2 * 11008, dtype=bf16, device=dev)
+    us = timeit(lambda: ops.ext().swiglu_packed(yp))
+    out.append(row("swiglu_packed", "2048x2x11008", us,
+                   1.5 * yp.numel() * 2))
+    us = timeit(lambda: ops.ext().geglu_packed(yp))
+    out.append(row("geglu_packed", "2048x2x11008", us,
+                   1.5 * yp.numel() * 2))
+
     # cross-entropy fwd+bwd [2048, 32000]
     logits = torch.randn(4, 512, 32000, dtype=bf16, device=dev,
                          requires_grad=True)
