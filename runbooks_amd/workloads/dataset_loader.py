@@ -57,6 +57,8 @@ def main():
         print(f"dataset-loader: fetch failed ({e}); writing synthetic data",
               file=sys.stderr)
         write_synthetic(out_dir)
+    (out_dir / "completed.json").write_text(
+        json.dumps({"completed": True}))
     print(f"dataset-loader: wrote {sorted(p.name for p in out_dir.iterdir())}")
     return 0
 

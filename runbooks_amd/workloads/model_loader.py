@@ -58,6 +58,8 @@ def main():
             print(f"model-loader: hub download failed ({e}); "
                   f"trying synthetic registry model", file=sys.stderr)
             materialize_synthetic(name.split("/")[-1], out_dir)
+    (out_dir / "completed.json").write_text(
+        json.dumps({"completed": True}))
     print(f"model-loader: wrote {sorted(p.name for p in out_dir.iterdir())}")
     return 0
 

@@ -125,6 +125,11 @@ def main():
     trainer.resume()
     trainer.fit(dataset, eval_dataset=eval_dataset)
     if comm.rank() == 0:
+        # artifact-completeness marker (reference docs/design.md
+        # "Buckets": reconcile logic can check completed.json in the
+        # bucket after cluster re-creation)
+        (out_dir / "completed.json").write_text(json.dumps(
+            {"completed": True, "step": trainer.step_num}))
         print(f"trainer: done at step {trainer.step_num}; "
               f"artifacts in {out_dir}")
     if comm.is_dist():

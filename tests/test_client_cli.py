@@ -212,6 +212,10 @@ def test_trainer_main_end_to_end(tmp_path, monkeypatch):
     ckpts = list(out_dir.glob("checkpoint-*"))
     assert ckpts, "trainer wrote no checkpoint"
     assert (ckpts[0] / "model.safetensors").exists()
+    # artifact-completeness markers (reference design.md "Buckets")
+    import json as _json
+    for d in (model_dir, data_dir, out_dir):
+        assert _json.loads((d / "completed.json").read_text())["completed"]
 
 
 def test_bench_contract_cpu(tmp_path):
