@@ -128,7 +128,7 @@ def main():
         # artifact-completeness marker (reference docs/design.md
         # "Buckets": reconcile logic can check completed.json in the
         # bucket after cluster re-creation)
-        (out_dir / "completed.json").write_text(json.dumps(
+        (Path(out_dir) / "completed.json").write_text(json.dumps(
             {"completed": True, "step": trainer.step_num}))
         print(f"trainer: done at step {trainer.step_num}; "
               f"artifacts in {out_dir}")
