@@ -591,3 +591,19 @@ def test_fp8_quant_roundtrip_cpu():
     assert (err <= bound + 1e-6).all(), float((err - bound).max())
     assert w.data_ptr() in _FP8_REGISTRY
     _FP8_REGISTRY.pop(w.data_ptr())
+
+
+def test_admission_does_not_starve_decode():
+    """Merged-step scheduling: while new prompts are being admitted,
+    already-running sequences still produce a token every step."""
+    m = build_model("tiny-llama", dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64,
+                 max_batch=8)
+    first = eng.submit([1, 2, 3], max_new_tokens=32)
+    eng.step()  # prefill + first decode
+    produced = len(first.output_ids)
+    for i in range(4):  # four more admissions, one per step
+        eng.submit([5 + i] * 4, max_new_tokens=8)
+        eng.step()
+        assert len(first.output_ids) > produced, "decode starved"
+        produced = len(first.output_ids)
