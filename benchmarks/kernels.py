@@ -133,6 +133,17 @@ def main():
     out.append(row("paged_decode", f"b{B} len{L} h32 d128", us,
                    2 * B * 32 * L * 128 * 2))
 
+    # gemma-width decode (Dh=256 instantiation; first validated round 2)
+    kc2 = torch.randn(nb // 2, 16, BS, 256, dtype=bf16, device=dev)
+    vc2 = torch.randn_like(kc2)
+    bt2 = torch.arange(B // 2 * (L // BS), dtype=torch.int32,
+                       device=dev).reshape(B // 2, -1)
+    sl2 = torch.full((B // 2,), L, dtype=torch.int32, device=dev)
+    qd2 = torch.randn(B // 2, 16, 256, dtype=bf16, device=dev)
+    us = timeit(lambda: ops.paged_decode(qd2, kc2, vc2, bt2, sl2))
+    out.append(row("paged_decode_dh256", f"b{B//2} len{L} h16 d256", us,
+                   2 * (B // 2) * 16 * L * 256 * 2))
+
     # decode GEMMs: fresh weights per call (no L3 reuse)
     for N, K in [(12288, 4096), (4096, 4096), (22016, 4096), (4096, 11008),
                  (32000, 4096)]:
