@@ -607,3 +607,15 @@ def test_admission_does_not_starve_decode():
         eng.step()
         assert len(first.output_ids) > produced, "decode starved"
         produced = len(first.output_ids)
+
+
+def test_prefix_cache_env_gate(monkeypatch):
+    m = build_model("tiny-llama", dtype=torch.float32)
+    assert not Engine(m, device="cpu", dtype=torch.float32,
+                      kv_blocks=32).prefix_cache_enabled
+    monkeypatch.setenv("RB_PREFIX_CACHE", "1")
+    assert Engine(m, device="cpu", dtype=torch.float32,
+                  kv_blocks=32).prefix_cache_enabled
+    # explicit arg beats the env
+    assert not Engine(m, device="cpu", dtype=torch.float32, kv_blocks=32,
+                      prefix_cache=False).prefix_cache_enabled
