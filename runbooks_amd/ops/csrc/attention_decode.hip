@@ -44,6 +44,12 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   const int Hq = hkv * G;
 
   const int seq_len = seq_lens[b];
+  // Round-2 planned extension (strict sliding windows): add an optional
+  // per-sequence seq_start and compute
+  //     len = seq_len - start; chunk = ceil(len / nsplit);
+  //     t_begin = start + split * chunk; t_end = min(seq_len, ...)
+  // The index math is pre-validated on CPU in
+  // tests/test_decode_sim_cpu.py (simulate_decode seq_starts).
   int t_begin = 0, t_end = seq_len;
   if (SPLIT) {
     const int chunk = (seq_len + nsplit - 1) / nsplit;
