@@ -433,3 +433,23 @@ def test_queue_backpressure_429():
     with TestClient(app) as c:  # zero queue capacity -> always shedding
         r = c.post("/v1/completions", json={"prompt": "x"})
     assert r.status_code == 429
+
+
+def test_engine_loop_fail_all_releases_watchers():
+    """If engine.step keeps crashing, the loop fails all requests after
+    3 attempts and watchers unblock (HTTP returns instead of hanging)."""
+    eng = Engine("tiny-llama", device="cpu", dtype=torch.float32,
+                 kv_blocks=64, seed=6)
+
+    def boom():
+        raise RuntimeError("injected step failure")
+
+    eng.step = boom
+    app = build_app(eng, model_name="t")
+    with TestClient(app) as c:
+        r = c.post("/v1/completions", json={"prompt": "x", "max_tokens": 4})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["usage"]["completion_tokens"] == 0
+    assert not app.state.engine_loop._watchers
+    assert not eng.waiting and not eng.running
