@@ -237,6 +237,14 @@ def build_app(engine: Engine, tokenizer=None,
                 "data": [{"id": model_name, "object": "model",
                           "owned_by": "runbooks-amd"}]}
 
+    @app.get("/v1/models/{model_id}")
+    def model_info(model_id: str):
+        if model_id != model_name:
+            return JSONResponse(status_code=404, content={"error": {
+                "message": f"model '{model_id}' not found"}})
+        return {"id": model_name, "object": "model",
+                "owned_by": "runbooks-amd"}
+
     def _usage(n_prompt, n_out):
         return {"prompt_tokens": n_prompt, "completion_tokens": n_out,
                 "total_tokens": n_prompt + n_out}

@@ -453,3 +453,8 @@ def test_engine_loop_fail_all_releases_watchers():
     assert body["usage"]["completion_tokens"] == 0
     assert not app.state.engine_loop._watchers
     assert not eng.waiting and not eng.running
+
+
+def test_model_info_endpoint(client):
+    assert client.get("/v1/models/tiny-llama").json()["id"] == "tiny-llama"
+    assert client.get("/v1/models/nope").status_code == 404
