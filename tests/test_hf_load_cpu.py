@@ -272,3 +272,26 @@ def test_gpt2_hf_parity():
         theirs = hf(tokens).logits
         got = ours(tokens)
     _logits_close(got, theirs)
+
+
+def test_llama_hf_parity_deep_tied():
+    """Stronger llama parity: 3 layers, 8 heads, GQA-4, tied embeddings
+    — exact to float noise (3e-7 observed; eps must match, which real
+    config.json imports do via rms_norm_eps)."""
+    from transformers import LlamaConfig, LlamaForCausalLM
+    torch.manual_seed(0)
+    hf = LlamaForCausalLM(LlamaConfig(
+        vocab_size=256, hidden_size=96, intermediate_size=192,
+        num_hidden_layers=3, num_attention_heads=8, num_key_value_heads=2,
+        max_position_embeddings=64, rms_norm_eps=1e-5,
+        tie_word_embeddings=True)).eval()
+    cfg = ModelConfig("x", vocab_size=256, hidden_size=96, num_layers=3,
+                      num_heads=8, num_kv_heads=2, intermediate_size=192,
+                      max_seq_len=64, tie_embeddings=True)
+    ours = build_model(cfg, dtype=torch.float32)
+    ours.load_state_dict(convert_hf_state_dict(hf.state_dict(), cfg),
+                         strict=False)
+    tokens = torch.randint(0, 256, (2, 23))
+    with torch.no_grad():
+        d = (ours(tokens) - hf(tokens).logits).abs().max()
+    assert float(d) < 1e-5, float(d)
