@@ -397,14 +397,15 @@ def test_build_job_recreated_when_out_of_date(env):
     assert job2["metadata"]["uid"] != job1["metadata"]["uid"]
 
 
-def test_controller_fuzz_random_object_graph(env):
+@pytest.mark.parametrize("seed", [7, 42, 123])
+def test_controller_fuzz_random_object_graph(env, seed):
     """Random Models/Datasets/Servers/Notebooks with random (possibly
     dangling) references: reconciliation must never wedge or corrupt —
     every object ends with a coherent status once its chain is unblocked."""
     import random
 
     kube, cloud, sci, mgr = env
-    rng = random.Random(7)
+    rng = random.Random(seed)
     names = [f"o{i}" for i in range(12)]
     for i, n in enumerate(names):
         kind = rng.choice(["Model", "Dataset", "Server", "Notebook"])
