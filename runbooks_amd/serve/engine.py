@@ -439,6 +439,9 @@ class Engine:
         """One engine iteration. Returns requests finished this step."""
         self.stats["steps"] += 1
         finished = []
+        # sweep first: requests already satisfied (cancel() capped their
+        # max_new_tokens between steps) must not decode one extra token
+        self._sweep_finished(finished)
         req = self._admit()
         if req is None and not self.running and self.waiting:
             head = self.waiting[0]
