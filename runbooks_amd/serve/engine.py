@@ -138,8 +138,12 @@ class Engine:
         seq_cap = self.cfg.max_seq_len
         if self.cfg.sliding_window:
             seq_cap = min(seq_cap, self.cfg.sliding_window + 2 * self.bs)
+        # +1: _admit allocates ceil(S/bs)+1 blocks (prompt + headroom for
+        # the first decoded token), so a near-max-length prompt can hold
+        # ceil(seq_cap/bs)+1 blocks — the graph staging table must be at
+        # least that wide or btn[i, :len(row)] raises mid-flight.
         self.max_blocks_per_seq = min(
-            kv_blocks, (seq_cap + self.bs - 1) // self.bs)
+            kv_blocks, (seq_cap + self.bs - 1) // self.bs + 1)
         self._graphed = None
         # hipGraphs drive the single-GPU decode; the TP>1 path runs eager
         # so the worker-follow protocol (serve/tp_worker.py) sees every
@@ -460,12 +464,22 @@ class Engine:
                     f"cache ({self.allocator.num_blocks} blocks)")
         tr = get_tracer()
         if req is not None:
-            if tr:
-                with tr.span("prefill", tokens=len(req.prompt_ids),
-                             request=req.request_id):
+            try:
+                if tr:
+                    with tr.span("prefill", tokens=len(req.prompt_ids),
+                                 request=req.request_id):
+                        first = self._prefill(req)
+                else:
                     first = self._prefill(req)
-            else:
-                first = self._prefill(req)
+            except Exception:
+                # _admit already popped req from waiting and allocated its
+                # blocks; releasing here (shared prefix blocks are
+                # refcounted, release decrements) keeps a failing request
+                # from leaking KV pool forever.
+                self.allocator.release(req.blocks)
+                req.blocks = []
+                req.finished = True
+                raise
             self.stats["prefills"] += 1
             if self.prefix_cache_enabled:
                 self._pc_insert(req)

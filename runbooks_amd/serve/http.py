@@ -257,7 +257,11 @@ def build_app(engine: Engine, tokenizer=None,
         return None
 
     def _encode(prompt: str, max_tokens: int) -> list[int]:
-        ids = tok.encode(prompt)[-engine.cfg.max_seq_len + max_tokens + 1:]
+        # keep at most max_seq_len - max_tokens - 1 prompt tokens so
+        # prompt + generation never exceeds the model's positional range
+        # (keep >= 1: a naive negative slice of 0 would keep EVERYTHING)
+        keep = max(1, engine.cfg.max_seq_len - max_tokens - 1)
+        ids = tok.encode(prompt)[-keep:]
         # guard: a fallback tokenizer may emit ids past a small model's
         # vocab (identity for any properly paired tokenizer)
         return [i % engine.cfg.vocab_size for i in ids]

@@ -131,7 +131,9 @@ class Trainer:
                 break
         self.model.train()
         if comm.world_size() > 1:
-            t = torch.tensor([total, float(n)])
+            # must live on self.device: the RCCL backend only reduces
+            # CUDA tensors (a CPU tensor here crashes multi-GPU eval)
+            t = torch.tensor([total, float(n)], device=self.device)
             torch.distributed.all_reduce(t)
             total, n = float(t[0]), int(t[1])
         return total / max(1, n)

@@ -71,7 +71,13 @@ def main():
         arch = meta.get("runbooks_amd_config", arch)
         if arch not in list_configs():
             from ..models.load import config_from_hf_json
+            from ..models import register
             arch = config_from_hf_json(marker)
+            # Trainer resolves models by name via the registry; a config
+            # derived from an out-of-registry HF checkpoint must be
+            # registered or build_model(name) raises KeyError at startup
+            # (server_main passes the config object through instead).
+            register(arch)
 
     cfg = TrainConfig(
         model=arch if isinstance(arch, str) else arch.name,

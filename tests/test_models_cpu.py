@@ -621,3 +621,44 @@ def test_prefix_cache_env_gate(monkeypatch):
     # explicit arg beats the env
     assert not Engine(m, device="cpu", dtype=torch.float32, kv_blocks=32,
                       prefix_cache=False).prefix_cache_enabled
+
+
+def test_near_max_length_prompt_fits_block_table():
+    """Regression: a prompt whose length lands in the
+    last block window must not overflow max_blocks_per_seq — _admit
+    allocates ceil(S/bs)+1 blocks, so the staging width needs the +1."""
+    m = build_model("tiny-llama", dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    bs, cap = eng.bs, eng.cfg.max_seq_len            # 16, 128
+    S = cap - 2                                       # 126: last block window
+    req = eng.submit(list(range(S % eng.cfg.vocab_size)) [:S] or [1],
+                     max_new_tokens=2)
+    req.prompt_ids = [i % eng.cfg.vocab_size for i in range(S)]
+    while not req.finished:
+        eng.step()
+    assert len(req.output_ids) >= 1
+    # block list may legitimately reach ceil(cap/bs)+1; staging must cover
+    assert eng.max_blocks_per_seq >= (cap + bs - 1) // bs + 1
+
+
+def test_prefill_failure_releases_blocks():
+    """Regression: a prefill that raises must not leak
+    the blocks _admit already allocated."""
+    m = build_model("tiny-llama", dtype=torch.float32)
+    eng = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64)
+    free_before = len(eng.allocator.free)
+    eng.submit([1, 2, 3], max_new_tokens=2)
+    orig = eng._prefill
+    def boom(req):
+        raise RuntimeError("injected prefill failure")
+    eng._prefill = boom
+    with pytest.raises(RuntimeError, match="injected"):
+        eng.step()
+    assert len(eng.allocator.free) == free_before
+    assert not eng.running and not eng.waiting
+    # engine recovers: next request goes through untouched
+    eng._prefill = orig
+    r = eng.submit([4, 5, 6], max_new_tokens=2)
+    while not r.finished:
+        eng.step()
+    assert len(r.output_ids) == 2
