@@ -1,12 +1,13 @@
 #!/usr/bin/env python3
 """Flagship benchmark for the driver contract.
 
-Default: llama2-7b LoRA fine-tune step (DP over RCCL/xGMI), bf16,
-synthetic token data, random-init weights — the BASELINE.json metric
-"finetune samples/sec, llama2-7b at 1/2/4/8 MI355X" (the reference
-publishes no numbers; this bench ESTABLISHES the baseline,
-BASELINE.md).  --mode serve measures the companion metric
-"Server tokens/sec" (paged-KV decode throughput on the same model).
+Default (--mode both) emits BOTH BASELINE.json headline metrics as
+sequential JSON lines: the llama2-7b LoRA fine-tune step (DP over
+RCCL/xGMI, "finetune samples/sec") first, then "Server tokens/sec"
+(paged-KV batch-32 decode on the same model), bf16, synthetic token
+data, random-init weights. The reference publishes no numbers; this
+bench ESTABLISHES the baseline. --mode train / serve
+select a single metric.
 
 Launch (driver):
   python bench.py --gpus 1 --steps K --warmup W
@@ -15,7 +16,8 @@ Launch (driver):
 
 Scaling: train is weak (per-GPU micro-batch fixed, DP over RCCL);
 serve under torchrun is strong (TP: rank 0 drives the engine, worker
-ranks follow the broadcast protocol). Rank 0 prints ONE JSON line.
+ranks follow the broadcast protocol). Rank 0 prints one JSON line per
+metric.
 """
 from __future__ import annotations
 
@@ -121,9 +123,12 @@ def bench_serve(args) -> dict:
             return {}
     vocab = eng.cfg.vocab_size
     g = torch.Generator().manual_seed(17)
+    # leave decode headroom within the model's positional range (matters
+    # for the tiny CPU default, max_seq_len=128)
+    prompt_len = min(args.prompt_len, eng.cfg.max_seq_len // 2)
 
     def new_req():
-        prompt = torch.randint(0, vocab, (args.prompt_len,), generator=g).tolist()
+        prompt = torch.randint(0, vocab, (prompt_len,), generator=g).tolist()
         eng.submit(prompt, max_new_tokens=1 << 30)  # run until bench ends
 
     for _ in range(args.serve_batch):
@@ -161,7 +166,7 @@ def bench_serve(args) -> dict:
                  if torch.cuda.is_available() else "float32",
         "data": "synthetic",
         "config": {"model": args.model, "global_batch": args.serve_batch,
-                   "seq_len": args.prompt_len,
+                   "seq_len": prompt_len,
                    "parallelism": f"tp{comm.world_size()}"},
     }
 
@@ -171,7 +176,8 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--mode", choices=["train", "serve"], default="train")
+    p.add_argument("--mode", choices=["both", "train", "serve"],
+                   default="both")
     p.add_argument("--model", default=None)
     p.add_argument("--seq-len", type=int, default=512)
     p.add_argument("--micro-batch", type=int, default=4)
@@ -182,14 +188,36 @@ def main():
                         "config; reported with dtype=bf16-act+fp8-w)")
     args = p.parse_args()
     if args.model is None:
-        args.model = "llama2-7b" if torch.cuda.is_available() else "tiny-llama"
+        if torch.cuda.is_available():
+            args.model = "llama2-7b"
+        else:
+            # CPU rehearsal: pick a tiny model whose heads split across
+            # the requested TP degree (driver scale run goes to ws=8)
+            ws = int(os.environ.get("WORLD_SIZE", "1"))
+            args.model = "tiny-llama" if ws <= 2 else "tiny-llama-8h"
 
     comm.init_from_env()
     torch.manual_seed(17)
 
-    result = bench_train(args) if args.mode == "train" else bench_serve(args)
+    # Default emits BOTH headline metrics sequentially (BASELINE.json:
+    # "Server tokens/sec + finetune samples/sec") — one JSON line each,
+    # finetune first, serve last.
+    results = []
+    if args.mode in ("both", "train"):
+        results.append(bench_train(args))
+    if args.mode in ("both", "serve"):
+        if args.mode == "both":
+            # release train model/optimizer before the engine allocates
+            # weights + KV cache
+            import gc
+            gc.collect()
+            if torch.cuda.is_available():
+                torch.cuda.empty_cache()
+        results.append(bench_serve(args))
     if comm.rank() == 0:
-        print(json.dumps(result), flush=True)
+        for r in results:
+            if r:
+                print(json.dumps(r), flush=True)
     if comm.is_dist():
         torch.distributed.destroy_process_group()
 

@@ -163,6 +163,11 @@ register(ModelConfig("tiny-falcon", vocab_size=256, hidden_size=64,
                      intermediate_size=128, head_dim=8, max_seq_len=128,
                      norm="layernorm", act="gelu", parallel_residual=True,
                      single_norm=True, tie_embeddings=True))
+# 8 heads / 8 kv-heads so CPU torchrun rehearsals of the driver's TP=8
+# scale command work on gloo (tiny-llama's 4 heads can't split 8 ways)
+register(ModelConfig("tiny-llama-8h", vocab_size=256, hidden_size=64,
+                     num_layers=2, num_heads=8, num_kv_heads=8,
+                     intermediate_size=128, head_dim=8, max_seq_len=128))
 register(ModelConfig("tiny-qwen", vocab_size=256, hidden_size=64,
                      num_layers=2, num_heads=4, num_kv_heads=2,
                      intermediate_size=128, max_seq_len=128,
