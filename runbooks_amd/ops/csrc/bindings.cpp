@@ -38,6 +38,9 @@ at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 at::Tensor mfma_probe_16x16x32(at::Tensor a, at::Tensor b);
 at::Tensor train_gemm_nt(at::Tensor a, at::Tensor b);
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
+at::Tensor decode_gemm(at::Tensor x, at::Tensor w);
+int64_t decode_gemm_split(int64_t N, int64_t K);
+bool decode_gemm_supported(int64_t M, int64_t N, int64_t K);
 at::Tensor skinny_gemm_fp8(at::Tensor x, at::Tensor w8, at::Tensor scale);
 int64_t skinny_gemm_mmax();
 at::Tensor qkv_rope_append(at::Tensor y, at::Tensor cos, at::Tensor sin,
@@ -75,6 +78,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm,
         "HBM-rate decode GEMM: y[M<=32,N] = x @ W^T (bf16)");
   m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
+  m.def("decode_gemm", &decode_gemm,
+        "v2 weight-stream decode GEMM: y[M<=32,N] = x @ W^T (bf16), "
+        "W straight to MFMA fragments with nt loads");
+  m.def("decode_gemm_split", &decode_gemm_split);
+  m.def("decode_gemm_supported", &decode_gemm_supported);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32,
         "layout probe: one 16x16x32 bf16 MFMA as a plain matmul");
   m.def("train_gemm_nt", &train_gemm_nt,

@@ -14,9 +14,11 @@ import torch.nn.functional as F
 
 from . import _backend
 
-# The custom kernel currently trails hipBLASLt on most decode shapes
-# (see profiles/README.md measurements); keep it opt-in until it wins.
+# Round-1 tile kernel; superseded by decode_gemm (kept for A/B runs).
 _USE_SKINNY = os.environ.get("RB_SKINNY_GEMM", "0") == "1"
+# v2 weight-stream kernel (csrc/decode_gemm.hip): default-on; RB_DECODE_GEMM=0
+# falls back to hipBLASLt.
+_USE_DECODE_GEMM = os.environ.get("RB_DECODE_GEMM", "1") == "1"
 
 
 # data_ptr(weight) -> (w8 uint8 view, f32 per-channel scale). Populated by
@@ -51,13 +53,19 @@ def fast_linear(x: torch.Tensor, weight: torch.Tensor,
         k = x.shape[-1]
         m = x.numel() // k
         n = weight.shape[0]
-        if m <= 32 and n % 64 == 0 and k % 256 == 0:
-            q = _FP8_REGISTRY.get(weight.data_ptr())
-            if q is not None:
-                y = _backend.ext().skinny_gemm_fp8(
-                    x.reshape(m, k).contiguous(), q[0], q[1])
+        if m <= 32:
+            if n % 64 == 0 and k % 256 == 0:
+                q = _FP8_REGISTRY.get(weight.data_ptr())
+                if q is not None:
+                    y = _backend.ext().skinny_gemm_fp8(
+                        x.reshape(m, k).contiguous(), q[0], q[1])
+                    return y.view(*x.shape[:-1], n)
+            if _USE_DECODE_GEMM and k >= 1024 and n % 32 == 0 and \
+                    k % 16 == 0:
+                y = _backend.ext().decode_gemm(
+                    x.reshape(m, k).contiguous(), weight)
                 return y.view(*x.shape[:-1], n)
-            if _USE_SKINNY:
+            if _USE_SKINNY and n % 64 == 0 and k % 256 == 0:
                 y = _backend.ext().skinny_gemm(x.reshape(m, k).contiguous(),
                                                weight)
                 return y.view(*x.shape[:-1], n)

@@ -80,3 +80,39 @@ def test_gemma_engine_decode_dh256():
                  seed=2).generate([5, 9, 2, 7], max_new_tokens=8)
     # greedy bf16-vs-fp32 may diverge late; the prefix must agree
     assert out[:4] == ref[:4], (out, ref)
+
+
+def test_prefix_cache_gpu_matches_uncached():
+    """Prefix caching on GPU: shared-prompt requests must decode the
+    same tokens as an uncached engine (the kernels only ever see a block
+    table, so this validates the bookkeeping end-to-end on silicon)."""
+    assert ops.has_hip()
+    m = build_model("smoke-llama", dtype=torch.bfloat16, seed=4)
+    sys_prompt = list(range(1, 33))          # two full 16-token chunks
+    tails = [[40 + i, 41 + i, 42 + i] for i in range(3)]
+
+    base = Engine(m, device=DEV, kv_blocks=128, seed=11, prefix_cache=False)
+    want = [base.generate(sys_prompt + t, max_new_tokens=6) for t in tails]
+
+    pc = Engine(m, device=DEV, kv_blocks=128, seed=11, prefix_cache=True)
+    got = [pc.generate(sys_prompt + t, max_new_tokens=6) for t in tails]
+    assert got == want, (got, want)
+    assert pc.stats["prefix_hits"] >= 2, pc.stats
+    assert pc.stats["prefix_hit_blocks"] >= 4, pc.stats
+
+
+def test_sliding_window_gpu_generates():
+    """Windowed decode on GPU: engine trims front KV blocks while the
+    graphed decode keeps producing valid tokens."""
+    import dataclasses
+    from runbooks_amd.models import get_config
+    from runbooks_amd.models.config import register
+    assert ops.has_hip()
+    cfg = dataclasses.replace(get_config("smoke-llama"),
+                              name="smoke-window", sliding_window=64)
+    register(cfg)
+    m = build_model("smoke-window", dtype=torch.bfloat16, seed=4)
+    eng = Engine(m, device=DEV, kv_blocks=256, seed=11)
+    out = eng.generate(list(range(1, 81)), max_new_tokens=48)
+    assert len(out) == 48 and all(0 <= t < m.cfg.vocab_size for t in out)
+    assert eng.stats["window_dropped_blocks"] > 0

@@ -502,3 +502,37 @@ def test_paged_decode_dh256():
                                vc.float().cpu(), bt.cpu(), sl.cpu(),
                                scale=1.0 / 16.0)
     assert torch.allclose(got.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("M,N,K", [
+    (32, 4096, 4096),     # o_proj: cross-WG split path (slab + combine)
+    (32, 12288, 4096),    # fused qkv: split=1, direct bf16 store
+    (32, 4096, 11008),    # down_proj: split=4, k-step tail (688 % 16 != 0)
+    (32, 22016, 4096),    # fused gate/up
+    (17, 11008, 4096),    # M < 32 (clamped x rows dropped at epilogue)
+    (1, 4096, 4096),      # single-sequence decode
+    (32, 32000, 4096),    # lm_head
+])
+def test_decode_gemm(M, N, K):
+    """csrc/decode_gemm.hip (v2 weight-stream kernel) vs fp32 matmul."""
+    _assert_hip()
+    torch.manual_seed(M * 31 + N)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+    y = ops.ext().decode_gemm(x, w)
+    assert y.shape == (M, N) and y.dtype == torch.bfloat16
+    ref = x.cpu().float() @ w.cpu().float().t()
+    d = (y.cpu().float() - ref).abs().max().item()
+    rel = d / ref.abs().max().item()
+    assert rel < 2e-2, f"max abs {d} rel {rel}"
+
+
+def test_decode_gemm_split_heuristic():
+    """Split choice covers the 256 CUs on every flagship decode shape."""
+    _assert_hip()
+    e = ops.ext()
+    for n, k in [(12288, 4096), (4096, 4096), (22016, 4096),
+                 (4096, 11008), (32000, 4096)]:
+        split = e.decode_gemm_split(n, k)
+        assert (n // 32) * split >= 256, (n, k, split)
+        assert (k // 16) % 1 == 0
