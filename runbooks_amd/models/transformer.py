@@ -162,10 +162,8 @@ class MLP(nn.Module):
                     y[:, :half].contiguous(), y[:, half:].contiguous()))
             return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
         if self.act == "gelu_glu":
-            # gemma GeGLU (tanh approximation). The fused gate/up path
-            # (csrc geglu_packed) exists but fuse_for_inference only
-            # enables it under RB_FUSED_GEGLU=1 until a GPU validation
-            # pass; eager torch is the default.
+            # gemma GeGLU (tanh approximation); fused gate/up GEMM +
+            # csrc geglu_packed epilogue (GPU-validated r2).
             if getattr(self, "_gateup_w", None) is not None and \
                     not torch.is_grad_enabled():
                 y = ops.fast_linear(x, self._gateup_w)
@@ -370,9 +368,11 @@ def fuse_for_inference(model: "Transformer",
                                               requires_grad=False)
         mlp = blk.mlp
         import os as _os
+        # geglu_packed GPU-validated (GPUTEST r2: vs fp32 tanh-gelu ref);
+        # default-on, RB_FUSED_GEGLU=0 reverts to eager
         glu_fusable = model.cfg.act == "silu_glu" or (
             model.cfg.act == "gelu_glu" and
-            _os.environ.get("RB_FUSED_GEGLU", "0") == "1")
+            _os.environ.get("RB_FUSED_GEGLU", "1") == "1")
         if glu_fusable and mlp.gate_proj.bias is None:
             fused = torch.cat([mlp.gate_proj.weight.data,
                                mlp.up_proj.weight.data],

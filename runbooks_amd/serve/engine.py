@@ -123,6 +123,16 @@ class Engine:
         self.seed = seed
         if kv_blocks is None:
             kv_blocks = self._auto_kv_blocks(mem_fraction)
+        if comm.is_dist() and comm.world_size() > 1 and model.tp > 1:
+            # Every rank must agree on the pool size: rank 0's scheduler
+            # hands out slot indices that worker ranks index their OWN
+            # caches with — a rank with less free memory would otherwise
+            # read/write out of bounds. Take the fleet minimum.
+            t = torch.tensor([kv_blocks], dtype=torch.int64,
+                             device=self.device if
+                             torch.cuda.is_available() else "cpu")
+            torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MIN)
+            kv_blocks = int(t.item())
         # last block is reserved as the dummy block hipGraph-padded decode
         # rows read/write (serve/graph.py)
         self.allocator = BlockAllocator(kv_blocks - 1)
@@ -157,9 +167,11 @@ class Engine:
         # identical; the kernels are untouched, only the block table
         # changes). Bounds cache memory for common-system-prompt serving.
         import os as _os
+        # default-on since the GPU validation pass (r2: shared-prompt
+        # decode parity vs an uncached engine); RB_PREFIX_CACHE=0 disables
         self.prefix_cache_enabled = (
             prefix_cache if prefix_cache is not None
-            else _os.environ.get("RB_PREFIX_CACHE", "0") == "1")
+            else _os.environ.get("RB_PREFIX_CACHE", "1") == "1")
         from collections import OrderedDict
         self._pc: "OrderedDict[tuple, int]" = OrderedDict()
         self.stats = {"steps": 0, "prefills": 0, "decode_tokens": 0,
@@ -500,6 +512,10 @@ class Engine:
                 if r not in self.running:
                     continue
                 while r.seq_len - r.dropped >= len(r.blocks) * self.bs:
+                    if not self.allocator.free and self.prefix_cache_enabled:
+                        # idle cached prefix blocks are reclaimable memory:
+                        # evict before resorting to preemption
+                        self._evict_prefix(1)
                     if self.allocator.free:
                         r.blocks.extend(self.allocator.alloc(1))
                     elif not self._preempt_newest(exclude=r):

@@ -219,8 +219,8 @@ def test_trainer_main_end_to_end(tmp_path, monkeypatch):
 
 
 def test_bench_contract_cpu(tmp_path):
-    """bench.py emits exactly one driver-contract JSON line for both
-    modes on CPU (tiny model)."""
+    """bench.py default emits BOTH driver-contract JSON lines (finetune
+    first, serve last — BASELINE.json names both headline metrics)."""
     import json as _json
     import subprocess
     import sys
@@ -230,12 +230,16 @@ def test_bench_contract_cpu(tmp_path):
          "--micro-batch", "2", "--seq-len", "16"],
         capture_output=True, text=True, timeout=240)
     assert out.returncode == 0, out.stderr[-800:]
-    line = out.stdout.strip().splitlines()[-1]
-    d = _json.loads(line)
+    lines = [ln for ln in out.stdout.strip().splitlines() if ln.startswith("{")]
+    assert len(lines) == 2
+    d = _json.loads(lines[0])
     assert d["metric"] == "finetune_samples_per_sec"
     assert d["n_gpus"] == 1 and d["steps"] == 2 and d["scaling"] == "weak"
     assert set(d) >= {"value", "unit", "ms_per_step", "higher_is_better",
                       "vs_baseline", "dtype", "data", "config"}
+    d2 = _json.loads(lines[1])
+    assert d2["metric"] == "serve_tokens_per_sec"
+    assert d2["config"]["parallelism"] == "tp1"
 
     out = subprocess.run(
         [sys.executable, "bench.py", "--mode", "serve", "--steps", "3",

@@ -130,6 +130,7 @@ def test_kv_preemption_under_pressure():
     # total generated = 24 per request even across preemptions
     assert all(len(r.prompt_ids) + r.max_new_tokens == 30 + 24 or
                len(r.output_ids) == r.max_new_tokens for r in reqs)
+    eng.flush_prefix_cache()   # cache refs are intentional retention
     assert len(eng.allocator.free) == eng.allocator.num_blocks
 
 
@@ -221,11 +222,20 @@ def test_engine_scheduler_fuzz(seed):
                 eng.step()
             except RuntimeError:
                 pass  # oversize prompt rejected
-        # invariant: block ownership disjoint and in-range
+        # invariants (refcount-aware: prefix cache shares blocks):
+        # every reference is accounted for, free list = refs==0 exactly
         owned = [b for r in eng.running for b in r.blocks]
-        assert len(owned) == len(set(owned)), "double-owned KV block"
         assert all(0 <= b < eng.allocator.num_blocks for b in owned)
-        assert len(owned) + len(eng.allocator.free) == \
+        from collections import Counter
+        held = Counter(owned)
+        held.update(eng._pc.values())
+        for b, n in held.items():
+            assert eng.allocator.refs[b] == n, f"refcount drift on {b}"
+        live_set = set(held)
+        assert not live_set & set(eng.allocator.free), "freed live block"
+        assert len(eng.allocator.free) == \
+            sum(1 for r in eng.allocator.refs if r == 0), "free-list drift"
+        assert len(eng.allocator.free) + len(live_set) == \
             eng.allocator.num_blocks, "block leak"
     # drain
     for _ in range(3000):
@@ -236,6 +246,7 @@ def test_engine_scheduler_fuzz(seed):
         except RuntimeError:
             pass
     assert not eng.has_work(), "scheduler did not terminate"
+    eng.flush_prefix_cache()
     assert len(eng.allocator.free) == eng.allocator.num_blocks
     for r in live:
         assert r.finished
@@ -613,11 +624,12 @@ def test_admission_does_not_starve_decode():
 
 def test_prefix_cache_env_gate(monkeypatch):
     m = build_model("tiny-llama", dtype=torch.float32)
-    assert not Engine(m, device="cpu", dtype=torch.float32,
-                      kv_blocks=32).prefix_cache_enabled
-    monkeypatch.setenv("RB_PREFIX_CACHE", "1")
+    # default ON since the r2 GPU validation pass
     assert Engine(m, device="cpu", dtype=torch.float32,
                   kv_blocks=32).prefix_cache_enabled
+    monkeypatch.setenv("RB_PREFIX_CACHE", "0")
+    assert not Engine(m, device="cpu", dtype=torch.float32,
+                      kv_blocks=32).prefix_cache_enabled
     # explicit arg beats the env
     assert not Engine(m, device="cpu", dtype=torch.float32, kv_blocks=32,
                       prefix_cache=False).prefix_cache_enabled

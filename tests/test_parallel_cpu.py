@@ -399,3 +399,126 @@ def test_dp_save_resume():
 
 def test_tp_engine_prefix_cache():
     _run_dist(_tp_engine_prefix_cache, port=PORT + 13)
+
+
+# --- round-2: ws=4/8 hardening (driver scale-run rehearsal shapes) ----------
+
+def _dp_grads_average_ws8(rank, world):
+    """DP bucket overlap correctness at world_size 8 (the driver's
+    N=8 scale shape): all-reduced grads equal the 8-way average."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.parallel import DataParallel
+
+    torch.manual_seed(0)
+    model = build_model("tiny-llama-8h", dtype=torch.float32, tp=1)
+    ddp = DataParallel(model, bucket_mb=1)
+    torch.manual_seed(900 + rank)
+    x = torch.randint(0, 256, (2, 9))
+    logits = ddp(x[:, :-1])
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, logits.shape[-1]), x[:, 1:].reshape(-1))
+    loss.backward()
+    ddp.finish_backward()
+
+    ref = build_model("tiny-llama-8h", dtype=torch.float32, tp=1)
+    ref.load_state_dict(model.state_dict())
+    grads = {}
+    for r in range(world):
+        ref.zero_grad()
+        torch.manual_seed(900 + r)
+        xr = torch.randint(0, 256, (2, 9))
+        lg = ref(xr[:, :-1])
+        ls = torch.nn.functional.cross_entropy(
+            lg.reshape(-1, lg.shape[-1]), xr[:, 1:].reshape(-1))
+        ls.backward()
+        for n, p in ref.named_parameters():
+            grads[n] = grads.get(n, 0) + p.grad / world
+    for n, p in model.named_parameters():
+        assert torch.allclose(p.grad, grads[n], atol=1e-5), n
+
+
+def _tp4_engine_uneven_admission(rank, world):
+    """TP=4 serving under a ragged request pattern: requests of different
+    lengths admitted/finished across steps (prefills interleave with
+    decodes), then a second wave after the first drains — exercises the
+    broadcast protocol through admission, completion and idle gaps."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.serve import Engine
+    from runbooks_amd.serve.tp_worker import broadcast_shutdown, worker_loop
+
+    torch.manual_seed(0)
+    tp_model = build_model("tiny-llama-8h", dtype=torch.float32, tp=world,
+                           seed=8)
+    single = build_model("tiny-llama-8h", dtype=torch.float32, tp=1, seed=8)
+    sd = single.state_dict()
+    for name, t in tp_model.state_dict().items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:
+            t.copy_(full[rank * t.shape[0]:(rank + 1) * t.shape[0]])
+        else:
+            t.copy_(full[:, rank * t.shape[1]:(rank + 1) * t.shape[1]])
+
+    eng = Engine(tp_model, device="cpu", kv_blocks=64, seed=31, max_batch=3)
+    if rank != 0:
+        worker_loop(eng)
+        return
+    prompts = [[1 + i, 2, 3 + i] * (1 + i % 3) for i in range(5)]
+    lens = [3 + i for i in range(5)]
+    reqs = [eng.submit(list(p), max_new_tokens=n)
+            for p, n in zip(prompts[:3], lens[:3])]
+    # drain the first wave, then submit the second mid-flight
+    for step in range(30):
+        eng.step()
+        if step == 4:
+            reqs += [eng.submit(list(p), max_new_tokens=n)
+                     for p, n in zip(prompts[3:], lens[3:])]
+        if all(r.finished for r in reqs) and not eng.has_work():
+            break
+    broadcast_shutdown(torch.device("cpu"))
+    assert all(r.finished for r in reqs)
+    ref = Engine(single, device="cpu", kv_blocks=64, seed=31, max_batch=3)
+    ref.tp = 1
+    ref_out = []
+    rreqs = [ref.submit(list(p), max_new_tokens=n)
+             for p, n in zip(prompts[:3], lens[:3])]
+    for step in range(30):
+        ref.step()
+        if step == 4:
+            rreqs += [ref.submit(list(p), max_new_tokens=n)
+                      for p, n in zip(prompts[3:], lens[3:])]
+        if all(r.finished for r in rreqs) and not ref.has_work():
+            break
+    for r, rr in zip(reqs, rreqs):
+        assert r.output_ids == rr.output_ids, (r.request_id, r.output_ids,
+                                               rr.output_ids)
+
+
+def _tp_kv_pool_agreement(rank, world):
+    """Ranks that would auto-size different KV pools must agree on the
+    minimum (slot indices from rank 0 index every rank's own cache)."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.serve import Engine
+
+    torch.manual_seed(0)
+    m = build_model("tiny-llama-8h", dtype=torch.float32, tp=world, seed=3)
+    # simulate uneven free memory: each rank asks for a different size
+    eng = Engine(m, device="cpu", kv_blocks=48 + 8 * rank, seed=1)
+    sizes = torch.tensor([eng.allocator.num_blocks + 1], dtype=torch.int64)
+    gathered = [torch.empty_like(sizes) for _ in range(world)]
+    dist.all_gather(gathered, sizes)
+    assert all(int(g) == int(gathered[0]) for g in gathered), gathered
+    assert int(gathered[0]) == 48  # the fleet minimum
+
+
+def test_dp_ws8_gradient_allreduce():
+    _run_dist(_dp_grads_average_ws8, world=8, port=PORT + 20)
+
+
+def test_tp4_engine_uneven_admission():
+    _run_dist(_tp4_engine_uneven_admission, world=4, port=PORT + 21)
+
+
+def test_tp4_kv_pool_agreement():
+    _run_dist(_tp_kv_pool_agreement, world=4, port=PORT + 22)

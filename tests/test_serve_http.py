@@ -210,6 +210,7 @@ def test_concurrent_mixed_requests_stress(client):
         import time as _t
         _t.sleep(0.02)
     assert not eng.has_work()
+    eng.flush_prefix_cache()   # cached prefixes are retained on purpose
     assert len(eng.allocator.free) == eng.allocator.num_blocks
     assert not client.app.state.engine_loop._watchers
 
