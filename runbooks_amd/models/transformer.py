@@ -384,6 +384,34 @@ def fuse_for_inference(model: "Transformer",
             mlp.up_proj.weight = nn.Parameter(fused[half:],
                                               requires_grad=False)
 
+    if not load_in_8bit and torch.cuda.is_available() and ops.has_hip() and \
+            _os.environ.get("RB_DECODE_GEMM", "1") == "1":
+        # Register fragment-lane-major weight copies for the v2 decode
+        # GEMM (csrc/decode_gemm.hip): doubles decode-weight memory, which
+        # 288 GB HBM3E affords for every BASELINE config at TP=1 except
+        # llama2-70b (which serves via MODEL_LOAD_IN_8BIT / TP>1). Budget
+        # guard: keep at least ~35% of the GPU free for KV cache.
+        from ..ops.linear import register_decode_weight
+        ext = ops.ext()
+        cands = []
+        for blk in model.blocks:
+            for w in (getattr(blk.attn, "_qkv_w", None),
+                      getattr(blk.mlp, "_gateup_w", None),
+                      blk.attn.o_proj.weight,
+                      blk.mlp.down_proj.weight):
+                if w is not None and ext.decode_gemm_supported(
+                        32, w.shape[0], w.shape[1]):
+                    cands.append(w)
+        if model.lm_head.weight.shape[0] % 32 == 0 and \
+                ext.decode_gemm_supported(32, model.lm_head.weight.shape[0],
+                                          model.lm_head.weight.shape[1]):
+            cands.append(model.lm_head.weight)
+        need = sum(2 * w.numel() for w in cands)
+        free, total = torch.cuda.mem_get_info()
+        if free - need >= int(0.35 * total):
+            for w in cands:
+                register_decode_weight(w)
+
     if load_in_8bit and torch.cuda.is_available():
         # MODEL_LOAD_IN_8BIT: decode weights additionally stored as OCP
         # e4m3 + per-channel scales — the decode GEMM streams half the

@@ -38,7 +38,9 @@ at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 at::Tensor mfma_probe_16x16x32(at::Tensor a, at::Tensor b);
 at::Tensor train_gemm_nt(at::Tensor a, at::Tensor b);
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
-at::Tensor decode_gemm(at::Tensor x, at::Tensor w);
+at::Tensor decode_gemm(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N, int64_t K);
+at::Tensor decode_swizzle_w(at::Tensor w);
+at::Tensor decode_swizzle_x(at::Tensor x);
 int64_t decode_gemm_split(int64_t N, int64_t K);
 bool decode_gemm_supported(int64_t M, int64_t N, int64_t K);
 at::Tensor skinny_gemm_fp8(at::Tensor x, at::Tensor w8, at::Tensor scale);
@@ -79,8 +81,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "HBM-rate decode GEMM: y[M<=32,N] = x @ W^T (bf16)");
   m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
   m.def("decode_gemm", &decode_gemm,
-        "v2 weight-stream decode GEMM: y[M<=32,N] = x @ W^T (bf16), "
-        "W straight to MFMA fragments with nt loads");
+        "v2 weight-stream decode GEMM over pre-swizzled operands: "
+        "y[M<=32,N] = x @ W^T (bf16), 1KiB coalesced nt weight bursts");
+  m.def("decode_swizzle_w", &decode_swizzle_w,
+        "one-time [N,K] -> fragment-lane-major weight layout");
+  m.def("decode_swizzle_x", &decode_swizzle_x,
+        "per-call x -> B-fragment lane order (zero-padded to M=32)");
   m.def("decode_gemm_split", &decode_gemm_split);
   m.def("decode_gemm_supported", &decode_gemm_supported);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32,

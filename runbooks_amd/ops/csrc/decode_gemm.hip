@@ -1,39 +1,39 @@
 // Decode GEMM v2 for CDNA4 (gfx950): y[M,N] = x[M,K] @ W[N,K]^T,
 // M <= 32 (decode batch), bf16 in / bf16 out, fp32 accumulate.
 //
-// Replaces the round-1 skinny_gemm tile design (64x256 LDS-staged tiles,
-// K/256-way cross-WG split with fp32 slab round trips) with the shape the
-// hardware actually wants for a pure weight stream (the M<=32 decode GEMM
-// is >99% W traffic; MI355X_MICROARCH "GEMV / M <= 16 decode weights:
-// load straight to VGPRs, deep unroll, late vmcnt"):
+// The M<=32 decode GEMM is >99% W traffic, so the ONLY thing that
+// matters is shaping the weight read as perfectly coalesced nontemporal
+// bursts with enough in flight (MI355X ~6.3 TB/s achievable; hipBLASLt
+// measured ~2.4 TB/s in the round-1 decode loop, and a fragment-shaped
+// direct-load variant measured ~1.5 TB/s because every wave instruction
+// became 32 scattered 32 B requests). This version therefore streams
+// BOTH operands from layouts pre-swizzled into MFMA fragment-lane order:
 //
+//  * W is stored fragment-major at model-load time (decode_swizzle_w:
+//    [N/32][K/16][lane][8] bf16) so each wave instruction is one 1 KiB
+//    contiguous nontemporal read. 288 GB HBM3E per GPU makes the second
+//    weight copy the right trade (same pattern as the fp8 registry);
+//    prefill/training keep using the original [N,K] tensor via hipBLASLt.
+//  * x (tiny: M*K <= 0.7 MB) is swizzled per call by decode_swizzle_x —
+//    one extra ~microsecond kernel — into [K/16][lane][8] with zero
+//    padding for m >= M; after that every wave reads the same contiguous
+//    L2-resident stream.
 //  * One wave owns 32 W rows x a DEEP contiguous k-range (K/(4*SPLIT)),
-//    so split-K slab traffic collapses: SPLIT is 1 for N>=8192 shapes and
-//    2-4 for the N=4096 shapes (vs K/256 = 16..43 slices before).
-//  * W streams straight into MFMA A-fragments with nontemporal 16 B/lane
-//    loads (nt: streamed-once data must not displace L2/L1 — the
-//    "nt-weights" row of the microarch price list). No LDS round trip,
-//    no __syncthreads in the main loop, so hipcc pipelines the 2x8-deep
-//    load ring freely (in-flight bytes per CU ~= waves x 16 x 1KB >> the
-//    ~9 KB Little's-law requirement at 24.6 GB/s/CU).
-//  * x fragments load from global per chunk: x is <=0.7 MB total and
-//    L2/L3-resident, and the 32x32x16 MFMA shape halves x traffic per W
-//    byte vs 16x16x32 (1 KB x per 1 KB W per instruction).
-//  * 4 waves of a WG take adjacent k-quarters; one 16 KB LDS reduction
-//    at the end combines them (still inside the workgroup: no
-//    inter-workgroup visibility protocol needed). SPLIT>1 adds fp32
-//    slabs [SPLIT,M,N] reduced by decode_gemm_combine (1-4 slabs, ~1 MB:
-//    negligible next to the 32-90 MB W stream).
+//    no LDS and no barrier in the main loop (hipcc pipelines the 2x8
+//    ping-pong register ring freely; ~16 KB in flight per wave vs the
+//    ~9 KB/CU Little's-law requirement at 24.6 GB/s/CU).
+//  * 4 waves per WG take adjacent k-quarters; one 16 KB LDS reduction
+//    combines them. SPLIT (1 for N>=16k, 2-4 for N=4096 shapes) adds
+//    fp32 slabs [SPLIT,M,N] reduced by a combine kernel (~1 MB next to
+//    the 32-90 MB W stream).
 //
-// Fragment maps (A = W rows so 16 B/lane stays row-contiguous; validated
-// on-GPU by the round-1 skinny_gemm tests and test_decode_gemm):
-//   v_mfma_f32_32x32x16_bf16: A lane l -> A[row=l&31][k=(l>>5)*8+i]
-//   B lane l -> B[k=(l>>5)*8+i][col=l&31];  C lane l ->
-//   C[row=(r&3)+8*(r>>2)+4*(l>>5)][col=l&31], r = 0..15.
-// With A=W (row=n), B=x^T (col=m): acc[r] = y[m=l&31][n=n_local(r)].
+// Fragment maps (v_mfma_f32_32x32x16_bf16, A=W so C cols = m):
+//   A lane l -> A[row=l&31][k=(l>>5)*8+i];  B lane l -> B[k][col=l&31];
+//   C lane l -> C[row=(r&3)+8*(r>>2)+4*(l>>5)][col=l&31], r = 0..15.
+// The swizzles bake exactly these maps into the storage order.
 //
 // Reference parity: replaces hipBLASLt for the decode hot loop the way
-// the reference's external server image relies on cuBLAS
+// the reference's external server image leans on cuBLAS
 // (substratusai/runbooks docs/container-contract.md serving contract).
 
 #include <torch/extension.h>
@@ -63,11 +63,13 @@ __device__ __forceinline__ bf16x8v load8v(const uint16_t *p) {
   return c.v;
 }
 
-// One chunk of U k-steps: W via nt (A operand), x via plain load (B).
+// One chunk of U k-steps. W nontemporal (streamed once, must not evict
+// L2); x plain (re-read by every n-block: L2 is exactly where it wants
+// to live). Both streams are lane*16B contiguous per instruction.
 #define RB_LOAD_CHUNK(WB, XB, SBASE)                                   \
   _Pragma("unroll") for (int u = 0; u < U; ++u) {                      \
-    WB[u] = nt_load8v(wrow + (int64_t)((SBASE) + u) * 16);             \
-    XB[u] = load8v(xrow + (int64_t)((SBASE) + u) * 16);                \
+    WB[u] = nt_load8v(wseg + (int64_t)((SBASE) + u) * 512 + lane8);    \
+    XB[u] = load8v(xs + (int64_t)((SBASE) + u) * 512 + lane8);         \
   }
 
 #define RB_MFMA_CHUNK(WB, XB)                                          \
@@ -77,10 +79,10 @@ __device__ __forceinline__ bf16x8v load8v(const uint16_t *p) {
   }
 
 // STORE_BF16: write y bf16 directly (SPLIT == 1). Otherwise store an
-// fp32 slab slice at slab + kslice*M*N for decode_gemm_combine.
+// fp32 slab slice at slabs + kslice*M*N for decode_gemm_combine.
 template <bool STORE_BF16>
 __global__ __launch_bounds__(BLOCK, 1) void decode_gemm_kernel(
-    const uint16_t *__restrict__ xp, const uint16_t *__restrict__ wp,
+    const uint16_t *__restrict__ xs, const uint16_t *__restrict__ ws,
     uint16_t *__restrict__ yp, float *__restrict__ slabs,
     int M, int N, int K) {
   const int tid = threadIdx.x;
@@ -88,13 +90,14 @@ __global__ __launch_bounds__(BLOCK, 1) void decode_gemm_kernel(
   const int lane = tid & 63;
   const int hi = lane >> 5;
   const int col = lane & 31;
+  const int lane8 = lane * 8;
 
   const int nblk = blockIdx.x;           // 32 W rows per WG
   const int kslice = blockIdx.y;
   const int split = gridDim.y;
 
   // k-step range (steps of 16) for this (kslice, wave) group; contiguous
-  // per group so each W row is one long sequential stream.
+  // per group so the W stream is one long sequential burst.
   const int steps_total = K / 16;
   const int ngroups = split * 4;
   const int spg = (steps_total + ngroups - 1) / ngroups;
@@ -102,12 +105,7 @@ __global__ __launch_bounds__(BLOCK, 1) void decode_gemm_kernel(
   const int s0 = g * spg;
   const int s1 = min(steps_total, s0 + spg);
 
-  const int n = nblk * 32 + col;
-  const int m = col;                     // B operand col = m
-  // m >= M reads row M-1 (clamped, in-bounds); its outputs are dropped
-  // in the epilogue, so no masking cost in the hot loop.
-  const uint16_t *wrow = wp + (int64_t)n * K + hi * 8;
-  const uint16_t *xrow = xp + (int64_t)min(m, M - 1) * K + hi * 8;
+  const uint16_t *wseg = ws + (int64_t)nblk * steps_total * 512;
 
   f32x16v acc = (f32x16v)(0.0f);
 
@@ -126,8 +124,8 @@ __global__ __launch_bounds__(BLOCK, 1) void decode_gemm_kernel(
   }
   // tail: single k-steps, no prefetch (<= 2U-1 iterations)
   for (; s < s1; ++s) {
-    bf16x8v w1 = nt_load8v(wrow + (int64_t)s * 16);
-    bf16x8v x1 = load8v(xrow + (int64_t)s * 16);
+    bf16x8v w1 = nt_load8v(wseg + (int64_t)s * 512 + lane8);
+    bf16x8v x1 = load8v(xs + (int64_t)s * 512 + lane8);
     acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(w1, x1, acc, 0, 0, 0);
   }
 
@@ -164,7 +162,7 @@ __global__ __launch_bounds__(BLOCK, 1) void decode_gemm_kernel(
   }
 }
 
-// slab [SPLIT, M, N] f32 -> y [M, N] bf16
+// slabs [SPLIT, M, N] f32 -> y [M, N] bf16
 __global__ void decode_gemm_combine_kernel(
     const float *__restrict__ slabs, uint16_t *__restrict__ yp,
     int split, int64_t mn) {
@@ -180,6 +178,33 @@ __global__ void decode_gemm_combine_kernel(
     o.u[0] = rb::f32_to_bf16(s.x); o.u[1] = rb::f32_to_bf16(s.y);
     o.u[2] = rb::f32_to_bf16(s.z); o.u[3] = rb::f32_to_bf16(s.w);
     *reinterpret_cast<uint64_t *>(yp + i) = o.q;
+  }
+}
+
+// x [M, K] row-major -> xs [K/16][64 lanes][8] bf16 (B-fragment lane
+// order: lane = hi*32 + m, elems = x[m][s*16 + hi*8 + i]; zeros for
+// m >= M). One scattered read pass of <= 0.7 MB, coalesced writes.
+__global__ void decode_swizzle_x_kernel(
+    const uint16_t *__restrict__ xp, uint16_t *__restrict__ xs,
+    int M, int K) {
+  const int steps = K / 16;
+  const int64_t total = (int64_t)steps * 64;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const int s = (int)(idx >> 6);
+    const int l = (int)(idx & 63);
+    const int m = l & 31;
+    const int hi = l >> 5;
+    rb::bf16x8 v;
+    if (m < M) {
+      v = *reinterpret_cast<const rb::bf16x8 *>(
+          xp + (int64_t)m * K + s * 16 + hi * 8);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v.v[e] = 0;
+    }
+    *reinterpret_cast<rb::bf16x8 *>(xs + idx * 8) = v;
   }
 }
 
@@ -201,33 +226,62 @@ bool decode_gemm_supported(int64_t M, int64_t N, int64_t K) {
   return M >= 1 && M <= MMAX && N % 32 == 0 && K % 16 == 0 && K >= 1024;
 }
 
-at::Tensor decode_gemm(at::Tensor x, at::Tensor w) {
-  TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.is_contiguous() &&
-              w.is_contiguous(), "decode_gemm: contiguous GPU tensors");
-  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
-              w.scalar_type() == at::kBFloat16, "decode_gemm: bf16 only");
-  const int M = x.size(0), K = x.size(1), N = w.size(0);
-  TORCH_CHECK((int)w.size(1) == K, "decode_gemm: K mismatch");
+// One-time weight swizzle (host-side tensor ops): [N,K] ->
+// [N/32][K/16][2][32][8] = fragment-lane-major chunks of 1 KiB.
+at::Tensor decode_swizzle_w(at::Tensor w) {
+  TORCH_CHECK(w.dim() == 2 && w.scalar_type() == at::kBFloat16 &&
+              w.is_contiguous(), "decode_swizzle_w: contiguous bf16 [N,K]");
+  const int64_t N = w.size(0), K = w.size(1);
+  TORCH_CHECK(N % 32 == 0 && K % 16 == 0, "decode_swizzle_w: shape");
+  return w.view({N / 32, 32, K / 16, 2, 8})
+      .permute({0, 2, 3, 1, 4}).contiguous();
+}
+
+at::Tensor decode_swizzle_x(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() &&
+              x.scalar_type() == at::kBFloat16, "decode_swizzle_x: x");
+  const int M = x.size(0), K = x.size(1);
+  TORCH_CHECK(M <= MMAX && K % 16 == 0, "decode_swizzle_x: shape");
+  auto xs = at::empty({(int64_t)(K / 16) * 512}, x.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int grid = rb::rb_grid_1d((int64_t)(K / 16) * 64, 256);
+  hipLaunchKernelGGL(decode_swizzle_x_kernel, dim3(grid), dim3(256), 0,
+                     stream, (const uint16_t *)x.data_ptr(),
+                     (uint16_t *)xs.data_ptr(), M, K);
+  return xs;
+}
+
+// xs from decode_swizzle_x, ws from decode_swizzle_w; M/N/K of the
+// ORIGINAL y[M,N] = x[M,K] @ W[N,K]^T problem.
+at::Tensor decode_gemm(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N,
+                       int64_t K) {
+  TORCH_CHECK(xs.is_cuda() && ws.is_cuda() && xs.is_contiguous() &&
+              ws.is_contiguous(), "decode_gemm: contiguous GPU tensors");
+  TORCH_CHECK(xs.scalar_type() == at::kBFloat16 &&
+              ws.scalar_type() == at::kBFloat16, "decode_gemm: bf16 only");
+  TORCH_CHECK(xs.numel() == (K / 16) * 512, "decode_gemm: xs size");
+  TORCH_CHECK(ws.numel() == N * K, "decode_gemm: ws size");
   TORCH_CHECK(decode_gemm_supported(M, N, K),
               "decode_gemm: unsupported shape ", M, "x", N, "x", K);
 
   auto stream = at::cuda::getCurrentHIPStream();
-  auto y = at::empty({M, N}, x.options());
+  auto y = at::empty({M, N}, xs.options());
   const int split = (int)decode_gemm_split(N, K);
   if (split == 1) {
     hipLaunchKernelGGL((decode_gemm_kernel<true>), dim3(N / 32, 1),
                        dim3(BLOCK), 0, stream,
-                       (const uint16_t *)x.data_ptr(),
-                       (const uint16_t *)w.data_ptr(),
-                       (uint16_t *)y.data_ptr(), nullptr, M, N, K);
+                       (const uint16_t *)xs.data_ptr(),
+                       (const uint16_t *)ws.data_ptr(),
+                       (uint16_t *)y.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K);
   } else {
-    auto slabs = at::empty({split, M, N}, x.options().dtype(at::kFloat));
+    auto slabs = at::empty({split, M, N}, xs.options().dtype(at::kFloat));
     hipLaunchKernelGGL((decode_gemm_kernel<false>), dim3(N / 32, split),
                        dim3(BLOCK), 0, stream,
-                       (const uint16_t *)x.data_ptr(),
-                       (const uint16_t *)w.data_ptr(), nullptr,
-                       (float *)slabs.data_ptr(), M, N, K);
-    const int64_t mn = (int64_t)M * N;
+                       (const uint16_t *)xs.data_ptr(),
+                       (const uint16_t *)ws.data_ptr(), nullptr,
+                       (float *)slabs.data_ptr(), (int)M, (int)N, (int)K);
+    const int64_t mn = M * N;
     const int grid = rb::rb_grid_1d(mn / 4, 256);
     hipLaunchKernelGGL(decode_gemm_combine_kernel, dim3(grid), dim3(256),
                        0, stream, (const float *)slabs.data_ptr(),

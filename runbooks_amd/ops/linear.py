@@ -26,6 +26,19 @@ _USE_DECODE_GEMM = os.environ.get("RB_DECODE_GEMM", "1") == "1"
 # fast path.
 _FP8_REGISTRY: dict[int, tuple[torch.Tensor, torch.Tensor]] = {}
 
+# data_ptr(weight) -> fragment-lane-major copy ([N/32][K/16][lane][8])
+# consumed by csrc/decode_gemm.hip. A second full-precision weight copy:
+# 288 GB HBM3E per GPU makes layout-specialized decode weights the right
+# trade on MI355X (prefill/training keep the original [N,K] tensor).
+_DECODE_W_REGISTRY: dict[int, tuple[torch.Tensor, tuple[int, int]]] = {}
+
+
+def register_decode_weight(weight: torch.Tensor) -> None:
+    """Build + register the decode-GEMM weight layout for `weight`."""
+    ws = _backend.ext().decode_swizzle_w(weight.data)
+    _DECODE_W_REGISTRY[weight.data_ptr()] = (
+        ws, (int(weight.shape[0]), int(weight.shape[1])))
+
 E4M3_MAX = 448.0
 
 
@@ -60,11 +73,13 @@ def fast_linear(x: torch.Tensor, weight: torch.Tensor,
                     y = _backend.ext().skinny_gemm_fp8(
                         x.reshape(m, k).contiguous(), q[0], q[1])
                     return y.view(*x.shape[:-1], n)
-            if _USE_DECODE_GEMM and k >= 1024 and n % 32 == 0 and \
-                    k % 16 == 0:
-                y = _backend.ext().decode_gemm(
-                    x.reshape(m, k).contiguous(), weight)
-                return y.view(*x.shape[:-1], n)
+            if _USE_DECODE_GEMM:
+                dw = _DECODE_W_REGISTRY.get(weight.data_ptr())
+                if dw is not None:
+                    xs = _backend.ext().decode_swizzle_x(
+                        x.reshape(m, k).contiguous())
+                    y = _backend.ext().decode_gemm(xs, dw[0], m, n, k)
+                    return y.view(*x.shape[:-1], n)
             if _USE_SKINNY and n % 64 == 0 and k % 256 == 0:
                 y = _backend.ext().skinny_gemm(x.reshape(m, k).contiguous(),
                                                weight)

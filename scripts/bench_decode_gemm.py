@@ -14,6 +14,11 @@ import argparse
 import json
 import time
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 import torch.nn.functional as F
 
@@ -55,8 +60,14 @@ def main():
         x = torch.randn(args.m, K, dtype=torch.bfloat16, device=dev)
 
         t_lib = bench(lambda i: F.linear(x, ws[i % n_w]), args.iters)
-        t_v2 = bench(lambda i: ops.ext().decode_gemm(x, ws[i % n_w]),
-                     args.iters)
+        swz = [ops.ext().decode_swizzle_w(w) for w in ws]
+        xs = ops.ext().decode_swizzle_x(x)
+        t_v2 = bench(lambda i: ops.ext().decode_gemm(
+            xs, swz[i % n_w], args.m, N, K), args.iters)
+        t_v2_full = bench(lambda i: ops.ext().decode_gemm(
+            ops.ext().decode_swizzle_x(x), swz[i % n_w], args.m, N, K),
+            args.iters)
+        del swz
         gb = N * K * 2 / 1e9
         row = {
             "N": N, "K": K, "M": args.m, "weights_gb": round(gb, 3),
@@ -64,7 +75,8 @@ def main():
             "hipblaslt_tbps": round(gb / t_lib / 1e3, 3),
             "decode_gemm_ms": round(t_v2 * 1e3, 4),
             "decode_gemm_tbps": round(gb / t_v2 / 1e3, 3),
-            "speedup": round(t_lib / t_v2, 3),
+            "with_xswizzle_ms": round(t_v2_full * 1e3, 4),
+            "speedup": round(t_lib / t_v2_full, 3),
             "split": int(ops.ext().decode_gemm_split(N, K)),
         }
         results.append(row)

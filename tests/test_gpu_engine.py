@@ -63,23 +63,52 @@ def test_fp8_engine_decode_runs():
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(
-    __import__("os").environ.get("RB_EXPERIMENTAL") != "1",
-    reason="RB_EXPERIMENTAL=1 only (gemma Dh=256 path awaits first "
-           "on-GPU validation: decode kernel + fused qkv epilogue)")
 def test_gemma_engine_decode_dh256():
-    """End-to-end gemma-family decode on GPU (Dh=256 instantiation +
-    fused qkv_rope_append + GeGLU MLP) vs the CPU fp32 engine."""
+    """gemma-family decode on GPU (Dh=256 paged decode + fused
+    qkv_rope_append + GeGLU MLP) vs the CPU fp32 model: compares decode
+    LOGITS with a tolerance. (Greedy token-prefix comparison is not
+    stable for this config: gemma's sqrt(hidden) embed scaling on
+    random-init weights makes the bf16 argmax coin-flip.)"""
+    from runbooks_amd.models.transformer import fuse_for_inference
+    assert ops.has_hip()
+    prompt = [5, 9, 2, 7]
     m = build_model("smoke-gemma", dtype=torch.bfloat16, device="cuda:0",
                     seed=11)
-    eng = Engine(m, device="cuda:0", kv_blocks=64, seed=2)
-    out = eng.generate([5, 9, 2, 7], max_new_tokens=8)
-
+    fuse_for_inference(m)
     cpu = build_model("smoke-gemma", dtype=torch.float32, seed=11)
-    ref = Engine(cpu, device="cpu", dtype=torch.float32, kv_blocks=64,
-                 seed=2).generate([5, 9, 2, 7], max_new_tokens=8)
-    # greedy bf16-vs-fp32 may diverge late; the prefix must agree
-    assert out[:4] == ref[:4], (out, ref)
+    bs = ops.BLOCK_SIZE
+    caches = m.alloc_caches(8, "cuda:0")
+    caches_c = cpu.alloc_caches(8, "cpu")
+    S = len(prompt)
+
+    def run(model, caches, dev, dt):
+        tokens = torch.tensor([prompt], dtype=torch.long, device=dev)
+        pos = torch.arange(S, dtype=torch.int32, device=dev)
+        slots = torch.arange(S, dtype=torch.int32, device=dev)
+        lp = model.prefill(tokens, pos, caches, slots)
+        # one decode step at position S
+        t = torch.tensor([3], dtype=torch.long, device=dev)
+        p = torch.tensor([S], dtype=torch.int32, device=dev)
+        sl = torch.tensor([S], dtype=torch.int32, device=dev)
+        bt = torch.tensor([[0, 1]], dtype=torch.int32, device=dev)
+        seq = torch.tensor([S + 1], dtype=torch.int32, device=dev)
+        ld = model.decode(t, p, caches, sl, bt, seq)
+        return lp.float().cpu(), ld.float().cpu()
+
+    with torch.no_grad():
+        lp_g, ld_g = run(m, caches, "cuda:0", torch.bfloat16)
+        lp_c, ld_c = run(cpu, caches_c, "cpu", torch.float32)
+    for got, ref in ((lp_g, lp_c), (ld_g, ld_c)):
+        scale = ref.abs().max().item()
+        assert (got - ref).abs().max().item() / scale < 3e-2, \
+            (got[:, :8], ref[:, :8])
+
+    # engine-level plumbing: Dh=256 graphed decode produces valid tokens
+    eng = Engine(build_model("smoke-gemma", dtype=torch.bfloat16,
+                             device="cuda:0", seed=11),
+                 device="cuda:0", kv_blocks=64, seed=2)
+    out = eng.generate(list(prompt), max_new_tokens=8)
+    assert len(out) == 8 and all(0 <= t < eng.cfg.vocab_size for t in out)
 
 
 def test_prefix_cache_gpu_matches_uncached():

@@ -514,17 +514,40 @@ def test_paged_decode_dh256():
     (32, 32000, 4096),    # lm_head
 ])
 def test_decode_gemm(M, N, K):
-    """csrc/decode_gemm.hip (v2 weight-stream kernel) vs fp32 matmul."""
+    """csrc/decode_gemm.hip (v2 swizzled weight-stream kernel) vs fp32
+    matmul, via the same swizzle helpers the serving path uses."""
     _assert_hip()
     torch.manual_seed(M * 31 + N)
     x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
     w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
-    y = ops.ext().decode_gemm(x, w)
+    ws = ops.ext().decode_swizzle_w(w)
+    xs = ops.ext().decode_swizzle_x(x)
+    y = ops.ext().decode_gemm(xs, ws, M, N, K)
     assert y.shape == (M, N) and y.dtype == torch.bfloat16
     ref = x.cpu().float() @ w.cpu().float().t()
     d = (y.cpu().float() - ref).abs().max().item()
     rel = d / ref.abs().max().item()
     assert rel < 2e-2, f"max abs {d} rel {rel}"
+
+
+def test_decode_gemm_via_fast_linear():
+    """fast_linear dispatches through the decode-weight registry and
+    matches F.linear."""
+    _assert_hip()
+    from runbooks_amd.ops.linear import _DECODE_W_REGISTRY, \
+        register_decode_weight
+    torch.manual_seed(3)
+    x = torch.randn(8, 4096, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(11008, 4096, dtype=torch.bfloat16, device=DEV)
+    register_decode_weight(w)
+    try:
+        with torch.no_grad():
+            y = ops.fast_linear(x, w)
+        ref = torch.nn.functional.linear(x, w)
+        d = (y.float() - ref.float()).abs().max() / ref.float().abs().max()
+        assert d < 2e-2, d
+    finally:
+        _DECODE_W_REGISTRY.clear()
 
 
 def test_decode_gemm_split_heuristic():
