@@ -156,6 +156,12 @@ class MLP(nn.Module):
                     not torch.is_grad_enabled():
                 y = ops.fast_linear(x, self._gateup_w)
                 if ops.use_hip(y):
+                    I = y.shape[-1] // 2
+                    if (y.dtype == torch.bfloat16 and I % 16 == 0 and
+                            y.numel() // y.shape[-1] <= 32):
+                        h, swz = ops.ext().swiglu_packed_dec(y)
+                        h._rb_swz = swz
+                        return self.down_proj(h)
                     return self.down_proj(ops.ext().swiglu_packed(y))
                 half = y.shape[-1] // 2
                 return self.down_proj(ops.swiglu(
@@ -168,6 +174,12 @@ class MLP(nn.Module):
                     not torch.is_grad_enabled():
                 y = ops.fast_linear(x, self._gateup_w)
                 if ops.use_hip(y):
+                    I = y.shape[-1] // 2
+                    if (y.dtype == torch.bfloat16 and I % 16 == 0 and
+                            y.numel() // y.shape[-1] <= 32):
+                        h, swz = ops.ext().geglu_packed_dec(y)
+                        h._rb_swz = swz
+                        return self.down_proj(h)
                     return self.down_proj(ops.ext().geglu_packed(y))
                 half = y.shape[-1] // 2
                 return self.down_proj(
@@ -207,6 +219,28 @@ class Block(nn.Module):
             return x + attn_fn(self.norm1(x)) + self.mlp(self.norm2(x))
         x = x + attn_fn(self.norm1(x))
         return x + self.mlp(self.norm2(x))
+
+    def forward_decode_fused(self, x_base, delta, attn_fn):
+        """Decode hot path (sequential-residual RMSNorm models): the
+        residual add is fused into the next norm's kernel
+        (rmsnorm_res_fwd_dec), so the layer's residual chain costs zero
+        standalone elementwise launches. Returns (new_base, new_delta)
+        with the true hidden state = new_base + new_delta (the caller
+        folds the final pending delta into norm_f)."""
+        ext = ops.ext()
+        eps = self.norm1.eps
+        if delta is None:
+            y1 = self.norm1(x_base)          # attaches _rb_swz itself
+            xr1 = x_base
+        else:
+            xr1, y1, s1 = ext.rmsnorm_res_fwd_dec(
+                x_base, delta, self.norm1.weight, eps)
+            y1._rb_swz = s1
+        a = attn_fn(y1)
+        xr2, y2, s2 = ext.rmsnorm_res_fwd_dec(xr1, a, self.norm2.weight,
+                                              self.norm2.eps)
+        y2._rb_swz = s2
+        return xr2, self.mlp(y2)
 
 
 class Transformer(nn.Module):
@@ -321,6 +355,25 @@ class Transformer(nn.Module):
         hipGraph capture, serve/graph.py)."""
         B = tokens.shape[0]
         x = self._embed(tokens.view(B, 1), positions)
+        import os as _os
+        if (ops.use_hip(x) and x.dtype == torch.bfloat16 and B <= 32 and
+                self.cfg.norm == "rmsnorm" and
+                not self.cfg.parallel_residual and not self.cfg.single_norm
+                and self.cfg.hidden_size % 16 == 0 and
+                _os.environ.get("RB_FUSED_RESID", "1") == "1"):
+            # fused residual chain: zero standalone adds in the hot loop
+            delta = None
+            for blk, (kc, vc) in zip(self.blocks, caches):
+                fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
+                    h, self.rope_cos, self.rope_sin, positions, kc, vc,
+                    slot_mapping, block_tables, seq_lens, nsplit=nsplit)
+                x, delta = blk.forward_decode_fused(x, delta, fn)
+            _, y, swz = ops.ext().rmsnorm_res_fwd_dec(
+                x, delta, self.norm_f.weight, self.norm_f.eps)
+            y._rb_swz = swz
+            logits = self.lm_head(y)
+            return (gather_from_tp(logits) if self._lm_vocab_parallel
+                    else logits)
         for blk, (kc, vc) in zip(self.blocks, caches):
             fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
                 h, self.rope_cos, self.rope_sin, positions, kc, vc, slot_mapping,

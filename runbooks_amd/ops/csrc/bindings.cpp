@@ -3,6 +3,10 @@
 #include <torch/extension.h>
 
 std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_fwd_dec(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_res_fwd_dec(at::Tensor x, at::Tensor res, at::Tensor w, double eps);
+std::vector<at::Tensor> swiglu_packed_dec(at::Tensor y);
+std::vector<at::Tensor> geglu_packed_dec(at::Tensor y);
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
                                     at::Tensor inv_rms);
 at::Tensor rope_fwd(at::Tensor x, at::Tensor cos, at::Tensor sin,
@@ -55,6 +59,14 @@ at::Tensor geglu_packed(at::Tensor y);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "runbooks_amd gfx950 HIP kernels";
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm forward (bf16/f32)");
+  m.def("rmsnorm_fwd_dec", &rmsnorm_fwd_dec,
+        "RMSNorm forward + decode-GEMM operand swizzle in one pass");
+  m.def("rmsnorm_res_fwd_dec", &rmsnorm_res_fwd_dec,
+        "residual add + RMSNorm + decode operand swizzle in one pass");
+  m.def("swiglu_packed_dec", &swiglu_packed_dec,
+        "packed SwiGLU + decode-GEMM operand swizzle in one pass");
+  m.def("geglu_packed_dec", &geglu_packed_dec,
+        "packed GeGLU + decode-GEMM operand swizzle in one pass");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm backward");
   m.def("rope_fwd", &rope_fwd, "RoPE rotate-half forward");
   m.def("rope_bwd", &rope_bwd, "RoPE rotate-half backward (inverse rotation)");

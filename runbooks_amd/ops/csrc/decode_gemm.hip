@@ -210,12 +210,14 @@ __global__ void decode_swizzle_x_kernel(
 
 }  // namespace
 
-// Pick a cross-WG k-split so the grid covers the 256 CUs (>=2 WGs per CU
-// where the shape allows; each WG is one CU-resident 4-wave block).
+// Pick a cross-WG k-split: just enough WGs to cover the 256 CUs. Split
+// costs a slab round trip + a combine kernel (~1.2-1.5 us boundary each
+// inside the captured decode graph), so stop as soon as the chip is
+// covered — N >= 8192 shapes run split-free.
 int64_t decode_gemm_split(int64_t N, int64_t K) {
   const int64_t nblocks = N / 32;
   int64_t split = 1;
-  while (split < 8 && nblocks * split < 512 &&
+  while (split < 8 && nblocks * split < 256 &&
          (K / 16) % (split * 2) == 0 && K / (split * 2) >= 512) {
     split *= 2;
   }

@@ -99,6 +99,7 @@ __global__ void qkv_rope_append_kernel(
 template <typename T>
 __global__ void swiglu_packed_kernel(const T *__restrict__ y,
                                      T *__restrict__ out,
+                                     uint16_t *__restrict__ swz,
                                      int64_t n_rows, int inter) {
   constexpr int W = rb::VIO<T>::W;
   const int nvec = inter / W;
@@ -117,6 +118,13 @@ __global__ void swiglu_packed_kernel(const T *__restrict__ y,
       g[k] = s * u[k];
     }
     rb::VIO<T>::store(out + row * inter + v * W, g);
+    if constexpr (sizeof(T) == 2) {
+      if (swz != nullptr) {
+        const int k0 = v * 8;
+        rb::VIO<T>::store(reinterpret_cast<T *>(swz) + (k0 >> 4) * 512 +
+                          ((k0 >> 3) & 1) * 256 + row * 8, g);
+      }
+    }
   }
 }
 
@@ -127,6 +135,7 @@ __global__ void swiglu_packed_kernel(const T *__restrict__ y,
 template <typename T>
 __global__ void geglu_packed_kernel(const T *__restrict__ y,
                                     T *__restrict__ out,
+                                    uint16_t *__restrict__ swz,
                                     int64_t n_rows, int inter) {
   constexpr int W = rb::VIO<T>::W;
   const int nvec = inter / W;
@@ -147,6 +156,13 @@ __global__ void geglu_packed_kernel(const T *__restrict__ y,
       g[k] = 0.5f * x * (1.0f + t) * u[k];
     }
     rb::VIO<T>::store(out + row * inter + v * W, g);
+    if constexpr (sizeof(T) == 2) {
+      if (swz != nullptr) {
+        const int k0 = v * 8;
+        rb::VIO<T>::store(reinterpret_cast<T *>(swz) + (k0 >> 4) * 512 +
+                          ((k0 >> 3) & 1) * 256 + row * 8, g);
+      }
+    }
   }
 }
 
@@ -197,12 +213,14 @@ at::Tensor swiglu_packed(at::Tensor y) {
     hipLaunchKernelGGL(swiglu_packed_kernel<uint16_t>, dim3(grid),
                        dim3(BLOCK), 0, stream,
                        (const uint16_t *)y.data_ptr(),
-                       (uint16_t *)out.data_ptr(), n_rows, inter);
+                       (uint16_t *)out.data_ptr(), (uint16_t *)nullptr,
+                       n_rows, inter);
   } else if (y.scalar_type() == at::kFloat) {
     TORCH_CHECK(inter % 4 == 0, "swiglu_packed f32: I % 4");
     hipLaunchKernelGGL(swiglu_packed_kernel<float>, dim3(grid), dim3(BLOCK),
                        0, stream, (const float *)y.data_ptr(),
-                       (float *)out.data_ptr(), n_rows, inter);
+                       (float *)out.data_ptr(), (uint16_t *)nullptr,
+                       n_rows, inter);
   } else {
     TORCH_CHECK(false, "swiglu_packed: dtype");
   }
@@ -223,14 +241,56 @@ at::Tensor geglu_packed(at::Tensor y) {
     hipLaunchKernelGGL(geglu_packed_kernel<uint16_t>, dim3(grid),
                        dim3(BLOCK), 0, stream,
                        (const uint16_t *)y.data_ptr(),
-                       (uint16_t *)out.data_ptr(), n_rows, inter);
+                       (uint16_t *)out.data_ptr(), (uint16_t *)nullptr,
+                       n_rows, inter);
   } else if (y.scalar_type() == at::kFloat) {
     TORCH_CHECK(inter % 4 == 0, "geglu_packed f32: I % 4");
     hipLaunchKernelGGL(geglu_packed_kernel<float>, dim3(grid), dim3(BLOCK),
                        0, stream, (const float *)y.data_ptr(),
-                       (float *)out.data_ptr(), n_rows, inter);
+                       (float *)out.data_ptr(), (uint16_t *)nullptr,
+                       n_rows, inter);
   } else {
     TORCH_CHECK(false, "geglu_packed: dtype");
   }
   return out;
+}
+
+// Decode variants: also emit the decode-GEMM pre-swizzled operand
+// ([I/16]x512 bf16; rows m >= n_rows uninitialized — dropped by the
+// GEMM epilogue). Saves the standalone x-swizzle launch per MLP.
+static std::vector<at::Tensor> glu_packed_dec(at::Tensor y, bool gelu) {
+  TORCH_CHECK(y.is_cuda() && y.is_contiguous() &&
+              y.scalar_type() == at::kBFloat16, "glu_packed_dec: bf16");
+  const int64_t inter2 = y.size(-1);
+  const int inter = (int)(inter2 / 2);
+  const int64_t n_rows = y.numel() / inter2;
+  TORCH_CHECK(n_rows <= 32 && inter % 16 == 0, "glu_packed_dec: shape");
+  auto sizes = y.sizes().vec();
+  sizes.back() = inter;
+  auto out = at::empty(sizes, y.options());
+  auto swz = at::empty({(int64_t)(inter / 16) * 512}, y.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int grid = rb::rb_grid_1d(n_rows * (inter / 8), BLOCK);
+  if (gelu) {
+    hipLaunchKernelGGL(geglu_packed_kernel<uint16_t>, dim3(grid),
+                       dim3(BLOCK), 0, stream,
+                       (const uint16_t *)y.data_ptr(),
+                       (uint16_t *)out.data_ptr(),
+                       (uint16_t *)swz.data_ptr(), n_rows, inter);
+  } else {
+    hipLaunchKernelGGL(swiglu_packed_kernel<uint16_t>, dim3(grid),
+                       dim3(BLOCK), 0, stream,
+                       (const uint16_t *)y.data_ptr(),
+                       (uint16_t *)out.data_ptr(),
+                       (uint16_t *)swz.data_ptr(), n_rows, inter);
+  }
+  return {out, swz};
+}
+
+std::vector<at::Tensor> swiglu_packed_dec(at::Tensor y) {
+  return glu_packed_dec(y, false);
+}
+
+std::vector<at::Tensor> geglu_packed_dec(at::Tensor y) {
+  return glu_packed_dec(y, true);
 }

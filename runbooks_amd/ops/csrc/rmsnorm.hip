@@ -18,11 +18,22 @@ namespace {
 
 constexpr int BLOCK = 256;
 
+// swz (optional, bf16 only): ALSO write the normed row into the decode
+// GEMM's B-fragment lane order ([D/16][2][32][8], csrc/decode_gemm.hip)
+// so the decode path needs no separate x-swizzle launch. Each thread's
+// 8-element vector IS one swizzled 16 B slot: pos = (k/16)*512 +
+// ((k/8)&1)*256 + row*8.
+// res/sum_out (optional, together): normalize x + res instead of x and
+// also write the sum — fuses the decode loop's residual add into the
+// norm (one launch instead of two; the sum feeds the NEXT residual).
 template <typename T>
 __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
+                                   const T *__restrict__ res,
+                                   T *__restrict__ sum_out,
                                    const T *__restrict__ w,
                                    T *__restrict__ y,
                                    float *__restrict__ inv_rms,
+                                   uint16_t *__restrict__ swz,
                                    int64_t n_rows, int D, float eps) {
   constexpr int W = rb::VIO<T>::W;
   __shared__ float red[BLOCK / RB_WAVE];
@@ -36,6 +47,14 @@ __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
     for (int i = threadIdx.x; i < nvec; i += BLOCK) {
       float f[W];
       rb::VIO<T>::load(xr + i * W, f);
+      if (res != nullptr) {
+        float r[W];
+        rb::VIO<T>::load(res + row * D + i * W, r);
+#pragma unroll
+        for (int k = 0; k < W; ++k) f[k] += r[k];
+        if (sum_out != nullptr)
+          rb::VIO<T>::store(sum_out + row * D + i * W, f);
+      }
 #pragma unroll
       for (int k = 0; k < W; ++k) ss += f[k] * f[k];
     }
@@ -46,10 +65,24 @@ __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
     for (int i = threadIdx.x; i < nvec; i += BLOCK) {
       float f[W], g[W];
       rb::VIO<T>::load(xr + i * W, f);     // L1-resident second read
+      if (res != nullptr) {
+        float r[W];
+        rb::VIO<T>::load(res + row * D + i * W, r);
+#pragma unroll
+        for (int k = 0; k < W; ++k) f[k] += r[k];
+      }
       rb::VIO<T>::load(w + i * W, g);
 #pragma unroll
       for (int k = 0; k < W; ++k) f[k] = f[k] * ir * g[k];
       rb::VIO<T>::store(yr + i * W, f);
+      if constexpr (sizeof(T) == 2) {
+        if (swz != nullptr) {
+          const int k0 = i * 8;
+          rb::VIO<T>::store(reinterpret_cast<T *>(swz) +
+                            (k0 >> 4) * 512 + ((k0 >> 3) & 1) * 256 +
+                            row * 8, f);
+        }
+      }
     }
     __syncthreads();
   }
@@ -131,19 +164,77 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
     TORCH_CHECK(D % 8 == 0, "rmsnorm bf16: D must be a multiple of 8");
     hipLaunchKernelGGL(rmsnorm_fwd_kernel<uint16_t>, dim3(std::min<int64_t>(n_rows, 2048)),
                        dim3(BLOCK), 0, stream,
-                       (const uint16_t *)x.data_ptr(), (const uint16_t *)w.data_ptr(),
-                       (uint16_t *)y.data_ptr(), inv_rms.data_ptr<float>(), n_rows, D,
-                       (float)eps);
+                       (const uint16_t *)x.data_ptr(),
+                       (const uint16_t *)nullptr, (uint16_t *)nullptr,
+                       (const uint16_t *)w.data_ptr(),
+                       (uint16_t *)y.data_ptr(), inv_rms.data_ptr<float>(),
+                       (uint16_t *)nullptr, n_rows, D, (float)eps);
   } else if (x.scalar_type() == at::kFloat) {
     TORCH_CHECK(D % 4 == 0, "rmsnorm f32: D must be a multiple of 4");
     hipLaunchKernelGGL(rmsnorm_fwd_kernel<float>, dim3(std::min<int64_t>(n_rows, 2048)),
                        dim3(BLOCK), 0, stream,
-                       x.data_ptr<float>(), w.data_ptr<float>(), y.data_ptr<float>(),
-                       inv_rms.data_ptr<float>(), n_rows, D, (float)eps);
+                       x.data_ptr<float>(), (const float *)nullptr,
+                       (float *)nullptr, w.data_ptr<float>(),
+                       y.data_ptr<float>(),
+                       inv_rms.data_ptr<float>(), (uint16_t *)nullptr,
+                       n_rows, D, (float)eps);
   } else {
     TORCH_CHECK(false, "rmsnorm: unsupported dtype");
   }
   return {y, inv_rms};
+}
+
+// Decode-path variant: returns {y, y_swz} where y_swz is the decode
+// GEMM's pre-swizzled B-operand ([D/16]x512 bf16, rows m >= n_rows left
+// uninitialized: decode_gemm drops their outputs at the epilogue).
+std::vector<at::Tensor> rmsnorm_fwd_dec(at::Tensor x, at::Tensor w,
+                                        double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous(),
+              "rmsnorm_fwd_dec: contiguous GPU tensors");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "rmsnorm_fwd_dec: bf16");
+  const int64_t n_rows = x.numel() / x.size(-1);
+  const int D = (int)x.size(-1);
+  TORCH_CHECK(n_rows <= 32 && D % 16 == 0, "rmsnorm_fwd_dec: shape");
+  auto y = at::empty_like(x);
+  auto swz = at::empty({(int64_t)(D / 16) * 512}, x.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  auto inv_rms = at::empty({n_rows}, x.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel<uint16_t>, dim3(n_rows),
+                     dim3(BLOCK), 0, stream,
+                     (const uint16_t *)x.data_ptr(),
+                     (const uint16_t *)nullptr, (uint16_t *)nullptr,
+                     (const uint16_t *)w.data_ptr(),
+                     (uint16_t *)y.data_ptr(), inv_rms.data_ptr<float>(),
+                     (uint16_t *)swz.data_ptr(), n_rows, D, (float)eps);
+  return {y, swz};
+}
+
+// Fused residual + norm + swizzle for the decode loop: returns
+// {sum = x + res, y = rmsnorm(sum) * w, y_swz}. One launch replaces the
+// eager residual add + the norm + the operand swizzle.
+std::vector<at::Tensor> rmsnorm_res_fwd_dec(at::Tensor x, at::Tensor res,
+                                            at::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && res.is_contiguous() &&
+              w.is_contiguous(), "rmsnorm_res_fwd_dec: contiguous");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              res.scalar_type() == at::kBFloat16, "rmsnorm_res_fwd_dec: bf16");
+  TORCH_CHECK(x.sizes() == res.sizes(), "rmsnorm_res_fwd_dec: shape");
+  const int64_t n_rows = x.numel() / x.size(-1);
+  const int D = (int)x.size(-1);
+  TORCH_CHECK(n_rows <= 32 && D % 16 == 0, "rmsnorm_res_fwd_dec: shape");
+  auto sum = at::empty_like(x);
+  auto y = at::empty_like(x);
+  auto swz = at::empty({(int64_t)(D / 16) * 512}, x.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel<uint16_t>, dim3(n_rows),
+                     dim3(BLOCK), 0, stream,
+                     (const uint16_t *)x.data_ptr(),
+                     (const uint16_t *)res.data_ptr(),
+                     (uint16_t *)sum.data_ptr(),
+                     (const uint16_t *)w.data_ptr(),
+                     (uint16_t *)y.data_ptr(), (float *)nullptr,
+                     (uint16_t *)swz.data_ptr(), n_rows, D, (float)eps);
+  return {sum, y, swz};
 }
 
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,

@@ -76,8 +76,13 @@ def fast_linear(x: torch.Tensor, weight: torch.Tensor,
             if _USE_DECODE_GEMM:
                 dw = _DECODE_W_REGISTRY.get(weight.data_ptr())
                 if dw is not None:
-                    xs = _backend.ext().decode_swizzle_x(
-                        x.reshape(m, k).contiguous())
+                    # producers (rmsnorm_fwd_dec, *_packed_dec) attach the
+                    # pre-swizzled operand; otherwise one tiny swizzle
+                    # kernel builds it here
+                    xs = getattr(x, "_rb_swz", None)
+                    if xs is None:
+                        xs = _backend.ext().decode_swizzle_x(
+                            x.reshape(m, k).contiguous())
                     y = _backend.ext().decode_gemm(xs, dw[0], m, n, k)
                     return y.view(*x.shape[:-1], n)
             if _USE_SKINNY and n % 64 == 0 and k % 256 == 0:

@@ -30,6 +30,16 @@ class _RMSNormFn(torch.autograd.Function):
 
 def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
     if _backend.use_hip(x):
+        if (not torch.is_grad_enabled() and x.dtype == torch.bfloat16
+                and x.numel() // x.shape[-1] <= 32
+                and x.shape[-1] % 16 == 0):
+            # decode-shaped rows: one pass also emits the decode GEMM's
+            # pre-swizzled operand (csrc/decode_gemm.hip), saving the
+            # standalone x-swizzle launch on the consumer side
+            y, swz = _backend.ext().rmsnorm_fwd_dec(
+                x.contiguous(), weight.contiguous(), eps)
+            y._rb_swz = swz
+            return y
         return _RMSNormFn.apply(x.contiguous(), weight.contiguous(), eps)
     return rmsnorm_ref(x, weight, eps)
 

@@ -76,6 +76,10 @@ def test_gemma_engine_decode_dh256():
                     seed=11)
     fuse_for_inference(m)
     cpu = build_model("smoke-gemma", dtype=torch.float32, seed=11)
+    # share the exact bf16 weight values: otherwise bf16-vs-fp32 INIT
+    # rounding (amplified by gemma's sqrt(hidden)=22.6 embed scale over 2
+    # layers) dominates the comparison and hides real kernel bugs
+    cpu.load_state_dict({k: v.float().cpu() for k, v in m.state_dict().items()})
     bs = ops.BLOCK_SIZE
     caches = m.alloc_caches(8, "cuda:0")
     caches_c = cpu.alloc_caches(8, "cpu")
@@ -145,3 +149,33 @@ def test_sliding_window_gpu_generates():
     out = eng.generate(list(range(1, 81)), max_new_tokens=48)
     assert len(out) == 48 and all(0 <= t < m.cfg.vocab_size for t in out)
     assert eng.stats["window_dropped_blocks"] > 0
+
+
+def test_fused_residual_decode_matches_unfused(monkeypatch):
+    """RB_FUSED_RESID decode path (residual folded into rmsnorm_res) vs
+    the generic block path: same logits to bf16 noise."""
+    assert ops.has_hip()
+    from runbooks_amd.models.transformer import fuse_for_inference
+    m = build_model("smoke-llama", dtype=torch.bfloat16, device=DEV, seed=4)
+    fuse_for_inference(m)
+    caches = m.alloc_caches(8, DEV)
+    prompt = [5, 3, 8, 1]
+    S = len(prompt)
+    tokens = torch.tensor([prompt], dtype=torch.long, device=DEV)
+    pos = torch.arange(S, dtype=torch.int32, device=DEV)
+    slots = torch.arange(S, dtype=torch.int32, device=DEV)
+    with torch.no_grad():
+        m.prefill(tokens, pos, caches, slots)
+        args = (torch.tensor([2], dtype=torch.long, device=DEV),
+                torch.tensor([S], dtype=torch.int32, device=DEV),
+                caches,
+                torch.tensor([S], dtype=torch.int32, device=DEV),
+                torch.tensor([[0, 1]], dtype=torch.int32, device=DEV),
+                torch.tensor([S + 1], dtype=torch.int32, device=DEV))
+        monkeypatch.setenv("RB_FUSED_RESID", "1")
+        l_fused = m.decode(*args).float().cpu()
+        monkeypatch.setenv("RB_FUSED_RESID", "0")
+        l_plain = m.decode(*args).float().cpu()
+    scale = l_plain.abs().max().item()
+    assert (l_fused - l_plain).abs().max().item() / scale < 1e-2, \
+        (l_fused[:, :8], l_plain[:, :8])
