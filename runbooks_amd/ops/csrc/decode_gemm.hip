@@ -210,18 +210,30 @@ __global__ void decode_swizzle_x_kernel(
 
 }  // namespace
 
-// Pick a cross-WG k-split: just enough WGs to cover the 256 CUs. Split
-// costs a slab round trip + a combine kernel (~1.2-1.5 us boundary each
-// inside the captured decode graph), so stop as soon as the chip is
-// covered — N >= 8192 shapes run split-free.
+// Pick the cross-WG k-split by LOAD BALANCE: the kernel is a pure
+// weight stream, so wall time = (WG waves over the 256 CUs) = ceil(
+// nWG/256) rounds, and utilization = nWG / (256 * rounds). Measured
+// r5 microbench tracks this exactly: 384 WGs -> 75% -> 4.2 TB/s,
+// 688 -> 90% -> 4.7, 1000 -> 98% -> 5.2. Score each legal split and
+// keep the best; ties go to fewer splits (a split adds a slab round
+// trip + combine kernel ~5-6 us inside the captured graph).
 int64_t decode_gemm_split(int64_t N, int64_t K) {
   const int64_t nblocks = N / 32;
-  int64_t split = 1;
-  while (split < 8 && nblocks * split < 256 &&
-         (K / 16) % (split * 2) == 0 && K / (split * 2) >= 512) {
-    split *= 2;
+  int64_t best = 1;
+  double best_util = 0.0;
+  for (int64_t s = 1; s <= 8; s *= 2) {
+    if (s > 1 && ((K / 16) % (s * 4) != 0 || K / s < 1024)) break;
+    const int64_t nwg = nblocks * s;
+    const int64_t rounds = (nwg + 255) / 256;
+    double util = (double)nwg / (double)(256 * rounds);
+    // discount the fixed slab+combine cost of splitting (~3% each)
+    util -= 0.03 * (s > 1 ? 1 : 0) + 0.005 * s;
+    if (util > best_util + 1e-9) {
+      best_util = util;
+      best = s;
+    }
   }
-  return split;
+  return best;
 }
 
 bool decode_gemm_supported(int64_t M, int64_t N, int64_t K) {
