@@ -116,7 +116,7 @@ class Attention(nn.Module):
         return self.o_proj(o.reshape(B * S, self.hq * self.dh))
 
     def forward_decode(self, x, cos, sin, positions, kc, vc, slot_mapping,
-                       block_tables, seq_lens, nsplit=None):
+                       block_tables, seq_lens, nsplit=None, slab_ok=False):
         B = x.shape[0]
         if getattr(self, "_qkv_w", None) is not None and \
                 self.cfg.pos == "rope" and ops.use_hip(x) and \
@@ -131,7 +131,11 @@ class Attention(nn.Module):
             ops.kv_append(k, v, kc, vc, slot_mapping)
         o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale,
                              nsplit=nsplit)
-        return self.o_proj(o.reshape(B, self.hq * self.dh))
+        o2 = o.reshape(B, self.hq * self.dh)
+        if slab_ok and self.tp == 1 and self.o_proj.bias is None:
+            out, _ = ops.decode_linear_raw(o2, self.o_proj.weight)
+            return out
+        return self.o_proj(o2)
 
 
 class MLP(nn.Module):
@@ -150,7 +154,7 @@ class MLP(nn.Module):
         self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size,
                                            bias=cfg.mlp_bias, tp_size=tp, dtype=dtype)
 
-    def forward(self, x):
+    def forward(self, x, slab_ok=False):
         if self.act == "silu_glu":
             if getattr(self, "_gateup_w", None) is not None and \
                     not torch.is_grad_enabled():
@@ -161,6 +165,10 @@ class MLP(nn.Module):
                             y.numel() // y.shape[-1] <= 32):
                         h, swz = ops.ext().swiglu_packed_dec(y)
                         h._rb_swz = swz
+                        if slab_ok and self.down_proj.bias is None:
+                            out, _ = ops.decode_linear_raw(
+                                h, self.down_proj.weight)
+                            return out
                         return self.down_proj(h)
                     return self.down_proj(ops.ext().swiglu_packed(y))
                 half = y.shape[-1] // 2
@@ -179,6 +187,10 @@ class MLP(nn.Module):
                             y.numel() // y.shape[-1] <= 32):
                         h, swz = ops.ext().geglu_packed_dec(y)
                         h._rb_swz = swz
+                        if slab_ok and self.down_proj.bias is None:
+                            out, _ = ops.decode_linear_raw(
+                                h, self.down_proj.weight)
+                            return out
                         return self.down_proj(h)
                     return self.down_proj(ops.ext().geglu_packed(y))
                 half = y.shape[-1] // 2
@@ -229,18 +241,22 @@ class Block(nn.Module):
         folds the final pending delta into norm_f)."""
         ext = ops.ext()
         eps = self.norm1.eps
+
+        def res_norm(base, d, w, e):
+            if d.dtype == torch.float32:      # uncombined split-K slab
+                return ext.rmsnorm_res_slab_fwd_dec(base, d, w, e)
+            return ext.rmsnorm_res_fwd_dec(base, d, w, e)
+
         if delta is None:
             y1 = self.norm1(x_base)          # attaches _rb_swz itself
             xr1 = x_base
         else:
-            xr1, y1, s1 = ext.rmsnorm_res_fwd_dec(
-                x_base, delta, self.norm1.weight, eps)
+            xr1, y1, s1 = res_norm(x_base, delta, self.norm1.weight, eps)
             y1._rb_swz = s1
         a = attn_fn(y1)
-        xr2, y2, s2 = ext.rmsnorm_res_fwd_dec(xr1, a, self.norm2.weight,
-                                              self.norm2.eps)
+        xr2, y2, s2 = res_norm(xr1, a, self.norm2.weight, self.norm2.eps)
         y2._rb_swz = s2
-        return xr2, self.mlp(y2)
+        return xr2, self.mlp(y2, slab_ok=True)
 
 
 class Transformer(nn.Module):
@@ -366,10 +382,15 @@ class Transformer(nn.Module):
             for blk, (kc, vc) in zip(self.blocks, caches):
                 fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
                     h, self.rope_cos, self.rope_sin, positions, kc, vc,
-                    slot_mapping, block_tables, seq_lens, nsplit=nsplit)
+                    slot_mapping, block_tables, seq_lens, nsplit=nsplit,
+                    slab_ok=True)
                 x, delta = blk.forward_decode_fused(x, delta, fn)
-            _, y, swz = ops.ext().rmsnorm_res_fwd_dec(
-                x, delta, self.norm_f.weight, self.norm_f.eps)
+            if delta.dtype == torch.float32:
+                _, y, swz = ops.ext().rmsnorm_res_slab_fwd_dec(
+                    x, delta, self.norm_f.weight, self.norm_f.eps)
+            else:
+                _, y, swz = ops.ext().rmsnorm_res_fwd_dec(
+                    x, delta, self.norm_f.weight, self.norm_f.eps)
             y._rb_swz = swz
             logits = self.lm_head(y)
             return (gather_from_tp(logits) if self._lm_vocab_parallel

@@ -9,7 +9,7 @@ from .activations import cross_entropy, swiglu  # noqa: F401
 from .adamw import FusedAdamW  # noqa: F401
 from .attention import causal_attention, flash_prefill, paged_decode, paged_decode_ref  # noqa: F401
 from .kvcache import BLOCK_SIZE, alloc_kv_cache, kv_append, kv_append_ref  # noqa: F401
-from .linear import Linear, fast_linear  # noqa: F401
+from .linear import Linear, decode_linear_raw, fast_linear  # noqa: F401
 from .norm import RMSNorm, rmsnorm, rmsnorm_ref  # noqa: F401
 from .rope import rope, rope_ref, rope_tables  # noqa: F401
 from .sampling import sample_tokens  # noqa: F401

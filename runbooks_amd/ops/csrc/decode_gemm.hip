@@ -218,22 +218,17 @@ __global__ void decode_swizzle_x_kernel(
 // keep the best; ties go to fewer splits (a split adds a slab round
 // trip + combine kernel ~5-6 us inside the captured graph).
 int64_t decode_gemm_split(int64_t N, int64_t K) {
+  // Split only when the grid cannot cover the 256 CUs (o_proj/down_proj
+  // N=4096 -> 128 WGs). Measured r5/r6: splitting an already-covering
+  // grid loses (gateup 688 WGs split 4: -10%; qkv 384 split 2 gained
+  // ~2% on the kernel but pays a combine launch inside the graph).
   const int64_t nblocks = N / 32;
-  int64_t best = 1;
-  double best_util = 0.0;
-  for (int64_t s = 1; s <= 8; s *= 2) {
-    if (s > 1 && ((K / 16) % (s * 4) != 0 || K / s < 1024)) break;
-    const int64_t nwg = nblocks * s;
-    const int64_t rounds = (nwg + 255) / 256;
-    double util = (double)nwg / (double)(256 * rounds);
-    // discount the fixed slab+combine cost of splitting (~3% each)
-    util -= 0.03 * (s > 1 ? 1 : 0) + 0.005 * s;
-    if (util > best_util + 1e-9) {
-      best_util = util;
-      best = s;
-    }
+  int64_t split = 1;
+  while (split < 8 && nblocks * split < 256 &&
+         (K / 16) % (split * 4) == 0 && K / (split * 2) >= 1024) {
+    split *= 2;
   }
-  return best;
+  return split;
 }
 
 bool decode_gemm_supported(int64_t M, int64_t N, int64_t K) {
@@ -263,6 +258,38 @@ at::Tensor decode_swizzle_x(at::Tensor x) {
                      stream, (const uint16_t *)x.data_ptr(),
                      (uint16_t *)xs.data_ptr(), M, K);
   return xs;
+}
+
+// Raw variant for fused consumers: split == 1 returns y bf16 [M, N];
+// split > 1 returns the fp32 slab [split, M, N] UNCOMBINED — the
+// consumer kernel (rmsnorm_res_slab_fwd_dec) folds the slices while it
+// reads, saving the combine launch inside the decode graph.
+at::Tensor decode_gemm_raw(at::Tensor xs, at::Tensor ws, int64_t M,
+                           int64_t N, int64_t K) {
+  TORCH_CHECK(xs.is_cuda() && ws.is_cuda() && xs.is_contiguous() &&
+              ws.is_contiguous() && xs.scalar_type() == at::kBFloat16 &&
+              ws.scalar_type() == at::kBFloat16, "decode_gemm_raw: inputs");
+  TORCH_CHECK(xs.numel() == (K / 16) * 512 && ws.numel() == N * K &&
+              decode_gemm_supported(M, N, K), "decode_gemm_raw: shape");
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int split = (int)decode_gemm_split(N, K);
+  if (split == 1) {
+    auto y = at::empty({M, N}, xs.options());
+    hipLaunchKernelGGL((decode_gemm_kernel<true>), dim3(N / 32, 1),
+                       dim3(BLOCK), 0, stream,
+                       (const uint16_t *)xs.data_ptr(),
+                       (const uint16_t *)ws.data_ptr(),
+                       (uint16_t *)y.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K);
+    return y;
+  }
+  auto slabs = at::empty({split, M, N}, xs.options().dtype(at::kFloat));
+  hipLaunchKernelGGL((decode_gemm_kernel<false>), dim3(N / 32, split),
+                     dim3(BLOCK), 0, stream,
+                     (const uint16_t *)xs.data_ptr(),
+                     (const uint16_t *)ws.data_ptr(), nullptr,
+                     (float *)slabs.data_ptr(), (int)M, (int)N, (int)K);
+  return slabs;
 }
 
 // xs from decode_swizzle_x, ws from decode_swizzle_w; M/N/K of the

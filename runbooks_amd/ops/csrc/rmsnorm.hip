@@ -26,9 +26,14 @@ constexpr int BLOCK = 256;
 // res/sum_out (optional, together): normalize x + res instead of x and
 // also write the sum — fuses the decode loop's residual add into the
 // norm (one launch instead of two; the sum feeds the NEXT residual).
+// res32/rsplit (optional, bf16 path): the residual arrives as the
+// UNCOMBINED fp32 split-K slab [rsplit, n_rows, D] straight from
+// decode_gemm_raw — folding the slices here saves the combine launch.
 template <typename T>
 __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
                                    const T *__restrict__ res,
+                                   const float *__restrict__ res32,
+                                   int rsplit,
                                    T *__restrict__ sum_out,
                                    const T *__restrict__ w,
                                    T *__restrict__ y,
@@ -47,11 +52,22 @@ __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
     for (int i = threadIdx.x; i < nvec; i += BLOCK) {
       float f[W];
       rb::VIO<T>::load(xr + i * W, f);
-      if (res != nullptr) {
-        float r[W];
-        rb::VIO<T>::load(res + row * D + i * W, r);
+      if (res != nullptr || res32 != nullptr) {
+        if (res != nullptr) {
+          float r[W];
+          rb::VIO<T>::load(res + row * D + i * W, r);
 #pragma unroll
-        for (int k = 0; k < W; ++k) f[k] += r[k];
+          for (int k = 0; k < W; ++k) f[k] += r[k];
+        } else {
+          for (int sl = 0; sl < rsplit; ++sl) {
+            const float *rp = res32 + ((int64_t)sl * n_rows + row) * D + i * W;
+#pragma unroll
+            for (int k = 0; k < W; k += 4) {
+              float4 a = *reinterpret_cast<const float4 *>(rp + k);
+              f[k] += a.x; f[k + 1] += a.y; f[k + 2] += a.z; f[k + 3] += a.w;
+            }
+          }
+        }
         if (sum_out != nullptr)
           rb::VIO<T>::store(sum_out + row * D + i * W, f);
       }
@@ -70,6 +86,15 @@ __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
         rb::VIO<T>::load(res + row * D + i * W, r);
 #pragma unroll
         for (int k = 0; k < W; ++k) f[k] += r[k];
+      } else if (res32 != nullptr) {
+        for (int sl = 0; sl < rsplit; ++sl) {
+          const float *rp = res32 + ((int64_t)sl * n_rows + row) * D + i * W;
+#pragma unroll
+          for (int k = 0; k < W; k += 4) {
+            float4 a = *reinterpret_cast<const float4 *>(rp + k);
+            f[k] += a.x; f[k + 1] += a.y; f[k + 2] += a.z; f[k + 3] += a.w;
+          }
+        }
       }
       rb::VIO<T>::load(w + i * W, g);
 #pragma unroll
@@ -165,7 +190,8 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
     hipLaunchKernelGGL(rmsnorm_fwd_kernel<uint16_t>, dim3(std::min<int64_t>(n_rows, 2048)),
                        dim3(BLOCK), 0, stream,
                        (const uint16_t *)x.data_ptr(),
-                       (const uint16_t *)nullptr, (uint16_t *)nullptr,
+                       (const uint16_t *)nullptr, (const float *)nullptr, 0,
+                       (uint16_t *)nullptr,
                        (const uint16_t *)w.data_ptr(),
                        (uint16_t *)y.data_ptr(), inv_rms.data_ptr<float>(),
                        (uint16_t *)nullptr, n_rows, D, (float)eps);
@@ -174,6 +200,7 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
     hipLaunchKernelGGL(rmsnorm_fwd_kernel<float>, dim3(std::min<int64_t>(n_rows, 2048)),
                        dim3(BLOCK), 0, stream,
                        x.data_ptr<float>(), (const float *)nullptr,
+                       (const float *)nullptr, 0,
                        (float *)nullptr, w.data_ptr<float>(),
                        y.data_ptr<float>(),
                        inv_rms.data_ptr<float>(), (uint16_t *)nullptr,
@@ -202,7 +229,8 @@ std::vector<at::Tensor> rmsnorm_fwd_dec(at::Tensor x, at::Tensor w,
   hipLaunchKernelGGL(rmsnorm_fwd_kernel<uint16_t>, dim3(n_rows),
                      dim3(BLOCK), 0, stream,
                      (const uint16_t *)x.data_ptr(),
-                     (const uint16_t *)nullptr, (uint16_t *)nullptr,
+                     (const uint16_t *)nullptr, (const float *)nullptr, 0,
+                     (uint16_t *)nullptr,
                      (const uint16_t *)w.data_ptr(),
                      (uint16_t *)y.data_ptr(), inv_rms.data_ptr<float>(),
                      (uint16_t *)swz.data_ptr(), n_rows, D, (float)eps);
@@ -230,6 +258,37 @@ std::vector<at::Tensor> rmsnorm_res_fwd_dec(at::Tensor x, at::Tensor res,
                      dim3(BLOCK), 0, stream,
                      (const uint16_t *)x.data_ptr(),
                      (const uint16_t *)res.data_ptr(),
+                     (const float *)nullptr, 0,
+                     (uint16_t *)sum.data_ptr(),
+                     (const uint16_t *)w.data_ptr(),
+                     (uint16_t *)y.data_ptr(), (float *)nullptr,
+                     (uint16_t *)swz.data_ptr(), n_rows, D, (float)eps);
+  return {sum, y, swz};
+}
+
+// Same, with the residual as an UNCOMBINED fp32 split-K slab
+// [split, n_rows, D] from decode_gemm_raw.
+std::vector<at::Tensor> rmsnorm_res_slab_fwd_dec(at::Tensor x,
+                                                 at::Tensor slabs,
+                                                 at::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && slabs.is_contiguous() &&
+              w.is_contiguous(), "rmsnorm_res_slab_fwd_dec: contiguous");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              slabs.scalar_type() == at::kFloat &&
+              slabs.dim() == 3, "rmsnorm_res_slab_fwd_dec: dtypes");
+  const int64_t n_rows = x.numel() / x.size(-1);
+  const int D = (int)x.size(-1);
+  TORCH_CHECK(slabs.size(1) == n_rows && slabs.size(2) == D &&
+              n_rows <= 32 && D % 16 == 0, "rmsnorm_res_slab_fwd_dec: shape");
+  auto sum = at::empty_like(x);
+  auto y = at::empty_like(x);
+  auto swz = at::empty({(int64_t)(D / 16) * 512}, x.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel<uint16_t>, dim3(n_rows),
+                     dim3(BLOCK), 0, stream,
+                     (const uint16_t *)x.data_ptr(),
+                     (const uint16_t *)nullptr,
+                     (const float *)slabs.data_ptr(), (int)slabs.size(0),
                      (uint16_t *)sum.data_ptr(),
                      (const uint16_t *)w.data_ptr(),
                      (uint16_t *)y.data_ptr(), (float *)nullptr,

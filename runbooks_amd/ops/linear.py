@@ -92,6 +92,25 @@ def fast_linear(x: torch.Tensor, weight: torch.Tensor,
     return F.linear(x, weight, bias)
 
 
+def decode_linear_raw(x, weight):
+    """Decode GEMM that may return an UNCOMBINED fp32 split-K slab
+    [split, M, N] (consumer kernels fold it — rmsnorm_res_slab_fwd_dec),
+    or a plain bf16 [M, N]. Falls back to fast_linear when the weight
+    has no registered decode layout. Returns (out, is_slab)."""
+    dw = _DECODE_W_REGISTRY.get(weight.data_ptr())
+    if dw is None or not _USE_DECODE_GEMM:
+        with torch.no_grad():
+            return fast_linear(x, weight), False
+    k = x.shape[-1]
+    m = x.numel() // k
+    n = weight.shape[0]
+    xs = getattr(x, "_rb_swz", None)
+    if xs is None:
+        xs = _backend.ext().decode_swizzle_x(x.reshape(m, k).contiguous())
+    out = _backend.ext().decode_gemm_raw(xs, dw[0], m, n, k)
+    return out, out.dtype == torch.float32
+
+
 class Linear(torch.nn.Linear):
     """nn.Linear with the decode-GEMM fast path."""
 
