@@ -496,14 +496,14 @@ def fuse_for_inference(model: "Transformer",
         for blk in model.blocks:
             for w in (getattr(blk.attn, "_qkv_w", None),
                       getattr(blk.mlp, "_gateup_w", None),
-                      blk.attn.o_proj.weight.data,
-                      blk.mlp.down_proj.weight.data):
+                      blk.attn.o_proj.weight,
+                      blk.mlp.down_proj.weight):
                 if w is not None and w.shape[0] % 64 == 0 and \
                         w.shape[1] % 256 == 0:
                     quantize_fp8(w)
         if model.lm_head.weight.shape[0] % 64 == 0 and \
                 model.lm_head.weight.shape[1] % 256 == 0:
-            quantize_fp8(model.lm_head.weight.data)
+            quantize_fp8(model.lm_head.weight)
     return model
 
 

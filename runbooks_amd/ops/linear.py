@@ -8,6 +8,7 @@ on hipBLASLt via F.linear — the library is the right tool for big GEMMs.
 from __future__ import annotations
 
 import os
+import weakref
 
 import torch
 import torch.nn.functional as F
@@ -30,25 +31,46 @@ _FP8_REGISTRY: dict[int, tuple[torch.Tensor, torch.Tensor]] = {}
 # consumed by csrc/decode_gemm.hip. A second full-precision weight copy:
 # 288 GB HBM3E per GPU makes layout-specialized decode weights the right
 # trade on MI355X (prefill/training keep the original [N,K] tensor).
-_DECODE_W_REGISTRY: dict[int, tuple[torch.Tensor, tuple[int, int]]] = {}
+# value: (swizzled copy, (N, K), weakref to the registered tensor).
+# data_ptr alone is NOT a safe key — a freed weight's address gets
+# recycled by later allocations, so every lookup verifies tensor
+# identity through the weakref (stale entries are dropped lazily).
+_DECODE_W_REGISTRY: dict[int, tuple] = {}
+
+
+def _registry_get(reg: dict, weight: torch.Tensor):
+    entry = reg.get(weight.data_ptr())
+    if entry is None:
+        return None
+    if entry[-1]() is not weight:
+        del reg[weight.data_ptr()]
+        return None
+    return entry
 
 
 def register_decode_weight(weight: torch.Tensor) -> None:
     """Build + register the decode-GEMM weight layout for `weight`."""
+    for ptr in [p for p, e in _DECODE_W_REGISTRY.items() if e[-1]() is None]:
+        del _DECODE_W_REGISTRY[ptr]          # purge dead entries
     ws = _backend.ext().decode_swizzle_w(weight.data)
     _DECODE_W_REGISTRY[weight.data_ptr()] = (
-        ws, (int(weight.shape[0]), int(weight.shape[1])))
+        ws, (int(weight.shape[0]), int(weight.shape[1])),
+        weakref.ref(weight))
 
 E4M3_MAX = 448.0
 
 
+@torch.no_grad()
 def quantize_fp8(weight: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     """Row-wise (per-output-channel) OCP e4m3 quantization; registers the
-    pair so fast_linear dispatches to the fp8 decode GEMM."""
+    pair so fast_linear dispatches to the fp8 decode GEMM. Pass the
+    PERSISTENT tensor object (Parameter / fused buffer), not .data —
+    lookups verify identity via weakref."""
     scale = weight.float().abs().amax(dim=1).clamp(min=1e-8) / E4M3_MAX
     w8 = (weight.float() / scale[:, None]).to(torch.float8_e4m3fn)
     w8_bytes = w8.view(torch.uint8).contiguous()
-    _FP8_REGISTRY[weight.data_ptr()] = (w8_bytes, scale.contiguous())
+    _FP8_REGISTRY[weight.data_ptr()] = (w8_bytes, scale.contiguous(),
+                                        weakref.ref(weight))
     return w8_bytes, scale
 
 
@@ -68,13 +90,13 @@ def fast_linear(x: torch.Tensor, weight: torch.Tensor,
         n = weight.shape[0]
         if m <= 32:
             if n % 64 == 0 and k % 256 == 0:
-                q = _FP8_REGISTRY.get(weight.data_ptr())
+                q = _registry_get(_FP8_REGISTRY, weight)
                 if q is not None:
                     y = _backend.ext().skinny_gemm_fp8(
                         x.reshape(m, k).contiguous(), q[0], q[1])
                     return y.view(*x.shape[:-1], n)
             if _USE_DECODE_GEMM:
-                dw = _DECODE_W_REGISTRY.get(weight.data_ptr())
+                dw = _registry_get(_DECODE_W_REGISTRY, weight)
                 if dw is not None:
                     # producers (rmsnorm_fwd_dec, *_packed_dec) attach the
                     # pre-swizzled operand; otherwise one tiny swizzle
@@ -97,7 +119,7 @@ def decode_linear_raw(x, weight):
     [split, M, N] (consumer kernels fold it — rmsnorm_res_slab_fwd_dec),
     or a plain bf16 [M, N]. Falls back to fast_linear when the weight
     has no registered decode layout. Returns (out, is_slab)."""
-    dw = _DECODE_W_REGISTRY.get(weight.data_ptr())
+    dw = _registry_get(_DECODE_W_REGISTRY, weight)
     if dw is None or not _USE_DECODE_GEMM:
         with torch.no_grad():
             return fast_linear(x, weight), False
