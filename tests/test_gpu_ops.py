@@ -455,7 +455,10 @@ def test_mfma_16x16x32_layout_probe():
     (256, 256, 128),      # single tile, 2 K-tiles (minimum)
     (256, 256, 512),
     (512, 768, 1024),     # multi-tile, XCD remap with nwg%8 != 0
-    (2048, 4096, 4096),   # the training shape
+    (512, 768, 4096),     # bisect: deep K alone
+    (2048, 4096, 1024),   # bisect: big M*N alone
+    (256, 256, 4096),     # bisect: single tile, deep K
+    (2048, 4096, 4096),   # the training shape (failed r1: rel 1.29)
 ])
 def test_train_gemm_nt(M, N, K):
     _assert_hip()
