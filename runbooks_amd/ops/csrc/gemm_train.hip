@@ -225,7 +225,6 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         bf[1][fn][ks] = lds_read_frag(B, b_byte(2 + fn, ks));
-    prefetch_one(1);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_setprio(1);
@@ -246,7 +245,6 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         af[fm][ks] = lds_read_frag(A, a_byte(4 + fm, ks));
-    prefetch_one(2);
     asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -263,6 +261,15 @@ __global__ __launch_bounds__(THREADS, 1) void gemm_nt_8phase_kernel(
     __builtin_amdgcn_s_barrier();
 
     // ---- phase 4: reuse A(fmh1) + B(fnh0); vmcnt gate for next K-tile ---
+    // The three stages of tile kt+2 land in db(kt)'s LDS image, which
+    // phases 1-3 of THIS tile still read — issuing them any earlier is a
+    // data race (r1 shipped them in phases 2/3: wrong outputs on every
+    // shape big enough for the glds to land before phase 3's reads,
+    // GPUTEST r9 bisect). Phase 4 reads no LDS (registers only) and
+    // starts after the phase-3-end barrier, so every wave's reads of
+    // db(kt) are complete here.
+    prefetch_one(1);
+    prefetch_one(2);
     prefetch_one(3);
     if (kt + 1 < ktiles) {
       // gate the NEXT K-tile's data: leave in flight only half-tiles
