@@ -114,11 +114,89 @@ __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
 }
 
 // Backward: one workgroup per row (grid sized to give every row its own
-// WG), dx in one pass; dw accumulated per-block in LDS (each thread owns
-// fixed columns across its rows -> no LDS contention), one global fp32
-// atomicAdd per column per block at the end. A wave-per-row variant with
-// register caching measured 2.4x SLOWER (105 vs 43 us on [2048, 4096])
-// — likely spilling the cached vectors — so this shape stays.
+// WG), dx in one pass. dw: each thread's column set is FIXED across its
+// rows (i strides by BLOCK), so the per-thread dw partials live in
+// REGISTERS (<= RB_RN_MAXV vectors = 32 floats at D=8192) and hit
+// global memory once via atomicAdd at the end. The round-1 version
+// accumulated dw element-wise in LDS — 16 scalar ds ops per 16 B of
+// input made LDS the bottleneck (65 us on [2048,4096] vs ~14 ideal).
+// D > 8*BLOCK*RB_RN_MAXV falls back to the LDS path.
+#define RB_RN_MAXV 4
+
+// Register-dw variant (D <= 8 * BLOCK * RB_RN_MAXV).
+template <typename T>
+__global__ void rmsnorm_bwd_reg_kernel(const T *__restrict__ x,
+                                       const T *__restrict__ w,
+                                       const T *__restrict__ dy,
+                                       const float *__restrict__ inv_rms,
+                                       T *__restrict__ dx,
+                                       float *__restrict__ dw,
+                                       int64_t n_rows, int D) {
+  constexpr int W = rb::VIO<T>::W;
+  __shared__ float red[BLOCK / RB_WAVE];
+  const int nvec = D / W;
+
+  float dwacc[RB_RN_MAXV][W];
+#pragma unroll
+  for (int v = 0; v < RB_RN_MAXV; ++v)
+#pragma unroll
+    for (int k = 0; k < W; ++k) dwacc[v][k] = 0.0f;
+
+  // cache w once per thread (fixed columns)
+  float wf[RB_RN_MAXV][W];
+#pragma unroll
+  for (int v = 0; v < RB_RN_MAXV; ++v) {
+    const int i = v * BLOCK + threadIdx.x;
+    if (i < nvec) rb::VIO<T>::load(w + i * W, wf[v]);
+  }
+
+  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const T *xr = x + row * D;
+    const T *dyr = dy + row * D;
+    T *dxr = dx + row * D;
+    const float ir = inv_rms[row];
+
+    float xf[RB_RN_MAXV][W], df[RB_RN_MAXV][W];
+    float s = 0.0f;
+#pragma unroll
+    for (int v = 0; v < RB_RN_MAXV; ++v) {
+      const int i = v * BLOCK + threadIdx.x;
+      if (i < nvec) {
+        rb::VIO<T>::load(xr + i * W, xf[v]);
+        rb::VIO<T>::load(dyr + i * W, df[v]);
+#pragma unroll
+        for (int k = 0; k < W; ++k) s += df[v][k] * wf[v][k] * xf[v][k];
+      }
+    }
+    s = rb::block_reduce_sum(s, red);
+    const float c = ir * ir * ir * s / (float)D;
+
+#pragma unroll
+    for (int v = 0; v < RB_RN_MAXV; ++v) {
+      const int i = v * BLOCK + threadIdx.x;
+      if (i < nvec) {
+        float o[W];
+#pragma unroll
+        for (int k = 0; k < W; ++k) {
+          o[k] = ir * wf[v][k] * df[v][k] - c * xf[v][k];
+          dwacc[v][k] += df[v][k] * xf[v][k] * ir;
+        }
+        rb::VIO<T>::store(dxr + i * W, o);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int v = 0; v < RB_RN_MAXV; ++v) {
+    const int i = v * BLOCK + threadIdx.x;
+    if (i < nvec) {
+#pragma unroll
+      for (int k = 0; k < W; ++k)
+        if (dwacc[v][k] != 0.0f) atomicAdd(dw + i * W + k, dwacc[v][k]);
+    }
+  }
+}
+
 template <typename T>
 __global__ void rmsnorm_bwd_kernel(const T *__restrict__ x,
                                            const T *__restrict__ w,
@@ -309,15 +387,31 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
   TORCH_CHECK(shmem <= 160 * 1024, "rmsnorm_bwd: D too large for LDS accumulation");
 
   if (x.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL(rmsnorm_bwd_kernel<uint16_t>, dim3(nwg), dim3(BLOCK), shmem, stream,
-                       (const uint16_t *)x.data_ptr(), (const uint16_t *)w.data_ptr(),
-                       (const uint16_t *)dy.data_ptr(), inv_rms.data_ptr<float>(),
-                       (uint16_t *)dx.data_ptr(), dw.data_ptr<float>(), n_rows, D);
+    if (D <= 8 * BLOCK * RB_RN_MAXV) {
+      hipLaunchKernelGGL(rmsnorm_bwd_reg_kernel<uint16_t>, dim3(nwg),
+                         dim3(BLOCK), 0, stream,
+                         (const uint16_t *)x.data_ptr(), (const uint16_t *)w.data_ptr(),
+                         (const uint16_t *)dy.data_ptr(), inv_rms.data_ptr<float>(),
+                         (uint16_t *)dx.data_ptr(), dw.data_ptr<float>(), n_rows, D);
+    } else {
+      hipLaunchKernelGGL(rmsnorm_bwd_kernel<uint16_t>, dim3(nwg), dim3(BLOCK), shmem, stream,
+                         (const uint16_t *)x.data_ptr(), (const uint16_t *)w.data_ptr(),
+                         (const uint16_t *)dy.data_ptr(), inv_rms.data_ptr<float>(),
+                         (uint16_t *)dx.data_ptr(), dw.data_ptr<float>(), n_rows, D);
+    }
   } else if (x.scalar_type() == at::kFloat) {
-    hipLaunchKernelGGL(rmsnorm_bwd_kernel<float>, dim3(nwg), dim3(BLOCK), shmem, stream,
-                       x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
-                       inv_rms.data_ptr<float>(), dx.data_ptr<float>(), dw.data_ptr<float>(),
-                       n_rows, D);
+    if (D <= 4 * BLOCK * RB_RN_MAXV) {
+      hipLaunchKernelGGL(rmsnorm_bwd_reg_kernel<float>, dim3(nwg),
+                         dim3(BLOCK), 0, stream,
+                         x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
+                         inv_rms.data_ptr<float>(), dx.data_ptr<float>(), dw.data_ptr<float>(),
+                         n_rows, D);
+    } else {
+      hipLaunchKernelGGL(rmsnorm_bwd_kernel<float>, dim3(nwg), dim3(BLOCK), shmem, stream,
+                         x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
+                         inv_rms.data_ptr<float>(), dx.data_ptr<float>(), dw.data_ptr<float>(),
+                         n_rows, D);
+    }
   } else {
     TORCH_CHECK(false, "rmsnorm_bwd: unsupported dtype");
   }
