@@ -121,10 +121,13 @@ __global__ void rmsnorm_fwd_kernel(const T *__restrict__ x,
 // accumulated dw element-wise in LDS — 16 scalar ds ops per 16 B of
 // input made LDS the bottleneck (65 us on [2048,4096] vs ~14 ideal).
 // D > 8*BLOCK*RB_RN_MAXV falls back to the LDS path.
-#define RB_RN_MAXV 4
-
-// Register-dw variant (D <= 8 * BLOCK * RB_RN_MAXV).
-template <typename T>
+// Register-dw variant, D == NV * BLOCK * W exactly (llama 4096 -> NV=2
+// bf16). NV is a template int so every load is UNGUARDED — a runtime
+// `if (i < nvec)` around unrolled loads makes hipcc branch per element
+// and drain vmcnt(0) each time (cdna_hip_programming.md §5 trap (c));
+// the first cut of this kernel did exactly that and ran 7x slower than
+// the LDS version it meant to replace (r11: 472 us vs 65).
+template <typename T, int NV>
 __global__ void rmsnorm_bwd_reg_kernel(const T *__restrict__ x,
                                        const T *__restrict__ w,
                                        const T *__restrict__ dy,
@@ -134,21 +137,17 @@ __global__ void rmsnorm_bwd_reg_kernel(const T *__restrict__ x,
                                        int64_t n_rows, int D) {
   constexpr int W = rb::VIO<T>::W;
   __shared__ float red[BLOCK / RB_WAVE];
-  const int nvec = D / W;
 
-  float dwacc[RB_RN_MAXV][W];
+  float dwacc[NV][W];
 #pragma unroll
-  for (int v = 0; v < RB_RN_MAXV; ++v)
+  for (int v = 0; v < NV; ++v)
 #pragma unroll
     for (int k = 0; k < W; ++k) dwacc[v][k] = 0.0f;
 
-  // cache w once per thread (fixed columns)
-  float wf[RB_RN_MAXV][W];
+  float wf[NV][W];
 #pragma unroll
-  for (int v = 0; v < RB_RN_MAXV; ++v) {
-    const int i = v * BLOCK + threadIdx.x;
-    if (i < nvec) rb::VIO<T>::load(w + i * W, wf[v]);
-  }
+  for (int v = 0; v < NV; ++v)
+    rb::VIO<T>::load(w + (v * BLOCK + threadIdx.x) * W, wf[v]);
 
   for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
     const T *xr = x + row * D;
@@ -156,44 +155,38 @@ __global__ void rmsnorm_bwd_reg_kernel(const T *__restrict__ x,
     T *dxr = dx + row * D;
     const float ir = inv_rms[row];
 
-    float xf[RB_RN_MAXV][W], df[RB_RN_MAXV][W];
+    float xf[NV][W], df[NV][W];
     float s = 0.0f;
 #pragma unroll
-    for (int v = 0; v < RB_RN_MAXV; ++v) {
+    for (int v = 0; v < NV; ++v) {
       const int i = v * BLOCK + threadIdx.x;
-      if (i < nvec) {
-        rb::VIO<T>::load(xr + i * W, xf[v]);
-        rb::VIO<T>::load(dyr + i * W, df[v]);
+      rb::VIO<T>::load(xr + i * W, xf[v]);
+      rb::VIO<T>::load(dyr + i * W, df[v]);
 #pragma unroll
-        for (int k = 0; k < W; ++k) s += df[v][k] * wf[v][k] * xf[v][k];
-      }
+      for (int k = 0; k < W; ++k) s += df[v][k] * wf[v][k] * xf[v][k];
     }
     s = rb::block_reduce_sum(s, red);
     const float c = ir * ir * ir * s / (float)D;
 
 #pragma unroll
-    for (int v = 0; v < RB_RN_MAXV; ++v) {
+    for (int v = 0; v < NV; ++v) {
       const int i = v * BLOCK + threadIdx.x;
-      if (i < nvec) {
-        float o[W];
+      float o[W];
 #pragma unroll
-        for (int k = 0; k < W; ++k) {
-          o[k] = ir * wf[v][k] * df[v][k] - c * xf[v][k];
-          dwacc[v][k] += df[v][k] * xf[v][k] * ir;
-        }
-        rb::VIO<T>::store(dxr + i * W, o);
+      for (int k = 0; k < W; ++k) {
+        o[k] = ir * wf[v][k] * df[v][k] - c * xf[v][k];
+        dwacc[v][k] += df[v][k] * xf[v][k] * ir;
       }
+      rb::VIO<T>::store(dxr + i * W, o);
     }
   }
 
 #pragma unroll
-  for (int v = 0; v < RB_RN_MAXV; ++v) {
+  for (int v = 0; v < NV; ++v) {
     const int i = v * BLOCK + threadIdx.x;
-    if (i < nvec) {
 #pragma unroll
-      for (int k = 0; k < W; ++k)
-        if (dwacc[v][k] != 0.0f) atomicAdd(dw + i * W + k, dwacc[v][k]);
-    }
+    for (int k = 0; k < W; ++k)
+      if (dwacc[v][k] != 0.0f) atomicAdd(dw + i * W + k, dwacc[v][k]);
   }
 }
 
@@ -386,32 +379,46 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
   const size_t shmem = (size_t)D * sizeof(float) + (BLOCK / RB_WAVE) * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "rmsnorm_bwd: D too large for LDS accumulation");
 
+  const int W0 = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+  const int nv = (D % (W0 * BLOCK) == 0) ? D / (W0 * BLOCK) : 0;
   if (x.scalar_type() == at::kBFloat16) {
-    if (D <= 8 * BLOCK * RB_RN_MAXV) {
-      hipLaunchKernelGGL(rmsnorm_bwd_reg_kernel<uint16_t>, dim3(nwg),
-                         dim3(BLOCK), 0, stream,
-                         (const uint16_t *)x.data_ptr(), (const uint16_t *)w.data_ptr(),
-                         (const uint16_t *)dy.data_ptr(), inv_rms.data_ptr<float>(),
-                         (uint16_t *)dx.data_ptr(), dw.data_ptr<float>(), n_rows, D);
-    } else {
-      hipLaunchKernelGGL(rmsnorm_bwd_kernel<uint16_t>, dim3(nwg), dim3(BLOCK), shmem, stream,
-                         (const uint16_t *)x.data_ptr(), (const uint16_t *)w.data_ptr(),
-                         (const uint16_t *)dy.data_ptr(), inv_rms.data_ptr<float>(),
-                         (uint16_t *)dx.data_ptr(), dw.data_ptr<float>(), n_rows, D);
-    }
+    auto a = (const uint16_t *)x.data_ptr();
+    auto b = (const uint16_t *)w.data_ptr();
+    auto c = (const uint16_t *)dy.data_ptr();
+    auto irp = inv_rms.data_ptr<float>();
+    auto dxp = (uint16_t *)dx.data_ptr();
+    auto dwp = dw.data_ptr<float>();
+    if (nv == 1)
+      hipLaunchKernelGGL((rmsnorm_bwd_reg_kernel<uint16_t, 1>), dim3(nwg),
+                         dim3(BLOCK), 0, stream, a, b, c, irp, dxp, dwp, n_rows, D);
+    else if (nv == 2)
+      hipLaunchKernelGGL((rmsnorm_bwd_reg_kernel<uint16_t, 2>), dim3(nwg),
+                         dim3(BLOCK), 0, stream, a, b, c, irp, dxp, dwp, n_rows, D);
+    else if (nv == 4)
+      hipLaunchKernelGGL((rmsnorm_bwd_reg_kernel<uint16_t, 4>), dim3(nwg),
+                         dim3(BLOCK), 0, stream, a, b, c, irp, dxp, dwp, n_rows, D);
+    else
+      hipLaunchKernelGGL(rmsnorm_bwd_kernel<uint16_t>, dim3(nwg), dim3(BLOCK),
+                         shmem, stream, a, b, c, irp, dxp, dwp, n_rows, D);
   } else if (x.scalar_type() == at::kFloat) {
-    if (D <= 4 * BLOCK * RB_RN_MAXV) {
-      hipLaunchKernelGGL(rmsnorm_bwd_reg_kernel<float>, dim3(nwg),
-                         dim3(BLOCK), 0, stream,
-                         x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
-                         inv_rms.data_ptr<float>(), dx.data_ptr<float>(), dw.data_ptr<float>(),
-                         n_rows, D);
-    } else {
-      hipLaunchKernelGGL(rmsnorm_bwd_kernel<float>, dim3(nwg), dim3(BLOCK), shmem, stream,
-                         x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
-                         inv_rms.data_ptr<float>(), dx.data_ptr<float>(), dw.data_ptr<float>(),
-                         n_rows, D);
-    }
+    auto a = x.data_ptr<float>();
+    auto b = w.data_ptr<float>();
+    auto c = dy.data_ptr<float>();
+    auto irp = inv_rms.data_ptr<float>();
+    auto dxp = dx.data_ptr<float>();
+    auto dwp = dw.data_ptr<float>();
+    if (nv == 1)
+      hipLaunchKernelGGL((rmsnorm_bwd_reg_kernel<float, 1>), dim3(nwg),
+                         dim3(BLOCK), 0, stream, a, b, c, irp, dxp, dwp, n_rows, D);
+    else if (nv == 2)
+      hipLaunchKernelGGL((rmsnorm_bwd_reg_kernel<float, 2>), dim3(nwg),
+                         dim3(BLOCK), 0, stream, a, b, c, irp, dxp, dwp, n_rows, D);
+    else if (nv == 4)
+      hipLaunchKernelGGL((rmsnorm_bwd_reg_kernel<float, 4>), dim3(nwg),
+                         dim3(BLOCK), 0, stream, a, b, c, irp, dxp, dwp, n_rows, D);
+    else
+      hipLaunchKernelGGL(rmsnorm_bwd_kernel<float>, dim3(nwg), dim3(BLOCK),
+                         shmem, stream, a, b, c, irp, dxp, dwp, n_rows, D);
   } else {
     TORCH_CHECK(false, "rmsnorm_bwd: unsupported dtype");
   }
