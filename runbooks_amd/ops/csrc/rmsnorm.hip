@@ -379,8 +379,11 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
   const size_t shmem = (size_t)D * sizeof(float) + (BLOCK / RB_WAVE) * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "rmsnorm_bwd: D too large for LDS accumulation");
 
-  const int W0 = x.scalar_type() == at::kBFloat16 ? 8 : 4;
-  const int nv = (D % (W0 * BLOCK) == 0) ? D / (W0 * BLOCK) : 0;
+  // The register-dw variant measured 7x SLOWER than this LDS path on
+  // [2048,4096] (472 vs 65 us, r11/r13) despite clean ISA (no spills,
+  // 95 VGPR, 5 waves/SIMD) — root cause unidentified; the LDS
+  // accumulate stays the shipped path.
+  const int nv = 0;
   if (x.scalar_type() == at::kBFloat16) {
     auto a = (const uint16_t *)x.data_ptr();
     auto b = (const uint16_t *)w.data_ptr();
