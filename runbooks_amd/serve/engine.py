@@ -105,10 +105,15 @@ class Engine:
             model = build_model(model, dtype=dtype, seed=seed,
                                 device=self.device)
         self.model = model.to(self.device).eval()
-        if torch.cuda.is_available() and model.tp == 1:
+        if torch.cuda.is_available():
+            # TP>1 included: each rank fuses ITS OWN qkv / gate-up weight
+            # shards and registers decode-GEMM layouts for them — the
+            # collective order (o/down all-reduce per layer) is untouched,
+            # so the worker-follow protocol stays aligned.
             from ..models.transformer import fuse_for_inference
-            fuse_for_inference(self.model, load_in_8bit=load_in_8bit)
-        elif load_in_8bit and model.tp > 1:
+            fuse_for_inference(self.model,
+                               load_in_8bit=load_in_8bit and model.tp == 1)
+        if load_in_8bit and model.tp > 1:
             print("engine: MODEL_LOAD_IN_8BIT ignored at TP>1 "
                   "(fp8 decode path is single-GPU; 288 GB/GPU rarely "
                   "needs 8-bit at TP>1)", flush=True)

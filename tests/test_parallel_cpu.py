@@ -522,3 +522,45 @@ def test_tp4_engine_uneven_admission():
 
 def test_tp4_kv_pool_agreement():
     _run_dist(_tp_kv_pool_agreement, world=4, port=PORT + 22)
+
+
+def _tp_engine_fused_weights(rank, world):
+    """TP=2 serving with fuse_for_inference applied per rank (the GPU
+    engine now fuses shards at TP>1 too): outputs must still match the
+    single-rank engine exactly."""
+    from runbooks_amd.models import build_model
+    from runbooks_amd.models.transformer import fuse_for_inference
+    from runbooks_amd.serve import Engine
+    from runbooks_amd.serve.tp_worker import broadcast_shutdown, worker_loop
+
+    torch.manual_seed(0)
+    tp_model = build_model("tiny-llama", dtype=torch.float32, tp=world,
+                           seed=8)
+    single = build_model("tiny-llama", dtype=torch.float32, tp=1, seed=8)
+    sd = single.state_dict()
+    for name, t in tp_model.state_dict().items():
+        full = sd[name]
+        if t.shape == full.shape:
+            t.copy_(full)
+        elif t.shape[0] * world == full.shape[0]:
+            t.copy_(full[rank * t.shape[0]:(rank + 1) * t.shape[0]])
+        else:
+            t.copy_(full[:, rank * t.shape[1]:(rank + 1) * t.shape[1]])
+    fuse_for_inference(tp_model)
+    assert getattr(tp_model.blocks[0].attn, "_qkv_w", None) is not None
+
+    eng = Engine(tp_model, device="cpu", kv_blocks=64, seed=21)
+    prompts = [[2, 7, 1, 8], [4, 4, 9]]
+    if rank == 0:
+        outs = [eng.generate(list(p), max_new_tokens=5) for p in prompts]
+        broadcast_shutdown(torch.device("cpu"))
+        ref_eng = Engine(single, device="cpu", kv_blocks=64, seed=21)
+        ref_eng.tp = 1
+        for p, o in zip(prompts, outs):
+            assert o == ref_eng.generate(list(p), max_new_tokens=5), (p, o)
+    else:
+        worker_loop(eng)
+
+
+def test_tp_engine_fused_weights():
+    _run_dist(_tp_engine_fused_weights, world=2, port=PORT + 23)
