@@ -1,0 +1,117 @@
+// Fused LoRA rank-r delta accumulate for CDNA4 (gfx950):
+//
+//   y[T, N] += scale * t[T, r] @ W[N, r]^T        (r <= 32)
+//
+// covers BOTH LoRA merge points of a train step (train/lora.py):
+// forward  y += s * (x A^T) B^T   with t = x A^T, W = B
+// backward dx += s * (dy B) A     with t = dy B,  W = A
+//
+// hipBLASLt runs these [2048, r<=32] @ [r, 4096..22016] accumulates at
+// 12-20 us each (~700 calls per llama2-7b LoRA step = ~10% of the step,
+// profiles/train_*_top_kernels.csv): tiny-K GEMMs are latency-bound in
+// a library tiled for big K. Here the whole thing is one
+// bandwidth-shaped pass: y is read+written once (the unavoidable
+// traffic), t rows sit in registers, W panels are L2-resident (N*r*2 <=
+// 1.4 MB), and the per-element work is r<=32 FMAs on the VALU — at
+// [2048, 4096] r=16 the kernel is a pure 67 MB stream.
+//
+// Grid: one wave per 64-element y span; each lane owns one f32x?-wide
+// column strip. No LDS, no barriers.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int RMAX = 32;
+
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4v;
+
+// WT: W stored [N, r] (y += t @ W^T, the B merge). !WT: W stored
+// [r, N] (y += t @ W, the A merge in backward) — there each k reads a
+// 16 B row chunk W[k, col8:col8+8], an even friendlier pattern.
+template <bool WT>
+__global__ void lora_delta_kernel(const uint16_t *__restrict__ t,
+                                  const uint16_t *__restrict__ w,
+                                  uint16_t *__restrict__ y,
+                                  int64_t T, int N, int r, float scale) {
+  // element-vector id: 8 bf16 per thread
+  const int64_t nvec_row = N / 8;
+  const int64_t total = T * nvec_row;
+  const int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t idx = (int64_t)blockIdx.x * BLOCK + threadIdx.x; idx < total;
+       idx += stride) {
+    const int64_t row = idx / nvec_row;
+    const int col8 = (int)(idx % nvec_row) * 8;
+
+    // t row (<= 32 bf16 = 64 B) — L2-hot, read per thread
+    float tv[RMAX];
+    const uint16_t *tr = t + row * r;
+    for (int k = 0; k < r; ++k) tv[k] = rb::bf16_to_f32(tr[k]);
+
+    float acc[8];
+    rb::bf16x8 yv = *reinterpret_cast<const rb::bf16x8 *>(y + row * N + col8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = rb::bf16_to_f32(yv.v[e]);
+
+    if (WT) {
+      // W rows col8..col8+7, r bf16 each (L2-resident panel)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const uint16_t *wr = w + (int64_t)(col8 + e) * r;
+        float s = 0.0f;
+        for (int k = 0; k < r; ++k) s += tv[k] * rb::bf16_to_f32(wr[k]);
+        acc[e] += scale * s;
+      }
+    } else {
+      float s8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      for (int k = 0; k < r; ++k) {
+        rb::bf16x8 wv =
+            *reinterpret_cast<const rb::bf16x8 *>(w + (int64_t)k * N + col8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          s8[e] += tv[k] * rb::bf16_to_f32(wv.v[e]);
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += scale * s8[e];
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) yv.v[e] = rb::f32_to_bf16(acc[e]);
+    *reinterpret_cast<rb::bf16x8 *>(y + row * N + col8) = yv;
+  }
+}
+
+}  // namespace
+
+// In-place: y += scale * t @ w^T (w [N,r], w_transposed=true) or
+// y += scale * t @ w (w [r,N], w_transposed=false). bf16, r <= 32.
+at::Tensor lora_delta_(at::Tensor y, at::Tensor t, at::Tensor w,
+                       double scale, bool w_transposed) {
+  TORCH_CHECK(y.is_cuda() && y.is_contiguous() && t.is_contiguous() &&
+              w.is_contiguous(), "lora_delta_: contiguous GPU tensors");
+  TORCH_CHECK(y.scalar_type() == at::kBFloat16 &&
+              t.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16, "lora_delta_: bf16");
+  const int64_t T = t.size(0);
+  const int r = (int)t.size(1);
+  const int N = (int)(w_transposed ? w.size(0) : w.size(1));
+  const int wr = (int)(w_transposed ? w.size(1) : w.size(0));
+  TORCH_CHECK(r <= RMAX && wr == r && y.size(0) == T &&
+              (int)y.size(1) == N && N % 8 == 0, "lora_delta_: shape");
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int grid = rb::rb_grid_1d(T * (N / 8), BLOCK);
+  if (w_transposed)
+    hipLaunchKernelGGL(lora_delta_kernel<true>, dim3(grid), dim3(BLOCK), 0,
+                       stream, (const uint16_t *)t.data_ptr(),
+                       (const uint16_t *)w.data_ptr(),
+                       (uint16_t *)y.data_ptr(), T, N, r, (float)scale);
+  else
+    hipLaunchKernelGGL(lora_delta_kernel<false>, dim3(grid), dim3(BLOCK), 0,
+                       stream, (const uint16_t *)t.data_ptr(),
+                       (const uint16_t *)w.data_ptr(),
+                       (uint16_t *)y.data_ptr(), T, N, r, (float)scale);
+  return y;
+}

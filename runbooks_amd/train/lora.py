@@ -19,6 +19,20 @@ from torch import nn
 
 from ..parallel.tp import ColumnParallelLinear, RowParallelLinear
 
+
+def _delta_add_(y, t, w, scale, w_transposed=True):
+    """y += scale * t @ (w^T if w_transposed else w) — the gfx950
+    lora_delta_ kernel on GPU bf16 (hipBLASLt runs these r<=32
+    accumulates latency-bound; the kernel is one read+write pass of y),
+    addmm_ elsewhere."""
+    if (y.is_cuda and y.dtype == torch.bfloat16 and y.shape[1] % 8 == 0
+            and t.shape[1] <= 32):
+        from ..ops import _backend
+        _backend.ext().lora_delta_(y, t.contiguous(), w.contiguous(),
+                                   scale, w_transposed)
+        return y
+    return y.addmm_(t, w.t() if w_transposed else w, alpha=scale)
+
 DEFAULT_TARGETS = ("q_proj", "k_proj", "v_proj", "o_proj",
                    "gate_proj", "up_proj", "down_proj", "fc1", "fc2")
 
@@ -71,7 +85,7 @@ class _LoRAFused(torch.autograd.Function):
         # torch.profiler); y is fresh from the base GEMM so accumulating
         # into it is safe
         y = _nt(x, w)
-        y.addmm_(t, b.t(), alpha=scale)
+        _delta_add_(y, t, b, scale)
         ctx.save_for_backward(x, w, a, b, t)
         ctx.scale = scale
         return y
@@ -86,7 +100,7 @@ class _LoRAFused(torch.autograd.Function):
             dx = _nt(dy, _wt(w))
         else:
             dx = dy @ w
-        dx.addmm_(t2, a, alpha=s)
+        _delta_add_(dx, t2, a, s, w_transposed=False)
         da = torch.mm(t2.t(), x).mul_(s)
         db = torch.mm(dy.t(), t).mul_(s)
         return dx, None, da, db, None

@@ -562,3 +562,48 @@ def test_decode_gemm_split_heuristic():
         split = e.decode_gemm_split(n, k)
         assert (n // 32) * split >= 256, (n, k, split)
         assert (k // 16) % 1 == 0
+
+
+@pytest.mark.parametrize("wt", [True, False])
+def test_lora_delta(wt):
+    """csrc/lora.hip fused LoRA merge vs torch addmm (both W layouts)."""
+    _assert_hip()
+    torch.manual_seed(5)
+    T, r, N = 2048, 16, 4096
+    t = torch.randn(T, r, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn((N, r) if wt else (r, N), dtype=torch.bfloat16,
+                    device=DEV)
+    y0 = torch.randn(T, N, dtype=torch.bfloat16, device=DEV)
+    ref = (y0.float() + 0.5 * (t.float() @ (w.float().t() if wt
+                                            else w.float()))).to(torch.bfloat16)
+    y = y0.clone()
+    ops.ext().lora_delta_(y, t, w, 0.5, wt)
+    d = (y.float() - ref.float()).abs().max() / ref.float().abs().max()
+    assert d < 2e-2, d
+
+
+def test_lora_fused_train_step_matches_cpu_math():
+    """One _LoRAFused fwd+bwd on GPU bf16 vs fp32 torch reference."""
+    from runbooks_amd.train.lora import _LoRAFused
+    torch.manual_seed(2)
+    T, K, N, r, s = 256, 512, 1024, 16, 2.0
+    x = torch.randn(T, K, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+    a = torch.randn(r, K, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    b = (torch.randn(N, r, dtype=torch.bfloat16, device=DEV) / 100
+         ).requires_grad_()
+    y = _LoRAFused.apply(x, w, a, b, s)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.detach().float().requires_grad_()
+    af = a.detach().float().requires_grad_()
+    bf = b.detach().float().requires_grad_()
+    yf = xf @ w.float().t() + s * (xf @ af.t()) @ bf.t()
+    yf.backward(dy.float())
+    for got, ref, name in ((y, yf, "y"), (x.grad, xf.grad, "dx"),
+                           (a.grad, af.grad, "da"), (b.grad, bf.grad, "db")):
+        rel = (got.float() - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
+        assert rel < 5e-2, (name, rel)
