@@ -26,7 +26,7 @@ def _delta_add_(y, t, w, scale, w_transposed=True):
     accumulates latency-bound; the kernel is one read+write pass of y),
     addmm_ elsewhere."""
     if (y.is_cuda and y.dtype == torch.bfloat16 and y.shape[1] % 8 == 0
-            and t.shape[1] <= 32):
+            and t.shape[1] <= 32 and t.shape[1] % 8 == 0):
         from ..ops import _backend
         _backend.ext().lora_delta_(y, t.contiguous(), w.contiguous(),
                                    scale, w_transposed)
