@@ -21,11 +21,15 @@ from ..parallel.tp import ColumnParallelLinear, RowParallelLinear
 
 
 def _delta_add_(y, t, w, scale, w_transposed=True):
-    """y += scale * t @ (w^T if w_transposed else w) — the gfx950
-    lora_delta_ kernel on GPU bf16 (hipBLASLt runs these r<=32
-    accumulates latency-bound; the kernel is one read+write pass of y),
-    addmm_ elsewhere."""
-    if (y.is_cuda and y.dtype == torch.bfloat16 and y.shape[1] % 8 == 0
+    """y += scale * t @ (w^T if w_transposed else w). In-place addmm_
+    (hipBLASLt accumulate) is the shipped path: the hand-written
+    lora_delta_ kernel is numerically right but its per-lane W gather
+    (64 distinct addresses per instruction) measured SLOWER in-step
+    (96 -> 124 ms, gpurun r17) — opt in with RB_LORA_KERNEL=1 for
+    A/B work only."""
+    if (os.environ.get("RB_LORA_KERNEL", "0") == "1"
+            and y.is_cuda and y.dtype == torch.bfloat16
+            and y.shape[1] % 8 == 0
             and t.shape[1] <= 32 and t.shape[1] % 8 == 0):
         from ..ops import _backend
         _backend.ext().lora_delta_(y, t.contiguous(), w.contiguous(),
