@@ -4,10 +4,13 @@ Parity: reference internal/client/sync.go:28-135 (SyncFilesFromNotebook:
 push the nbwatch agent into the pod, exec it, mirror WRITE/CREATE/REMOVE
 events to the local dir) and port_forward.go:21-46.
 
-The reference talks SPDY to the kubelet through client-go; here the same
-operations go through `kubectl` subprocesses (`cp`, `exec`,
-`port-forward`) — the reference itself shells out to kubectl for cp
-(internal/cp/kubectl.go:15-26).
+The reference talks SPDY to the kubelet through client-go. Here the
+NATIVE path (`client/remote.py`) speaks the same channel protocol over
+the API server's WebSocket endpoints (v4.channel.k8s.io exec +
+portforward) via aiohttp, used whenever a configured HTTPKubeClient is
+passed; the kubectl subprocess path remains the kubeconfig-only
+fallback (the reference itself shells out to kubectl for cp,
+internal/cp/kubectl.go:15-26).
 """
 from __future__ import annotations
 
@@ -29,14 +32,29 @@ def _kubectl() -> str:
 
 
 def cp_to_pod(namespace: str, pod: str, src: str, dst: str,
-              container: str = "notebook") -> None:
+              container: str = "notebook", client=None) -> None:
+    if client is not None:
+        from . import remote
+        with open(src, "rb") as f:
+            res = remote.cp_to_pod_native(client, namespace, pod, f.read(),
+                                          dst, container=container)
+        if res.returncode != 0:
+            raise RuntimeError(f"cp_to_pod: {res.status}")
+        return
     subprocess.run([_kubectl(), "cp", src, f"{namespace}/{pod}:{dst}",
                     "-c", container], check=True)
 
 
 def cp_from_pod(namespace: str, pod: str, src: str, dst: str,
-                container: str = "notebook") -> None:
+                container: str = "notebook", client=None) -> None:
     os.makedirs(os.path.dirname(dst) or ".", exist_ok=True)
+    if client is not None:
+        from . import remote
+        data = remote.cp_from_pod_native(client, namespace, pod, src,
+                                         container=container)
+        with open(dst, "wb") as f:
+            f.write(data)
+        return
     subprocess.run([_kubectl(), "cp", f"{namespace}/{pod}:{src}", dst,
                     "-c", container], check=True)
 
