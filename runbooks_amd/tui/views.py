@@ -21,10 +21,21 @@ except ImportError:  # pragma: no cover - rich ships in the image
 
 
 def select_manifest(objs: list, assume_first: bool = False):
-    """Pick one substratus object from discovered manifests
-    (reference manifests picker). Non-TTY or single match -> first."""
+    """Pick one substratus object from discovered manifests — the
+    interactive arrow-key picker (tui/core.py event loop, reference
+    manifests.go) on a TTY; first match otherwise."""
     if len(objs) == 1 or assume_first or not sys.stdin.isatty():
         return objs[0]
+    try:
+        from .core import Program, SelectModel
+        items = [f"{o.kind}/{o.name}  ({getattr(o, '_source_file', '?')})"
+                 for o in objs]
+        model = Program(SelectModel("select a manifest:", items)).run()
+        if model.chosen is None:
+            raise SystemExit(1)
+        return objs[model.chosen]
+    except (ImportError, OSError):  # no termios (exotic terminal): prompt
+        pass
     for i, o in enumerate(objs):
         src = getattr(o, "_source_file", "?")
         print(f"  [{i}] {o.kind}/{o.name}  ({src})")
