@@ -120,9 +120,12 @@ class Attention(nn.Module):
         B = x.shape[0]
         if getattr(self, "_qkv_w", None) is not None and \
                 self.cfg.pos == "rope" and ops.use_hip(x) and \
-                x.dtype == torch.bfloat16:
+                x.dtype == torch.bfloat16 and kc.dtype == torch.bfloat16:
             # fused decode hot path: one wide GEMM, then one kernel doing
-            # rope(q), rope(k)->cache, v->cache (csrc/qkv_fused.hip)
+            # rope(q), rope(k)->cache, v->cache (csrc/qkv_fused.hip).
+            # fp8 caches take the generic path below (rope + fp8-quant
+            # kv_append): that mode targets long-context KV-bound decode
+            # where the attention read, not the epilogue, dominates.
             y = ops.fast_linear(x, self._qkv_w)
             q = ops.ext().qkv_rope_append(y, cos, sin, positions, kc, vc,
                                           slot_mapping, self.hq)
@@ -408,9 +411,10 @@ class Transformer(nn.Module):
     def local_kv_heads(self) -> int:
         return self.blocks[0].attn.hkv
 
-    def alloc_caches(self, num_blocks: int, device):
+    def alloc_caches(self, num_blocks: int, device, fp8: bool = False):
         return [ops.alloc_kv_cache(num_blocks, self.local_kv_heads(),
-                                   self.cfg.head_dim, device, dtype=self.dtype)
+                                   self.cfg.head_dim, device,
+                                   dtype=self.dtype, fp8=fp8)
                 for _ in range(self.cfg.num_layers)]
 
 

@@ -112,7 +112,12 @@ def flash_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):
-    """fp32 reference decode: q [B, Hq, Dh]."""
+    """fp32 reference decode: q [B, Hq, Dh]. fp8 caches (uint8 rows of
+    Dh bytes + f32 scale) are dequantized up front."""
+    if k_cache.dtype == torch.uint8:
+        from .kvcache import fp8_dequant_cache_ref
+        k_cache = fp8_dequant_cache_ref(k_cache)
+        v_cache = fp8_dequant_cache_ref(v_cache)
     B, Hq, Dh = q.shape
     _, Hkv, BS, _ = k_cache.shape
     g = Hq // Hkv

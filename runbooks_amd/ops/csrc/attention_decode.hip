@@ -29,7 +29,12 @@ namespace {
 constexpr int BLOCK = 256;    // 4 waves
 constexpr int NW = 4;
 
-template <int DH, int G, bool SPLIT>
+// FP8: caches are e4m3 rows of fp8_row_bytes(DH) = DH bytes + f32
+// scale (+pad): decode attention is KV-bandwidth bound at long context,
+// so halving cache bytes halves the dominant stream AND doubles KV
+// capacity within 288 GB (BASELINE config #5). Dequant rides
+// v_cvt_pk_f32_fp8 in the inner loop.
+template <int DH, int G, bool SPLIT, bool FP8 = false>
 __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const uint16_t *__restrict__ q, const uint16_t *__restrict__ k_cache,
     const uint16_t *__restrict__ v_cache, const int32_t *__restrict__ block_tables,
@@ -97,17 +102,39 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const bool valid = true;
     const int tt = t;
     const int blk = bt[tt / bs];
-    const int64_t base = (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * DH + d0;
 
     float kf[VE], vf[VE];
-    if (VE == 8) {
-      rb::VIO<uint16_t>::load(k_cache + base, kf);
-      rb::VIO<uint16_t>::load(v_cache + base, vf);
-    } else {
+    if (FP8) {
+      constexpr int RB8 = DH + 8;
+      const uint8_t *k8 = reinterpret_cast<const uint8_t *>(k_cache);
+      const uint8_t *v8 = reinterpret_cast<const uint8_t *>(v_cache);
+      const int64_t row = (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * RB8;
 #pragma unroll
-      for (int e = 0; e < VE; ++e) {
-        kf[e] = rb::bf16_to_f32(k_cache[base + e]);
-        vf[e] = rb::bf16_to_f32(v_cache[base + e]);
+      for (int e8 = 0; e8 < VE; e8 += 8) {
+        if (VE - e8 >= 8) {
+          rb::fp8x8_to_f32(k8 + row + d0 + e8, kf + e8);
+          rb::fp8x8_to_f32(v8 + row + d0 + e8, vf + e8);
+        } else {
+          rb::fp8x4_to_f32(k8 + row + d0 + e8, kf + e8);
+          rb::fp8x4_to_f32(v8 + row + d0 + e8, vf + e8);
+        }
+      }
+      const float ks = *reinterpret_cast<const float *>(k8 + row + DH);
+      const float vs = *reinterpret_cast<const float *>(v8 + row + DH);
+#pragma unroll
+      for (int e = 0; e < VE; ++e) { kf[e] *= ks; vf[e] *= vs; }
+    } else {
+      const int64_t base =
+          (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * DH + d0;
+      if (VE == 8) {
+        rb::VIO<uint16_t>::load(k_cache + base, kf);
+        rb::VIO<uint16_t>::load(v_cache + base, vf);
+      } else {
+#pragma unroll
+        for (int e = 0; e < VE; ++e) {
+          kf[e] = rb::bf16_to_f32(k_cache[base + e]);
+          vf[e] = rb::bf16_to_f32(v_cache[base + e]);
+        }
       }
     }
 
@@ -222,7 +249,7 @@ __global__ void decode_combine_kernel(const float *__restrict__ partial,
   }
 }
 
-template <int DH, int G>
+template <int DH, int G, bool FP8>
 void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                    const at::Tensor &v_cache, const at::Tensor &block_tables,
                    const at::Tensor &seq_lens, at::Tensor &out, int nsplit,
@@ -232,7 +259,8 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
   const int bs = (int)k_cache.size(2);
   const int max_blocks = (int)block_tables.size(1);
   if (nsplit <= 1) {
-    hipLaunchKernelGGL((paged_decode_kernel<DH, G, false>), dim3(B, hkv, 1),
+    hipLaunchKernelGGL((paged_decode_kernel<DH, G, false, FP8>),
+                       dim3(B, hkv, 1),
                        dim3(BLOCK), 0, stream, (const uint16_t *)q.data_ptr(),
                        (const uint16_t *)k_cache.data_ptr(),
                        (const uint16_t *)v_cache.data_ptr(),
@@ -242,7 +270,8 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
     const int Hq = hkv * G;
     auto partial = at::empty({B, Hq, nsplit, DH + 2},
                              q.options().dtype(at::kFloat));
-    hipLaunchKernelGGL((paged_decode_kernel<DH, G, true>), dim3(B, hkv, nsplit),
+    hipLaunchKernelGGL((paged_decode_kernel<DH, G, true, FP8>),
+                       dim3(B, hkv, nsplit),
                        dim3(BLOCK), 0, stream, (const uint16_t *)q.data_ptr(),
                        (const uint16_t *)k_cache.data_ptr(),
                        (const uint16_t *)v_cache.data_ptr(),
@@ -271,9 +300,23 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   auto out = at::empty_like(q);
   auto stream = at::hip::getCurrentHIPStream();
 
+  const bool fp8 = k_cache.scalar_type() == at::kByte;
+  if (fp8) {
+    TORCH_CHECK((int)k_cache.size(3) == dh + 8 &&
+                v_cache.scalar_type() == at::kByte,
+                "paged_decode: fp8 cache rows must be dh+8 bytes");
+  }
 #define RB_DEC(DHV, GV)                                                        \
-  launch_decode<DHV, GV>(q, k_cache, v_cache, block_tables, seq_lens, out,     \
-                         (int)nsplit, (float)scale, stream)
+  do {                                                                         \
+    if (fp8)                                                                   \
+      launch_decode<DHV, GV, true>(q, k_cache, v_cache, block_tables,          \
+                                   seq_lens, out, (int)nsplit, (float)scale,   \
+                                   stream);                                    \
+    else                                                                       \
+      launch_decode<DHV, GV, false>(q, k_cache, v_cache, block_tables,         \
+                                    seq_lens, out, (int)nsplit, (float)scale,  \
+                                    stream);                                   \
+  } while (0)
   if (dh == 128) {
     switch (G) {
       case 1: RB_DEC(128, 1); break;

@@ -5,6 +5,7 @@
 // loops capped so the scheduler has room (256 CUs x 8 XCDs).
 #pragma once
 
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 
@@ -34,6 +35,38 @@ RB_DEV uint16_t f32_to_bf16(float f) {
 
 // 8 x bf16 = 16 bytes: the coalescing sweet spot for wave64 streams.
 struct bf16x8 { uint16_t v[8]; };
+
+// ---------------------------------------------------------------------------
+// OCP fp8-e4m3 (KV-cache compression; gfx950 has native cvt both ways)
+// ---------------------------------------------------------------------------
+RB_DEV uint8_t f32_to_fp8(float f) {
+  __hip_fp8_e4m3 v(f);
+  return v.__x;
+}
+
+// 8 packed e4m3 (one u64) -> 8 f32 via v_cvt_pk_f32_fp8
+RB_DEV void fp8x8_to_f32(const uint8_t *p, float *f) {
+  const uint32_t w0 = *reinterpret_cast<const uint32_t *>(p);
+  const uint32_t w1 = *reinterpret_cast<const uint32_t *>(p + 4);
+  auto a = __builtin_amdgcn_cvt_pk_f32_fp8(w0, false);
+  auto b = __builtin_amdgcn_cvt_pk_f32_fp8(w0, true);
+  auto c = __builtin_amdgcn_cvt_pk_f32_fp8(w1, false);
+  auto d = __builtin_amdgcn_cvt_pk_f32_fp8(w1, true);
+  f[0] = a[0]; f[1] = a[1]; f[2] = b[0]; f[3] = b[1];
+  f[4] = c[0]; f[5] = c[1]; f[6] = d[0]; f[7] = d[1];
+}
+
+// 4 packed e4m3 (one u32) -> 4 f32
+RB_DEV void fp8x4_to_f32(const uint8_t *p, float *f) {
+  const uint32_t w = *reinterpret_cast<const uint32_t *>(p);
+  auto a = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  auto b = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  f[0] = a[0]; f[1] = a[1]; f[2] = b[0]; f[3] = b[1];
+}
+
+// fp8 KV-cache row layout: dh e4m3 bytes, then the f32 scale, then pad
+// to 8 bytes. Row stride for head_dim dh:
+RB_DEV constexpr int fp8_row_bytes(int dh) { return dh + 8; }
 struct f32x8 { float v[8]; };
 
 RB_DEV f32x8 to_f32(const bf16x8 &a) {

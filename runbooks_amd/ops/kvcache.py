@@ -9,14 +9,57 @@ BLOCK_SIZE = 16  # tokens per cache block
 
 
 def alloc_kv_cache(num_blocks: int, num_kv_heads: int, head_dim: int,
-                   device, dtype=torch.bfloat16, block_size: int = BLOCK_SIZE):
+                   device, dtype=torch.bfloat16, block_size: int = BLOCK_SIZE,
+                   fp8: bool = False):
+    """fp8=True: e4m3 cache rows of head_dim bytes + f32 scale (+pad) —
+    half the KV bytes of bf16 (decode attention is KV-bandwidth bound at
+    long context) and double the capacity within 288 GB."""
+    if fp8:
+        shape = (num_blocks, num_kv_heads, block_size, head_dim + 8)
+        k = torch.zeros(shape, device=device, dtype=torch.uint8)
+        v = torch.zeros(shape, device=device, dtype=torch.uint8)
+        return k, v
     shape = (num_blocks, num_kv_heads, block_size, head_dim)
     k = torch.zeros(shape, device=device, dtype=dtype)
     v = torch.zeros(shape, device=device, dtype=dtype)
     return k, v
 
 
+def fp8_quant_row_ref(x: torch.Tensor):
+    """Reference per-row e4m3 quantization (rows = last dim): returns
+    (bytes uint8 [..., dh], scale f32 [...])."""
+    amax = x.float().abs().amax(dim=-1).clamp(min=1e-8)
+    scale = amax / 448.0
+    q = (x.float() / scale[..., None]).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale
+
+
+def fp8_dequant_cache_ref(cache: torch.Tensor) -> torch.Tensor:
+    """[blocks, hkv, bs, dh+8] uint8 -> [blocks, hkv, bs, dh] f32."""
+    dh = cache.shape[-1] - 8
+    data = cache[..., :dh].contiguous().view(torch.float8_e4m3fn).float()
+    scale = cache[..., dh:dh + 4].contiguous().view(torch.float32)
+    return data * scale
+
+
+def _kv_append_fp8_ref(k, v, k_cache, v_cache, slot_mapping):
+    bs = k_cache.shape[2]
+    dh = k_cache.shape[-1] - 8
+    for t in range(k.shape[0]):
+        slot = int(slot_mapping[t])
+        if slot < 0:
+            continue
+        blk, off = divmod(slot, bs)
+        for cache, src in ((k_cache, k), (v_cache, v)):
+            q, scale = fp8_quant_row_ref(src[t])      # [hkv, dh], [hkv]
+            cache[blk, :, off, :dh] = q
+            cache[blk, :, off, dh:dh + 4] = scale.float().view(
+                torch.uint8).reshape(-1, 4)
+
+
 def kv_append_ref(k, v, k_cache, v_cache, slot_mapping):
+    if k_cache.dtype == torch.uint8:
+        return _kv_append_fp8_ref(k, v, k_cache, v_cache, slot_mapping)
     bs = k_cache.shape[2]
     for t in range(k.shape[0]):
         slot = int(slot_mapping[t])

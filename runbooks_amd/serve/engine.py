@@ -98,7 +98,8 @@ class Engine:
     def __init__(self, model: Transformer | str, device=None,
                  dtype=torch.bfloat16, kv_blocks: int | None = None,
                  max_batch: int = 64, mem_fraction: float = 0.85, seed: int = 0,
-                 load_in_8bit: bool = False, prefix_cache: bool | None = None):
+                 load_in_8bit: bool = False, prefix_cache: bool | None = None,
+                 kv_fp8: bool | None = None):
         self.device = device if device is not None else (
             f"cuda:{comm.local_rank()}" if torch.cuda.is_available() else "cpu")
         if isinstance(model, str):
@@ -126,6 +127,12 @@ class Engine:
         self.bs = ops.BLOCK_SIZE
         self.max_batch = max_batch
         self.seed = seed
+        # fp8-e4m3 KV cache (RB_KV_FP8=1 or kv_fp8=True): halves the
+        # decode-attention read stream and doubles KV capacity — the
+        # long-context lever for the 288 GB/GPU KV story (BASELINE #5).
+        import os as _os0
+        self.kv_fp8 = (kv_fp8 if kv_fp8 is not None
+                       else _os0.environ.get("RB_KV_FP8", "0") == "1")
         if kv_blocks is None:
             kv_blocks = self._auto_kv_blocks(mem_fraction)
         if comm.is_dist() and comm.world_size() > 1 and model.tp > 1:
@@ -141,7 +148,8 @@ class Engine:
         # last block is reserved as the dummy block hipGraph-padded decode
         # rows read/write (serve/graph.py)
         self.allocator = BlockAllocator(kv_blocks - 1)
-        self.caches = self.model.alloc_caches(kv_blocks, self.device)
+        self.caches = self.model.alloc_caches(kv_blocks, self.device,
+                                              fp8=self.kv_fp8)
         self.dummy_block = kv_blocks - 1
         self.waiting: list[Request] = []
         self.running: list[Request] = []
@@ -184,8 +192,10 @@ class Engine:
                       "prefix_hit_blocks": 0, "window_dropped_blocks": 0}
 
     def _auto_kv_blocks(self, mem_fraction: float) -> int:
+        elem = (self.cfg.head_dim + 8) if self.kv_fp8 \
+            else self.cfg.head_dim * 2
         bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
-                           * self.bs * self.cfg.head_dim * 2)
+                           * self.bs * elem)
         if torch.cuda.is_available():
             free, _total = torch.cuda.mem_get_info(self.device)
             budget = int(free * mem_fraction) - (2 << 30)  # activations headroom

@@ -133,3 +133,40 @@ def test_fp8_rowwise_quantization_roundtrip():
     assert rel < 0.08, rel  # e4m3 has ~2 mantissa bits at full scale
     assert _FP8_REGISTRY.get(w.data_ptr()) is not None
     _FP8_REGISTRY.clear()
+
+
+def test_fp8_kv_quant_roundtrip_cpu():
+    """e4m3 KV rows: per-(token,head) scale bounds relative error ~2^-3."""
+    import torch
+    from runbooks_amd.ops import kvcache as kc
+    torch.manual_seed(0)
+    k, v = kc.alloc_kv_cache(8, 4, 32, "cpu", fp8=True)
+    assert k.shape == (8, 4, 16, 40) and k.dtype == torch.uint8
+    x = torch.randn(5, 4, 32) * 3.0
+    kc.kv_append_ref(x, x * 0.5, k, v, torch.tensor([3, 19, 48, 0, 127]))
+    dq_k = kc.fp8_dequant_cache_ref(k)
+    blk, off = 19 // 16, 19 % 16
+    rel = (dq_k[blk, :, off] - x[1].float()).abs().max() / x[1].abs().max()
+    assert rel < 0.08, rel
+    dq_v = kc.fp8_dequant_cache_ref(v)
+    rel_v = (dq_v[blk, :, off] - 0.5 * x[1].float()).abs().max() / \
+        (0.5 * x[1]).abs().max()
+    assert rel_v < 0.08, rel_v
+
+
+def test_fp8_kv_engine_cpu_decodes():
+    """Engine with kv_fp8=True on CPU (reference fp8 cache path): decode
+    logits stay close to the bf16-cache engine's."""
+    import torch
+    from runbooks_amd.models import build_model
+    from runbooks_amd.serve import Engine
+    m = build_model("tiny-llama", dtype=torch.float32, seed=3)
+    e8 = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=1,
+                kv_fp8=True)
+    assert e8.caches[0][0].dtype == torch.uint8
+    out = e8.generate([5, 1, 9, 2], max_new_tokens=6)
+    assert len(out) == 6 and all(0 <= t < 256 for t in out)
+    # same prompt on the plain engine: greedy paths agree at the start
+    e16 = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=1)
+    ref = e16.generate([5, 1, 9, 2], max_new_tokens=6)
+    assert out[0] == ref[0], (out, ref)
