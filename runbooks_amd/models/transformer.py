@@ -116,7 +116,8 @@ class Attention(nn.Module):
         return self.o_proj(o.reshape(B * S, self.hq * self.dh))
 
     def forward_decode(self, x, cos, sin, positions, kc, vc, slot_mapping,
-                       block_tables, seq_lens, nsplit=None, slab_ok=False):
+                       block_tables, seq_lens, nsplit=None, slab_ok=False,
+                       seq_starts=None):
         B = x.shape[0]
         if getattr(self, "_qkv_w", None) is not None and \
                 self.cfg.pos == "rope" and ops.use_hip(x) and \
@@ -133,7 +134,7 @@ class Attention(nn.Module):
             q, k, v = self._qkv(x, cos, sin, positions)
             ops.kv_append(k, v, kc, vc, slot_mapping)
         o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale,
-                             nsplit=nsplit)
+                             nsplit=nsplit, seq_starts=seq_starts)
         o2 = o.reshape(B, self.hq * self.dh)
         if slab_ok and self.tp == 1 and self.o_proj.bias is None:
             out, _ = ops.decode_linear_raw(o2, self.o_proj.weight)
@@ -368,7 +369,7 @@ class Transformer(nn.Module):
 
     @torch.no_grad()
     def decode(self, tokens, positions, caches, slot_mapping, block_tables,
-               seq_lens, nsplit=None):
+               seq_lens, nsplit=None, seq_starts=None):
         """tokens [B] (one new token per sequence) -> logits [B, V].
         ``nsplit`` fixes the paged-decode work split (required under
         hipGraph capture, serve/graph.py)."""
@@ -386,7 +387,7 @@ class Transformer(nn.Module):
                 fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
                     h, self.rope_cos, self.rope_sin, positions, kc, vc,
                     slot_mapping, block_tables, seq_lens, nsplit=nsplit,
-                    slab_ok=True)
+                    slab_ok=True, seq_starts=seq_starts)
                 x, delta = blk.forward_decode_fused(x, delta, fn)
             if delta.dtype == torch.float32:
                 _, y, swz = ops.ext().rmsnorm_res_slab_fwd_dec(
@@ -401,7 +402,7 @@ class Transformer(nn.Module):
         for blk, (kc, vc) in zip(self.blocks, caches):
             fn = lambda h, b=blk, kc=kc, vc=vc: b.attn.forward_decode(  # noqa: E731
                 h, self.rope_cos, self.rope_sin, positions, kc, vc, slot_mapping,
-                block_tables, seq_lens, nsplit=nsplit)
+                block_tables, seq_lens, nsplit=nsplit, seq_starts=seq_starts)
             x = blk(x, fn)
         x = self.norm_f(x)
         logits = self.lm_head(x)

@@ -111,9 +111,12 @@ def flash_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return causal_attention_ref(q, k, v, scale=scale)
 
 
-def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):
+def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
+                     seq_starts=None):
     """fp32 reference decode: q [B, Hq, Dh]. fp8 caches (uint8 rows of
-    Dh bytes + f32 scale) are dequantized up front."""
+    Dh bytes + f32 scale) are dequantized up front. seq_starts (strict
+    sliding window) bounds attention to virtual positions
+    [start, seq_len)."""
     if k_cache.dtype == torch.uint8:
         from .kvcache import fp8_dequant_cache_ref
         k_cache = fp8_dequant_cache_ref(k_cache)
@@ -124,9 +127,10 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):
     out = torch.empty_like(q)
     for b in range(B):
         n = int(seq_lens[b])
+        st = int(seq_starts[b]) if seq_starts is not None else 0
         blocks = block_tables[b, : (n + BS - 1) // BS].long()
-        k = k_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[:n].float()
-        v = v_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[:n].float()
+        k = k_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[st:n].float()
+        v = v_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[st:n].float()
         for h in range(Hq):
             hk = h // g
             s = (k[:, hk] @ q[b, h].float()) * scale
@@ -136,8 +140,10 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale):
 
 
 def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
-                 scale: float | None = None, nsplit: int | None = None):
-    """Decode attention over the paged cache. q [B, Hq, Dh] bf16."""
+                 scale: float | None = None, nsplit: int | None = None,
+                 seq_starts=None):
+    """Decode attention over the paged cache. q [B, Hq, Dh] bf16.
+    seq_starts (int32 [B], optional): strict-window start positions."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
@@ -153,6 +159,7 @@ def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
                 nsplit = max(nsplit, 1)
         return _backend.ext().paged_decode(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
-            int(nsplit), float(scale),
+            int(nsplit), float(scale), seq_starts=seq_starts,
         )
-    return paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale)
+    return paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens,
+                            scale, seq_starts=seq_starts)

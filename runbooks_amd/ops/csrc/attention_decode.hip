@@ -38,7 +38,8 @@ template <int DH, int G, bool SPLIT, bool FP8 = false>
 __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const uint16_t *__restrict__ q, const uint16_t *__restrict__ k_cache,
     const uint16_t *__restrict__ v_cache, const int32_t *__restrict__ block_tables,
-    const int32_t *__restrict__ seq_lens, uint16_t *__restrict__ out,
+    const int32_t *__restrict__ seq_lens,
+    const int32_t *__restrict__ seq_starts, uint16_t *__restrict__ out,
     float *__restrict__ partial, int hkv, int bs, int max_blocks, int nsplit,
     float scale) {
   constexpr int VE = DH / 16;              // elems per lane (16 lanes cover Dh)
@@ -49,16 +50,17 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   const int Hq = hkv * G;
 
   const int seq_len = seq_lens[b];
-  // Round-2 planned extension (strict sliding windows): add an optional
-  // per-sequence seq_start and compute
-  //     len = seq_len - start; chunk = ceil(len / nsplit);
-  //     t_begin = start + split * chunk; t_end = min(seq_len, ...)
-  // The index math is pre-validated on CPU in
-  // tests/test_decode_sim_cpu.py (simulate_decode seq_starts).
-  int t_begin = 0, t_end = seq_len;
+  // Strict sliding windows: seq_starts[b] (0 when null) is the first
+  // VIRTUAL position this query attends to — mistral attends to exactly
+  // W trailing tokens, not the engine's retained block-aligned W..W+15.
+  // Index math pre-validated on CPU: tests/test_decode_sim_cpu.py
+  // (simulate_decode seq_starts).
+  const int start = (seq_starts != nullptr) ? seq_starts[b] : 0;
+  int t_begin = start, t_end = seq_len;
   if (SPLIT) {
-    const int chunk = (seq_len + nsplit - 1) / nsplit;
-    t_begin = split * chunk;
+    const int len = seq_len - start;
+    const int chunk = (len + nsplit - 1) / nsplit;
+    t_begin = start + split * chunk;
     t_end = min(seq_len, t_begin + chunk);
   }
 
@@ -252,8 +254,12 @@ __global__ void decode_combine_kernel(const float *__restrict__ partial,
 template <int DH, int G, bool FP8>
 void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                    const at::Tensor &v_cache, const at::Tensor &block_tables,
-                   const at::Tensor &seq_lens, at::Tensor &out, int nsplit,
+                   const at::Tensor &seq_lens,
+                   const c10::optional<at::Tensor> &seq_starts,
+                   at::Tensor &out, int nsplit,
                    float scale, hipStream_t stream) {
+  const int32_t *starts = seq_starts.has_value()
+      ? seq_starts->data_ptr<int32_t>() : nullptr;
   const int B = (int)q.size(0);
   const int hkv = (int)k_cache.size(1);
   const int bs = (int)k_cache.size(2);
@@ -265,6 +271,7 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                        (const uint16_t *)k_cache.data_ptr(),
                        (const uint16_t *)v_cache.data_ptr(),
                        block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+                       starts,
                        (uint16_t *)out.data_ptr(), nullptr, hkv, bs, max_blocks, 1, scale);
   } else {
     const int Hq = hkv * G;
@@ -276,6 +283,7 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                        (const uint16_t *)k_cache.data_ptr(),
                        (const uint16_t *)v_cache.data_ptr(),
                        block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
+                       starts,
                        nullptr, partial.data_ptr<float>(), hkv, bs, max_blocks, nsplit,
                        scale);
     hipLaunchKernelGGL(decode_combine_kernel, dim3(B * Hq), dim3(256), 0, stream,
@@ -287,7 +295,12 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
 
 at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                         at::Tensor block_tables, at::Tensor seq_lens,
-                        int64_t nsplit, double scale) {
+                        int64_t nsplit, double scale,
+                        c10::optional<at::Tensor> seq_starts) {
+  if (seq_starts.has_value()) {
+    TORCH_CHECK(seq_starts->scalar_type() == at::kInt &&
+                seq_starts->is_contiguous(), "paged_decode: seq_starts");
+  }
   TORCH_CHECK(q.is_cuda() && q.is_contiguous(), "paged_decode: q");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "paged_decode: bf16 only");
   TORCH_CHECK(block_tables.scalar_type() == at::kInt && seq_lens.scalar_type() == at::kInt,
@@ -310,12 +323,12 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   do {                                                                         \
     if (fp8)                                                                   \
       launch_decode<DHV, GV, true>(q, k_cache, v_cache, block_tables,          \
-                                   seq_lens, out, (int)nsplit, (float)scale,   \
-                                   stream);                                    \
+                                   seq_lens, seq_starts, out, (int)nsplit,     \
+                                   (float)scale, stream);                      \
     else                                                                       \
       launch_decode<DHV, GV, false>(q, k_cache, v_cache, block_tables,         \
-                                    seq_lens, out, (int)nsplit, (float)scale,  \
-                                    stream);                                   \
+                                    seq_lens, seq_starts, out, (int)nsplit,    \
+                                    (float)scale, stream);                     \
   } while (0)
   if (dh == 128) {
     switch (G) {

@@ -361,6 +361,11 @@ class Engine:
         for r, v in zip(reqs, vpos):
             blk = r.blocks[v // self.bs]
             slots.append(blk * self.bs + v % self.bs)
+        # strict sliding window: attend to exactly the last W virtual
+        # positions (block dropping keeps memory bounded; seq_starts
+        # removes the residual 0..15-token block slack)
+        W = self.cfg.sliding_window
+        starts = [max(0, v + 1 - W) if W else 0 for v in vpos]
 
         if self.use_graphs and B <= self.max_batch:
             if self._graphed is None:
@@ -370,7 +375,7 @@ class Engine:
                     self.max_blocks_per_seq, self.dummy_block, dev)
             logits = self._graphed.decode(last, pos, slots,
                                           [r.blocks for r in reqs],
-                                          [v + 1 for v in vpos])
+                                          [v + 1 for v in vpos], starts)
         else:
             maxb = max(len(r.blocks) for r in reqs)
             bt = torch.zeros(B, maxb, dtype=torch.int32)
@@ -382,12 +387,14 @@ class Engine:
             slot_t = torch.tensor(slots, dtype=torch.int32, device=dev)
             seq_lens = torch.tensor([v + 1 for v in vpos], dtype=torch.int32,
                                     device=dev)
+            start_t = torch.tensor(starts, dtype=torch.int32, device=dev)
             bt = bt.to(dev)
             if self.tp > 1:
                 from .tp_worker import broadcast_decode
-                broadcast_decode(tokens, positions, slot_t, bt, seq_lens)
+                broadcast_decode(tokens, positions, slot_t, bt, seq_lens,
+                                 start_t)
             logits = self.model.decode(tokens, positions, self.caches, slot_t,
-                                       bt, seq_lens)
+                                       bt, seq_lens, seq_starts=start_t)
         return self._sample_batch(logits, reqs)
 
     def _sample_batch(self, logits, reqs: list[Request]) -> list[int]:

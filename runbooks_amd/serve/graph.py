@@ -52,9 +52,10 @@ class GraphedDecoder:
         self.block_tables = torch.full((B, max_blocks), dummy_block,
                                        dtype=torch.int32, device=dev)
         self.seq_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        self.seq_starts = torch.zeros(B, dtype=torch.int32, device=dev)
         pin = dict(dtype=torch.int64,
                    pin_memory=torch.cuda.is_available())
-        self.h_staging = torch.zeros(B * 4 + B * max_blocks, **pin)
+        self.h_staging = torch.zeros(B * 5 + B * max_blocks, **pin)
         self._np = self.h_staging.numpy()
         self.graphs = {}     # bucket -> (CUDAGraph, logits_out)
         self._pool = None
@@ -65,26 +66,28 @@ class GraphedDecoder:
         nsplit = fixed_nsplit(b, hkv)
         args = (self.tokens[:b], self.positions[:b], self.caches,
                 self.slots[:b], self.block_tables[:b], self.seq_lens[:b])
+        kw = dict(nsplit=nsplit, seq_starts=self.seq_starts[:b])
         # warmup outside the graph (allocator + autotune settle)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                self.model.decode(*args, nsplit=nsplit)
+                self.model.decode(*args, **kw)
         torch.cuda.current_stream().wait_stream(s)
 
         g = torch.cuda.CUDAGraph()
         if self._pool is None:
             with torch.cuda.graph(g):
-                out = self.model.decode(*args, nsplit=nsplit)
+                out = self.model.decode(*args, **kw)
             self._pool = g.pool()
         else:
             with torch.cuda.graph(g, pool=self._pool):
-                out = self.model.decode(*args, nsplit=nsplit)
+                out = self.model.decode(*args, **kw)
         self.graphs[b] = (g, out)
 
     # -- replay ------------------------------------------------------------
-    def decode(self, tokens, positions, slots, block_rows, seq_lens):
+    def decode(self, tokens, positions, slots, block_rows, seq_lens,
+               seq_starts=None):
         """All args host lists; block_rows is a list of per-seq block lists.
         Returns logits [len(tokens), V] (a view into the static output)."""
         n = len(tokens)
@@ -92,18 +95,21 @@ class GraphedDecoder:
         if b not in self.graphs:
             self._capture(b)
         g, out = self.graphs[b]
-        bt = self._stage(b, tokens, positions, slots, block_rows, seq_lens)
+        bt = self._stage(b, tokens, positions, slots, block_rows, seq_lens,
+                         seq_starts or [0] * n)
 
         h = self.h_staging
         self.tokens[:b].copy_(h[0:b], non_blocking=True)
         self.positions[:b].copy_(h[b:2 * b], non_blocking=True)
         self.slots[:b].copy_(h[2 * b:3 * b], non_blocking=True)
         self.seq_lens[:b].copy_(h[3 * b:4 * b], non_blocking=True)
+        self.seq_starts[:b].copy_(h[4 * b:5 * b], non_blocking=True)
         self.block_tables[:b].copy_(bt, non_blocking=True)
         g.replay()
         return out[:n]
 
-    def _stage(self, b, tokens, positions, slots, block_rows, seq_lens):
+    def _stage(self, b, tokens, positions, slots, block_rows, seq_lens,
+               seq_starts):
         """Fill the pinned staging buffer for bucket size b (numpy view:
         C-speed fills instead of one small torch.tensor per row). Unit-
         tested on CPU (the rest of this class needs a GPU)."""
@@ -114,7 +120,8 @@ class GraphedDecoder:
         hn[b:b + n] = positions
         hn[2 * b:2 * b + n] = slots
         hn[3 * b:3 * b + n] = seq_lens
-        btn = hn[4 * b:4 * b + b * mb].reshape(b, mb)
+        hn[4 * b:4 * b + n] = seq_starts
+        btn = hn[5 * b:5 * b + b * mb].reshape(b, mb)
         btn.fill(self.dummy_block)
         for i, row in enumerate(block_rows):
             btn[i, :len(row)] = row
@@ -124,7 +131,8 @@ class GraphedDecoder:
             hn[b + n:2 * b] = 0
             hn[2 * b + n:3 * b] = self.dummy_block * _bs()
             hn[3 * b + n:4 * b] = 1
-        return self.h_staging[4 * b:4 * b + b * mb].view(b, mb)
+            hn[4 * b + n:5 * b] = 0
+        return self.h_staging[5 * b:5 * b + b * mb].view(b, mb)
 
 
 def _bs() -> int:

@@ -37,7 +37,8 @@ def broadcast_prefill(tokens, positions, slots):
     _bcast(slots)
 
 
-def broadcast_decode(tokens, positions, slots, block_tables, seq_lens):
+def broadcast_decode(tokens, positions, slots, block_tables, seq_lens,
+                     seq_starts):
     dev = tokens.device
     hdr = torch.tensor([OP_DECODE, tokens.shape[0], 0,
                         block_tables.shape[1]], dtype=torch.int64,
@@ -48,6 +49,7 @@ def broadcast_decode(tokens, positions, slots, block_tables, seq_lens):
     _bcast(slots)
     _bcast(block_tables)
     _bcast(seq_lens)
+    _bcast(seq_starts)
 
 
 def broadcast_shutdown(device):
@@ -77,4 +79,6 @@ def worker_loop(engine) -> None:
             slots = _bcast(torch.zeros(b, dtype=torch.int32, device=dev))
             bt = _bcast(torch.zeros(b, maxb, dtype=torch.int32, device=dev))
             seq_lens = _bcast(torch.zeros(b, dtype=torch.int32, device=dev))
-            model.decode(tokens, positions, caches, slots, bt, seq_lens)
+            seq_starts = _bcast(torch.zeros(b, dtype=torch.int32, device=dev))
+            model.decode(tokens, positions, caches, slots, bt, seq_lens,
+                         seq_starts=seq_starts)

@@ -182,13 +182,15 @@ def test_graphed_decoder_staging():
     gd = GraphedDecoder(model=None, caches=None, max_batch=8, max_blocks=4,
                         dummy_block=99, device="cpu")
     bt = gd._stage(4, tokens=[7, 8], positions=[3, 5], slots=[12, 20],
-                   block_rows=[[1, 2], [4]], seq_lens=[4, 6])
+                   block_rows=[[1, 2], [4]], seq_lens=[4, 6],
+                   seq_starts=[0, 2])
     h = gd.h_staging
     assert h[0:4].tolist() == [7, 8, 0, 0]          # tokens + pad
     assert h[4:8].tolist() == [3, 5, 0, 0]          # positions + pad
     assert h[8:10].tolist() == [12, 20]             # live slots
     assert (h[10:12] == 99 * 16).all()              # dummy slots
     assert h[12:16].tolist() == [4, 6, 1, 1]        # seq_lens + pad
+    assert h[16:20].tolist() == [0, 2, 0, 0]        # seq_starts + pad
     assert bt.shape == (4, 4)
     assert bt[0].tolist() == [1, 2, 99, 99]
     assert bt[1].tolist() == [4, 99, 99, 99]
@@ -348,7 +350,10 @@ def test_sliding_window_matches_masked_forward():
 
     mask = torch.zeros(L, L, dtype=torch.bool)
     for i in range(L):
-        start = 0 if i < P else max(0, bs * ((i + 1 - w) // bs))
+        # strict W: generated row i attends exactly [i+1-w, i] (the
+        # engine passes seq_starts to the decode kernel; block dropping
+        # only bounds memory)
+        start = 0 if i < P else max(0, i + 1 - w)
         mask[i, start:i + 1] = True
 
     def masked_attn(q, k, v, scale=None, q_block=256):
