@@ -129,7 +129,8 @@ def bench_serve(args) -> dict:
     eng = Engine(args.model, dtype=dtype,
                  kv_blocks=None if torch.cuda.is_available() else 4096,
                  max_batch=args.serve_batch, seed=17,
-                 load_in_8bit=args.serve_8bit)
+                 load_in_8bit=args.serve_8bit,
+                 kv_fp8=args.serve_fp8kv or None)
     if comm.is_dist() and comm.world_size() > 1:
         # TP serving: rank 0 drives the engine + measures; workers follow
         # the broadcast protocol until shutdown (serve/tp_worker.py).
@@ -178,8 +179,9 @@ def bench_serve(args) -> dict:
         # TP serving keeps total work fixed as ranks grow
         "scaling": "strong" if comm.world_size() > 1 else "weak",
         "vs_baseline": None,
-        "dtype": ("bf16-act+fp8-w" if args.serve_8bit else "bf16")
-                 if torch.cuda.is_available() else "float32",
+        "dtype": (("bf16-act+fp8-w" if args.serve_8bit else
+                   "bf16+fp8-kv" if args.serve_fp8kv else "bf16")
+                  if torch.cuda.is_available() else "float32"),
         "data": "synthetic",
         "config": {"model": args.model, "global_batch": args.serve_batch,
                    "seq_len": prompt_len,
@@ -202,6 +204,10 @@ def main():
     p.add_argument("--serve-8bit", action="store_true",
                    help="fp8-e4m3 weight-only decode (NOT the headline "
                         "config; reported with dtype=bf16-act+fp8-w)")
+    p.add_argument("--serve-fp8kv", action="store_true",
+                   help="fp8-e4m3 KV cache (2x KV capacity; NOT the "
+                        "headline config — reported with "
+                        "dtype=bf16+fp8-kv)")
     args = p.parse_args()
     if args.model is None:
         if torch.cuda.is_available():

@@ -69,7 +69,11 @@ def main():
         print(f"server: loaded weights from {model_dir}")
 
     in_8bit = os.environ.get("MODEL_LOAD_IN_8BIT", "").lower() in ("1", "true")
-    engine = Engine(model, load_in_8bit=in_8bit)
+    # PARAM_KV_FP8 / KV_FP8: e4m3 KV cache (2x capacity, long-context)
+    kv_fp8 = os.environ.get("PARAM_KV_FP8",
+                            os.environ.get("KV_FP8", "")).lower() in \
+        ("1", "true")
+    engine = Engine(model, load_in_8bit=in_8bit, kv_fp8=kv_fp8)
     tok = load_tokenizer(model_dir if model_dir.exists() else None)
     if comm.rank() == 0:
         serve_forever(engine, tok, port=int(os.environ.get("PORT", "8080")),
