@@ -36,6 +36,8 @@
 // the reference's external server image leans on cuBLAS
 // (substratusai/runbooks docs/container-contract.md serving contract).
 
+#include <stdlib.h>
+
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -45,7 +47,6 @@ namespace {
 
 constexpr int BLOCK = 256;   // 4 waves
 constexpr int MMAX = 32;
-constexpr int U = 8;         // k-steps (of 16) per unrolled chunk
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 typedef __attribute__((ext_vector_type(16))) float f32x16v;
@@ -80,7 +81,9 @@ __device__ __forceinline__ bf16x8v load8v(const uint16_t *p) {
 
 // STORE_BF16: write y bf16 directly (SPLIT == 1). Otherwise store an
 // fp32 slab slice at slabs + kslice*M*N for decode_gemm_combine.
-template <bool STORE_BF16>
+// U = k-steps (of 16) per unrolled chunk: 8 keeps 3 waves/SIMD, 12
+// trades occupancy (2/SIMD) for a deeper in-flight ring (RB_DG_U=12).
+template <bool STORE_BF16, int U = 8>
 __global__ __launch_bounds__(BLOCK, 1) void decode_gemm_kernel(
     const uint16_t *__restrict__ xs, const uint16_t *__restrict__ ws,
     uint16_t *__restrict__ yp, float *__restrict__ slabs,
@@ -264,6 +267,14 @@ at::Tensor decode_swizzle_x(at::Tensor x) {
 // split > 1 returns the fp32 slab [split, M, N] UNCOMBINED — the
 // consumer kernel (rmsnorm_res_slab_fwd_dec) folds the slices while it
 // reads, saving the combine launch inside the decode graph.
+static int dg_u() {
+  static int u = [] {
+    const char *e = getenv("RB_DG_U");
+    return (e != nullptr && atoi(e) == 12) ? 12 : 8;
+  }();
+  return u;
+}
+
 at::Tensor decode_gemm_raw(at::Tensor xs, at::Tensor ws, int64_t M,
                            int64_t N, int64_t K) {
   TORCH_CHECK(xs.is_cuda() && ws.is_cuda() && xs.is_contiguous() &&
@@ -275,20 +286,35 @@ at::Tensor decode_gemm_raw(at::Tensor xs, at::Tensor ws, int64_t M,
   const int split = (int)decode_gemm_split(N, K);
   if (split == 1) {
     auto y = at::empty({M, N}, xs.options());
-    hipLaunchKernelGGL((decode_gemm_kernel<true>), dim3(N / 32, 1),
-                       dim3(BLOCK), 0, stream,
-                       (const uint16_t *)xs.data_ptr(),
-                       (const uint16_t *)ws.data_ptr(),
-                       (uint16_t *)y.data_ptr(), nullptr,
-                       (int)M, (int)N, (int)K);
+    if (dg_u() == 12)
+      hipLaunchKernelGGL((decode_gemm_kernel<true, 12>), dim3(N / 32, 1),
+                         dim3(BLOCK), 0, stream,
+                         (const uint16_t *)xs.data_ptr(),
+                         (const uint16_t *)ws.data_ptr(),
+                         (uint16_t *)y.data_ptr(), nullptr,
+                         (int)M, (int)N, (int)K);
+    else
+      hipLaunchKernelGGL((decode_gemm_kernel<true, 8>), dim3(N / 32, 1),
+                         dim3(BLOCK), 0, stream,
+                         (const uint16_t *)xs.data_ptr(),
+                         (const uint16_t *)ws.data_ptr(),
+                         (uint16_t *)y.data_ptr(), nullptr,
+                         (int)M, (int)N, (int)K);
     return y;
   }
   auto slabs = at::empty({split, M, N}, xs.options().dtype(at::kFloat));
-  hipLaunchKernelGGL((decode_gemm_kernel<false>), dim3(N / 32, split),
-                     dim3(BLOCK), 0, stream,
-                     (const uint16_t *)xs.data_ptr(),
-                     (const uint16_t *)ws.data_ptr(), nullptr,
-                     (float *)slabs.data_ptr(), (int)M, (int)N, (int)K);
+  if (dg_u() == 12)
+    hipLaunchKernelGGL((decode_gemm_kernel<false, 12>), dim3(N / 32, split),
+                       dim3(BLOCK), 0, stream,
+                       (const uint16_t *)xs.data_ptr(),
+                       (const uint16_t *)ws.data_ptr(), nullptr,
+                       (float *)slabs.data_ptr(), (int)M, (int)N, (int)K);
+  else
+    hipLaunchKernelGGL((decode_gemm_kernel<false, 8>), dim3(N / 32, split),
+                       dim3(BLOCK), 0, stream,
+                       (const uint16_t *)xs.data_ptr(),
+                       (const uint16_t *)ws.data_ptr(), nullptr,
+                       (float *)slabs.data_ptr(), (int)M, (int)N, (int)K);
   return slabs;
 }
 
