@@ -165,19 +165,27 @@ class ControllerManager:
     def _watch_kind(self, api: str, kind: str, owned: bool) -> None:
         while not self._stop.is_set():
             try:
+                # Level-triggered re-list on every watch (re)establishment:
+                # anything created/updated while the watch was down would
+                # otherwise never reconcile (controller-runtime's informer
+                # does the same list+watch dance; reference
+                # internal/controller relies on it implicitly).
+                for obj in self.kube.list(api, kind):
+                    self._enqueue(kind, obj, owned)
                 for ev in self.kube.watch(api, kind, stop=self._stop):
-                    obj = ev["object"]
-                    if owned:
-                        req = self._owner_request(obj)
-                        if req:
-                            self._queue.put(req)
-                    else:
-                        m = obj["metadata"]
-                        self._queue.put(
-                            (kind, m.get("namespace", "default"), m["name"]))
+                    self._enqueue(kind, ev["object"], owned)
             except Exception:
                 log.exception("watch %s restarting", kind)
                 self._stop.wait(1.0)
+
+    def _enqueue(self, kind: str, obj: dict, owned: bool) -> None:
+        if owned:
+            req = self._owner_request(obj)
+            if req:
+                self._queue.put(req)
+        else:
+            m = obj["metadata"]
+            self._queue.put((kind, m.get("namespace", "default"), m["name"]))
 
 
 class LeaderElector:
