@@ -30,3 +30,7 @@ images:           ## build all workload container images
 installation-manifests: manifests  ## render install bundles
 	kubectl kustomize config/install-kind > install/kind/manifests.yaml || true
 	kubectl kustomize config/install-gcp > install/gcp/manifests.yaml || true
+
+# API reference docs (reference: crd-ref-docs target, Makefile:204-214)
+docs: ## regenerate docs/api.md from the CRD dataclasses
+	python scripts/gen_api_docs.py
