@@ -42,7 +42,16 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const int32_t *__restrict__ seq_starts, uint16_t *__restrict__ out,
     float *__restrict__ partial, int hkv, int bs, int max_blocks, int nsplit,
     float scale) {
-  constexpr int VE = DH / 16;              // elems per lane (16 lanes cover Dh)
+  // Lane-group geometry: GL lanes cover one token's Dh. bf16 reads 2 B
+  // per element, so 16 lanes x VE=DH/16 elems = 16 B/lane; fp8 reads
+  // 1 B per element, so 8-lane groups with VE=DH/8 keep the loads at
+  // the full 16 B width (8 B accesses run 0.54-0.70x on gfx950 — the
+  // first fp8 cut used 16-lane groups and showed ZERO long-context win).
+  // (G > 4 keeps 16-lane groups even for fp8: qreg[G][VE]+acc[G][VE]
+  // at VE=16 would spill past 256 VGPRs)
+  constexpr int GL = (FP8 && G <= 4) ? 8 : 16;
+  constexpr int VE = DH / GL;              // elems per lane
+  constexpr int TPW = 64 / GL;             // tokens per wave pass
   const int b = blockIdx.x;
   const int h_kv = blockIdx.y;
   const int split = SPLIT ? blockIdx.z : 0;
@@ -66,8 +75,8 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int grp = lane >> 4;               // 16-lane group: token sub-index
-  const int gl = lane & 15;                // lane within group
+  const int grp = lane / GL;               // lane group: token sub-index
+  const int gl = lane % GL;                // lane within group
   const int d0 = gl * VE;                  // this lane's Dh slice
 
   // Q for all G heads, pre-scaled (softmax scale folded into q).
@@ -79,10 +88,14 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     for (int e = 0; e < VE; ++e) qreg[g][e] = 0.f;
     {
       float tmp[VE];
-      if (VE == 8) rb::VIO<uint16_t>::load(qp, tmp);
-      else {
 #pragma unroll
-        for (int e = 0; e < VE; ++e) tmp[e] = rb::bf16_to_f32(qp[e]);
+      for (int e8 = 0; e8 < VE; e8 += 8) {
+        if (VE - e8 >= 8) rb::VIO<uint16_t>::load(qp + e8, tmp + e8);
+        else {
+#pragma unroll
+          for (int e = 0; e < VE % 8; ++e)
+            tmp[e8 + e] = rb::bf16_to_f32(qp[e8 + e]);
+        }
       }
 #pragma unroll
       for (int e = 0; e < VE; ++e) qreg[g][e] = tmp[e] * scale;
@@ -99,26 +112,29 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 
   const int32_t *bt = block_tables + (int64_t)b * max_blocks;
 
-  // Each wave covers tokens t_begin + i*16 + wid*4 + grp for i = 0,1,...
-  for (int t = t_begin + wid * 4 + grp; t < t_end; t += NW * 4) {
+  // Each wave covers tokens t_begin + i*NW*TPW + wid*TPW + grp.
+  for (int t = t_begin + wid * TPW + grp; t < t_end; t += NW * TPW) {
     const bool valid = true;
     const int tt = t;
     const int blk = bt[tt / bs];
 
     float kf[VE], vf[VE];
     if (FP8) {
-      constexpr int RB8 = DH + 8;
+      constexpr int RB8 = DH + 16;
       const uint8_t *k8 = reinterpret_cast<const uint8_t *>(k_cache);
       const uint8_t *v8 = reinterpret_cast<const uint8_t *>(v_cache);
       const int64_t row = (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * RB8;
 #pragma unroll
-      for (int e8 = 0; e8 < VE; e8 += 8) {
-        if (VE - e8 >= 8) {
-          rb::fp8x8_to_f32(k8 + row + d0 + e8, kf + e8);
-          rb::fp8x8_to_f32(v8 + row + d0 + e8, vf + e8);
+      for (int e16 = 0; e16 < VE; e16 += 16) {
+        if (VE - e16 >= 16) {
+          rb::fp8x16_to_f32(k8 + row + d0 + e16, kf + e16);
+          rb::fp8x16_to_f32(v8 + row + d0 + e16, vf + e16);
+        } else if (VE - e16 >= 8) {
+          rb::fp8x8_to_f32(k8 + row + d0 + e16, kf + e16);
+          rb::fp8x8_to_f32(v8 + row + d0 + e16, vf + e16);
         } else {
-          rb::fp8x4_to_f32(k8 + row + d0 + e8, kf + e8);
-          rb::fp8x4_to_f32(v8 + row + d0 + e8, vf + e8);
+          rb::fp8x4_to_f32(k8 + row + d0 + e16, kf + e16);
+          rb::fp8x4_to_f32(v8 + row + d0 + e16, vf + e16);
         }
       }
       const float ks = *reinterpret_cast<const float *>(k8 + row + DH);
@@ -145,9 +161,10 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
       float s = 0.f;
 #pragma unroll
       for (int e = 0; e < VE; ++e) s += qreg[g][e] * kf[e];
-      // reduce across the 16-lane group
+      // reduce across the lane group
 #pragma unroll
-      for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+      for (int off = GL / 2; off > 0; off >>= 1)
+        s += __shfl_xor(s, off, 64);
       if (!valid) s = -INFINITY;
       const float mn = fmaxf(m[g], s);
       if (mn != -INFINITY) {
@@ -161,10 +178,10 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
   }
 
-  // Merge the 4 groups inside each wave: lanes l and l^16 / l^32 hold the
+  // Merge the TPW groups inside each wave: lanes l^GL ... l^32 hold the
   // same d0 slice, so shfl_xor merges matching elements.
 #pragma unroll
-  for (int off = 16; off <= 32; off <<= 1) {
+  for (int off = GL; off <= 32; off <<= 1) {
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       const float mo = __shfl_xor(m[g], off, 64);
@@ -186,7 +203,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   // Merge across the 4 waves via LDS. Layout per (wave, head):
   // [DH acc][m][l] floats.
   __shared__ float lds[NW][G][DH + 2];
-  if (grp == 0) {   // lanes 0..15 of each wave hold the wave's merged state
+  if (grp == 0) {   // lanes 0..GL-1 of each wave hold the wave's merged state
 #pragma unroll
     for (int g = 0; g < G; ++g) {
 #pragma unroll
@@ -315,9 +332,9 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
 
   const bool fp8 = k_cache.scalar_type() == at::kByte;
   if (fp8) {
-    TORCH_CHECK((int)k_cache.size(3) == dh + 8 &&
+    TORCH_CHECK((int)k_cache.size(3) == dh + 16 &&
                 v_cache.scalar_type() == at::kByte,
-                "paged_decode: fp8 cache rows must be dh+8 bytes");
+                "paged_decode: fp8 cache rows must be dh+16 bytes");
   }
 #define RB_DEC(DHV, GV)                                                        \
   do {                                                                         \

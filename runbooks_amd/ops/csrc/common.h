@@ -64,9 +64,23 @@ RB_DEV void fp8x4_to_f32(const uint8_t *p, float *f) {
   f[0] = a[0]; f[1] = a[1]; f[2] = b[0]; f[3] = b[1];
 }
 
-// fp8 KV-cache row layout: dh e4m3 bytes, then the f32 scale, then pad
-// to 8 bytes. Row stride for head_dim dh:
-RB_DEV constexpr int fp8_row_bytes(int dh) { return dh + 8; }
+// 16 packed e4m3 (one 16 B vector) -> 16 f32
+RB_DEV void fp8x16_to_f32(const uint8_t *p, float *f) {
+  typedef __attribute__((ext_vector_type(4))) unsigned int u32x4i;
+  const u32x4i w = *reinterpret_cast<const u32x4i *>(p);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    auto a = __builtin_amdgcn_cvt_pk_f32_fp8(w[i], false);
+    auto b = __builtin_amdgcn_cvt_pk_f32_fp8(w[i], true);
+    f[i * 4 + 0] = a[0]; f[i * 4 + 1] = a[1];
+    f[i * 4 + 2] = b[0]; f[i * 4 + 3] = b[1];
+  }
+}
+
+// fp8 KV-cache row layout: dh e4m3 bytes, the f32 scale, pad to 16 B
+// (16-aligned rows keep the decode loop on full 16 B/lane loads —
+// 8 B accesses run at 0.54-0.70x the 16 B rate on gfx950).
+RB_DEV constexpr int fp8_row_bytes(int dh) { return dh + 16; }
 struct f32x8 { float v[8]; };
 
 RB_DEV f32x8 to_f32(const bf16x8 &a) {

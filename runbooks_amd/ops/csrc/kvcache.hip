@@ -55,7 +55,7 @@ __global__ void kv_append_fp8_kernel(const uint16_t *__restrict__ k,
                                      int64_t n_tokens, int hkv, int bs,
                                      int dh) {
   const int ve = dh / 16;                   // elems per lane (<= 16)
-  const int rb8 = dh + 8;
+  const int rb8 = dh + 16;
   const int64_t units = n_tokens * hkv * 2; // (token, head, k-or-v)
   const int64_t groups_per_block = BLOCK / 16;
   const int gl = threadIdx.x & 15;
@@ -102,7 +102,7 @@ void kv_append(at::Tensor k, at::Tensor v, at::Tensor k_cache, at::Tensor v_cach
   const int bs = (int)k_cache.size(2);
   auto stream = at::hip::getCurrentHIPStream();
   if (k_cache.scalar_type() == at::kByte) {
-    const int dh = (int)k_cache.size(3) - 8;
+    const int dh = (int)k_cache.size(3) - 16;
     TORCH_CHECK(dh % 16 == 0 && dh <= 256, "kv_append fp8: Dh % 16");
     const int grid = rb::rb_grid_1d(n_tokens * hkv * 2 * 16, BLOCK);
     hipLaunchKernelGGL(kv_append_fp8_kernel, dim3(grid), dim3(BLOCK), 0,

@@ -15,7 +15,9 @@ def alloc_kv_cache(num_blocks: int, num_kv_heads: int, head_dim: int,
     half the KV bytes of bf16 (decode attention is KV-bandwidth bound at
     long context) and double the capacity within 288 GB."""
     if fp8:
-        shape = (num_blocks, num_kv_heads, block_size, head_dim + 8)
+        # dh e4m3 bytes + f32 scale + pad to 16 (16-aligned rows: the
+        # decode loop stays on full 16 B/lane loads)
+        shape = (num_blocks, num_kv_heads, block_size, head_dim + 16)
         k = torch.zeros(shape, device=device, dtype=torch.uint8)
         v = torch.zeros(shape, device=device, dtype=torch.uint8)
         return k, v
@@ -35,8 +37,8 @@ def fp8_quant_row_ref(x: torch.Tensor):
 
 
 def fp8_dequant_cache_ref(cache: torch.Tensor) -> torch.Tensor:
-    """[blocks, hkv, bs, dh+8] uint8 -> [blocks, hkv, bs, dh] f32."""
-    dh = cache.shape[-1] - 8
+    """[blocks, hkv, bs, dh+16] uint8 -> [blocks, hkv, bs, dh] f32."""
+    dh = cache.shape[-1] - 16
     data = cache[..., :dh].contiguous().view(torch.float8_e4m3fn).float()
     scale = cache[..., dh:dh + 4].contiguous().view(torch.float32)
     return data * scale
@@ -44,7 +46,7 @@ def fp8_dequant_cache_ref(cache: torch.Tensor) -> torch.Tensor:
 
 def _kv_append_fp8_ref(k, v, k_cache, v_cache, slot_mapping):
     bs = k_cache.shape[2]
-    dh = k_cache.shape[-1] - 8
+    dh = k_cache.shape[-1] - 16
     for t in range(k.shape[0]):
         slot = int(slot_mapping[t])
         if slot < 0:

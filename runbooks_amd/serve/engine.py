@@ -192,7 +192,7 @@ class Engine:
                       "prefix_hit_blocks": 0, "window_dropped_blocks": 0}
 
     def _auto_kv_blocks(self, mem_fraction: float) -> int:
-        elem = (self.cfg.head_dim + 8) if self.kv_fp8 \
+        elem = (self.cfg.head_dim + 16) if self.kv_fp8 \
             else self.cfg.head_dim * 2
         bytes_per_block = (2 * self.cfg.num_layers * self.model.local_kv_heads()
                            * self.bs * elem)

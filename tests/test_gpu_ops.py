@@ -660,13 +660,13 @@ def test_fp8_kv_engine_gpu():
     e8 = Engine(m, device=DEV, kv_blocks=128, seed=2, kv_fp8=True)
     out = e8.generate([3, 1, 4, 1, 5], max_new_tokens=8)
     assert len(out) == 8 and all(0 <= t < m.cfg.vocab_size for t in out)
-    # capacity: fp8 blocks cost (dh+8)/2dh of bf16 bytes. Compare the
+    # capacity: fp8 blocks cost (dh+16)/2dh of bf16 bytes. Compare the
     # auto-sizing math directly (constructing two engines back to back
     # would let the first one's allocations shrink the second's budget).
     bf16_bpb = (2 * m.cfg.num_layers * m.cfg.num_kv_heads * e8.bs *
                 m.cfg.head_dim * 2)
     fp8_bpb = (2 * m.cfg.num_layers * m.cfg.num_kv_heads * e8.bs *
-               (m.cfg.head_dim + 8))
+               (m.cfg.head_dim + 16))
     assert fp8_bpb * 1.5 < bf16_bpb
     del e8
     torch.cuda.empty_cache()

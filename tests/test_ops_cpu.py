@@ -141,7 +141,7 @@ def test_fp8_kv_quant_roundtrip_cpu():
     from runbooks_amd.ops import kvcache as kc
     torch.manual_seed(0)
     k, v = kc.alloc_kv_cache(8, 4, 32, "cpu", fp8=True)
-    assert k.shape == (8, 4, 16, 40) and k.dtype == torch.uint8
+    assert k.shape == (8, 4, 16, 48) and k.dtype == torch.uint8
     x = torch.randn(5, 4, 32) * 3.0
     kc.kv_append_ref(x, x * 0.5, k, v, torch.tensor([3, 19, 48, 0, 127]))
     dq_k = kc.fp8_dequant_cache_ref(k)
