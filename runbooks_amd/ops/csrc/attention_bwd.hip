@@ -200,6 +200,9 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dq_kernel(
       }
     }
     // ---- stage K transposed ([d][key]) -----------------------------------
+    // from the LDS row image (same fix as dkdv: the global column-
+    // strided scalar re-read was latency-bound)
+    __syncthreads();
     {
       constexpr int GRPS = BLOCK / DH;
       constexpr int KPG = TILE / GRPS;
@@ -208,10 +211,8 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dq_kernel(
       uint16_t tmp[KPG];
 #pragma unroll
       for (int e = 0; e < KPG; ++e) {
-        const int gk = kbase + kg0 + e;
-        tmp[e] = (gk < S)
-            ? kp[((int64_t)(b * S + gk) * Hkv + h_kv) * DH + d]
-            : (uint16_t)0;
+        tmp[e] = *reinterpret_cast<const uint16_t *>(
+            k_img + (kg0 + e) * ROW_STRIDE + d * 2);
       }
 #pragma unroll
       for (int c8 = 0; c8 < KPG / 8; ++c8)
@@ -380,6 +381,11 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dkdv_kernel(
       }
     }
     // ---- stage Q^T and dO^T ([d][qrow]) ----------------------------------
+    // Sourced from the just-staged LDS row images (50-cycle LDS reads)
+    // instead of re-reading global column-strided with scalar loads
+    // (~900-cycle HBM/L2 round trips per element — measured r9: dkdv at
+    // 235 us was the largest single train kernel).
+    __syncthreads();
     {
       constexpr int GRPS = BLOCK / DH;
       constexpr int KPG = TILE / GRPS;
@@ -388,15 +394,11 @@ __global__ __launch_bounds__(BLOCK, 2) void fa_bwd_dkdv_kernel(
       uint16_t tq[KPG], td[KPG];
 #pragma unroll
       for (int e = 0; e < KPG; ++e) {
-        const int gq = qbase + qg0 + e;
-        if (gq < S) {
-          const int64_t base = ((int64_t)(b * S + gq) * Hq + h) * DH + d;
-          tq[e] = qp[base];
-          td[e] = dop[base];
-        } else {
-          tq[e] = 0;
-          td[e] = 0;
-        }
+        const int qr = qg0 + e;
+        tq[e] = *reinterpret_cast<const uint16_t *>(
+            q_img + qr * ROW_STRIDE + d * 2);
+        td[e] = *reinterpret_cast<const uint16_t *>(
+            do_img + qr * ROW_STRIDE + d * 2);
       }
 #pragma unroll
       for (int c8 = 0; c8 < KPG / 8; ++c8) {
