@@ -133,9 +133,12 @@ class Attention(nn.Module):
         else:
             q, k, v = self._qkv(x, cos, sin, positions)
             ops.kv_append(k, v, kc, vc, slot_mapping)
-        o = ops.paged_decode(q, kc, vc, block_tables, seq_lens, scale=self.scale,
-                             nsplit=nsplit, seq_starts=seq_starts)
+        o, oswz = ops.paged_decode_with_operand(
+            q, kc, vc, block_tables, seq_lens, scale=self.scale,
+            nsplit=nsplit, seq_starts=seq_starts)
         o2 = o.reshape(B, self.hq * self.dh)
+        if oswz is not None:
+            o2._rb_swz = oswz
         if slab_ok and self.tp == 1 and self.o_proj.bias is None:
             out, _ = ops.decode_linear_raw(o2, self.o_proj.weight)
             return out

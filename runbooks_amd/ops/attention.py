@@ -163,3 +163,29 @@ def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
         )
     return paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens,
                             scale, seq_starts=seq_starts)
+
+
+def paged_decode_with_operand(q, k_cache, v_cache, block_tables, seq_lens,
+                              scale: float | None = None,
+                              nsplit: int | None = None, seq_starts=None):
+    """paged_decode that ALSO returns the decode-GEMM operand layout of
+    the attention output (the o_proj input swizzle fused into the
+    epilogue) when the kernel emits it; (out, swz_or_None)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _backend.use_hip(q):
+        if nsplit is None:
+            B, hkv = q.shape[0], k_cache.shape[1]
+            base = B * hkv
+            max_len = int(seq_lens.max())
+            nsplit = 1
+            if base < 256 and max_len > 256:
+                nsplit = min(16, max(1, (2 * 256) // max(base, 1)))
+                nsplit = min(nsplit, (max_len + 255) // 256)
+                nsplit = max(nsplit, 1)
+        res = _backend.ext().paged_decode_swz(
+            q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
+            int(nsplit), float(scale), seq_starts=seq_starts)
+        return res[0], (res[1] if len(res) > 1 else None)
+    return paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens,
+                            scale, seq_starts=seq_starts), None

@@ -40,6 +40,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const uint16_t *__restrict__ v_cache, const int32_t *__restrict__ block_tables,
     const int32_t *__restrict__ seq_lens,
     const int32_t *__restrict__ seq_starts, uint16_t *__restrict__ out,
+    uint16_t *__restrict__ out_swz,
     float *__restrict__ partial, int hkv, int bs, int max_blocks, int nsplit,
     float scale) {
   // Lane-group geometry: GL lanes cover one token's Dh. bf16 reads 2 B
@@ -241,13 +242,29 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
       uint16_t *op = out + ((int64_t)b * Hq + hq0 + g) * DH + d0;
 #pragma unroll
       for (int e = 0; e < VE; ++e) op[e] = rb::f32_to_bf16(av[e] * inv_l);
+      if (out_swz != nullptr) {
+        // also emit the decode-GEMM operand layout for o_proj
+        // ([K/16][2][32][8] over K = Hq*DH, m = b): saves the standalone
+        // decode_swizzle_x launch per layer. VE >= 8, d0 8-aligned.
+#pragma unroll
+        for (int e8 = 0; e8 < VE; e8 += 8) {
+          const int kk = (hq0 + g) * DH + d0 + e8;
+          uint16_t *sp = out_swz + (kk >> 4) * 512 + ((kk >> 3) & 1) * 256 +
+                         b * 8;
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            sp[e] = rb::f32_to_bf16(av[e8 + e] * inv_l);
+        }
+      }
     }
   }
 }
 
 // Second pass for SPLIT mode: merge nsplit partials per (b, hq) row.
 __global__ void decode_combine_kernel(const float *__restrict__ partial,
-                                      uint16_t *__restrict__ out, int nsplit,
+                                      uint16_t *__restrict__ out,
+                                      uint16_t *__restrict__ out_swz,
+                                      int Hq, int nsplit,
                                       int dh) {
   const int64_t row = blockIdx.x;          // b * Hq + hq
   const float *p = partial + row * (int64_t)nsplit * (dh + 2);
@@ -258,13 +275,21 @@ __global__ void decode_combine_kernel(const float *__restrict__ partial,
     if (p[s * (dh + 2) + dh] != -INFINITY)
       ll += p[s * (dh + 2) + dh + 1] * __expf(p[s * (dh + 2) + dh] - mm);
   const float inv_l = (ll > 0.f) ? 1.0f / ll : 0.f;
+  const int b = (int)(row / Hq);
+  const int h = (int)(row % Hq);
   for (int d = threadIdx.x; d < dh; d += blockDim.x) {
     float o = 0.f;
     for (int s = 0; s < nsplit; ++s) {
       const float ms = p[s * (dh + 2) + dh];
       if (ms != -INFINITY) o += p[s * (dh + 2) + d] * __expf(ms - mm);
     }
-    out[row * dh + d] = rb::f32_to_bf16(o * inv_l);
+    const uint16_t ob = rb::f32_to_bf16(o * inv_l);
+    out[row * dh + d] = ob;
+    if (out_swz != nullptr) {
+      const int kk = h * dh + d;
+      out_swz[(kk >> 4) * 512 + ((kk >> 3) & 1) * 256 + b * 8 +
+              (kk & 7)] = ob;
+    }
   }
 }
 
@@ -273,7 +298,8 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                    const at::Tensor &v_cache, const at::Tensor &block_tables,
                    const at::Tensor &seq_lens,
                    const c10::optional<at::Tensor> &seq_starts,
-                   at::Tensor &out, int nsplit,
+                   at::Tensor &out, const c10::optional<at::Tensor> &out_swz,
+                   int nsplit,
                    float scale, hipStream_t stream) {
   const int32_t *starts = seq_starts.has_value()
       ? seq_starts->data_ptr<int32_t>() : nullptr;
@@ -281,6 +307,8 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
   const int hkv = (int)k_cache.size(1);
   const int bs = (int)k_cache.size(2);
   const int max_blocks = (int)block_tables.size(1);
+  uint16_t *swz = out_swz.has_value() ? (uint16_t *)out_swz->data_ptr()
+                                      : nullptr;
   if (nsplit <= 1) {
     hipLaunchKernelGGL((paged_decode_kernel<DH, G, false, FP8>),
                        dim3(B, hkv, 1),
@@ -289,7 +317,8 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                        (const uint16_t *)v_cache.data_ptr(),
                        block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
                        starts,
-                       (uint16_t *)out.data_ptr(), nullptr, hkv, bs, max_blocks, 1, scale);
+                       (uint16_t *)out.data_ptr(), swz, nullptr, hkv, bs,
+                       max_blocks, 1, scale);
   } else {
     const int Hq = hkv * G;
     auto partial = at::empty({B, Hq, nsplit, DH + 2},
@@ -301,19 +330,39 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                        (const uint16_t *)v_cache.data_ptr(),
                        block_tables.data_ptr<int32_t>(), seq_lens.data_ptr<int32_t>(),
                        starts,
-                       nullptr, partial.data_ptr<float>(), hkv, bs, max_blocks, nsplit,
-                       scale);
+                       nullptr, nullptr, partial.data_ptr<float>(), hkv, bs,
+                       max_blocks, nsplit, scale);
     hipLaunchKernelGGL(decode_combine_kernel, dim3(B * Hq), dim3(256), 0, stream,
-                       partial.data_ptr<float>(), (uint16_t *)out.data_ptr(), nsplit, DH);
+                       partial.data_ptr<float>(), (uint16_t *)out.data_ptr(),
+                       swz, Hq, nsplit, DH);
   }
 }
 
 }  // namespace
 
+std::vector<at::Tensor> paged_decode_swz(at::Tensor q, at::Tensor k_cache,
+                                         at::Tensor v_cache,
+                                         at::Tensor block_tables,
+                                         at::Tensor seq_lens, int64_t nsplit,
+                                         double scale,
+                                         c10::optional<at::Tensor> seq_starts);
+
 at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                         at::Tensor block_tables, at::Tensor seq_lens,
                         int64_t nsplit, double scale,
                         c10::optional<at::Tensor> seq_starts) {
+  return paged_decode_swz(q, k_cache, v_cache, block_tables, seq_lens,
+                          nsplit, scale, seq_starts)[0];
+}
+
+// Variant also returning the decode-GEMM operand layout of the output
+// (o_proj input swizzle fused into the attention epilogue).
+std::vector<at::Tensor> paged_decode_swz(at::Tensor q, at::Tensor k_cache,
+                                         at::Tensor v_cache,
+                                         at::Tensor block_tables,
+                                         at::Tensor seq_lens, int64_t nsplit,
+                                         double scale,
+                                         c10::optional<at::Tensor> seq_starts) {
   if (seq_starts.has_value()) {
     TORCH_CHECK(seq_starts->scalar_type() == at::kInt &&
                 seq_starts->is_contiguous(), "paged_decode: seq_starts");
@@ -328,6 +377,12 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
   TORCH_CHECK(Hq % hkv == 0, "paged_decode: Hq % Hkv");
   const int G = Hq / hkv;
   auto out = at::empty_like(q);
+  // emit the o_proj operand layout only for decode-batch shapes
+  c10::optional<at::Tensor> out_swz;
+  const int B0 = (int)q.size(0);
+  if (B0 <= 32 && (Hq * dh) % 16 == 0) {
+    out_swz = at::empty({(int64_t)(Hq * dh / 16) * 512}, q.options());
+  }
   auto stream = at::hip::getCurrentHIPStream();
 
   const bool fp8 = k_cache.scalar_type() == at::kByte;
@@ -339,13 +394,13 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
 #define RB_DEC(DHV, GV)                                                        \
   do {                                                                         \
     if (fp8)                                                                   \
-      launch_decode<DHV, GV, true>(q, k_cache, v_cache, block_tables,          \
-                                   seq_lens, seq_starts, out, (int)nsplit,     \
-                                   (float)scale, stream);                      \
+      launch_decode<DHV, GV, true>(q, k_cache, v_cache, block_tables,         \
+                                   seq_lens, seq_starts, out, out_swz,        \
+                                   (int)nsplit, (float)scale, stream);        \
     else                                                                       \
-      launch_decode<DHV, GV, false>(q, k_cache, v_cache, block_tables,         \
-                                    seq_lens, seq_starts, out, (int)nsplit,    \
-                                    (float)scale, stream);                     \
+      launch_decode<DHV, GV, false>(q, k_cache, v_cache, block_tables,        \
+                                    seq_lens, seq_starts, out, out_swz,       \
+                                    (int)nsplit, (float)scale, stream);       \
   } while (0)
   if (dh == 128) {
     switch (G) {
@@ -376,5 +431,6 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
     TORCH_CHECK(false, "paged_decode: Dh must be 64/128/256, got ", dh);
   }
 #undef RB_DEC
-  return out;
+  if (out_swz.has_value()) return {out, *out_swz};
+  return {out};
 }

@@ -30,6 +30,12 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor k_cache, at::Tensor v_cache,
                         at::Tensor block_tables, at::Tensor seq_lens,
                         int64_t nsplit, double scale,
                         c10::optional<at::Tensor> seq_starts);
+std::vector<at::Tensor> paged_decode_swz(at::Tensor q, at::Tensor k_cache,
+                                         at::Tensor v_cache,
+                                         at::Tensor block_tables,
+                                         at::Tensor seq_lens, int64_t nsplit,
+                                         double scale,
+                                         c10::optional<at::Tensor> seq_starts);
 at::Tensor flash_prefill(at::Tensor q, at::Tensor k, at::Tensor v, double scale);
 std::vector<at::Tensor> flash_fwd_train(at::Tensor q, at::Tensor k, at::Tensor v,
                                         double scale);
@@ -88,6 +94,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_tokens", &sample_tokens, "greedy / Gumbel-max sampling");
   m.def("kv_append", &kv_append, "paged KV-cache append");
   m.def("paged_decode", &paged_decode, "paged GQA/MQA decode attention",
+        py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+        py::arg("block_tables"), py::arg("seq_lens"), py::arg("nsplit"),
+        py::arg("scale"), py::arg("seq_starts") = py::none());
+  m.def("paged_decode_swz", &paged_decode_swz,
+        "paged decode also emitting the o_proj decode-GEMM operand",
         py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
         py::arg("block_tables"), py::arg("seq_lens"), py::arg("nsplit"),
         py::arg("scale"), py::arg("seq_starts") = py::none());

@@ -693,3 +693,28 @@ def test_paged_decode_seq_starts():
                                seq_starts=st)
         d = (got.float().cpu() - ref).abs().max() / ref.abs().max()
         assert d < 3e-2, (nsplit, d)
+
+
+def test_paged_decode_with_operand_swz():
+    """The fused o_proj operand emit matches decode_swizzle_x of the
+    attention output."""
+    _assert_hip()
+    torch.manual_seed(4)
+    B, hkv, G, bs, nblk, dh = 4, 8, 4, 16, 16, 128
+    kc = torch.randn(nblk, hkv, bs, dh, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    q = torch.randn(B, hkv * G, dh, dtype=torch.bfloat16, device=DEV)
+    bt = torch.arange(B * 4, dtype=torch.int32, device=DEV).reshape(B, 4)
+    sl = torch.tensor([60, 33, 12, 7], dtype=torch.int32, device=DEV)
+    for nsplit in (1, 4):
+        out, swz = ops.paged_decode_with_operand(q, kc, vc, bt, sl,
+                                                 nsplit=nsplit)
+        assert swz is not None
+        o2 = out.reshape(B, hkv * G * dh).contiguous()
+        ref = ops.ext().decode_swizzle_x(o2)
+        # rows m >= B are undefined in both layouts; compare live rows
+        K = hkv * G * dh
+        s_v = swz.view(K // 16, 2, 32, 8)
+        r_v = ref.view(K // 16, 2, 32, 8)
+        d = (s_v[:, :, :B].float() - r_v[:, :, :B].float()).abs().max()
+        assert d == 0, (nsplit, d)
