@@ -18,7 +18,8 @@ from safetensors.torch import load_file, save_file
 
 
 def save_checkpoint(out_dir: str | Path, step: int, model_state: dict,
-                    optim_state: dict | None = None, keep: int = 3) -> Path:
+                    optim_state: dict | None = None, keep: int = 3,
+                    model_cfg=None) -> Path:
     out_dir = Path(out_dir)
     out_dir.mkdir(parents=True, exist_ok=True)
     ckpt = out_dir / f"checkpoint-{step}"
@@ -28,6 +29,14 @@ def save_checkpoint(out_dir: str | Path, step: int, model_state: dict,
     if optim_state is not None:
         torch.save(optim_state, tmp / "optimizer.pt")
     (tmp / "meta.json").write_text(json.dumps({"step": step}))
+    if model_cfg is not None:
+        # the serving contract: a Server pointed at these artifacts must
+        # be able to reconstruct the architecture (reference flow:
+        # finetuned Model -> Server mounts its artifacts)
+        import dataclasses
+        (tmp / "config.json").write_text(json.dumps({
+            "runbooks_amd_config": model_cfg.name,
+            "runbooks_amd_fields": dataclasses.asdict(model_cfg)}))
     if ckpt.exists():
         import shutil
         shutil.rmtree(ckpt)

@@ -190,6 +190,7 @@ class Trainer:
         state = (self.model.state_dict() if self.cfg.full_finetune
                  else lora_state_dict(self.model))
         ckpt_mod.save_checkpoint(self.cfg.output_dir, self.step_num, state,
+                                 model_cfg=self.model.cfg,
                                  optim_state=self.optimizer.state_dict())
 
     def resume(self) -> bool:

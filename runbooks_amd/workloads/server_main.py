@@ -48,6 +48,16 @@ def main():
 
     comm.init_from_env()
     model_dir = Path(os.environ.get("MODEL_DIR", "/content/model"))
+    # A fine-tuned Model's artifacts hold trainer checkpoints
+    # (checkpoint-N/model.safetensors + meta.json) rather than top-level
+    # files: serve the latest checkpoint (the Server CRD points at the
+    # Model's artifacts — reference server_controller.go model mount).
+    if not list(model_dir.glob("*.safetensors")):
+        ckpts = sorted(model_dir.glob("checkpoint-*"),
+                       key=lambda p: int(p.name.split("-")[-1]))
+        if ckpts:
+            model_dir = ckpts[-1]
+            print(f"server: serving latest checkpoint {model_dir}")
     arch = os.environ.get("PARAM_MODEL") or "llama2-7b"
     marker = model_dir / "config.json"
     cfg = None
@@ -56,6 +66,10 @@ def main():
         name = meta.get("runbooks_amd_config")
         if name and name in list_configs():
             cfg = get_config(name)
+        elif "runbooks_amd_fields" in meta:
+            from ..models import ModelConfig
+            from ..models.config import register
+            cfg = register(ModelConfig(**meta["runbooks_amd_fields"]))
         elif "model_type" in meta or "architectures" in meta:
             cfg = config_from_hf_json(marker)
     if cfg is None:

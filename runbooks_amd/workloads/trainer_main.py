@@ -131,13 +131,33 @@ def main():
     trainer.resume()
     trainer.fit(dataset, eval_dataset=eval_dataset)
     if comm.rank() == 0:
+        # Final SERVABLE model at the artifacts root: merge LoRA into
+        # the base and save full weights + the architecture marker —
+        # a Server pointed at this Model's artifacts loads it directly
+        # (reference flow: finetuned-model.yaml -> server.yaml mounts
+        # the model at /content/model). Intermediate checkpoint-N dirs
+        # stay adapter-only for cheap resume.
+        import dataclasses
+
+        from safetensors.torch import save_file
+
+        from ..train.lora import merge_lora
+        if not cfg.full_finetune:
+            merge_lora(trainer.model)
+        save_file({k: v.contiguous().cpu()
+                   for k, v in trainer.model.state_dict().items()
+                   if not k.startswith("rope_")},
+                  str(Path(out_dir) / "model.safetensors"))
+        (Path(out_dir) / "config.json").write_text(json.dumps({
+            "runbooks_amd_config": trainer.model.cfg.name,
+            "runbooks_amd_fields": dataclasses.asdict(trainer.model.cfg)}))
         # artifact-completeness marker (reference docs/design.md
         # "Buckets": reconcile logic can check completed.json in the
         # bucket after cluster re-creation)
         (Path(out_dir) / "completed.json").write_text(json.dumps(
             {"completed": True, "step": trainer.step_num}))
         print(f"trainer: done at step {trainer.step_num}; "
-              f"artifacts in {out_dir}")
+              f"merged servable model + artifacts in {out_dir}")
     if comm.is_dist():
         torch.distributed.destroy_process_group()
     return 0
