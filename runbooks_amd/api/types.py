@@ -376,8 +376,15 @@ class Model(_Base):
         spec = d.get("spec") or {}
         if spec.get("model"):
             o.model = ObjectRef.from_dict(spec["model"])
+        elif spec.get("modelName"):
+            # legacy flat form used by two of the reference's own
+            # examples (facebook-opt-125m/finetuned-{notebook,server});
+            # the CRD field is model: ObjectRef (server_types.go:27)
+            o.model = ObjectRef(str(spec["modelName"]))
         if spec.get("dataset"):
             o.dataset = ObjectRef.from_dict(spec["dataset"])
+        elif spec.get("datasetName"):
+            o.dataset = ObjectRef(str(spec["datasetName"]))
         return o
 
 
@@ -406,6 +413,8 @@ class Server(_Base):
         spec = d.get("spec") or {}
         if spec.get("model"):
             o.model = ObjectRef.from_dict(spec["model"])
+        elif spec.get("modelName"):
+            o.model = ObjectRef(str(spec["modelName"]))
         return o
 
 
@@ -439,8 +448,15 @@ class Notebook(_Base):
         o.suspend = spec.get("suspend")
         if spec.get("model"):
             o.model = ObjectRef.from_dict(spec["model"])
+        elif spec.get("modelName"):
+            # legacy flat form used by two of the reference's own
+            # examples (facebook-opt-125m/finetuned-{notebook,server});
+            # the CRD field is model: ObjectRef (server_types.go:27)
+            o.model = ObjectRef(str(spec["modelName"]))
         if spec.get("dataset"):
             o.dataset = ObjectRef.from_dict(spec["dataset"])
+        elif spec.get("datasetName"):
+            o.dataset = ObjectRef(str(spec["datasetName"]))
         return o
 
 
