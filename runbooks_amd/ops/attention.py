@@ -56,7 +56,9 @@ def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
     Differentiable. GPU bf16 (Dh 64/128) runs the MFMA flash kernels;
     otherwise the chunked fp32 reference below (also the numerics oracle
-    for the GPU test).
+    for the GPU test). Dh=256 has an MFMA INFERENCE prefill
+    (flash_prefill) but trains on the reference (fa_bwd has no 256
+    tile yet).
     """
     if _backend.use_hip(q) and q.shape[-1] in (64, 128) \
             and q.dtype == torch.bfloat16:
@@ -104,7 +106,7 @@ def flash_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _backend.use_hip(q) and q.shape[-1] in (64, 128) \
+    if _backend.use_hip(q) and q.shape[-1] in (64, 128, 256) \
             and q.dtype == torch.bfloat16:
         return _backend.ext().flash_prefill(
             q.contiguous(), k.contiguous(), v.contiguous(), float(scale))

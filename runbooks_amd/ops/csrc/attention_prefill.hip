@@ -267,8 +267,17 @@ static at::Tensor flash_fwd_impl(at::Tensor q, at::Tensor k, at::Tensor v,
                        (const uint16_t *)k.data_ptr(),
                        (const uint16_t *)v.data_ptr(), (uint16_t *)out.data_ptr(),
                        lsep, B, S, Hq, Hkv, (float)scale);
+  } else if (DH == 256) {
+    // gemma-7b wide heads: 70.7 KB LDS, fits; closes the last eager
+    // prefill fallback (round-1 NOTES item).
+    constexpr size_t shmem = KVB * (256 * 2 + 16) + 256 * (KVB * 2 + 16);
+    hipLaunchKernelGGL((flash_prefill_kernel<256>), grid, dim3(BLOCK), shmem,
+                       stream, (const uint16_t *)q.data_ptr(),
+                       (const uint16_t *)k.data_ptr(),
+                       (const uint16_t *)v.data_ptr(), (uint16_t *)out.data_ptr(),
+                       lsep, B, S, Hq, Hkv, (float)scale);
   } else {
-    TORCH_CHECK(false, "flash_prefill: DH must be 64 or 128, got ", DH);
+    TORCH_CHECK(false, "flash_prefill: DH must be 64, 128 or 256, got ", DH);
   }
   return out;
 }

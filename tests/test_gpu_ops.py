@@ -718,3 +718,19 @@ def test_paged_decode_with_operand_swz():
         r_v = ref.view(K // 16, 2, 32, 8)
         d = (s_v[:, :, :B].float() - r_v[:, :, :B].float()).abs().max()
         assert d == 0, (nsplit, d)
+
+
+def test_flash_prefill_dh256():
+    """gemma-width MFMA prefill (Dh=256) vs the fp32 chunked reference —
+    the last eager inference fallback closed."""
+    _assert_hip()
+    torch.manual_seed(6)
+    B, S, Hq, Hkv, DH = 2, 192, 4, 2, 256
+    q = torch.randn(B, S, Hq, DH, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(B, S, Hkv, DH, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn_like(k)
+    got = ops.flash_prefill(q, k, v)
+    ref = ops.causal_attention(q.float().cpu(), k.float().cpu(),
+                               v.float().cpu())
+    d = (got.float().cpu() - ref).abs().max() / ref.abs().max()
+    assert d < 3e-2, d
