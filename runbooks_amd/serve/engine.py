@@ -126,6 +126,11 @@ class Engine:
                 f"{model.cfg.name} has {model.cfg.head_dim}")
         self.bs = ops.BLOCK_SIZE
         self.max_batch = max_batch
+        # Admission budget per step: the default 1 matches the classic
+        # one-prefill-then-decode cadence (decode latency bounded); the
+        # HTTP server raises it for bursty arrival patterns where
+        # admission throughput dominates (scripts/bench_http.py).
+        self.max_prefills_per_step = 1
         self.seed = seed
         # fp8-e4m3 KV cache (RB_KV_FP8=1 or kv_fp8=True): halves the
         # decode-attention read stream and doubles KV capacity — the
@@ -497,7 +502,8 @@ class Engine:
                     f"prompt of {len(r.prompt_ids)} tokens cannot fit the KV "
                     f"cache ({self.allocator.num_blocks} blocks)")
         tr = get_tracer()
-        if req is not None:
+        n_admitted = 0
+        while req is not None:
             try:
                 if tr:
                     with tr.span("prefill", tokens=len(req.prompt_ids),
@@ -521,6 +527,9 @@ class Engine:
             self.running.append(req)
             # the prefill token may already satisfy the request
             self._sweep_finished(finished)
+            n_admitted += 1
+            req = (self._admit()
+                   if n_admitted < self.max_prefills_per_step else None)
         # decode every step — admissions must not starve running
         # sequences (a freshly prefilled request decodes its second
         # token in the same step it was admitted)

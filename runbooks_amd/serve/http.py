@@ -190,6 +190,7 @@ def build_app(engine: Engine, tokenizer=None,
               max_queue: int = 512) -> FastAPI:
     import contextlib
 
+    engine.max_prefills_per_step = max(engine.max_prefills_per_step, 4)
     loop = EngineLoop(engine)
 
     @contextlib.asynccontextmanager

@@ -679,3 +679,24 @@ def test_prefill_failure_releases_blocks():
     while not r.finished:
         eng.step()
     assert len(r.output_ids) == 2
+
+
+def test_multi_prefill_admission():
+    """max_prefills_per_step > 1: a burst admits several prompts in one
+    step (HTTP serving path) with identical outputs to serial admission."""
+    m = build_model("tiny-llama", dtype=torch.float32)
+    fast = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=5)
+    fast.max_prefills_per_step = 4
+    slow = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=5)
+    prompts = [[1 + i, 9, 2] for i in range(6)]
+    fr = [fast.submit(list(p), max_new_tokens=4) for p in prompts]
+    sr = [slow.submit(list(p), max_new_tokens=4) for p in prompts]
+    fast.step()
+    assert len(fast.running) == 4      # burst admitted in one step
+    while fast.has_work():
+        fast.step()
+    while slow.has_work():
+        slow.step()
+    for a, b in zip(fr, sr):
+        assert a.output_ids == b.output_ids, (a.request_id, a.output_ids,
+                                              b.output_ids)
