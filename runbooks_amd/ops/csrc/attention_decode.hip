@@ -113,47 +113,84 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 
   const int32_t *bt = block_tables + (int64_t)b * max_blocks;
 
-  // Each wave covers tokens t_begin + i*NW*TPW + wid*TPW + grp.
-  for (int t = t_begin + wid * TPW + grp; t < t_end; t += NW * TPW) {
-    const bool valid = true;
-    const int tt = t;
-    const int blk = bt[tt / bs];
+  // Stage this WG's slice of the block table through LDS: all four
+  // waves walk the same row, and a per-iteration global bt[] read
+  // sits on the K/V address chain (its vmcnt gates the row loads).
+  // 512 entries cover 8K tokens at bs=16; longer ranges fall back to
+  // global reads past the staged prefix.
+  constexpr int BT_LDS = 512;
+  __shared__ int32_t bt_lds[BT_LDS];
+  const int blk0 = t_begin / bs;
+  if (t_begin < t_end) {
+    const int nbt = min((t_end - 1) / bs - blk0 + 1, BT_LDS);
+    for (int i = (int)threadIdx.x; i < nbt; i += BLOCK)
+      bt_lds[i] = bt[blk0 + i];
+  }
+  __syncthreads();
 
-    float kf[VE], vf[VE];
+  // Software-pipelined KV stream: iteration i+1's K/V rows are issued
+  // as RAW bytes while i is computed. (Converting at load time puts an
+  // s_waitcnt vmcnt(0) between every row and its dot product — the
+  // unpipelined loop measured ~3.9 TB/s effective in-situ vs ~6 TB/s
+  // streaming.) Tail iterations clamp to the last row; the duplicate
+  // rows are masked out of the softmax with s = -inf, so the trip
+  // count is wave-uniform and the prefetch is branch-free inside.
+  constexpr int KBPL = FP8 ? VE : VE * 2;  // K/V bytes per lane
+  constexpr int RW = KBPL / 4;             // u32 words of that
+  const int stride = NW * TPW;
+  const int tw0 = t_begin + wid * TPW;     // wave's first token
+  const int t_base = tw0 + grp;            // this lane group's first
+  const int niter = (t_end > tw0) ? (t_end - tw0 + stride - 1) / stride : 0;
+
+  uint32_t kraw[2][RW], vraw[2][RW];
+  float kscale[2], vscale[2];
+
+  auto fetch = [&](int i, int slot) {
+    const int tt = min(t_base + i * stride, t_end - 1);
+    const int bi = tt / bs - blk0;
+    const int blk = (bi < BT_LDS) ? bt_lds[bi] : bt[tt / bs];
     if (FP8) {
       constexpr int RB8 = DH + 16;
       const uint8_t *k8 = reinterpret_cast<const uint8_t *>(k_cache);
       const uint8_t *v8 = reinterpret_cast<const uint8_t *>(v_cache);
       const int64_t row = (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * RB8;
-#pragma unroll
-      for (int e16 = 0; e16 < VE; e16 += 16) {
-        if (VE - e16 >= 16) {
-          rb::fp8x16_to_f32(k8 + row + d0 + e16, kf + e16);
-          rb::fp8x16_to_f32(v8 + row + d0 + e16, vf + e16);
-        } else if (VE - e16 >= 8) {
-          rb::fp8x8_to_f32(k8 + row + d0 + e16, kf + e16);
-          rb::fp8x8_to_f32(v8 + row + d0 + e16, vf + e16);
-        } else {
-          rb::fp8x4_to_f32(k8 + row + d0 + e16, kf + e16);
-          rb::fp8x4_to_f32(v8 + row + d0 + e16, vf + e16);
-        }
-      }
-      const float ks = *reinterpret_cast<const float *>(k8 + row + DH);
-      const float vs = *reinterpret_cast<const float *>(v8 + row + DH);
-#pragma unroll
-      for (int e = 0; e < VE; ++e) { kf[e] *= ks; vf[e] *= vs; }
+      rb::ld_words<RW>(kraw[slot], k8 + row + d0);
+      rb::ld_words<RW>(vraw[slot], v8 + row + d0);
+      kscale[slot] = *reinterpret_cast<const float *>(k8 + row + DH);
+      vscale[slot] = *reinterpret_cast<const float *>(v8 + row + DH);
     } else {
       const int64_t base =
           (((int64_t)blk * hkv + h_kv) * bs + (tt % bs)) * DH + d0;
-      if (VE == 8) {
-        rb::VIO<uint16_t>::load(k_cache + base, kf);
-        rb::VIO<uint16_t>::load(v_cache + base, vf);
-      } else {
+      rb::ld_words<RW>(kraw[slot],
+                       reinterpret_cast<const uint8_t *>(k_cache + base));
+      rb::ld_words<RW>(vraw[slot],
+                       reinterpret_cast<const uint8_t *>(v_cache + base));
+    }
+  };
+
+  // cur/nxt are compile-time constants at every call site below, so
+  // the double buffer stays in registers (no dynamic indexing).
+  auto step = [&](int i, int cur, int nxt) {
+    if (i + 1 < niter) fetch(i + 1, nxt);
+    const bool valid = t_base + i * stride < t_end;
+
+    float kf[VE], vf[VE];
+    if (FP8) {
 #pragma unroll
-        for (int e = 0; e < VE; ++e) {
-          kf[e] = rb::bf16_to_f32(k_cache[base + e]);
-          vf[e] = rb::bf16_to_f32(v_cache[base + e]);
-        }
+      for (int w = 0; w < RW; ++w) {
+        rb::fp8w_to_f32(kraw[cur][w], kf + w * 4);
+        rb::fp8w_to_f32(vraw[cur][w], vf + w * 4);
+      }
+      const float ks = kscale[cur], vs = vscale[cur];
+#pragma unroll
+      for (int e = 0; e < VE; ++e) { kf[e] *= ks; vf[e] *= vs; }
+    } else {
+      const uint16_t *kr = reinterpret_cast<const uint16_t *>(kraw[cur]);
+      const uint16_t *vr = reinterpret_cast<const uint16_t *>(vraw[cur]);
+#pragma unroll
+      for (int e = 0; e < VE; ++e) {
+        kf[e] = rb::bf16_to_f32(kr[e]);
+        vf[e] = rb::bf16_to_f32(vr[e]);
       }
     }
 
@@ -177,6 +214,12 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
         m[g] = mn;
       }
     }
+  };
+
+  if (niter > 0) fetch(0, 0);
+  for (int i = 0; i < niter; i += 2) {
+    step(i, 0, 1);
+    if (i + 1 < niter) step(i + 1, 1, 0);
   }
 
   // Merge the TPW groups inside each wave: lanes l^GL ... l^32 hold the

@@ -64,6 +64,35 @@ RB_DEV void fp8x4_to_f32(const uint8_t *p, float *f) {
   f[0] = a[0]; f[1] = a[1]; f[2] = b[0]; f[3] = b[1];
 }
 
+// one u32 of packed e4m3 (already in a register) -> 4 f32; lets a
+// pipelined consumer keep raw bytes in registers and convert at use
+RB_DEV void fp8w_to_f32(uint32_t w, float *f) {
+  auto a = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  auto b = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  f[0] = a[0]; f[1] = a[1]; f[2] = b[0]; f[3] = b[1];
+}
+
+// raw vector load of RW u32 words (8/16/32 B) — no conversion, so the
+// compiler can keep the s_waitcnt away from the issue point
+template <int RW>
+RB_DEV void ld_words(uint32_t *dst, const uint8_t *src) {
+  typedef __attribute__((ext_vector_type(2))) unsigned int u32x2i;
+  typedef __attribute__((ext_vector_type(4))) unsigned int u32x4i;
+  if constexpr (RW == 1) {
+    *dst = *reinterpret_cast<const uint32_t *>(src);
+  } else if constexpr (RW == 2) {
+    *reinterpret_cast<u32x2i *>(dst) = *reinterpret_cast<const u32x2i *>(src);
+  } else if constexpr (RW == 4) {
+    *reinterpret_cast<u32x4i *>(dst) = *reinterpret_cast<const u32x4i *>(src);
+  } else {
+    static_assert(RW == 8, "unsupported row width");
+    reinterpret_cast<u32x4i *>(dst)[0] =
+        reinterpret_cast<const u32x4i *>(src)[0];
+    reinterpret_cast<u32x4i *>(dst)[1] =
+        reinterpret_cast<const u32x4i *>(src)[1];
+  }
+}
+
 // 16 packed e4m3 (one 16 B vector) -> 16 f32
 RB_DEV void fp8x16_to_f32(const uint8_t *p, float *f) {
   typedef __attribute__((ext_vector_type(4))) unsigned int u32x4i;
