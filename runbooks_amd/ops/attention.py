@@ -141,6 +141,19 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
     return out
 
 
+def _auto_nsplit(B, hkv, seq_lens):
+    """Work-split: the GQA kernels are VGPR-capped at 2 waves/SIMD, so
+    ~512 WGs (2 per CU) fills the machine; splitting further only adds
+    combine traffic (measured sweep: profiles/decode_attn_pipeline.md)."""
+    base = max(1, B * hkv)
+    if base >= 512:
+        return 1
+    max_len = int(seq_lens.max())
+    if max_len <= 256:
+        return 1
+    return min(16, max(1, 512 // base), (max_len + 63) // 64)
+
+
 def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
                  scale: float | None = None, nsplit: int | None = None,
                  seq_starts=None):
@@ -150,15 +163,7 @@ def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
         if nsplit is None:
-            # Fill 256 CUs: B * Hkv workgroups per split.
-            B, hkv = q.shape[0], k_cache.shape[1]
-            base = B * hkv
-            max_len = int(seq_lens.max())
-            nsplit = 1
-            if base < 256 and max_len > 256:
-                nsplit = min(16, max(1, (2 * 256) // max(base, 1)))
-                nsplit = min(nsplit, (max_len + 255) // 256)
-                nsplit = max(nsplit, 1)
+            nsplit = _auto_nsplit(q.shape[0], k_cache.shape[1], seq_lens)
         return _backend.ext().paged_decode(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
             int(nsplit), float(scale), seq_starts=seq_starts,
@@ -177,14 +182,7 @@ def paged_decode_with_operand(q, k_cache, v_cache, block_tables, seq_lens,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
         if nsplit is None:
-            B, hkv = q.shape[0], k_cache.shape[1]
-            base = B * hkv
-            max_len = int(seq_lens.max())
-            nsplit = 1
-            if base < 256 and max_len > 256:
-                nsplit = min(16, max(1, (2 * 256) // max(base, 1)))
-                nsplit = min(nsplit, (max_len + 255) // 256)
-                nsplit = max(nsplit, 1)
+            nsplit = _auto_nsplit(q.shape[0], k_cache.shape[1], seq_lens)
         res = _backend.ext().paged_decode_swz(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
             int(nsplit), float(scale), seq_starts=seq_starts)

@@ -137,13 +137,19 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   // count is wave-uniform and the prefetch is branch-free inside.
   constexpr int KBPL = FP8 ? VE : VE * 2;  // K/V bytes per lane
   constexpr int RW = KBPL / 4;             // u32 words of that
+  // Ring depth: the G>=4 instantiations are VGPR-capped at 2 waves/SIMD
+  // (qreg+acc alone are 2*G*VE floats), so two waves must hide the full
+  // HBM latency — a 4-deep ring keeps 3 iterations of K/V in flight for
+  // only +4*RW VGPRs. G<=2 runs 5-8 waves/SIMD where depth 2 suffices
+  // (deeper costs occupancy: 62->78 VGPRs drops the MHA kernel 8->6).
+  constexpr int DEPTH = (G >= 4) ? 4 : 2;
   const int stride = NW * TPW;
   const int tw0 = t_begin + wid * TPW;     // wave's first token
   const int t_base = tw0 + grp;            // this lane group's first
   const int niter = (t_end > tw0) ? (t_end - tw0 + stride - 1) / stride : 0;
 
-  uint32_t kraw[2][RW], vraw[2][RW];
-  float kscale[2], vscale[2];
+  uint32_t kraw[DEPTH][RW], vraw[DEPTH][RW];
+  float kscale[DEPTH], vscale[DEPTH];
 
   auto fetch = [&](int i, int slot) {
     const int tt = min(t_base + i * stride, t_end - 1);
@@ -168,10 +174,10 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
   };
 
-  // cur/nxt are compile-time constants at every call site below, so
-  // the double buffer stays in registers (no dynamic indexing).
-  auto step = [&](int i, int cur, int nxt) {
-    if (i + 1 < niter) fetch(i + 1, nxt);
+  // cur is a compile-time constant at every call site below, so the
+  // ring buffer stays in registers (no dynamic indexing).
+  auto step = [&](int i, int cur) {
+    if (i + DEPTH - 1 < niter) fetch(i + DEPTH - 1, (cur + DEPTH - 1) % DEPTH);
     const bool valid = t_base + i * stride < t_end;
 
     float kf[VE], vf[VE];
@@ -216,10 +222,14 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
   };
 
-  if (niter > 0) fetch(0, 0);
-  for (int i = 0; i < niter; i += 2) {
-    step(i, 0, 1);
-    if (i + 1 < niter) step(i + 1, 1, 0);
+#pragma unroll
+  for (int d = 0; d < DEPTH - 1; ++d)
+    if (d < niter) fetch(d, d);
+  for (int i = 0; i < niter; i += DEPTH) {
+    step(i, 0);
+    if (i + 1 < niter) step(i + 1, 1 % DEPTH);
+    if (DEPTH > 2 && i + 2 < niter) step(i + 2, 2 % DEPTH);
+    if (DEPTH > 2 && i + 3 < niter) step(i + 3, 3 % DEPTH);
   }
 
   // Merge the TPW groups inside each wave: lanes l^GL ... l^32 hold the
