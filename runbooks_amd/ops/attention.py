@@ -142,16 +142,15 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
 
 
 def _auto_nsplit(B, hkv, seq_lens):
-    """Work-split: the GQA kernels are VGPR-capped at 2 waves/SIMD, so
-    ~512 WGs (2 per CU) fills the machine; splitting further only adds
-    combine traffic (measured sweep: profiles/decode_attn_pipeline.md)."""
+    """Work-split: the head-split GQA kernels run 4-5 waves/SIMD, so
+    ~1024 WGs (4 per CU) fills the machine; each split chunk should
+    still cover >= ~64 tokens or the combine outweighs it (measured
+    sweep: profiles/decode_attn_pipeline.md)."""
     base = max(1, B * hkv)
-    if base >= 512:
+    if base >= 1024:
         return 1
     max_len = int(seq_lens.max())
-    if max_len <= 256:
-        return 1
-    return min(16, max(1, 512 // base), (max_len + 63) // 64)
+    return max(1, min(16, 1024 // base, (max_len + 63) // 64))
 
 
 def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,

@@ -53,6 +53,15 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   constexpr int GL = (FP8 && G <= 4) ? 8 : 16;
   constexpr int VE = DH / GL;              // elems per lane
   constexpr int TPW = 64 / GL;             // tokens per wave pass
+  // Head-split mode (G >= NW): each wave owns GW = G/NW query heads
+  // over ALL tokens — the waves re-read the same K/V rows (L1/L2-
+  // served; HBM bytes unchanged) — instead of all G heads over a
+  // quarter of the tokens. Cuts per-wave state (qreg+acc = 2*G*VE
+  // floats) by NW, unlocking occupancy 2 -> 4-5 on the GQA/MQA
+  // kernels, shortens the serial dot->reduce->exp chain per
+  // iteration, and removes the cross-wave LDS merge entirely.
+  constexpr bool HS = (G >= NW);
+  constexpr int GW = HS ? G / NW : G;
   const int b = blockIdx.x;
   const int h_kv = blockIdx.y;
   const int split = SPLIT ? blockIdx.z : 0;
@@ -80,11 +89,12 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   const int gl = lane % GL;                // lane within group
   const int d0 = gl * VE;                  // this lane's Dh slice
 
-  // Q for all G heads, pre-scaled (softmax scale folded into q).
-  float qreg[G][VE];
+  // Q for this wave's GW heads, pre-scaled (softmax scale folded in).
+  const int hq_base = hq0 + (HS ? wid * GW : 0);
+  float qreg[GW][VE];
 #pragma unroll
-  for (int g = 0; g < G; ++g) {
-    const uint16_t *qp = q + ((int64_t)b * Hq + hq0 + g) * DH + d0;
+  for (int g = 0; g < GW; ++g) {
+    const uint16_t *qp = q + ((int64_t)b * Hq + hq_base + g) * DH + d0;
 #pragma unroll
     for (int e = 0; e < VE; ++e) qreg[g][e] = 0.f;
     {
@@ -103,9 +113,9 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
   }
 
-  float m[G], l[G], acc[G][VE];
+  float m[GW], l[GW], acc[GW][VE];
 #pragma unroll
-  for (int g = 0; g < G; ++g) {
+  for (int g = 0; g < GW; ++g) {
     m[g] = -INFINITY; l[g] = 0.f;
 #pragma unroll
     for (int e = 0; e < VE; ++e) acc[g][e] = 0.f;
@@ -143,8 +153,8 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   // only +4*RW VGPRs. G<=2 runs 5-8 waves/SIMD where depth 2 suffices
   // (deeper costs occupancy: 62->78 VGPRs drops the MHA kernel 8->6).
   constexpr int DEPTH = (G >= 4) ? 4 : 2;
-  const int stride = NW * TPW;
-  const int tw0 = t_begin + wid * TPW;     // wave's first token
+  const int stride = HS ? TPW : NW * TPW;
+  const int tw0 = t_begin + (HS ? 0 : wid * TPW);  // wave's first token
   const int t_base = tw0 + grp;            // this lane group's first
   const int niter = (t_end > tw0) ? (t_end - tw0 + stride - 1) / stride : 0;
 
@@ -201,7 +211,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
 
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < GW; ++g) {
       float s = 0.f;
 #pragma unroll
       for (int e = 0; e < VE; ++e) s += qreg[g][e] * kf[e];
@@ -237,7 +247,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 #pragma unroll
   for (int off = GL; off <= 32; off <<= 1) {
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < GW; ++g) {
       const float mo = __shfl_xor(m[g], off, 64);
       const float lo = __shfl_xor(l[g], off, 64);
       const float mn = fmaxf(m[g], mo);
@@ -254,12 +264,48 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
   }
 
+  if constexpr (HS) {
+    // Head-split: each wave owns its GW heads outright — write out
+    // (or the split partial) straight from registers, no LDS merge.
+    if (grp == 0) {
+#pragma unroll
+      for (int g = 0; g < GW; ++g) {
+        const int h = hq_base + g;
+        if (SPLIT) {
+          float *pp =
+              partial + (((int64_t)b * Hq + h) * nsplit + split) * (DH + 2);
+#pragma unroll
+          for (int e = 0; e < VE; ++e) pp[d0 + e] = acc[g][e];
+          if (gl == 0) { pp[DH] = m[g]; pp[DH + 1] = l[g]; }
+        } else {
+          const float inv_l = (l[g] > 0.f) ? 1.0f / l[g] : 0.f;
+          uint16_t *op = out + ((int64_t)b * Hq + h) * DH + d0;
+#pragma unroll
+          for (int e = 0; e < VE; ++e)
+            op[e] = rb::f32_to_bf16(acc[g][e] * inv_l);
+          if (VE >= 8 && out_swz != nullptr) {
+#pragma unroll
+            for (int e8 = 0; e8 < VE; e8 += 8) {
+              const int kk = h * DH + d0 + e8;
+              uint16_t *sp = out_swz + (kk >> 4) * 512 +
+                             ((kk >> 3) & 1) * 256 + b * 8;
+#pragma unroll
+              for (int e = 0; e < 8 && e < VE; ++e)
+                sp[e] = rb::f32_to_bf16(acc[g][e8 + e] * inv_l);
+            }
+          }
+        }
+      }
+    }
+    return;
+  }
+
   // Merge across the 4 waves via LDS. Layout per (wave, head):
   // [DH acc][m][l] floats.
-  __shared__ float lds[NW][G][DH + 2];
+  __shared__ float lds[HS ? 1 : NW][HS ? 1 : G][DH + 2];
   if (grp == 0) {   // lanes 0..GL-1 of each wave hold the wave's merged state
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < GW; ++g) {
 #pragma unroll
       for (int e = 0; e < VE; ++e) lds[wid][g][d0 + e] = acc[g][e];
       if (gl == 0) { lds[wid][g][DH] = m[g]; lds[wid][g][DH + 1] = l[g]; }

@@ -21,14 +21,15 @@ def fixed_nsplit(batch: int, hkv: int) -> int:
     """Work-split for the paged-decode kernel, chosen per bucket at capture
     time (the eager heuristic in ops/attention.py reads seq_lens.max() —
     a host sync, impossible inside a graph). base = batch*hkv workgroups
-    per split; split until ~2x256 CUs are covered — the GQA kernels are
-    VGPR-capped at 2 waves/SIMD, so 2 WGs/CU (512 WGs) is what fills the
-    machine (measured: llama2-70b B=32 len-2048 1.65 vs 0.93 TB/s,
-    falcon MQA 0.54 vs 0.28 — profiles/decode_attn_pipeline.md)."""
+    per split; split until ~4x256 CUs are covered — the head-split GQA
+    kernels run 4-5 waves/SIMD, so ~4 WGs/CU (1024 WGs) fills the
+    machine (measured sweep: llama2-70b B=32 len-2048 1.75 vs 0.67
+    TB/s, falcon MQA 0.72 vs 0.23, B=8 0.99-1.24 vs 0.16 —
+    profiles/decode_attn_pipeline.md)."""
     base = max(1, batch * hkv)
-    if base >= 512:
+    if base >= 1024:
         return 1
-    return min(16, max(1, 512 // base))
+    return min(16, max(1, 1024 // base))
 
 
 class GraphedDecoder:
