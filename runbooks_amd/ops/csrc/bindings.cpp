@@ -52,7 +52,7 @@ at::Tensor mfma_probe_32x32x16(at::Tensor a, at::Tensor b);
 at::Tensor mfma_probe_16x16x32(at::Tensor a, at::Tensor b);
 at::Tensor train_gemm_nt(at::Tensor a, at::Tensor b);
 at::Tensor skinny_gemm(at::Tensor x, at::Tensor w);
-at::Tensor decode_gemm(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N, int64_t K);
+at::Tensor decode_gemm(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N, int64_t K, int64_t force_split);
 at::Tensor decode_swizzle_w(at::Tensor w);
 at::Tensor decode_swizzle_x(at::Tensor x);
 int64_t decode_gemm_split(int64_t N, int64_t K);
@@ -118,7 +118,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm_mmax", &skinny_gemm_mmax);
   m.def("decode_gemm", &decode_gemm,
         "v2 weight-stream decode GEMM over pre-swizzled operands: "
-        "y[M<=32,N] = x @ W^T (bf16), 1KiB coalesced nt weight bursts");
+        "y[M<=32,N] = x @ W^T (bf16), 1KiB coalesced nt weight bursts",
+        py::arg("xs"), py::arg("ws"), py::arg("M"), py::arg("N"),
+        py::arg("K"), py::arg("force_split") = -1);
   m.def("decode_swizzle_w", &decode_swizzle_w,
         "one-time [N,K] -> fragment-lane-major weight layout");
   m.def("decode_swizzle_x", &decode_swizzle_x,

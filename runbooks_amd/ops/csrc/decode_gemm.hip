@@ -321,7 +321,7 @@ at::Tensor decode_gemm_raw(at::Tensor xs, at::Tensor ws, int64_t M,
 // xs from decode_swizzle_x, ws from decode_swizzle_w; M/N/K of the
 // ORIGINAL y[M,N] = x[M,K] @ W[N,K]^T problem.
 at::Tensor decode_gemm(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N,
-                       int64_t K) {
+                       int64_t K, int64_t force_split) {
   TORCH_CHECK(xs.is_cuda() && ws.is_cuda() && xs.is_contiguous() &&
               ws.is_contiguous(), "decode_gemm: contiguous GPU tensors");
   TORCH_CHECK(xs.scalar_type() == at::kBFloat16 &&
@@ -333,7 +333,8 @@ at::Tensor decode_gemm(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N,
 
   auto stream = at::cuda::getCurrentHIPStream();
   auto y = at::empty({M, N}, xs.options());
-  const int split = (int)decode_gemm_split(N, K);
+  const int split = force_split > 0 ? (int)force_split
+                                    : (int)decode_gemm_split(N, K);
   if (split == 1) {
     hipLaunchKernelGGL((decode_gemm_kernel<true>), dim3(N / 32, 1),
                        dim3(BLOCK), 0, stream,

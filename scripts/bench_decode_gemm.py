@@ -67,6 +67,18 @@ def main():
         t_v2_full = bench(lambda i: ops.ext().decode_gemm(
             ops.ext().decode_swizzle_x(x), swz[i % n_w], args.m, N, K),
             args.iters)
+        if os.environ.get("RB_BENCH_SPLIT_SWEEP") == "1":
+            for fs in (1, 2, 4):
+                if (K // 16) % (fs * 4) or K // (fs * 2) < 1024:
+                    continue
+                t_fs = bench(lambda i: ops.ext().decode_gemm(
+                    xs, swz[i % n_w], args.m, N, K, force_split=fs),
+                    args.iters)
+                print(json.dumps({
+                    "N": N, "K": K, "force_split": fs,
+                    "ms": round(t_fs * 1e3, 4),
+                    "tbps": round(N * K * 2 / 1e9 / t_fs / 1e3, 3)}),
+                    flush=True)
         del swz
         gb = N * K * 2 / 1e9
         row = {
