@@ -392,6 +392,265 @@ __global__ void decode_combine_kernel(const float *__restrict__ partial,
   }
 }
 
+// ---------------------------------------------------------------------------
+// MFMA decode for GQA/MQA (G >= 4), bf16 cache, DH in {64, 128}, bs == 16.
+//
+// The scalar formulation above spends ~30 VALU ops per (token, head) on
+// dot -> shfl-reduce -> exp -> acc; at G >= 4 that serial chain is the
+// bind (PMC: 57% WAIT_ANY / 28% VALU at 2 waves/SIMD pre-head-split,
+// and the head-split variant re-reads K/V NW times through L2). Here
+// the score block S[16 tok][G] is ONE mfma_f32_16x16x32_bf16 chain over
+// DH, and O^T accumulates via mfma_f32_32x32x16_bf16 — each K/V row is
+// read once, and per-block VALU drops ~60x.
+//
+//  * wave owns whole 16-token cache blocks (bi = blk_lo + wid, step NW).
+//  * K is read DIRECTLY in A-fragment order from the unmodified cache
+//    layout: lane l reads K[token l&15][dh (l>>4)*8 + 32*ks] — 16 B per
+//    lane from 16 adjacent 256 B rows inside one contiguous 4 KB block
+//    (64 B segments, L2-friendly), no staging.
+//  * V needs the transpose -> tiny per-wave LDS image [DH][16+8pad]
+//    written b16-scattered, read back as b128 A-fragments. Wave-private
+//    region: no barrier, just lgkmcnt ordering.
+//  * fragment maps (guide + mfma_probe_16x16x32 GPU test):
+//      16x16x32: A[i=l&15][k=(l>>4)*8+e]; B[k][j=l&15]; C[j=l&15][i=(l>>4)*4+r]
+//      32x32x16: A[i=l&31][k=(l>>5)*8+e]; C[j=l&31][i=(r&3)+8*(r>>2)+4*(l>>5)]
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) __bf16 rb_bf16x8v;
+typedef __attribute__((ext_vector_type(16))) float rb_f32x16v;
+typedef __attribute__((ext_vector_type(4))) float rb_f32x4v;
+
+RB_DEV unsigned rb_pack_bf16(float lo, float hi) {
+  union { __bf16 b; unsigned short u; } a, b;
+  a.b = (__bf16)lo;
+  b.b = (__bf16)hi;
+  return ((unsigned)b.u << 16) | a.u;
+}
+
+template <int DH, int G, bool SPLIT>
+__global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
+    const uint16_t *__restrict__ q, const uint16_t *__restrict__ k_cache,
+    const uint16_t *__restrict__ v_cache,
+    const int32_t *__restrict__ block_tables,
+    const int32_t *__restrict__ seq_lens,
+    const int32_t *__restrict__ seq_starts, uint16_t *__restrict__ out,
+    uint16_t *__restrict__ out_swz, float *__restrict__ partial, int hkv,
+    int max_blocks, int nsplit, float scale) {
+  constexpr int BS = 16;                  // cache block = one S tile
+  constexpr int KS32 = DH / 32;           // QK^T contraction steps
+  constexpr int DT = DH / 32;             // O^T 32-row d-tiles
+  constexpr int VROW = BS + 8;            // V^T LDS row: 16 tok + 8 pad (48 B)
+
+  const int b = blockIdx.x;
+  const int h_kv = blockIdx.y;
+  const int split = SPLIT ? blockIdx.z : 0;
+  const int hq0 = h_kv * G;
+  const int Hq = hkv * G;
+
+  const int seq_len = seq_lens[b];
+  const int start = (seq_starts != nullptr) ? seq_starts[b] : 0;
+  int t_begin = start, t_end = seq_len;
+  if (SPLIT) {
+    const int len = seq_len - start;
+    const int chunk = (len + nsplit - 1) / nsplit;
+    t_begin = start + split * chunk;
+    t_end = min(seq_len, t_begin + chunk);
+  }
+
+  const int tid = (int)threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int i16 = lane & 15;
+  const int h16 = lane >> 4;              // 0..3
+  const int h32 = lane >> 5;              // 0..1
+  const int c32 = lane & 31;
+
+  const int32_t *bt = block_tables + (int64_t)b * max_blocks;
+
+  // block-table staging (shared by all waves; same scheme as scalar)
+  constexpr int BT_LDS = 512;
+  __shared__ int32_t bt_lds[BT_LDS];
+  const int blk0 = t_begin / BS;
+  if (t_begin < t_end) {
+    const int nbt = min((t_end - 1) / BS - blk0 + 1, BT_LDS);
+    for (int i = tid; i < nbt; i += BLOCK) bt_lds[i] = bt[blk0 + i];
+  }
+  // per-wave V^T image + cross-wave merge region
+  __shared__ __attribute__((aligned(16))) uint16_t vlds[NW][DH][VROW];
+  __shared__ float mlds[NW][G][DH + 2];
+  __syncthreads();
+
+  // Q fragments (B of the swapped QK^T), pre-scaled; head = i16.
+  rb_bf16x8v qf[KS32];
+  {
+    const bool hv = i16 < G;
+    const uint16_t *qp =
+        q + ((int64_t)b * Hq + hq0 + (hv ? i16 : 0)) * DH;
+#pragma unroll
+    for (int ks = 0; ks < KS32; ++ks) {
+      float f[8];
+      rb::VIO<uint16_t>::load(qp + ks * 32 + h16 * 8, f);
+      union { unsigned short u[8]; rb_bf16x8v v; } c;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        c.u[e] = rb::f32_to_bf16(hv ? f[e] * scale : 0.0f);
+      qf[ks] = c.v;
+    }
+  }
+
+  rb_f32x16v acc_o[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (rb_f32x16v)(0.0f);
+  float m_run = -INFINITY, l_run = 0.0f;
+
+  const int blk_lo = t_begin / BS;
+  const int blk_hi = (t_end > t_begin) ? (t_end - 1) / BS : blk_lo - 1;
+
+  for (int bi = blk_lo + wid; bi <= blk_hi; bi += NW) {
+    const int bi_l = bi - blk0;
+    const int blk = (bi_l < BT_LDS) ? bt_lds[bi_l] : bt[bi];
+    const uint16_t *kb = k_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
+    const uint16_t *vb = v_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
+
+    // ---- stage V transposed into this wave's LDS image ----------------
+    // lane reads token (lane*8+p*512)/DH's contiguous d-slice, writes it
+    // down a column of vlds (b16 scatter; wave-private so lgkm-ordered).
+#pragma unroll
+    for (int p = 0; p < BS * DH / (64 * 8); ++p) {
+      const int idx = p * 512 + lane * 8;
+      const int tok = idx / DH;
+      const int d0 = idx % DH;
+      rb::bf16x8 vv = *reinterpret_cast<const rb::bf16x8 *>(vb + idx);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vlds[wid][d0 + e][tok] = vv.v[e];
+    }
+
+    // ---- S[16 tok][16 heads] = K_blk @ Q^T ----------------------------
+    rb_f32x4v s4 = (rb_f32x4v)(0.0f);
+#pragma unroll
+    for (int ks = 0; ks < KS32; ++ks) {
+      const rb_bf16x8v kf = *reinterpret_cast<const rb_bf16x8v *>(
+          kb + i16 * DH + ks * 32 + h16 * 8);
+      s4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[ks], s4, 0, 0, 0);
+    }
+
+    // ---- mask + online softmax (state per head = per col = per lane) --
+    float p4[4];
+    float mx = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int t = bi * BS + h16 * 4 + r;
+      p4[r] = (t >= t_begin && t < t_end) ? s4[r] : -INFINITY;
+      mx = fmaxf(mx, p4[r]);
+    }
+    mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
+    mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+    const float mn = fmaxf(m_run, mx);
+    if (mn != -INFINITY) {
+      const float alpha = (m_run == -INFINITY) ? 0.0f : __expf(m_run - mn);
+      float psum = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        p4[r] = (p4[r] == -INFINITY) ? 0.0f : __expf(p4[r] - mn);
+        psum += p4[r];
+      }
+      psum += __shfl_xor(psum, 16, 64);
+      psum += __shfl_xor(psum, 32, 64);
+      l_run = l_run * alpha + psum;
+      m_run = mn;
+      if (alpha != 1.0f) {
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) acc_o[dt][r] *= alpha;
+      }
+
+      // ---- P[tok][head] -> B fragment of PV (k = tok, col = head) ----
+      // source lane s = head + 16*(tok>>2) holds tokens 4*(s>>4)+r.
+      // target lane l needs tokens (l>>5)*8..+8 of head l&31 (<16).
+      const unsigned w01 = rb_pack_bf16(p4[0], p4[1]);
+      const unsigned w23 = rb_pack_bf16(p4[2], p4[3]);
+      const int sh = c32 & 15;            // source head (cols 16+ unused)
+      const int s1 = sh + 16 * (h32 * 2);
+      const int s2 = s1 + 16;
+      union { unsigned u[4]; rb_bf16x8v v; } pf;
+      pf.u[0] = __shfl(w01, s1, 64);
+      pf.u[1] = __shfl(w23, s1, 64);
+      pf.u[2] = __shfl(w01, s2, 64);
+      pf.u[3] = __shfl(w23, s2, 64);
+
+      // ---- O^T[d][head] += V^T_blk @ P --------------------------------
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const rb_bf16x8v vf = *reinterpret_cast<const rb_bf16x8v *>(
+            &vlds[wid][dt * 32 + c32][h32 * 8]);
+        acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            vf, pf.v, acc_o[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- cross-wave merge via LDS (same scheme as the scalar kernel) ----
+  // acc_o lane map: head = c32, d = dt*32 + (r&3)+8*(r>>2)+4*h32.
+  if (c32 < G) {
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        mlds[wid][c32][dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * h32] =
+            acc_o[dt][r];
+  }
+  if (lane < 16 && i16 < G) {
+    mlds[wid][i16][DH] = m_run;
+    mlds[wid][i16][DH + 1] = l_run;
+  }
+  __syncthreads();
+
+  constexpr int VE = DH / 16;             // elems per merge lane
+  const int gl = lane & 15;
+  const int d0 = gl * VE;
+  for (int g = wid; g < G; g += NW) {
+    if (lane >= 16) continue;
+    float mm = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < NW; ++w) mm = fmaxf(mm, mlds[w][g][DH]);
+    float ll = 0.f, av[VE];
+#pragma unroll
+    for (int e = 0; e < VE; ++e) av[e] = 0.f;
+    if (mm != -INFINITY) {
+#pragma unroll
+      for (int w = 0; w < NW; ++w) {
+        const float a = __expf(mlds[w][g][DH] - mm);
+        ll += mlds[w][g][DH + 1] * a;
+#pragma unroll
+        for (int e = 0; e < VE; ++e) av[e] += mlds[w][g][d0 + e] * a;
+      }
+    }
+    if (SPLIT) {
+      float *pp =
+          partial + (((int64_t)b * Hq + hq0 + g) * nsplit + split) * (DH + 2);
+#pragma unroll
+      for (int e = 0; e < VE; ++e) pp[d0 + e] = av[e];
+      if (gl == 0) { pp[DH] = mm; pp[DH + 1] = ll; }
+    } else {
+      const float inv_l = (ll > 0.f) ? 1.0f / ll : 0.f;
+      uint16_t *op = out + ((int64_t)b * Hq + hq0 + g) * DH + d0;
+#pragma unroll
+      for (int e = 0; e < VE; ++e) op[e] = rb::f32_to_bf16(av[e] * inv_l);
+      if (VE >= 8 && out_swz != nullptr) {
+#pragma unroll
+        for (int e8 = 0; e8 < VE; e8 += 8) {
+          const int kk = (hq0 + g) * DH + d0 + e8;
+          uint16_t *sp = out_swz + (kk >> 4) * 512 + ((kk >> 3) & 1) * 256 +
+                         b * 8;
+#pragma unroll
+          for (int e = 0; e < 8 && e < VE; ++e)
+            sp[e] = rb::f32_to_bf16(av[e8 + e] * inv_l);
+        }
+      }
+    }
+  }
+}
+
 template <int DH, int G, bool FP8>
 void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
                    const at::Tensor &v_cache, const at::Tensor &block_tables,
@@ -408,6 +667,45 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
   const int max_blocks = (int)block_tables.size(1);
   uint16_t *swz = out_swz.has_value() ? (uint16_t *)out_swz->data_ptr()
                                       : nullptr;
+  // MFMA route (bf16 cache, G >= 4, DH <= 128, 16-token blocks): the
+  // GQA shapes where the scalar per-token VALU chain is the bind.
+  // RB_DECODE_MFMA=0 falls back to the scalar kernel for A/B runs.
+  if constexpr (!FP8 && G >= 4 && DH <= 128) {
+    static const bool mfma_on = []() {
+      const char *e = getenv("RB_DECODE_MFMA");
+      return !(e && e[0] == '0');
+    }();
+    if (bs == 16 && mfma_on) {
+      if (nsplit <= 1) {
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, false>),
+                           dim3(B, hkv, 1), dim3(BLOCK), 0, stream,
+                           (const uint16_t *)q.data_ptr(),
+                           (const uint16_t *)k_cache.data_ptr(),
+                           (const uint16_t *)v_cache.data_ptr(),
+                           block_tables.data_ptr<int32_t>(),
+                           seq_lens.data_ptr<int32_t>(), starts,
+                           (uint16_t *)out.data_ptr(), swz, nullptr, hkv,
+                           max_blocks, 1, scale);
+      } else {
+        const int Hq = hkv * G;
+        auto partial = at::empty({B, Hq, nsplit, DH + 2},
+                                 q.options().dtype(at::kFloat));
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, true>),
+                           dim3(B, hkv, nsplit), dim3(BLOCK), 0, stream,
+                           (const uint16_t *)q.data_ptr(),
+                           (const uint16_t *)k_cache.data_ptr(),
+                           (const uint16_t *)v_cache.data_ptr(),
+                           block_tables.data_ptr<int32_t>(),
+                           seq_lens.data_ptr<int32_t>(), starts,
+                           nullptr, nullptr, partial.data_ptr<float>(), hkv,
+                           max_blocks, nsplit, scale);
+        hipLaunchKernelGGL(decode_combine_kernel, dim3(B * Hq), dim3(256), 0,
+                           stream, partial.data_ptr<float>(),
+                           (uint16_t *)out.data_ptr(), swz, Hq, nsplit, DH);
+      }
+      return;
+    }
+  }
   if (nsplit <= 1) {
     hipLaunchKernelGGL((paged_decode_kernel<DH, G, false, FP8>),
                        dim3(B, hkv, 1),

@@ -111,6 +111,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE: per-row loss + lse");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward -> dlogits");
+  m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32,
+        "layout probe: one 16x16x32 bf16 MFMA as a plain matmul");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "layout probe: one 32x32x16 bf16 MFMA as a plain matmul");
   m.def("skinny_gemm", &skinny_gemm,

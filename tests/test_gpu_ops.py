@@ -437,8 +437,6 @@ def test_skinny_gemm_fp8():
 _EXPERIMENTAL = __import__("os").environ.get("RB_EXPERIMENTAL") == "1"
 
 
-@pytest.mark.skipif(not _EXPERIMENTAL, reason="RB_EXPERIMENTAL=1 only "
-                    "(drafted offline; validate before enabling)")
 def test_mfma_16x16x32_layout_probe():
     _assert_hip()
     torch.manual_seed(0)
