@@ -141,16 +141,17 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
     return out
 
 
-def _auto_nsplit(B, hkv, seq_lens):
-    """Work-split: the head-split GQA kernels run 4-5 waves/SIMD, so
-    ~1024 WGs (4 per CU) fills the machine; each split chunk should
-    still cover >= ~64 tokens or the combine outweighs it (measured
-    sweep: profiles/decode_attn_pipeline.md)."""
+def _auto_nsplit(B, hkv, seq_lens, g=1, dh=128, fp8=False):
+    """Work-split (measured sweeps: profiles/decode_attn_pipeline.md).
+    The MFMA GQA path (g>=4, bf16, dh<=128) is block-granular and
+    latency-light: ~512 WGs is the knee. The scalar kernels want ~1024
+    (4 WGs/CU). Chunks should still cover >= ~64 tokens each."""
     base = max(1, B * hkv)
-    if base >= 1024:
+    target = 512 if (g >= 4 and not fp8 and dh <= 128) else 1024
+    if base >= target:
         return 1
     max_len = int(seq_lens.max())
-    return max(1, min(16, 1024 // base, (max_len + 63) // 64))
+    return max(1, min(16, target // base, (max_len + 63) // 64))
 
 
 def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
@@ -162,7 +163,10 @@ def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
         if nsplit is None:
-            nsplit = _auto_nsplit(q.shape[0], k_cache.shape[1], seq_lens)
+            nsplit = _auto_nsplit(
+                q.shape[0], k_cache.shape[1], seq_lens,
+                g=q.shape[1] // k_cache.shape[1], dh=q.shape[-1],
+                fp8=k_cache.dtype == torch.uint8)
         return _backend.ext().paged_decode(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
             int(nsplit), float(scale), seq_starts=seq_starts,
@@ -181,7 +185,10 @@ def paged_decode_with_operand(q, k_cache, v_cache, block_tables, seq_lens,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
         if nsplit is None:
-            nsplit = _auto_nsplit(q.shape[0], k_cache.shape[1], seq_lens)
+            nsplit = _auto_nsplit(
+                q.shape[0], k_cache.shape[1], seq_lens,
+                g=q.shape[1] // k_cache.shape[1], dh=q.shape[-1],
+                fp8=k_cache.dtype == torch.uint8)
         res = _backend.ext().paged_decode_swz(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
             int(nsplit), float(scale), seq_starts=seq_starts)

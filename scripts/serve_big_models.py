@@ -46,14 +46,15 @@ def truncated_parity(name: str) -> dict:
     return {"engine": out, "full_forward": ref, "prefix_match": match}
 
 
-def serve_run(name: str, batch: int, steps: int) -> dict:
+def serve_run(name: str, batch: int, steps: int,
+              prompt_len: int = 96) -> dict:
     t0 = time.time()
     eng = Engine(name, dtype=torch.bfloat16, max_batch=batch, seed=17)
     build_s = time.time() - t0
     free, total = torch.cuda.mem_get_info()
     g = torch.Generator().manual_seed(5)
     for _ in range(batch):
-        eng.submit(torch.randint(0, eng.cfg.vocab_size, (96,),
+        eng.submit(torch.randint(0, eng.cfg.vocab_size, (prompt_len,),
                                  generator=g).tolist(),
                    max_new_tokens=1 << 30)
     for _ in range(batch + 5):
@@ -68,6 +69,7 @@ def serve_run(name: str, batch: int, steps: int) -> dict:
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     return {"model": name, "batch": batch, "steps": steps,
+            "prompt_len": prompt_len,
             "build_s": round(build_s, 1),
             "hbm_used_gb": round((total - free) / 2**30, 1),
             "kv_blocks": eng.allocator.num_blocks,
@@ -83,13 +85,15 @@ def main():
     p.add_argument("--model", default=None)
     p.add_argument("--batch", type=int, default=32)
     p.add_argument("--steps", type=int, default=40)
+    p.add_argument("--prompt-len", type=int, default=96)
     args = p.parse_args()
     models = [args.model] if args.model else ["falcon-40b", "llama2-70b"]
     for name in models:
         print(json.dumps({"parity": {name: truncated_parity(name)}}),
               flush=True)
         torch.cuda.empty_cache()
-        print(json.dumps(serve_run(name, args.batch, args.steps)),
+        print(json.dumps(serve_run(name, args.batch, args.steps,
+                                   args.prompt_len)),
               flush=True)
         torch.cuda.empty_cache()
 
