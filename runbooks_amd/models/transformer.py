@@ -416,9 +416,20 @@ class Transformer(nn.Module):
         return self.blocks[0].attn.hkv
 
     def alloc_caches(self, num_blocks: int, device, fp8: bool = False):
+        # Transposed-V blocks feed the MFMA decode kernel's direct V^T
+        # fragment reads (GQA/MQA models, bf16, Dh<=128). The cache
+        # shape is the routing signal all the way down (kv_append,
+        # paged_decode). RB_DECODE_MFMA=0 reverts to the scalar path.
+        import os
+        attn = self.blocks[0].attn
+        vt = (not fp8 and self.dtype == torch.bfloat16
+              and attn.hq // attn.hkv >= 4 and self.cfg.head_dim <= 128
+              and ops.BLOCK_SIZE == 16
+              and os.environ.get("RB_DECODE_MFMA", "1") != "0")
         return [ops.alloc_kv_cache(num_blocks, self.local_kv_heads(),
                                    self.cfg.head_dim, device,
-                                   dtype=self.dtype, fp8=fp8)
+                                   dtype=self.dtype, fp8=fp8,
+                                   v_transposed=vt)
                 for _ in range(self.cfg.num_layers)]
 
 

@@ -126,13 +126,20 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
     B, Hq, Dh = q.shape
     _, Hkv, BS, _ = k_cache.shape
     g = Hq // Hkv
+    vt = (v_cache.shape[2] == Dh and v_cache.shape[3] == BS
+          and Dh != BS)  # transposed-V layout (alloc_kv_cache v_transposed)
     out = torch.empty_like(q)
     for b in range(B):
         n = int(seq_lens[b])
         st = int(seq_starts[b]) if seq_starts is not None else 0
         blocks = block_tables[b, : (n + BS - 1) // BS].long()
         k = k_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[st:n].float()
-        v = v_cache[blocks].transpose(1, 2).reshape(-1, Hkv, Dh)[st:n].float()
+        if vt:
+            v = v_cache[blocks].permute(0, 3, 1, 2).reshape(
+                -1, Hkv, Dh)[st:n].float()
+        else:
+            v = v_cache[blocks].transpose(1, 2).reshape(
+                -1, Hkv, Dh)[st:n].float()
         for h in range(Hq):
             hk = h // g
             s = (k[:, hk] @ q[b, h].float()) * scale
@@ -141,13 +148,21 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
     return out
 
 
-def _auto_nsplit(B, hkv, seq_lens, g=1, dh=128, fp8=False):
+def _is_vt(k_cache, v_cache) -> bool:
+    """Transposed-V cache layout (alloc_kv_cache v_transposed) == the
+    MFMA decode path; the shape is the routing signal."""
+    return (v_cache.dim() == 4 and v_cache.shape[2] == k_cache.shape[3]
+            and v_cache.shape[3] == k_cache.shape[2]
+            and k_cache.shape[3] != k_cache.shape[2])
+
+
+def _auto_nsplit(B, hkv, seq_lens, mfma=False):
     """Work-split (measured sweeps: profiles/decode_attn_pipeline.md).
-    The MFMA GQA path (g>=4, bf16, dh<=128) is block-granular and
-    latency-light: ~512 WGs is the knee. The scalar kernels want ~1024
-    (4 WGs/CU). Chunks should still cover >= ~64 tokens each."""
+    The MFMA path is block-granular and latency-light: ~512 WGs is the
+    knee. The scalar kernels want ~1024 (4 WGs/CU). Chunks should still
+    cover >= ~64 tokens each."""
     base = max(1, B * hkv)
-    target = 512 if (g >= 4 and not fp8 and dh <= 128) else 1024
+    target = 512 if mfma else 1024
     if base >= target:
         return 1
     max_len = int(seq_lens.max())
@@ -163,10 +178,8 @@ def paged_decode(q, k_cache, v_cache, block_tables, seq_lens,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
         if nsplit is None:
-            nsplit = _auto_nsplit(
-                q.shape[0], k_cache.shape[1], seq_lens,
-                g=q.shape[1] // k_cache.shape[1], dh=q.shape[-1],
-                fp8=k_cache.dtype == torch.uint8)
+            nsplit = _auto_nsplit(q.shape[0], k_cache.shape[1], seq_lens,
+                                  mfma=_is_vt(k_cache, v_cache))
         return _backend.ext().paged_decode(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
             int(nsplit), float(scale), seq_starts=seq_starts,
@@ -185,10 +198,8 @@ def paged_decode_with_operand(q, k_cache, v_cache, block_tables, seq_lens,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _backend.use_hip(q):
         if nsplit is None:
-            nsplit = _auto_nsplit(
-                q.shape[0], k_cache.shape[1], seq_lens,
-                g=q.shape[1] // k_cache.shape[1], dh=q.shape[-1],
-                fp8=k_cache.dtype == torch.uint8)
+            nsplit = _auto_nsplit(q.shape[0], k_cache.shape[1], seq_lens,
+                                  mfma=_is_vt(k_cache, v_cache))
         res = _backend.ext().paged_decode_swz(
             q.contiguous(), k_cache, v_cache, block_tables, seq_lens,
             int(nsplit), float(scale), seq_starts=seq_starts)

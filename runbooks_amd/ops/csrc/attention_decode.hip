@@ -404,13 +404,16 @@ __global__ void decode_combine_kernel(const float *__restrict__ partial,
 // read once, and per-block VALU drops ~60x.
 //
 //  * wave owns whole 16-token cache blocks (bi = blk_lo + wid, step NW).
-//  * K is read DIRECTLY in A-fragment order from the unmodified cache
+//  * K is read DIRECTLY in A-fragment order from the [BS, Dh] block
 //    layout: lane l reads K[token l&15][dh (l>>4)*8 + 32*ks] — 16 B per
 //    lane from 16 adjacent 256 B rows inside one contiguous 4 KB block
 //    (64 B segments, L2-friendly), no staging.
-//  * V needs the transpose -> tiny per-wave LDS image [DH][16+8pad]
-//    written b16-scattered, read back as b128 A-fragments. Wave-private
-//    region: no barrier, just lgkmcnt ordering.
+//  * V comes from the TRANSPOSED cache layout ([Dh, BS] per block,
+//    alloc_kv_cache v_transposed — the framework owns the layout, so
+//    the PV A-fragment is a direct 16 B/lane read too. The first cut
+//    staged the transpose through per-wave LDS; PMC showed 26-43% of
+//    wave cycles burned in LDS bank conflicts (the b16 scatter's 8-row
+//    lane stride lands on 2 of 64 banks at any 16-aligned row pitch).
 //  * fragment maps (guide + mfma_probe_16x16x32 GPU test):
 //      16x16x32: A[i=l&15][k=(l>>4)*8+e]; B[k][j=l&15]; C[j=l&15][i=(l>>4)*4+r]
 //      32x32x16: A[i=l&31][k=(l>>5)*8+e]; C[j=l&31][i=(r&3)+8*(r>>2)+4*(l>>5)]
@@ -438,7 +441,6 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
   constexpr int BS = 16;                  // cache block = one S tile
   constexpr int KS32 = DH / 32;           // QK^T contraction steps
   constexpr int DT = DH / 32;             // O^T 32-row d-tiles
-  constexpr int VROW = BS + 8;            // V^T LDS row: 16 tok + 8 pad (48 B)
 
   const int b = blockIdx.x;
   const int h_kv = blockIdx.y;
@@ -474,8 +476,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
     const int nbt = min((t_end - 1) / BS - blk0 + 1, BT_LDS);
     for (int i = tid; i < nbt; i += BLOCK) bt_lds[i] = bt[blk0 + i];
   }
-  // per-wave V^T image + cross-wave merge region
-  __shared__ __attribute__((aligned(16))) uint16_t vlds[NW][DH][VROW];
+  // cross-wave merge region
   __shared__ float mlds[NW][G][DH + 2];
   __syncthreads();
 
@@ -509,20 +510,8 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
     const int bi_l = bi - blk0;
     const int blk = (bi_l < BT_LDS) ? bt_lds[bi_l] : bt[bi];
     const uint16_t *kb = k_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
-    const uint16_t *vb = v_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
-
-    // ---- stage V transposed into this wave's LDS image ----------------
-    // lane reads token (lane*8+p*512)/DH's contiguous d-slice, writes it
-    // down a column of vlds (b16 scatter; wave-private so lgkm-ordered).
-#pragma unroll
-    for (int p = 0; p < BS * DH / (64 * 8); ++p) {
-      const int idx = p * 512 + lane * 8;
-      const int tok = idx / DH;
-      const int d0 = idx % DH;
-      rb::bf16x8 vv = *reinterpret_cast<const rb::bf16x8 *>(vb + idx);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) vlds[wid][d0 + e][tok] = vv.v[e];
-    }
+    // v block is [DH][BS] (transposed layout)
+    const uint16_t *vb = v_cache + ((int64_t)blk * hkv + h_kv) * DH * BS;
 
     // ---- S[16 tok][16 heads] = K_blk @ Q^T ----------------------------
     rb_f32x4v s4 = (rb_f32x4v)(0.0f);
@@ -582,7 +571,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         const rb_bf16x8v vf = *reinterpret_cast<const rb_bf16x8v *>(
-            &vlds[wid][dt * 32 + c32][h32 * 8]);
+            vb + (dt * 32 + c32) * BS + h32 * 8);
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             vf, pf.v, acc_o[dt], 0, 0, 0);
       }
@@ -667,15 +656,14 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
   const int max_blocks = (int)block_tables.size(1);
   uint16_t *swz = out_swz.has_value() ? (uint16_t *)out_swz->data_ptr()
                                       : nullptr;
-  // MFMA route (bf16 cache, G >= 4, DH <= 128, 16-token blocks): the
-  // GQA shapes where the scalar per-token VALU chain is the bind.
-  // RB_DECODE_MFMA=0 falls back to the scalar kernel for A/B runs.
+  // MFMA route: keyed on the TRANSPOSED V layout ([Dh, BS] blocks) the
+  // allocator chooses for MFMA-eligible models (bf16, G >= 4,
+  // DH <= 128, 16-token blocks — RB_DECODE_MFMA=0 at alloc reverts).
+  // The layout is the single source of truth: a transposed cache can
+  // only be read by the MFMA kernel and vice versa.
+  const bool vt = v_cache.size(2) == (int64_t)DH && v_cache.size(3) == bs;
   if constexpr (!FP8 && G >= 4 && DH <= 128) {
-    static const bool mfma_on = []() {
-      const char *e = getenv("RB_DECODE_MFMA");
-      return !(e && e[0] == '0');
-    }();
-    if (bs == 16 && mfma_on) {
+    if (vt && bs == 16) {
       if (nsplit <= 1) {
         hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, false>),
                            dim3(B, hkv, 1), dim3(BLOCK), 0, stream,
@@ -706,6 +694,8 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
       return;
     }
   }
+  TORCH_CHECK(!vt, "paged_decode: transposed V cache requires the MFMA "
+              "path (bf16, G>=4, Dh<=128, bs==16); got G=", G, " Dh=", DH);
   if (nsplit <= 1) {
     hipLaunchKernelGGL((paged_decode_kernel<DH, G, false, FP8>),
                        dim3(B, hkv, 1),

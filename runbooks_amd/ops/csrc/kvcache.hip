@@ -17,6 +17,11 @@ namespace {
 
 constexpr int BLOCK = 256;
 
+// VT: v_cache blocks are TRANSPOSED ([Dh, BS] instead of [BS, Dh]) for
+// the MFMA decode kernel's direct V^T fragment reads — the v write
+// becomes 8 element stores at stride BS (append is write-once traffic,
+// ~1000x smaller than the decode-side reads it speeds up).
+template <bool VT>
 __global__ void kv_append_kernel(const uint16_t *__restrict__ k,
                                  const uint16_t *__restrict__ v,
                                  uint16_t *__restrict__ k_cache,
@@ -39,8 +44,16 @@ __global__ void kv_append_kernel(const uint16_t *__restrict__ k,
     const int64_t dst = (((int64_t)blk * hkv + h) * bs + off) * dh + c * 8;
     *reinterpret_cast<rb::bf16x8 *>(k_cache + dst) =
         *reinterpret_cast<const rb::bf16x8 *>(k + src);
-    *reinterpret_cast<rb::bf16x8 *>(v_cache + dst) =
-        *reinterpret_cast<const rb::bf16x8 *>(v + src);
+    if (VT) {
+      const int64_t vbase = ((int64_t)blk * hkv + h) * (int64_t)dh * bs;
+      const rb::bf16x8 vv = *reinterpret_cast<const rb::bf16x8 *>(v + src);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        v_cache[vbase + (int64_t)(c * 8 + e) * bs + off] = vv.v[e];
+    } else {
+      *reinterpret_cast<rb::bf16x8 *>(v_cache + dst) =
+          *reinterpret_cast<const rb::bf16x8 *>(v + src);
+    }
   }
 }
 
@@ -117,9 +130,22 @@ void kv_append(at::Tensor k, at::Tensor v, at::Tensor k_cache, at::Tensor v_cach
   TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16, "kv_append: bf16 cache");
   const int dh = (int)k_cache.size(3);
   TORCH_CHECK(dh % 8 == 0, "kv_append: Dh % 8 == 0");
+  const bool vt = v_cache.size(2) == (int64_t)dh &&
+                  v_cache.size(3) == (int64_t)bs && dh != bs;
   const int grid = rb::rb_grid_1d(n_tokens * hkv * (dh / 8), BLOCK);
-  hipLaunchKernelGGL(kv_append_kernel, dim3(grid), dim3(BLOCK), 0, stream,
-                     (const uint16_t *)k.data_ptr(), (const uint16_t *)v.data_ptr(),
-                     (uint16_t *)k_cache.data_ptr(), (uint16_t *)v_cache.data_ptr(),
-                     slot_mapping.data_ptr<int32_t>(), n_tokens, hkv, bs, dh);
+  if (vt) {
+    hipLaunchKernelGGL(kv_append_kernel<true>, dim3(grid), dim3(BLOCK), 0,
+                       stream, (const uint16_t *)k.data_ptr(),
+                       (const uint16_t *)v.data_ptr(),
+                       (uint16_t *)k_cache.data_ptr(),
+                       (uint16_t *)v_cache.data_ptr(),
+                       slot_mapping.data_ptr<int32_t>(), n_tokens, hkv, bs, dh);
+  } else {
+    hipLaunchKernelGGL(kv_append_kernel<false>, dim3(grid), dim3(BLOCK), 0,
+                       stream, (const uint16_t *)k.data_ptr(),
+                       (const uint16_t *)v.data_ptr(),
+                       (uint16_t *)k_cache.data_ptr(),
+                       (uint16_t *)v_cache.data_ptr(),
+                       slot_mapping.data_ptr<int32_t>(), n_tokens, hkv, bs, dh);
+  }
 }

@@ -34,6 +34,9 @@ __device__ __forceinline__ void rope_pair(
   }
 }
 
+// VT: v_cache blocks transposed ([Dh, BS]) for the MFMA decode kernel
+// (see kvcache.hip) — the v copy becomes 8 stride-BS element stores.
+template <bool VT>
 __global__ void qkv_rope_append_kernel(
     const uint16_t *__restrict__ y, uint16_t *__restrict__ q_out,
     uint16_t *__restrict__ k_cache, uint16_t *__restrict__ v_cache,
@@ -88,10 +91,19 @@ __global__ void qkv_rope_append_kernel(
       const int slot = slots[t];
       if (slot < 0) continue;
       const uint16_t *src = y + t * ystride + (hq + hkv + h) * dh + c * 8;
-      uint16_t *dst = v_cache +
-          ((((int64_t)(slot / bs) * hkv + h) * bs + slot % bs) * dh) + c * 8;
-      *reinterpret_cast<rb::bf16x8 *>(dst) =
-          *reinterpret_cast<const rb::bf16x8 *>(src);
+      if (VT) {
+        const int64_t vbase =
+            ((int64_t)(slot / bs) * hkv + h) * (int64_t)dh * bs;
+        const rb::bf16x8 vv = *reinterpret_cast<const rb::bf16x8 *>(src);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v_cache[vbase + (int64_t)(c * 8 + e) * bs + slot % bs] = vv.v[e];
+      } else {
+        uint16_t *dst = v_cache +
+            ((((int64_t)(slot / bs) * hkv + h) * bs + slot % bs) * dh) + c * 8;
+        *reinterpret_cast<rb::bf16x8 *>(dst) =
+            *reinterpret_cast<const rb::bf16x8 *>(src);
+      }
     }
   }
 }
@@ -187,15 +199,29 @@ at::Tensor qkv_rope_append(at::Tensor y, at::Tensor cos, at::Tensor sin,
   auto stream = at::hip::getCurrentHIPStream();
   const int64_t total = T * (hq + hkv) * (dh / 8) + T * hkv * (dh / 8);
   const int grid = rb::rb_grid_1d(total, BLOCK);
-  hipLaunchKernelGGL(qkv_rope_append_kernel, dim3(grid), dim3(BLOCK), 0,
-                     stream, (const uint16_t *)y.data_ptr(),
-                     (uint16_t *)q.data_ptr(),
-                     (uint16_t *)k_cache.data_ptr(),
-                     (uint16_t *)v_cache.data_ptr(),
-                     cos.data_ptr<float>(), sin.data_ptr<float>(),
-                     positions.data_ptr<int32_t>(),
-                     slot_mapping.data_ptr<int32_t>(),
-                     T, (int)hq, hkv, dh, bs);
+  const bool vt = v_cache.size(2) == (int64_t)dh &&
+                  v_cache.size(3) == (int64_t)bs && dh != bs;
+  if (vt) {
+    hipLaunchKernelGGL(qkv_rope_append_kernel<true>, dim3(grid), dim3(BLOCK),
+                       0, stream, (const uint16_t *)y.data_ptr(),
+                       (uint16_t *)q.data_ptr(),
+                       (uint16_t *)k_cache.data_ptr(),
+                       (uint16_t *)v_cache.data_ptr(),
+                       cos.data_ptr<float>(), sin.data_ptr<float>(),
+                       positions.data_ptr<int32_t>(),
+                       slot_mapping.data_ptr<int32_t>(),
+                       T, (int)hq, hkv, dh, bs);
+  } else {
+    hipLaunchKernelGGL(qkv_rope_append_kernel<false>, dim3(grid), dim3(BLOCK),
+                       0, stream, (const uint16_t *)y.data_ptr(),
+                       (uint16_t *)q.data_ptr(),
+                       (uint16_t *)k_cache.data_ptr(),
+                       (uint16_t *)v_cache.data_ptr(),
+                       cos.data_ptr<float>(), sin.data_ptr<float>(),
+                       positions.data_ptr<int32_t>(),
+                       slot_mapping.data_ptr<int32_t>(),
+                       T, (int)hq, hkv, dh, bs);
+  }
   return q;
 }
 

@@ -10,10 +10,15 @@ BLOCK_SIZE = 16  # tokens per cache block
 
 def alloc_kv_cache(num_blocks: int, num_kv_heads: int, head_dim: int,
                    device, dtype=torch.bfloat16, block_size: int = BLOCK_SIZE,
-                   fp8: bool = False):
+                   fp8: bool = False, v_transposed: bool = False):
     """fp8=True: e4m3 cache rows of head_dim bytes + f32 scale (+pad) —
     half the KV bytes of bf16 (decode attention is KV-bandwidth bound at
-    long context) and double the capacity within 288 GB."""
+    long context) and double the capacity within 288 GB.
+
+    v_transposed=True (bf16 only): v blocks are [head_dim, block_size]
+    so the MFMA decode kernel reads V^T fragments straight from HBM (the
+    LDS-transpose alternative measured 26-43% of wave cycles in bank
+    conflicts). kv_append and paged_decode key on the shape."""
     if fp8:
         # dh e4m3 bytes + f32 scale + pad to 16 (16-aligned rows: the
         # decode loop stays on full 16 B/lane loads)
@@ -23,7 +28,9 @@ def alloc_kv_cache(num_blocks: int, num_kv_heads: int, head_dim: int,
         return k, v
     shape = (num_blocks, num_kv_heads, block_size, head_dim)
     k = torch.zeros(shape, device=device, dtype=dtype)
-    v = torch.zeros(shape, device=device, dtype=dtype)
+    vshape = ((num_blocks, num_kv_heads, head_dim, block_size)
+              if v_transposed else shape)
+    v = torch.zeros(vshape, device=device, dtype=dtype)
     return k, v
 
 
@@ -63,13 +70,18 @@ def kv_append_ref(k, v, k_cache, v_cache, slot_mapping):
     if k_cache.dtype == torch.uint8:
         return _kv_append_fp8_ref(k, v, k_cache, v_cache, slot_mapping)
     bs = k_cache.shape[2]
+    vt = v_cache.shape[2] == k_cache.shape[3] and \
+        v_cache.shape[3] == bs and k_cache.shape[3] != bs
     for t in range(k.shape[0]):
         slot = int(slot_mapping[t])
         if slot < 0:
             continue
         blk, off = divmod(slot, bs)
         k_cache[blk, :, off] = k[t]
-        v_cache[blk, :, off] = v[t]
+        if vt:
+            v_cache[blk, :, :, off] = v[t]
+        else:
+            v_cache[blk, :, off] = v[t]
 
 
 def kv_append(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,

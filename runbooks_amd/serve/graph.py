@@ -17,8 +17,7 @@ from __future__ import annotations
 import torch
 
 
-def fixed_nsplit(batch: int, hkv: int, g: int = 1, dh: int = 128,
-                 fp8: bool = False) -> int:
+def fixed_nsplit(batch: int, hkv: int, mfma: bool = False) -> int:
     """Work-split for the paged-decode kernel, chosen per bucket at capture
     time (the eager heuristic in ops/attention.py reads seq_lens.max() —
     a host sync, impossible inside a graph). base = batch*hkv workgroups
@@ -28,11 +27,11 @@ def fixed_nsplit(batch: int, hkv: int, g: int = 1, dh: int = 128,
     TB/s, falcon MQA 0.72 vs 0.23, B=8 0.99-1.24 vs 0.16 —
     profiles/decode_attn_pipeline.md)."""
     base = max(1, batch * hkv)
-    # MFMA GQA path (g>=4, bf16, dh<=128): per-wave work is block-
-    # granular and latency-light — ~512 WGs is the knee (measured:
-    # llama2-70b B=32 len-2048 best at nsplit 2, B=8 at 4-8); the
-    # scalar kernels want ~1024.
-    target = 512 if (g >= 4 and not fp8 and dh <= 128) else 1024
+    # MFMA path (transposed-V caches): per-wave work is block-granular
+    # and latency-light — ~512 WGs is the knee (measured: llama2-70b
+    # B=32 len-2048 best at nsplit 2, B=8 at 4-8); the scalar kernels
+    # want ~1024.
+    target = 512 if mfma else 1024
     if base >= target:
         return 1
     return min(16, max(1, target // base))
@@ -73,10 +72,9 @@ class GraphedDecoder:
     # -- capture -----------------------------------------------------------
     def _capture(self, b: int):
         hkv = self.model.local_kv_heads()
-        attn = self.model.blocks[0].attn
-        nsplit = fixed_nsplit(b, hkv, g=attn.hq // attn.hkv,
-                              dh=self.model.cfg.head_dim,
-                              fp8=self.caches[0][0].dtype == torch.uint8)
+        from ..ops.attention import _is_vt
+        nsplit = fixed_nsplit(b, hkv,
+                              mfma=_is_vt(*self.caches[0][:2]))
         args = (self.tokens[:b], self.positions[:b], self.caches,
                 self.slots[:b], self.block_tables[:b], self.seq_lens[:b])
         kw = dict(nsplit=nsplit, seq_starts=self.seq_starts[:b])

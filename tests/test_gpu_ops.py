@@ -174,7 +174,12 @@ def test_kv_append():
 @pytest.mark.parametrize("hq,hkv,dh", [(8, 8, 128), (8, 2, 128), (16, 1, 64),
                                        (64, 8, 128), (4, 4, 64)])
 @pytest.mark.parametrize("nsplit", [1, 4])
-def test_paged_decode(hq, hkv, dh, nsplit):
+@pytest.mark.parametrize("vt", [False, True])
+def test_paged_decode(hq, hkv, dh, nsplit, vt):
+    # vt=True: transposed-V cache layout -> the MFMA decode kernel
+    # (only valid for its eligibility set G>=4 / dh<=128)
+    if vt and (hq // hkv < 4 or dh > 128):
+        pytest.skip("vt layout is MFMA-only (G>=4, dh<=128)")
     _assert_hip()
     torch.manual_seed(hq * 100 + hkv)
     B, BS = 3, 16
@@ -191,12 +196,43 @@ def test_paged_decode(hq, hkv, dh, nsplit):
         nxt += nb
     q = torch.randn(B, hq, dh, dtype=torch.bfloat16, device=DEV)
     scale = 1 / math.sqrt(dh)
-    out = ops.paged_decode(q, kc, vc, bt.to(DEV), seq_lens.to(DEV),
+    vc_dev = vc.permute(0, 1, 3, 2).contiguous() if vt else vc
+    out = ops.paged_decode(q, kc, vc_dev, bt.to(DEV), seq_lens.to(DEV),
                            scale=scale, nsplit=nsplit)
     ref = ops.paged_decode_ref(q.cpu().float(), kc.cpu().float(), vc.cpu().float(),
                                bt, seq_lens, scale)
     assert torch.allclose(out.cpu().float(), ref, atol=3e-2, rtol=3e-2), \
         (out.cpu().float() - ref).abs().max()
+
+
+@pytest.mark.parametrize("fused", [False, True])
+def test_kv_append_v_transposed(fused):
+    """kv_append / qkv_rope_append into the transposed-V layout match the
+    plain layout element-for-element."""
+    _assert_hip()
+    torch.manual_seed(3)
+    hkv, dh, BS, nb = 8, 128, 16, 6
+    T = 21
+    k = torch.randn(T, hkv, dh, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, hkv, dh, dtype=torch.bfloat16, device=DEV)
+    slots = torch.arange(3, 3 + T, dtype=torch.int32, device=DEV)
+    kc1, vc1 = ops.alloc_kv_cache(nb, hkv, dh, DEV)
+    kc2, vc2 = ops.alloc_kv_cache(nb, hkv, dh, DEV, v_transposed=True)
+    if fused:
+        hq = 16
+        y = torch.randn(T, (hq + 2 * hkv) * dh, dtype=torch.bfloat16,
+                        device=DEV)
+        pos = torch.arange(T, dtype=torch.int32, device=DEV)
+        from runbooks_amd.ops.rope import rope_tables
+        cos, sin = rope_tables(dh, 64, 10000.0, DEV)
+        q1 = ops.ext().qkv_rope_append(y, cos, sin, pos, kc1, vc1, slots, hq)
+        q2 = ops.ext().qkv_rope_append(y, cos, sin, pos, kc2, vc2, slots, hq)
+        assert torch.equal(q1, q2)
+    else:
+        ops.kv_append(k, v, kc1, vc1, slots)
+        ops.kv_append(k, v, kc2, vc2, slots)
+    assert torch.equal(kc1, kc2)
+    assert torch.equal(vc1, vc2.permute(0, 1, 3, 2).contiguous())
 
 
 def test_engine_gpu_matches_full_forward():
