@@ -40,6 +40,8 @@ def bench(B, Hq, Hkv, Dh, length, fp8, iters=50):
         kc = torch.randn(num_blocks, Hkv, BS, Dh, device="cuda",
                          dtype=torch.bfloat16)
         kv = torch.randn_like(kc)
+        if Hq // Hkv >= 4 and Dh <= 128:  # MFMA path: transposed-V layout
+            kv = kv.permute(0, 1, 3, 2).contiguous()
 
     q = torch.randn(B, Hq, Dh, device="cuda", dtype=torch.bfloat16)
     seq_lens = torch.full((B,), length, device="cuda", dtype=torch.int32)

@@ -32,6 +32,8 @@ def main():
     kc = torch.randn(num_blocks, Hkv, BS, Dh, device="cuda",
                      dtype=torch.bfloat16)
     kv = torch.randn_like(kc)
+    if Hq // Hkv >= 4 and Dh <= 128:   # MFMA path: transposed-V layout
+        kv = kv.permute(0, 1, 3, 2).contiguous()
     q = torch.randn(B, Hq, Dh, device="cuda", dtype=torch.bfloat16)
     seq_lens = torch.full((B,), length, device="cuda", dtype=torch.int32)
     t = torch.arange(1, 1 + B * bps, device="cuda",
