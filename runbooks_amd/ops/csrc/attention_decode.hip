@@ -506,20 +506,37 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
   const int blk_lo = t_begin / BS;
   const int blk_hi = (t_end > t_begin) ? (t_end - 1) / BS : blk_lo - 1;
 
-  for (int bi = blk_lo + wid; bi <= blk_hi; bi += NW) {
+  // 2-deep block ring: iteration i+1's K/V fragments are issued while
+  // i computes (PMC post-VT-layout: 74% WAIT_ANY at occupancy 2 —
+  // two waves/SIMD cannot hide HBM latency without in-flight depth).
+  // +KS32+DT fragment registers per slot keeps occupancy at 2.
+  rb_bf16x8v kfr[2][KS32], vfr[2][DT];
+
+  auto fetch = [&](int bi, int slot) {
     const int bi_l = bi - blk0;
     const int blk = (bi_l < BT_LDS) ? bt_lds[bi_l] : bt[bi];
     const uint16_t *kb = k_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
     // v block is [DH][BS] (transposed layout)
     const uint16_t *vb = v_cache + ((int64_t)blk * hkv + h_kv) * DH * BS;
+#pragma unroll
+    for (int ks = 0; ks < KS32; ++ks)
+      kfr[slot][ks] = *reinterpret_cast<const rb_bf16x8v *>(
+          kb + i16 * DH + ks * 32 + h16 * 8);
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+      vfr[slot][dt] = *reinterpret_cast<const rb_bf16x8v *>(
+          vb + (dt * 32 + c32) * BS + h32 * 8);
+  };
+
+  auto step = [&](int bi, int cur) {
+    if (bi + NW <= blk_hi) fetch(bi + NW, cur ^ 1);
 
     // ---- S[16 tok][16 heads] = K_blk @ Q^T ----------------------------
     rb_f32x4v s4 = (rb_f32x4v)(0.0f);
 #pragma unroll
     for (int ks = 0; ks < KS32; ++ks) {
-      const rb_bf16x8v kf = *reinterpret_cast<const rb_bf16x8v *>(
-          kb + i16 * DH + ks * 32 + h16 * 8);
-      s4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[ks], s4, 0, 0, 0);
+      s4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfr[cur][ks], qf[ks], s4,
+                                                   0, 0, 0);
     }
 
     // ---- mask + online softmax (state per head = per col = per lane) --
@@ -570,12 +587,17 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
       // ---- O^T[d][head] += V^T_blk @ P --------------------------------
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        const rb_bf16x8v vf = *reinterpret_cast<const rb_bf16x8v *>(
-            vb + (dt * 32 + c32) * BS + h32 * 8);
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            vf, pf.v, acc_o[dt], 0, 0, 0);
+            vfr[cur][dt], pf.v, acc_o[dt], 0, 0, 0);
       }
     }
+  };
+
+  const int bi0 = blk_lo + wid;
+  if (bi0 <= blk_hi) fetch(bi0, 0);
+  for (int bi = bi0; bi <= blk_hi; bi += 2 * NW) {
+    step(bi, 0);
+    if (bi + NW <= blk_hi) step(bi + NW, 1);
   }
 
   // ---- cross-wave merge via LDS (same scheme as the scalar kernel) ----
