@@ -70,6 +70,88 @@ def _wt(w: torch.Tensor) -> torch.Tensor:
     return wt
 
 
+class _LoRAGroupPrep(torch.autograd.Function):
+    """tcat = x @ cat(A_1..A_n)^T for sibling adapters sharing the same
+    input (q/k/v; gate/up): one activation stream instead of n.
+
+    Backward: dcat = dtcat^T x (one GEMM, x read once) sliced into the
+    per-adapter dA's, and dx = dtcat @ catA (one GEMM; autograd adds it
+    to the base-path dx's)."""
+
+    @staticmethod
+    def forward(ctx, x, *As):
+        catA = torch.cat(As)                    # [n*r, in]
+        t = x @ catA.t()                        # [T, n*r]
+        ctx.save_for_backward(x, catA)
+        ctx.rs = [a.shape[0] for a in As]
+        return t
+
+    @staticmethod
+    def backward(ctx, dt):
+        x, catA = ctx.saved_tensors
+        dcat = dt.t() @ x                       # [n*r, in]
+        dx = dt @ catA
+        das, off = [], 0
+        for r in ctx.rs:
+            das.append(dcat[off:off + r])
+            off += r
+        return (dx, *das)
+
+
+class _LoRAGroup:
+    """Shared-input sibling adapters + the per-forward tcat cache (keyed
+    on activation identity; dropped after every member consumed it so no
+    autograd graph outlives the step)."""
+
+    def __init__(self, members):
+        self.members = members
+        self._x_id = None
+        self._t = None
+        self._left = 0
+
+    def t_for(self, member, x):
+        if self._x_id != id(x) or self._left <= 0:
+            self._t = _LoRAGroupPrep.apply(
+                x, *[m.lora_a for m in self.members])
+            self._x_id = id(x)
+            self._left = len(self.members)
+        i = self.members.index(member)
+        off = sum(m.r for m in self.members[:i])
+        t = self._t[:, off:off + member.r]
+        self._left -= 1
+        if self._left == 0:
+            self._t = None
+            self._x_id = None
+        return t
+
+
+class _LoRAFusedT(torch.autograd.Function):
+    """y = x W^T + s t B^T with t precomputed by the group prep.
+    dt flows back through _LoRAGroupPrep (which owns the dx merge and
+    the dA GEMM for the whole group)."""
+
+    @staticmethod
+    def forward(ctx, x, w, t, b, scale):
+        y = _nt(x, w)
+        _delta_add_(y, t, b, scale)
+        ctx.save_for_backward(w, t, b, x)
+        ctx.scale = scale
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        w, t, b, x = ctx.saved_tensors
+        s = ctx.scale
+        if (_USE_CUSTOM_GEMM and dy.is_cuda and dy.dtype == torch.bfloat16
+                and _nt_ok(dy.shape[0], w.shape[1], dy.shape[1])):
+            dx = _nt(dy, _wt(w))
+        else:
+            dx = dy @ w
+        dt = (dy @ b).mul_(s)
+        db = torch.mm(dy.t(), t).mul_(s)
+        return dx, None, dt, db, None
+
+
 class _LoRAFused(torch.autograd.Function):
     """y = x W^T + s (x A^T) B^T with a GEMM-only backward.
 
@@ -148,6 +230,13 @@ class LoRALinear(nn.Module):
         if (isinstance(base, nn.Linear) or
                 (getattr(base, "tp", 1) == 1 and base.bias is None)) and \
                 x.dim() == 2 and getattr(base, "bias", None) is None:
+            group = getattr(self, "_group", None)
+            if group is not None:
+                # shared-input siblings (q/k/v, gate/up): the A-GEMM,
+                # its dA and the t-path dx merge run ONCE per group
+                t = group.t_for(self, x)
+                return _LoRAFusedT.apply(x, base.weight, t, self.lora_b,
+                                         self.scale)
             # fully-fused fwd+bwd path: every product is one hipBLASLt
             # call, the two gradient merges ride addmm epilogues
             return _LoRAFused.apply(x, base.weight, self.lora_a,
@@ -174,6 +263,7 @@ def apply_lora(model: nn.Module, r: int = 16, alpha: int = 32,
     for p in model.parameters():
         p.requires_grad_(False)
     wrapped = []
+    group_on = os.environ.get("RB_LORA_GROUP", "1") != "0"
     for name, module in model.named_modules():
         for child_name, child in list(module.named_children()):
             if child_name in targets and isinstance(
@@ -181,6 +271,19 @@ def apply_lora(model: nn.Module, r: int = 16, alpha: int = 32,
                 setattr(module, child_name,
                         LoRALinear(child, r=r, alpha=alpha, dropout=dropout))
                 wrapped.append(f"{name}.{child_name}" if name else child_name)
+        if group_on and dropout <= 0:
+            # group shared-input siblings for the one-A-GEMM path; only
+            # plain bias-free nn.Linear bases (the fused-path criteria)
+            for names in (("q_proj", "k_proj", "v_proj"),
+                          ("gate_proj", "up_proj")):
+                ms = [getattr(module, n, None) for n in names]
+                if all(isinstance(m, LoRALinear) and
+                       (isinstance(m.base, nn.Linear)
+                        or getattr(m.base, "tp", 1) == 1) and
+                       getattr(m.base, "bias", None) is None for m in ms):
+                    g = _LoRAGroup(ms)
+                    for m in ms:
+                        m._group = g
     if not wrapped:
         raise ValueError("apply_lora: no target modules found")
     return wrapped

@@ -263,3 +263,57 @@ def test_lora_dropout():
     with torch.no_grad():
         c, d = m(x), m(x)
     assert torch.equal(c, d)
+
+
+def test_lora_group_matches_ungrouped(monkeypatch):
+    """Grouped q/k/v + gate/up A-GEMMs (one activation stream per group)
+    produce the SAME loss and lora grads as the per-module fused path."""
+    import torch
+
+    from runbooks_amd.models import build_model
+    from runbooks_amd.train.lora import apply_lora
+
+    def run(grouped: bool):
+        monkeypatch.setenv("RB_LORA_GROUP", "1" if grouped else "0")
+        torch.manual_seed(11)
+        m = build_model("tiny-llama", dtype=torch.float32, seed=5)
+        apply_lora(m, r=4, alpha=8)
+        has_group = any(getattr(mm, "_group", None) is not None
+                        for mm in m.modules())
+        assert has_group == grouped
+        torch.manual_seed(2)
+        x = torch.randint(0, m.cfg.vocab_size, (2, 12))
+        logits = m(x)
+        loss = logits.float().pow(2).mean()
+        loss.backward()
+        grads = {k: p.grad.clone() for k, p in m.named_parameters()
+                 if p.grad is not None}
+        return float(loss), grads
+
+    l1, g1 = run(True)
+    l2, g2 = run(False)
+    assert abs(l1 - l2) < 1e-6, (l1, l2)
+    assert set(g1) == set(g2) and g1, "lora grads must exist"
+    for k in g1:
+        assert torch.allclose(g1[k], g2[k], atol=1e-5), (k, (g1[k] - g2[k]).abs().max())
+
+
+def test_lora_group_cache_does_not_leak_graph():
+    """The group tcat cache must be dropped after every member consumed
+    it (a retained autograd graph across steps is a memory leak)."""
+    import torch
+
+    from runbooks_amd.models import build_model
+    from runbooks_amd.train.lora import apply_lora, _LoRAGroup
+
+    torch.manual_seed(0)
+    m = build_model("tiny-llama", dtype=torch.float32, seed=5)
+    apply_lora(m, r=4, alpha=8)
+    x = torch.randint(0, m.cfg.vocab_size, (2, 8))
+    m(x).sum().backward()
+    groups = {id(mm._group): mm._group for mm in m.modules()
+              if getattr(mm, "_group", None) is not None}
+    assert groups
+    for g in groups.values():
+        assert isinstance(g, _LoRAGroup)
+        assert g._t is None and g._left == 0
