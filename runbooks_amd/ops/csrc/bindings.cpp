@@ -8,6 +8,7 @@ std::vector<at::Tensor> rmsnorm_res_fwd_dec(at::Tensor x, at::Tensor res, at::Te
 std::vector<at::Tensor> rmsnorm_res_slab_fwd_dec(at::Tensor x, at::Tensor slabs, at::Tensor w, double eps);
 at::Tensor decode_gemm_raw(at::Tensor xs, at::Tensor ws, int64_t M, int64_t N, int64_t K);
 at::Tensor lora_delta_(at::Tensor y, at::Tensor t, at::Tensor w, double scale, bool w_transposed);
+at::Tensor lora_badd_(at::Tensor y, at::Tensor t, at::Tensor w, double scale);
 std::vector<at::Tensor> swiglu_packed_dec(at::Tensor y);
 std::vector<at::Tensor> geglu_packed_dec(at::Tensor y);
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
@@ -79,6 +80,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "decode GEMM returning the uncombined fp32 slab at split>1");
   m.def("lora_delta_", &lora_delta_,
         "in-place y += scale * t[T,r<=32] @ W[N,r]^T (LoRA merge)");
+  m.def("lora_badd_", &lora_badd_,
+        "MFMA in-place y += scale * t[T,16] @ W[N,16]^T (LoRA B merge)");
   m.def("swiglu_packed_dec", &swiglu_packed_dec,
         "packed SwiGLU + decode-GEMM operand swizzle in one pass");
   m.def("geglu_packed_dec", &geglu_packed_dec,

@@ -21,12 +21,23 @@ from ..parallel.tp import ColumnParallelLinear, RowParallelLinear
 
 
 def _delta_add_(y, t, w, scale, w_transposed=True):
-    """y += scale * t @ (w^T if w_transposed else w). In-place addmm_
-    (hipBLASLt accumulate) is the shipped path: the hand-written
-    lora_delta_ kernel is numerically right but its per-lane W gather
-    (64 distinct addresses per instruction) measured SLOWER in-step
-    (96 -> 124 ms, gpurun r17) — opt in with RB_LORA_KERNEL=1 for
-    A/B work only."""
+    """y += scale * t @ (w^T if w_transposed else w).
+
+    r=16 w_transposed merges run the MFMA lora_badd_ kernel (both
+    operands are natural contiguous 32x32x16 fragments; hipBLASLt runs
+    these K=16 accumulates ~4x off the y-stream roofline — r33
+    torch.profiler). RB_LORA_MFMA=0 reverts to in-place addmm_.
+    The older scalar lora_delta_ kernel (per-lane W gather, measured
+    slower) stays behind RB_LORA_KERNEL=1 for A/B work only."""
+    if (w_transposed and os.environ.get("RB_LORA_MFMA", "1") == "1"
+            and y.is_cuda and y.dtype == torch.bfloat16
+            and t.shape[1] == 16 and y.shape[1] % 32 == 0
+            and w.dtype == torch.bfloat16):
+        from ..ops import _backend
+        if _backend.use_hip(y):
+            _backend.ext().lora_badd_(y, t.contiguous(), w.contiguous(),
+                                      scale)
+            return y
     if (os.environ.get("RB_LORA_KERNEL", "0") == "1"
             and y.is_cuda and y.dtype == torch.bfloat16
             and y.shape[1] % 8 == 0

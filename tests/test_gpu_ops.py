@@ -768,3 +768,18 @@ def test_flash_prefill_dh256():
                                v.float().cpu())
     d = (got.float().cpu() - ref).abs().max() / ref.abs().max()
     assert d < 3e-2, d
+
+
+@pytest.mark.parametrize("T,N", [(2048, 4096), (2048, 11008), (100, 352),
+                                 (33, 4096)])
+def test_lora_badd(T, N):
+    """MFMA y += s * t[T,16] @ W[N,16]^T vs torch addmm."""
+    _assert_hip()
+    torch.manual_seed(T + N)
+    y = torch.randn(T, N, dtype=torch.bfloat16, device=DEV)
+    t = torch.randn(T, 16, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(N, 16, dtype=torch.bfloat16, device=DEV)
+    ref = (y.float() + 2.0 * t.float() @ w.float().t())
+    ops.ext().lora_badd_(y, t, w, 2.0)
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2), \
+        (y.float() - ref).abs().max()
