@@ -29,13 +29,18 @@ def _delta_add_(y, t, w, scale, w_transposed=True):
     torch.profiler). RB_LORA_MFMA=0 reverts to in-place addmm_.
     The older scalar lora_delta_ kernel (per-lane W gather, measured
     slower) stays behind RB_LORA_KERNEL=1 for A/B work only."""
-    if (w_transposed and os.environ.get("RB_LORA_MFMA", "1") == "1"
+    if (os.environ.get("RB_LORA_MFMA", "1") == "1"
             and y.is_cuda and y.dtype == torch.bfloat16
             and t.shape[1] == 16 and y.shape[1] % 32 == 0
             and w.dtype == torch.bfloat16):
         from ..ops import _backend
         if _backend.use_hip(y):
-            _backend.ext().lora_badd_(y, t.contiguous(), w.contiguous(),
+            # w_transposed=False (y += t @ w, w [16, N]): transpose the
+            # TINY w into the kernel's [N, 16] fragment layout — ~5 us
+            # vs the 110 us hipBLASLt accumulate it replaces (r35
+            # profile, [2048, 11008] dx merge).
+            wk = w if w_transposed else w.t()
+            _backend.ext().lora_badd_(y, t.contiguous(), wk.contiguous(),
                                       scale)
             return y
     if (os.environ.get("RB_LORA_KERNEL", "0") == "1"
