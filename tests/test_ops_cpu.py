@@ -170,3 +170,81 @@ def test_fp8_kv_engine_cpu_decodes():
     e16 = Engine(m, device="cpu", dtype=torch.float32, kv_blocks=64, seed=1)
     ref = e16.generate([5, 1, 9, 2], max_new_tokens=6)
     assert out[0] == ref[0], (out, ref)
+
+
+def test_auto_nsplit_heuristic():
+    """Work-split targets: MFMA path ~512 WGs, scalar ~1024; chunks
+    cover >= ~64 tokens; no split for short sequences."""
+    import torch
+
+    from runbooks_amd.ops.attention import _auto_nsplit
+
+    lens = torch.tensor([2048])
+    # llama2-7b MHA B=32: base 1024 -> never split
+    assert _auto_nsplit(32, 32, lens) == 1
+    # llama2-70b GQA B=32 (base 256): scalar wants 4, MFMA wants 2
+    assert _auto_nsplit(32, 8, lens) == 4
+    assert _auto_nsplit(32, 8, lens, mfma=True) == 2
+    # B=8 (base 64): MFMA 8
+    assert _auto_nsplit(8, 8, lens, mfma=True) == 8
+    # short sequences: chunk >= ~64 tokens caps the split
+    assert _auto_nsplit(32, 8, torch.tensor([128]), mfma=True) == 2
+    assert _auto_nsplit(32, 8, torch.tensor([40]), mfma=True) == 1
+
+
+def test_is_vt_layout_detection():
+    import torch
+
+    from runbooks_amd.ops.attention import _is_vt
+    from runbooks_amd.ops.kvcache import alloc_kv_cache
+
+    k, v = alloc_kv_cache(4, 8, 128, "cpu")
+    assert not _is_vt(k, v)
+    k, v = alloc_kv_cache(4, 8, 128, "cpu", v_transposed=True)
+    assert _is_vt(k, v)
+    assert v.shape == (4, 8, 128, 16)
+    # fp8 rows are never vt
+    k, v = alloc_kv_cache(4, 8, 128, "cpu", fp8=True)
+    assert not _is_vt(k, v)
+
+
+def test_kv_append_ref_v_transposed():
+    """CPU reference append into the transposed-V layout round-trips."""
+    import torch
+
+    from runbooks_amd.ops.kvcache import alloc_kv_cache, kv_append_ref
+
+    torch.manual_seed(0)
+    hkv, dh, T = 2, 64, 5
+    k = torch.randn(T, hkv, dh)
+    v = torch.randn(T, hkv, dh)
+    slots = torch.arange(T, dtype=torch.int32)
+    k1, v1 = alloc_kv_cache(2, hkv, dh, "cpu", dtype=torch.float32)
+    k2, v2 = alloc_kv_cache(2, hkv, dh, "cpu", dtype=torch.float32,
+                            v_transposed=True)
+    kv_append_ref(k, v, k1, v1, slots)
+    kv_append_ref(k, v, k2, v2, slots)
+    assert torch.equal(k1, k2)
+    assert torch.equal(v1, v2.permute(0, 1, 3, 2))
+
+
+def test_paged_decode_ref_v_transposed_matches():
+    """The fp32 reference produces identical output for both V layouts."""
+    import math
+
+    import torch
+
+    from runbooks_amd.ops.attention import paged_decode_ref
+
+    torch.manual_seed(1)
+    B, hkv, hq, dh, BS = 2, 2, 8, 64, 16
+    kc = torch.randn(6, hkv, BS, dh)
+    vc = torch.randn_like(kc)
+    bt = torch.tensor([[0, 1, 2], [3, 4, 5]], dtype=torch.int32)
+    lens = torch.tensor([33, 17], dtype=torch.int32)
+    q = torch.randn(B, hq, dh)
+    s = 1 / math.sqrt(dh)
+    o1 = paged_decode_ref(q, kc, vc, bt, lens, s)
+    o2 = paged_decode_ref(q, kc, vc.permute(0, 1, 3, 2).contiguous(),
+                          bt, lens, s)
+    assert torch.allclose(o1, o2, atol=1e-6)
