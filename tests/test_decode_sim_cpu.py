@@ -155,3 +155,34 @@ def test_decode_sim_seq_start_strict_window():
             torch.tensor([[0, 1, 2, 3]], dtype=torch.int32),
             torch.tensor([n], dtype=torch.int32), scale).numpy()
         assert np.allclose(got[b], ref[0], atol=1e-5), b
+
+
+def test_engine_cpu_bf16_gqa_transposed_v_cache():
+    """A bf16 GQA (G=4) model on CPU allocates the transposed-V cache
+    and the reference decode path reads it: engine greedy decode must
+    match the full no-cache forward (end-to-end vt coverage without a
+    GPU)."""
+    import dataclasses
+
+    import torch
+
+    from runbooks_amd.models import build_model, get_config
+    from runbooks_amd.models.config import register
+    from runbooks_amd.ops.attention import _is_vt
+    from runbooks_amd.serve import Engine
+
+    cfg = dataclasses.replace(get_config("smoke-llama"), num_heads=8,
+                              num_kv_heads=2, head_dim=None, hidden_size=512,
+                              name="smoke-llama-gqa4")
+    register(cfg)
+    m = build_model(cfg.name, dtype=torch.bfloat16, seed=9)
+    eng = Engine(m, device="cpu", kv_blocks=64, seed=1)
+    assert _is_vt(*eng.caches[0][:2]), "GQA bf16 must allocate vt caches"
+    prompt = [3, 1, 4, 1, 5]
+    out = eng.generate(list(prompt), max_new_tokens=5)
+    seq = list(prompt)
+    for _ in range(5):
+        with torch.no_grad():
+            logits = m(torch.tensor([seq]))
+        seq.append(int(logits[0, -1].float().argmax()))
+    assert out == seq[len(prompt):], (out, seq[len(prompt):])
