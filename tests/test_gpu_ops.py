@@ -727,6 +727,14 @@ def test_paged_decode_seq_starts():
                                seq_starts=st)
         d = (got.float().cpu() - ref).abs().max() / ref.abs().max()
         assert d < 3e-2, (nsplit, d)
+    # the same windows through the MFMA kernel (transposed-V layout —
+    # the live config for mistral-style sliding windows at G>=4)
+    vt = vc.permute(0, 1, 3, 2).contiguous()
+    for nsplit in (1, 4):
+        got = ops.paged_decode(q, kc, vt, bt, sl, nsplit=nsplit,
+                               seq_starts=st)
+        d = (got.float().cpu() - ref).abs().max() / ref.abs().max()
+        assert d < 3e-2, ("vt", nsplit, d)
 
 
 def test_paged_decode_with_operand_swz():
@@ -740,18 +748,21 @@ def test_paged_decode_with_operand_swz():
     q = torch.randn(B, hkv * G, dh, dtype=torch.bfloat16, device=DEV)
     bt = torch.arange(B * 4, dtype=torch.int32, device=DEV).reshape(B, 4)
     sl = torch.tensor([60, 33, 12, 7], dtype=torch.int32, device=DEV)
-    for nsplit in (1, 4):
-        out, swz = ops.paged_decode_with_operand(q, kc, vc, bt, sl,
-                                                 nsplit=nsplit)
-        assert swz is not None
-        o2 = out.reshape(B, hkv * G * dh).contiguous()
-        ref = ops.ext().decode_swizzle_x(o2)
-        # rows m >= B are undefined in both layouts; compare live rows
-        K = hkv * G * dh
-        s_v = swz.view(K // 16, 2, 32, 8)
-        r_v = ref.view(K // 16, 2, 32, 8)
-        d = (s_v[:, :, :B].float() - r_v[:, :, :B].float()).abs().max()
-        assert d == 0, (nsplit, d)
+    # both kernel paths emit the operand: scalar (plain layout) and
+    # MFMA (transposed-V layout)
+    for v_dev in (vc, vc.permute(0, 1, 3, 2).contiguous()):
+        for nsplit in (1, 4):
+            out, swz = ops.paged_decode_with_operand(q, kc, v_dev, bt, sl,
+                                                     nsplit=nsplit)
+            assert swz is not None
+            o2 = out.reshape(B, hkv * G * dh).contiguous()
+            ref = ops.ext().decode_swizzle_x(o2)
+            # rows m >= B are undefined in both layouts; compare live rows
+            K = hkv * G * dh
+            s_v = swz.view(K // 16, 2, 32, 8)
+            r_v = ref.view(K // 16, 2, 32, 8)
+            d = (s_v[:, :, :B].float() - r_v[:, :, :B].float()).abs().max()
+            assert d == 0, (nsplit, d)
 
 
 def test_flash_prefill_dh256():
