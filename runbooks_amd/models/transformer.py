@@ -422,7 +422,7 @@ class Transformer(nn.Module):
         # paged_decode). RB_DECODE_MFMA=0 reverts to the scalar path.
         import os
         attn = self.blocks[0].attn
-        vt = (not fp8 and self.dtype == torch.bfloat16
+        vt = ((fp8 or self.dtype == torch.bfloat16)
               and attn.hq // attn.hkv >= 4 and self.cfg.head_dim <= 128
               and ops.BLOCK_SIZE == 16
               and os.environ.get("RB_DECODE_MFMA", "1") != "0")

@@ -150,10 +150,16 @@ def paged_decode_ref(q, k_cache, v_cache, block_tables, seq_lens, scale,
 
 def _is_vt(k_cache, v_cache) -> bool:
     """Transposed-V cache layout (alloc_kv_cache v_transposed) == the
-    MFMA decode path; the shape is the routing signal."""
-    return (v_cache.dim() == 4 and v_cache.shape[2] == k_cache.shape[3]
-            and v_cache.shape[3] == k_cache.shape[2]
-            and k_cache.shape[3] != k_cache.shape[2])
+    MFMA decode path; the shape is the routing signal. fp8 vt blocks
+    are [dh+4, bs] bytes (scale tail)."""
+    if v_cache.dim() != 4:
+        return False
+    bs = k_cache.shape[2]
+    if k_cache.dtype == torch.uint8:
+        dh = k_cache.shape[3] - 16
+        return v_cache.shape[2] == dh + 4 and v_cache.shape[3] == bs
+    return (v_cache.shape[2] == k_cache.shape[3]
+            and v_cache.shape[3] == bs and k_cache.shape[3] != bs)
 
 
 def _auto_nsplit(B, hkv, seq_lens, mfma=False):

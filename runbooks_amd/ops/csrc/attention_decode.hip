@@ -429,7 +429,13 @@ RB_DEV unsigned rb_pack_bf16(float lo, float hi) {
   return ((unsigned)b.u << 16) | a.u;
 }
 
-template <int DH, int G, bool SPLIT>
+// FP8 mode: K rows stay the scalar-path e4m3 layout ([BS][DH bytes +
+// f32 scale + pad], fp8_row_bytes); V blocks are TRANSPOSED e4m3
+// ([DH+4][BS] bytes — the 4 tail "rows" are 64 B = the 16 per-token
+// f32 scales). Fragments are 8 B/lane raw loads dequanted at use; the
+// per-token K scale folds into the S rows after the score MFMA and the
+// V scale into P before the PV MFMA (softmax l/m stay on unscaled P).
+template <int DH, int G, bool SPLIT, bool FP8 = false>
 __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
     const uint16_t *__restrict__ q, const uint16_t *__restrict__ k_cache,
     const uint16_t *__restrict__ v_cache,
@@ -441,6 +447,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
   constexpr int BS = 16;                  // cache block = one S tile
   constexpr int KS32 = DH / 32;           // QK^T contraction steps
   constexpr int DT = DH / 32;             // O^T 32-row d-tiles
+  constexpr int FRW = FP8 ? 2 : 4;        // u32 words per raw fragment
 
   const int b = blockIdx.x;
   const int h_kv = blockIdx.y;
@@ -506,26 +513,65 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
   const int blk_lo = t_begin / BS;
   const int blk_hi = (t_end > t_begin) ? (t_end - 1) / BS : blk_lo - 1;
 
-  // 2-deep block ring: iteration i+1's K/V fragments are issued while
-  // i computes (PMC post-VT-layout: 74% WAIT_ANY at occupancy 2 —
+  // 2-deep block ring: iteration i+1's K/V raw fragments are issued
+  // while i computes (PMC post-VT-layout: 74% WAIT_ANY at occupancy 2 —
   // two waves/SIMD cannot hide HBM latency without in-flight depth).
-  // +KS32+DT fragment registers per slot keeps occupancy at 2.
-  rb_bf16x8v kfr[2][KS32], vfr[2][DT];
+  // +(KS32+DT)*FRW words per slot keeps occupancy at 2.
+  uint32_t kfr[2][KS32][FRW], vfr[2][DT][FRW];
+  float ksc[2][4], vsc[2][4];            // fp8 per-token scales
 
   auto fetch = [&](int bi, int slot) {
     const int bi_l = bi - blk0;
     const int blk = (bi_l < BT_LDS) ? bt_lds[bi_l] : bt[bi];
-    const uint16_t *kb = k_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
-    // v block is [DH][BS] (transposed layout)
-    const uint16_t *vb = v_cache + ((int64_t)blk * hkv + h_kv) * DH * BS;
+    if (FP8) {
+      constexpr int RB8 = DH + 16;       // fp8 K row bytes
+      const uint8_t *kb = reinterpret_cast<const uint8_t *>(k_cache) +
+          ((int64_t)blk * hkv + h_kv) * BS * RB8;
+      const uint8_t *vb = reinterpret_cast<const uint8_t *>(v_cache) +
+          ((int64_t)blk * hkv + h_kv) * (DH + 4) * BS;
 #pragma unroll
-    for (int ks = 0; ks < KS32; ++ks)
-      kfr[slot][ks] = *reinterpret_cast<const rb_bf16x8v *>(
-          kb + i16 * DH + ks * 32 + h16 * 8);
+      for (int ks = 0; ks < KS32; ++ks)
+        rb::ld_words<2>(kfr[slot][ks], kb + i16 * RB8 + ks * 32 + h16 * 8);
 #pragma unroll
-    for (int dt = 0; dt < DT; ++dt)
-      vfr[slot][dt] = *reinterpret_cast<const rb_bf16x8v *>(
-          vb + (dt * 32 + c32) * BS + h32 * 8);
+      for (int dt = 0; dt < DT; ++dt)
+        rb::ld_words<2>(vfr[slot][dt], vb + (dt * 32 + c32) * BS + h32 * 8);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int tok = h16 * 4 + r;
+        ksc[slot][r] = *reinterpret_cast<const float *>(kb + tok * RB8 + DH);
+        vsc[slot][r] = *reinterpret_cast<const float *>(
+            vb + DH * BS + tok * 4);
+      }
+    } else {
+      const uint16_t *kb = k_cache + ((int64_t)blk * hkv + h_kv) * BS * DH;
+      // v block is [DH][BS] (transposed layout)
+      const uint16_t *vb = v_cache + ((int64_t)blk * hkv + h_kv) * DH * BS;
+#pragma unroll
+      for (int ks = 0; ks < KS32; ++ks)
+        rb::ld_words<4>(kfr[slot][ks], reinterpret_cast<const uint8_t *>(
+            kb + i16 * DH + ks * 32 + h16 * 8));
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt)
+        rb::ld_words<4>(vfr[slot][dt], reinterpret_cast<const uint8_t *>(
+            vb + (dt * 32 + c32) * BS + h32 * 8));
+    }
+  };
+
+  // raw fragment -> bf16 MFMA operand (fp8: dequant UNSCALED — scales
+  // fold into S / P outside the MFMA)
+  auto frag = [&](const uint32_t *raw) -> rb_bf16x8v {
+    union { unsigned u[4]; rb_bf16x8v v; } c;
+    if (FP8) {
+      float f[8];
+      rb::fp8w_to_f32(raw[0], f);
+      rb::fp8w_to_f32(raw[1], f + 4);
+#pragma unroll
+      for (int e = 0; e < 4; ++e)
+        c.u[e] = rb_pack_bf16(f[2 * e], f[2 * e + 1]);
+    } else {
+      c.u[0] = raw[0]; c.u[1] = raw[1]; c.u[2] = raw[2]; c.u[3] = raw[3];
+    }
+    return c.v;
   };
 
   auto step = [&](int bi, int cur) {
@@ -535,8 +581,12 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
     rb_f32x4v s4 = (rb_f32x4v)(0.0f);
 #pragma unroll
     for (int ks = 0; ks < KS32; ++ks) {
-      s4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfr[cur][ks], qf[ks], s4,
-                                                   0, 0, 0);
+      s4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag(kfr[cur][ks]),
+                                                   qf[ks], s4, 0, 0, 0);
+    }
+    if (FP8) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) s4[r] *= ksc[cur][r];
     }
 
     // ---- mask + online softmax (state per head = per col = per lane) --
@@ -573,8 +623,15 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
       // ---- P[tok][head] -> B fragment of PV (k = tok, col = head) ----
       // source lane s = head + 16*(tok>>2) holds tokens 4*(s>>4)+r.
       // target lane l needs tokens (l>>5)*8..+8 of head l&31 (<16).
-      const unsigned w01 = rb_pack_bf16(p4[0], p4[1]);
-      const unsigned w23 = rb_pack_bf16(p4[2], p4[3]);
+      // fp8: the per-token V scale rides on P here (l/m and psum above
+      // stay on the unscaled softmax numerators).
+      float pv0 = p4[0], pv1 = p4[1], pv2 = p4[2], pv3 = p4[3];
+      if (FP8) {
+        pv0 *= vsc[cur][0]; pv1 *= vsc[cur][1];
+        pv2 *= vsc[cur][2]; pv3 *= vsc[cur][3];
+      }
+      const unsigned w01 = rb_pack_bf16(pv0, pv1);
+      const unsigned w23 = rb_pack_bf16(pv2, pv3);
       const int sh = c32 & 15;            // source head (cols 16+ unused)
       const int s1 = sh + 16 * (h32 * 2);
       const int s2 = s1 + 16;
@@ -588,7 +645,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            vfr[cur][dt], pf.v, acc_o[dt], 0, 0, 0);
+            frag(vfr[cur][dt]), pf.v, acc_o[dt], 0, 0, 0);
       }
     }
   };
@@ -678,16 +735,18 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
   const int max_blocks = (int)block_tables.size(1);
   uint16_t *swz = out_swz.has_value() ? (uint16_t *)out_swz->data_ptr()
                                       : nullptr;
-  // MFMA route: keyed on the TRANSPOSED V layout ([Dh, BS] blocks) the
-  // allocator chooses for MFMA-eligible models (bf16, G >= 4,
-  // DH <= 128, 16-token blocks — RB_DECODE_MFMA=0 at alloc reverts).
-  // The layout is the single source of truth: a transposed cache can
-  // only be read by the MFMA kernel and vice versa.
-  const bool vt = v_cache.size(2) == (int64_t)DH && v_cache.size(3) == bs;
-  if constexpr (!FP8 && G >= 4 && DH <= 128) {
+  // MFMA route: keyed on the TRANSPOSED V layout (bf16 [Dh, BS] /
+  // fp8 [Dh+4, BS] blocks) the allocator chooses for MFMA-eligible
+  // models (G >= 4, DH <= 128, 16-token blocks — RB_DECODE_MFMA=0 at
+  // alloc reverts). The layout is the single source of truth: a
+  // transposed cache can only be read by the MFMA kernel and vice
+  // versa.
+  const bool vt = v_cache.size(3) == bs &&
+      v_cache.size(2) == (int64_t)(FP8 ? DH + 4 : DH);
+  if constexpr (G >= 4 && DH <= 128) {
     if (vt && bs == 16) {
       if (nsplit <= 1) {
-        hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, false>),
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, false, FP8>),
                            dim3(B, hkv, 1), dim3(BLOCK), 0, stream,
                            (const uint16_t *)q.data_ptr(),
                            (const uint16_t *)k_cache.data_ptr(),
@@ -700,7 +759,7 @@ void launch_decode(const at::Tensor &q, const at::Tensor &k_cache,
         const int Hq = hkv * G;
         auto partial = at::empty({B, Hq, nsplit, DH + 2},
                                  q.options().dtype(at::kFloat));
-        hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, true>),
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<DH, G, true, FP8>),
                            dim3(B, hkv, nsplit), dim3(BLOCK), 0, stream,
                            (const uint16_t *)q.data_ptr(),
                            (const uint16_t *)k_cache.data_ptr(),

@@ -60,6 +60,10 @@ __global__ void kv_append_kernel(const uint16_t *__restrict__ k,
 // fp8-e4m3 append: one 16-lane group quantizes one (token, head) row
 // for k and v each — absmax over Dh via group shfl-reduce, scale =
 // absmax/448 stored after the Dh bytes (row stride Dh + 8).
+// VT: the v cache is transposed e4m3 ([dh+4][bs] bytes per block; the
+// 4 tail rows are the 16 per-token f32 scales) for the fp8 MFMA decode
+// path. k rows keep the scalar layout either way.
+template <bool VT>
 __global__ void kv_append_fp8_kernel(const uint16_t *__restrict__ k,
                                      const uint16_t *__restrict__ v,
                                      uint8_t *__restrict__ k_cache,
@@ -95,11 +99,21 @@ __global__ void kv_append_fp8_kernel(const uint16_t *__restrict__ k,
       amax = fmaxf(amax, __shfl_xor(amax, off, 64));
     const float scale = amax / 448.0f;
     const float inv = 1.0f / scale;
-    uint8_t *row = (is_v ? v_cache : k_cache) +
-        (((int64_t)(slot / bs) * hkv + h) * bs + slot % bs) * rb8;
-    for (int e = 0; e < ve; ++e)
-      row[gl * ve + e] = rb::f32_to_fp8(f[e] * inv);
-    if (gl == 0) *reinterpret_cast<float *>(row + dh) = scale;
+    if (VT && is_v) {
+      const int off = slot % bs;
+      uint8_t *vb = v_cache +
+          ((int64_t)(slot / bs) * hkv + h) * (int64_t)(dh + 4) * bs;
+      for (int e = 0; e < ve; ++e)
+        vb[(int64_t)(gl * ve + e) * bs + off] = rb::f32_to_fp8(f[e] * inv);
+      if (gl == 0)
+        *reinterpret_cast<float *>(vb + (int64_t)dh * bs + off * 4) = scale;
+    } else {
+      uint8_t *row = (is_v ? v_cache : k_cache) +
+          (((int64_t)(slot / bs) * hkv + h) * bs + slot % bs) * rb8;
+      for (int e = 0; e < ve; ++e)
+        row[gl * ve + e] = rb::f32_to_fp8(f[e] * inv);
+      if (gl == 0) *reinterpret_cast<float *>(row + dh) = scale;
+    }
   }
 }
 
@@ -117,14 +131,26 @@ void kv_append(at::Tensor k, at::Tensor v, at::Tensor k_cache, at::Tensor v_cach
   if (k_cache.scalar_type() == at::kByte) {
     const int dh = (int)k_cache.size(3) - 16;
     TORCH_CHECK(dh % 16 == 0 && dh <= 256, "kv_append fp8: Dh % 16");
+    const bool vt = v_cache.size(2) == (int64_t)(dh + 4) &&
+                    v_cache.size(3) == (int64_t)bs;
     const int grid = rb::rb_grid_1d(n_tokens * hkv * 2 * 16, BLOCK);
-    hipLaunchKernelGGL(kv_append_fp8_kernel, dim3(grid), dim3(BLOCK), 0,
-                       stream, (const uint16_t *)k.data_ptr(),
-                       (const uint16_t *)v.data_ptr(),
-                       (uint8_t *)k_cache.data_ptr(),
-                       (uint8_t *)v_cache.data_ptr(),
-                       slot_mapping.data_ptr<int32_t>(), n_tokens, hkv, bs,
-                       dh);
+    if (vt) {
+      hipLaunchKernelGGL(kv_append_fp8_kernel<true>, dim3(grid), dim3(BLOCK),
+                         0, stream, (const uint16_t *)k.data_ptr(),
+                         (const uint16_t *)v.data_ptr(),
+                         (uint8_t *)k_cache.data_ptr(),
+                         (uint8_t *)v_cache.data_ptr(),
+                         slot_mapping.data_ptr<int32_t>(), n_tokens, hkv, bs,
+                         dh);
+    } else {
+      hipLaunchKernelGGL(kv_append_fp8_kernel<false>, dim3(grid), dim3(BLOCK),
+                         0, stream, (const uint16_t *)k.data_ptr(),
+                         (const uint16_t *)v.data_ptr(),
+                         (uint8_t *)k_cache.data_ptr(),
+                         (uint8_t *)v_cache.data_ptr(),
+                         slot_mapping.data_ptr<int32_t>(), n_tokens, hkv, bs,
+                         dh);
+    }
     return;
   }
   TORCH_CHECK(k_cache.scalar_type() == at::kBFloat16, "kv_append: bf16 cache");

@@ -24,7 +24,13 @@ def alloc_kv_cache(num_blocks: int, num_kv_heads: int, head_dim: int,
         # decode loop stays on full 16 B/lane loads)
         shape = (num_blocks, num_kv_heads, block_size, head_dim + 16)
         k = torch.zeros(shape, device=device, dtype=torch.uint8)
-        v = torch.zeros(shape, device=device, dtype=torch.uint8)
+        if v_transposed:
+            # fp8 MFMA path: v blocks [dh+4, bs] bytes — the 4 tail
+            # rows are 64 B = the 16 per-token f32 scales
+            v = torch.zeros((num_blocks, num_kv_heads, head_dim + 4,
+                             block_size), device=device, dtype=torch.uint8)
+        else:
+            v = torch.zeros(shape, device=device, dtype=torch.uint8)
         return k, v
     shape = (num_blocks, num_kv_heads, block_size, head_dim)
     k = torch.zeros(shape, device=device, dtype=dtype)
@@ -44,7 +50,17 @@ def fp8_quant_row_ref(x: torch.Tensor):
 
 
 def fp8_dequant_cache_ref(cache: torch.Tensor) -> torch.Tensor:
-    """[blocks, hkv, bs, dh+16] uint8 -> [blocks, hkv, bs, dh] f32."""
+    """[blocks, hkv, bs, dh+16] uint8 -> [blocks, hkv, bs, dh] f32;
+    transposed-V fp8 blocks ([dh+4, bs] bytes) -> [blocks, hkv, dh, bs]
+    f32 (the bf16-vt shape, so paged_decode_ref's vt branch applies)."""
+    bs = 16
+    if cache.shape[-1] == bs and cache.shape[-2] > bs + 16:
+        dh = cache.shape[-2] - 4
+        data = cache[..., :dh, :].contiguous().view(
+            torch.float8_e4m3fn).float()
+        scales = cache[..., dh:, :].reshape(*cache.shape[:-2], -1)
+        scales = scales.contiguous().view(torch.float32)  # [..., bs]
+        return data * scales[..., None, :]
     dh = cache.shape[-1] - 16
     data = cache[..., :dh].contiguous().view(torch.float8_e4m3fn).float()
     scale = cache[..., dh:dh + 4].contiguous().view(torch.float32)
@@ -54,15 +70,26 @@ def fp8_dequant_cache_ref(cache: torch.Tensor) -> torch.Tensor:
 def _kv_append_fp8_ref(k, v, k_cache, v_cache, slot_mapping):
     bs = k_cache.shape[2]
     dh = k_cache.shape[-1] - 16
+    vt = v_cache.shape[2] == dh + 4 and v_cache.shape[3] == bs
     for t in range(k.shape[0]):
         slot = int(slot_mapping[t])
         if slot < 0:
             continue
         blk, off = divmod(slot, bs)
-        for cache, src in ((k_cache, k), (v_cache, v)):
-            q, scale = fp8_quant_row_ref(src[t])      # [hkv, dh], [hkv]
-            cache[blk, :, off, :dh] = q
-            cache[blk, :, off, dh:dh + 4] = scale.float().view(
+        q, scale = fp8_quant_row_ref(k[t])            # [hkv, dh], [hkv]
+        k_cache[blk, :, off, :dh] = q
+        k_cache[blk, :, off, dh:dh + 4] = scale.float().view(
+            torch.uint8).reshape(-1, 4)
+        q, scale = fp8_quant_row_ref(v[t])
+        if vt:
+            v_cache[blk, :, :dh, off] = q
+            # scale of token `off` lives at flat bytes dh*bs + off*4
+            tail = v_cache[blk, :, dh:, :].reshape(v_cache.shape[1], -1)
+            tail[:, off * 4:off * 4 + 4] = scale.float().view(
+                torch.uint8).reshape(-1, 4)
+        else:
+            v_cache[blk, :, off, :dh] = q
+            v_cache[blk, :, off, dh:dh + 4] = scale.float().view(
                 torch.uint8).reshape(-1, 4)
 
 

@@ -794,3 +794,57 @@ def test_lora_badd(T, N):
     ops.ext().lora_badd_(y, t, w, 2.0)
     assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2), \
         (y.float() - ref).abs().max()
+
+
+@pytest.mark.parametrize("hq,hkv,dh,nsplit", [(32, 8, 128, 1), (32, 8, 128, 4),
+                                              (64, 4, 64, 1)])
+def test_fp8_vt_mfma_decode(hq, hkv, dh, nsplit):
+    """fp8-e4m3 transposed-V cache -> the fp8 MFMA decode kernel, append
+    via the VT fp8 kernel, vs the dequantized fp32 reference."""
+    _assert_hip()
+    from runbooks_amd.ops import kvcache as kc
+    from runbooks_amd.ops.attention import _is_vt
+    torch.manual_seed(hq + dh)
+    B, bs, nblk = 3, 16, 12
+    kcache, vcache = kc.alloc_kv_cache(nblk, hkv, dh, DEV, fp8=True,
+                                       v_transposed=True)
+    assert _is_vt(kcache, vcache)
+    T = 3 * 60
+    k = torch.randn(T, hkv, dh, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn_like(k)
+    slots = torch.arange(T, dtype=torch.int32, device=DEV)
+    ops.kv_append(k, v, kcache, vcache, slots)
+
+    q = torch.randn(B, hq, dh, dtype=torch.bfloat16, device=DEV)
+    bt = torch.arange(B * 4, dtype=torch.int32, device=DEV).reshape(B, 4)
+    sl = torch.tensor([60, 47, 12], dtype=torch.int32, device=DEV)
+    st = torch.tensor([5, 0, 0], dtype=torch.int32, device=DEV)
+    got = ops.paged_decode(q, kcache, vcache, bt, sl, nsplit=nsplit,
+                           seq_starts=st)
+    ref = ops.paged_decode_ref(q.float().cpu(), kcache.cpu(), vcache.cpu(),
+                               bt.cpu(), sl.cpu(),
+                               scale=1.0 / math.sqrt(dh),
+                               seq_starts=st.cpu())
+    d = (got.float().cpu() - ref).abs().max() / ref.abs().max()
+    assert d < 3e-2, d
+
+
+def test_fp8_vt_append_matches_plain_layout():
+    """VT fp8 append stores the same dequantized values as the plain
+    fp8 layout."""
+    _assert_hip()
+    from runbooks_amd.ops import kvcache as kc
+    torch.manual_seed(2)
+    hkv, dh, nblk, T = 4, 128, 4, 30
+    k = torch.randn(T, hkv, dh, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn_like(k)
+    slots = torch.arange(T, dtype=torch.int32, device=DEV)
+    k1, v1 = kc.alloc_kv_cache(nblk, hkv, dh, DEV, fp8=True)
+    k2, v2 = kc.alloc_kv_cache(nblk, hkv, dh, DEV, fp8=True,
+                               v_transposed=True)
+    ops.kv_append(k, v, k1, v1, slots)
+    ops.kv_append(k, v, k2, v2, slots)
+    assert torch.equal(k1, k2)
+    d1 = kc.fp8_dequant_cache_ref(v1.cpu())           # [nb,hkv,bs,dh]
+    d2 = kc.fp8_dequant_cache_ref(v2.cpu())           # [nb,hkv,dh,bs]
+    assert torch.allclose(d1, d2.permute(0, 1, 3, 2), atol=1e-6)
