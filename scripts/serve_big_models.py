@@ -55,9 +55,10 @@ def truncated_parity(name: str) -> dict:
 
 
 def serve_run(name: str, batch: int, steps: int,
-              prompt_len: int = 96) -> dict:
+              prompt_len: int = 96, kv_fp8: bool = False) -> dict:
     t0 = time.time()
-    eng = Engine(name, dtype=torch.bfloat16, max_batch=batch, seed=17)
+    eng = Engine(name, dtype=torch.bfloat16, max_batch=batch, seed=17,
+                 kv_fp8=kv_fp8 or None)
     build_s = time.time() - t0
     free, total = torch.cuda.mem_get_info()
     g = torch.Generator().manual_seed(5)
@@ -77,7 +78,7 @@ def serve_run(name: str, batch: int, steps: int,
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     return {"model": name, "batch": batch, "steps": steps,
-            "prompt_len": prompt_len,
+            "prompt_len": prompt_len, "kv_fp8": kv_fp8,
             "build_s": round(build_s, 1),
             "hbm_used_gb": round((total - free) / 2**30, 1),
             "kv_blocks": eng.allocator.num_blocks,
@@ -94,6 +95,7 @@ def main():
     p.add_argument("--batch", type=int, default=32)
     p.add_argument("--steps", type=int, default=40)
     p.add_argument("--prompt-len", type=int, default=96)
+    p.add_argument("--fp8kv", action="store_true")
     args = p.parse_args()
     models = [args.model] if args.model else ["falcon-40b", "llama2-70b"]
     for name in models:
@@ -101,7 +103,7 @@ def main():
               flush=True)
         torch.cuda.empty_cache()
         print(json.dumps(serve_run(name, args.batch, args.steps,
-                                   args.prompt_len)),
+                                   args.prompt_len, kv_fp8=args.fp8kv)),
               flush=True)
         torch.cuda.empty_cache()
 
