@@ -248,3 +248,33 @@ def test_paged_decode_ref_v_transposed_matches():
     o2 = paged_decode_ref(q, kc, vc.permute(0, 1, 3, 2).contiguous(),
                           bt, lens, s)
     assert torch.allclose(o1, o2, atol=1e-6)
+
+
+def test_fp8_vt_reference_roundtrip():
+    """CPU reference fp8 append into plain vs transposed-V layouts
+    dequantizes to the same values; _is_vt spots the fp8-vt shape."""
+    import torch
+
+    from runbooks_amd.ops.attention import _is_vt
+    from runbooks_amd.ops.kvcache import (alloc_kv_cache,
+                                          fp8_dequant_cache_ref,
+                                          kv_append_ref)
+
+    torch.manual_seed(4)
+    hkv, dh, T = 2, 64, 7
+    k = torch.randn(T, hkv, dh)
+    v = torch.randn(T, hkv, dh)
+    slots = torch.arange(T, dtype=torch.int32)
+    k1, v1 = alloc_kv_cache(2, hkv, dh, "cpu", fp8=True)
+    k2, v2 = alloc_kv_cache(2, hkv, dh, "cpu", fp8=True, v_transposed=True)
+    assert not _is_vt(k1, v1) and _is_vt(k2, v2)
+    assert v2.shape == (2, hkv, dh + 4, 16)
+    kv_append_ref(k, v, k1, v1, slots)
+    kv_append_ref(k, v, k2, v2, slots)
+    assert torch.equal(k1, k2)
+    d1 = fp8_dequant_cache_ref(v1)            # [nb, hkv, bs, dh]
+    d2 = fp8_dequant_cache_ref(v2)            # [nb, hkv, dh, bs]
+    assert torch.allclose(d1, d2.permute(0, 1, 3, 2), atol=1e-6)
+    # quantization error itself is bounded (e4m3 rows)
+    live = d1.permute(0, 2, 1, 3).reshape(-1, hkv, dh)[:T]
+    assert torch.allclose(live, v, atol=0.1, rtol=0.1)
