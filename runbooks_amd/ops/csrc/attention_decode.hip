@@ -29,6 +29,26 @@ namespace {
 constexpr int BLOCK = 256;    // 4 waves
 constexpr int NW = 4;
 
+typedef __attribute__((ext_vector_type(8))) __bf16 rb_bf16x8v;
+typedef __attribute__((ext_vector_type(2))) __bf16 rb_bf16x2v;
+typedef __attribute__((ext_vector_type(16))) float rb_f32x16v;
+typedef __attribute__((ext_vector_type(4))) float rb_f32x4v;
+
+RB_DEV unsigned rb_pack_bf16(float lo, float hi) {
+  union { __bf16 b; unsigned short u; } a, b;
+  a.b = (__bf16)lo;
+  b.b = (__bf16)hi;
+  return ((unsigned)b.u << 16) | a.u;
+}
+
+// one v_dot2_f32_bf16: c += a_pair . b_pair (packed bf16 pairs)
+RB_DEV float rb_dot2(unsigned a, unsigned b, float c) {
+  union { unsigned u; rb_bf16x2v v; } x, y;
+  x.u = a;
+  y.u = b;
+  return __builtin_amdgcn_fdot2_f32_bf16(x.v, y.v, c, false);
+}
+
 // FP8: caches are e4m3 rows of fp8_row_bytes(DH) = DH bytes + f32
 // scale (+pad): decode attention is KV-bandwidth bound at long context,
 // so halving cache bytes halves the dominant stream AND doubles KV
@@ -90,8 +110,13 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   const int d0 = gl * VE;                  // this lane's Dh slice
 
   // Q for this wave's GW heads, pre-scaled (softmax scale folded in).
+  // bf16 caches keep Q as PACKED bf16 pairs too: the K dot then runs on
+  // v_dot2_f32_bf16 straight from the raw K words — no per-element
+  // convert+FMA chain (the kernel is VALU-saturated at occupancy 8;
+  // the same q*scale->bf16 rounding the MFMA kernel already makes).
   const int hq_base = hq0 + (HS ? wid * GW : 0);
   float qreg[GW][VE];
+  unsigned qpk[GW][VE / 2];
 #pragma unroll
   for (int g = 0; g < GW; ++g) {
     const uint16_t *qp = q + ((int64_t)b * Hq + hq_base + g) * DH + d0;
@@ -108,8 +133,15 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
             tmp[e8 + e] = rb::bf16_to_f32(qp[e8 + e]);
         }
       }
+      if (FP8) {
 #pragma unroll
-      for (int e = 0; e < VE; ++e) qreg[g][e] = tmp[e] * scale;
+        for (int e = 0; e < VE; ++e) qreg[g][e] = tmp[e] * scale;
+      } else {
+#pragma unroll
+        for (int e2 = 0; e2 < VE / 2; ++e2)
+          qpk[g][e2] = rb_pack_bf16(tmp[2 * e2] * scale,
+                                    tmp[2 * e2 + 1] * scale);
+      }
     }
   }
 
@@ -201,20 +233,24 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 #pragma unroll
       for (int e = 0; e < VE; ++e) { kf[e] *= ks; vf[e] *= vs; }
     } else {
-      const uint16_t *kr = reinterpret_cast<const uint16_t *>(kraw[cur]);
+      // only V needs f32 elements (the p*v accumulate); the K dot runs
+      // packed below, straight from the raw words
       const uint16_t *vr = reinterpret_cast<const uint16_t *>(vraw[cur]);
 #pragma unroll
-      for (int e = 0; e < VE; ++e) {
-        kf[e] = rb::bf16_to_f32(kr[e]);
-        vf[e] = rb::bf16_to_f32(vr[e]);
-      }
+      for (int e = 0; e < VE; ++e) vf[e] = rb::bf16_to_f32(vr[e]);
     }
 
 #pragma unroll
     for (int g = 0; g < GW; ++g) {
       float s = 0.f;
+      if (FP8) {
 #pragma unroll
-      for (int e = 0; e < VE; ++e) s += qreg[g][e] * kf[e];
+        for (int e = 0; e < VE; ++e) s += qreg[g][e] * kf[e];
+      } else {
+#pragma unroll
+        for (int e2 = 0; e2 < VE / 2; ++e2)
+          s = rb_dot2(kraw[cur][e2], qpk[g][e2], s);
+      }
       // reduce across the lane group
 #pragma unroll
       for (int off = GL / 2; off > 0; off >>= 1)
@@ -418,17 +454,6 @@ __global__ void decode_combine_kernel(const float *__restrict__ partial,
 //      16x16x32: A[i=l&15][k=(l>>4)*8+e]; B[k][j=l&15]; C[j=l&15][i=(l>>4)*4+r]
 //      32x32x16: A[i=l&31][k=(l>>5)*8+e]; C[j=l&31][i=(r&3)+8*(r>>2)+4*(l>>5)]
 // ---------------------------------------------------------------------------
-typedef __attribute__((ext_vector_type(8))) __bf16 rb_bf16x8v;
-typedef __attribute__((ext_vector_type(16))) float rb_f32x16v;
-typedef __attribute__((ext_vector_type(4))) float rb_f32x4v;
-
-RB_DEV unsigned rb_pack_bf16(float lo, float hi) {
-  union { __bf16 b; unsigned short u; } a, b;
-  a.b = (__bf16)lo;
-  b.b = (__bf16)hi;
-  return ((unsigned)b.u << 16) | a.u;
-}
-
 // FP8 mode: K rows stay the scalar-path e4m3 layout ([BS][DH bytes +
 // f32 scale + pad], fp8_row_bytes); V blocks are TRANSPOSED e4m3
 // ([DH+4][BS] bytes — the 4 tail "rows" are 64 B = the 16 per-token
