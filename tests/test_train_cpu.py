@@ -317,3 +317,39 @@ def test_lora_group_cache_does_not_leak_graph():
     for g in groups.values():
         assert isinstance(g, _LoRAGroup)
         assert g._t is None and g._left == 0
+
+
+def test_grad_checkpointing_with_lora_groups():
+    """Activation checkpointing recomputes the block forward with a
+    FRESH activation tensor each time — the LoRA group tcat cache
+    (keyed on tensor identity) must not serve stale tensors across the
+    recompute; grads must match the non-checkpointed run exactly."""
+    import torch
+
+    from runbooks_amd.models import build_model
+    from runbooks_amd.train.lora import apply_lora
+
+    tokens = torch.randint(0, 256, (2, 17))
+
+    def run(ckpt):
+        torch.manual_seed(0)
+        m = build_model("tiny-llama", dtype=torch.float32, seed=7)
+        apply_lora(m, r=4, alpha=8)
+        assert any(getattr(mm, "_group", None) is not None
+                   for mm in m.modules())
+        if ckpt:
+            m.enable_grad_checkpointing()
+        logits = m(tokens[:, :-1])
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]), tokens[:, 1:].reshape(-1))
+        loss.backward()
+        grads = {n: p.grad.clone() for n, p in m.named_parameters()
+                 if p.grad is not None}
+        return float(loss.detach()), grads
+
+    l0, g0 = run(False)
+    l1, g1 = run(True)
+    assert l0 == l1
+    assert g0.keys() == g1.keys() and len(g0) > 0
+    for n in g0:
+        assert torch.equal(g0[n], g1[n]), n
