@@ -144,6 +144,24 @@ def main():
     out.append(row("paged_decode_dh256", f"b{B//2} len{L} h16 d256", us,
                    2 * (B // 2) * 16 * L * 256 * 2))
 
+    # MFMA GQA decode (llama2-70b shape) over the transposed-V layout
+    kc3 = torch.randn(nb, 8, BS, 128, dtype=bf16, device=dev)
+    vc3 = torch.randn_like(kc3).permute(0, 1, 3, 2).contiguous()
+    qd3 = torch.randn(B, 64, 128, dtype=bf16, device=dev)
+    us = timeit(lambda: ops.paged_decode(qd3, kc3, vc3, bt, sl))
+    out.append(row("paged_decode_mfma_gqa", f"b{B} len{L} 64/8 d128", us,
+                   2 * B * 8 * L * 128 * 2))
+
+    # MFMA LoRA B-merge vs the hipBLASLt K=16 accumulate
+    yb = torch.randn(2048, 11008, dtype=bf16, device=dev)
+    tb = torch.randn(2048, 16, dtype=bf16, device=dev)
+    wb = torch.randn(11008, 16, dtype=bf16, device=dev)
+    us = timeit(lambda: ops.ext().lora_badd_(yb, tb, wb, 2.0))
+    out.append(row("lora_badd", "2048x11008 r16", us, 2 * yb.numel() * 2))
+    us = timeit(lambda: yb.addmm_(tb, wb.t(), alpha=2.0))
+    out.append(row("blaslt_lora_addmm", "2048x11008 r16", us,
+                   2 * yb.numel() * 2))
+
     # decode GEMMs: fresh weights per call (no L3 reuse)
     for N, K in [(12288, 4096), (4096, 4096), (22016, 4096), (4096, 11008),
                  (32000, 4096)]:
