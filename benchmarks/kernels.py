@@ -185,6 +185,18 @@ def main():
         us = timeit(skinny)
         out.append(row(f"skinny_decode", f"32x{N}x{K}", us, N * K * 2,
                        flops=2 * 32 * N * K))
+        # v2 weight-stream kernel over pre-swizzled operands (the
+        # shipped decode path)
+        sws = [ops.ext().decode_swizzle_w(w) for w in ws]
+        sx = ops.ext().decode_swizzle_x(xs)
+
+        def dgv2():
+            i[0] = (i[0] + 1) % 10
+            return ops.ext().decode_gemm(sx, sws[i[0]], 32, N, K)
+        us = timeit(dgv2)
+        out.append(row(f"decode_gemm_v2", f"32x{N}x{K}", us, N * K * 2,
+                       flops=2 * 32 * N * K))
+        del sws
 
     # fp8 decode GEMM
     from runbooks_amd.ops.linear import quantize_fp8
