@@ -700,3 +700,49 @@ def test_multi_prefill_admission():
     for a, b in zip(fr, sr):
         assert a.output_ids == b.output_ids, (a.request_id, a.output_ids,
                                               b.output_ids)
+
+
+@pytest.mark.parametrize("seed", [11])
+def test_engine_fuzz_vt_gqa_bf16(seed):
+    """The same prefix-cache + sliding-window fuzz invariants over a
+    bf16 GQA model that allocates the TRANSPOSED-V cache on CPU (the
+    reference kernels read it): allocator refcounts, free-list and
+    window-drop bookkeeping are layout-independent."""
+    import dataclasses
+    import random
+
+    from runbooks_amd.ops.attention import _is_vt
+
+    cfg = dataclasses.replace(get_config("tiny-llama"), name="tiny-vt-gqa",
+                              num_heads=8, num_kv_heads=2, head_dim=None,
+                              hidden_size=512, sliding_window=32)
+    m = build_model(cfg, dtype=torch.bfloat16)
+    eng = Engine(m, device="cpu", dtype=torch.bfloat16, kv_blocks=28,
+                 max_batch=4, seed=5, prefix_cache=True)
+    assert _is_vt(*eng.caches[0][:2]), "GQA bf16 must allocate vt caches"
+    rng = random.Random(seed)
+    templates = [[i + 1] * 24 for i in range(2)]
+    live = []
+    for step in range(150):
+        if rng.random() < 0.3 and len(live) < 12:
+            p = list(rng.choice(templates)) + \
+                [rng.randrange(256) for _ in range(rng.randint(0, 8))]
+            live.append(eng.submit(p, max_new_tokens=rng.randint(1, 20),
+                                   temperature=rng.choice([0.0, 0.8])))
+        if eng.has_work():
+            eng.step()
+        counts = [0] * eng.allocator.num_blocks
+        for r in eng.running:
+            assert r.dropped % eng.bs == 0
+            for b in r.blocks:
+                counts[b] += 1
+        for b in eng._pc.values():
+            counts[b] += 1
+        assert counts == eng.allocator.refs, (step, counts)
+    for _ in range(2000):
+        if not eng.has_work():
+            break
+        eng.step()
+    assert not eng.has_work()
+    eng.flush_prefix_cache()
+    assert len(eng.allocator.free) == eng.allocator.num_blocks
