@@ -179,3 +179,51 @@ def test_fused_residual_decode_matches_unfused(monkeypatch):
     scale = l_plain.abs().max().item()
     assert (l_fused - l_plain).abs().max().item() / scale < 1e-2, \
         (l_fused[:, :8], l_plain[:, :8])
+
+
+def test_qwen2_style_gqa_bias_vt_decode():
+    """qwen2-family decode on GPU: qkv BIAS + GQA G=4 -> the MFMA
+    decode path over the transposed-V cache, vs the CPU fp32 model
+    (shared bf16 weight values; logit comparison)."""
+    import dataclasses
+
+    from runbooks_amd.models import get_config
+    from runbooks_amd.models.config import register
+    from runbooks_amd.models.transformer import fuse_for_inference
+    from runbooks_amd.ops.attention import _is_vt
+    assert ops.has_hip()
+    cfg = dataclasses.replace(get_config("smoke-llama"), name="smoke-qwen",
+                              num_heads=8, num_kv_heads=2, head_dim=None,
+                              hidden_size=512, qkv_bias=True)
+    register(cfg)
+    prompt = [5, 9, 2, 7]
+    m = build_model("smoke-qwen", dtype=torch.bfloat16, device="cuda:0",
+                    seed=3)
+    fuse_for_inference(m)
+    cpu = build_model("smoke-qwen", dtype=torch.float32, seed=3)
+    cpu.load_state_dict({k: v.float().cpu() for k, v in m.state_dict().items()})
+    caches = m.alloc_caches(8, "cuda:0")
+    caches_c = cpu.alloc_caches(8, "cpu")
+    assert _is_vt(*caches[0][:2]), "GQA bf16 must allocate the vt cache"
+    S = len(prompt)
+
+    def run(model, caches, dev):
+        tokens = torch.tensor([prompt], dtype=torch.long, device=dev)
+        pos = torch.arange(S, dtype=torch.int32, device=dev)
+        slots = torch.arange(S, dtype=torch.int32, device=dev)
+        lp = model.prefill(tokens, pos, caches, slots)
+        t = torch.tensor([3], dtype=torch.long, device=dev)
+        p = torch.tensor([S], dtype=torch.int32, device=dev)
+        sl = torch.tensor([S], dtype=torch.int32, device=dev)
+        bt = torch.tensor([[0, 1]], dtype=torch.int32, device=dev)
+        seq = torch.tensor([S + 1], dtype=torch.int32, device=dev)
+        ld = model.decode(t, p, caches, sl, bt, seq)
+        return lp.float().cpu(), ld.float().cpu()
+
+    with torch.no_grad():
+        lp_g, ld_g = run(m, caches, "cuda:0")
+        lp_c, ld_c = run(cpu, caches_c, "cpu")
+    for got, ref in ((lp_g, lp_c), (ld_g, ld_c)):
+        scale = ref.abs().max().item()
+        assert (got - ref).abs().max().item() / scale < 3e-2, \
+            (got[:, :8], ref[:, :8])
