@@ -326,5 +326,19 @@ def delete(kind, name, namespace):
         sys.exit(1)
 
 
+def kubectl_notebook():
+    """kubectl plugin shim (reference `kubectl notebook` UX): an
+    executable named kubectl-notebook on PATH is discovered by kubectl
+    as a plugin; it forwards to `sub notebook`."""
+    main(["notebook", *sys.argv[1:]], prog_name="kubectl notebook")
+
+
+def kubectl_applybuild():
+    """kubectl plugin shim (reference `kubectl applybuild`): forwards
+    to `sub apply` with the build-context upload. Usage:
+    kubectl applybuild -f manifest.yaml --build ./dir"""
+    main(["apply", *sys.argv[1:]], prog_name="kubectl applybuild")
+
+
 if __name__ == "__main__":
     main()

@@ -36,6 +36,10 @@ setup(
         "sci-gcp = runbooks_amd.sci.gcp_server:main",
         "sci-aws = runbooks_amd.sci.aws_server:main",
         "nbwatch = runbooks_amd.nbwatch:main",
+        # kubectl plugin executables (reference kubectl notebook /
+        # kubectl applybuild UX — kubectl discovers kubectl-* on PATH)
+        "kubectl-notebook = runbooks_amd.cli.main:kubectl_notebook",
+        "kubectl-applybuild = runbooks_amd.cli.main:kubectl_applybuild",
     ]},
     ext_modules=[
         cpp_extension.CUDAExtension(

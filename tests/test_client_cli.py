@@ -315,3 +315,22 @@ def test_server_main_builds_engine(tmp_path, monkeypatch):
     assert got["engine"].cfg.name == "tiny-llama"
     out = got["engine"].generate([1, 2, 3], max_new_tokens=2)
     assert len(out) == 2
+
+
+def test_kubectl_plugin_shims(monkeypatch, capsys):
+    """kubectl-notebook / kubectl-applybuild entry points forward to the
+    sub subcommands (kubectl plugin discovery contract)."""
+    import sys
+
+    import pytest as _pytest
+
+    import importlib
+    cli_main = importlib.import_module("runbooks_amd.cli.main")
+
+    for fn in (cli_main.kubectl_notebook, cli_main.kubectl_applybuild):
+        monkeypatch.setattr(sys, "argv", ["x", "--help"])
+        with _pytest.raises(SystemExit) as e:
+            fn()
+        assert e.value.code == 0
+    out = capsys.readouterr().out
+    assert "kubectl applybuild" in out
